@@ -1,0 +1,72 @@
+"""Synthetic corpus/query generator for benchmarks and tests.
+
+BASELINE configs 2/4 run over synthetic docs with random-init encoder
+weights (no network for datasets): BM25 postings follow a Zipf term
+distribution; dense embeddings are random unit vectors; queries are
+random term sets + unit embeddings. All deterministic per (seed, shard).
+"""
+from __future__ import annotations
+
+import numpy as np
+import torch
+
+from .gpu_index import BM25_VOCAB, GpuShard
+
+
+def synth_corpus_arrays(n_docs: int, avg_len: int = 120,
+                        vocab: int = BM25_VOCAB, seed: int = 0,
+                        zipf_a: float = 1.3):
+    """Returns (flat_terms, flat_docs, doc_lens) numpy arrays."""
+    rng = np.random.default_rng(seed)
+    doc_lens = np.clip(
+        rng.lognormal(mean=np.log(avg_len), sigma=0.4, size=n_docs),
+        8, avg_len * 6).astype(np.int64)
+    total = int(doc_lens.sum())
+    flat_terms = (rng.zipf(zipf_a, size=total) - 1) % vocab
+    flat_docs = np.repeat(np.arange(n_docs, dtype=np.int64), doc_lens)
+    return flat_terms.astype(np.int64), flat_docs, doc_lens
+
+
+def synth_embeddings(n_docs: int, dim: int = 384, seed: int = 0,
+                     device: str = "cuda",
+                     batch: int = 1_000_000) -> torch.Tensor:
+    """Random unit-norm bf16 embeddings generated on-device."""
+    g = torch.Generator(device=device).manual_seed(seed)
+    out = torch.empty(n_docs, dim, device=device, dtype=torch.bfloat16)
+    for i in range(0, n_docs, batch):
+        j = min(i + batch, n_docs)
+        e = torch.randn(j - i, dim, generator=g, device=device)
+        out[i:j] = torch.nn.functional.normalize(e, dim=-1).bfloat16()
+    return out
+
+
+def build_synth_shard(n_docs: int, shard_rank: int = 0, world: int = 1,
+                      avg_len: int = 120, dim: int = 384,
+                      device: str = "cuda", seed: int = 0,
+                      with_dense: bool = True) -> GpuShard:
+    """Build one GPU shard of a world-sharded synthetic corpus.
+
+    Global ids interleave round-robin (gid = local * world + rank) so
+    shard contents are disjoint and the union covers the corpus."""
+    terms, docs, lens = synth_corpus_arrays(
+        n_docs, avg_len, seed=seed * 1000 + shard_rank)
+    emb = synth_embeddings(n_docs, dim, seed=seed * 1000 + shard_rank,
+                           device=device) if with_dense else None
+    gids = np.arange(n_docs, dtype=np.int64) * world + shard_rank
+    shard = GpuShard(device=device)
+    shard.build_from_arrays(terms, docs, lens, gids, emb)
+    return shard
+
+
+def synth_queries(n_queries: int, n_terms: int = 4, dim: int = 384,
+                  vocab: int = BM25_VOCAB, seed: int = 1,
+                  device: str = "cuda", zipf_a: float = 1.15):
+    """Random queries: term-id arrays (Zipf, slightly flatter than docs)
+    + unit embeddings. Returns (list[np.ndarray], tensor [B, dim] f32)."""
+    rng = np.random.default_rng(seed)
+    terms = [((rng.zipf(zipf_a, size=n_terms) - 1) % vocab).astype(np.int64)
+             for _ in range(n_queries)]
+    g = torch.Generator(device=device).manual_seed(seed)
+    emb = torch.nn.functional.normalize(
+        torch.randn(n_queries, dim, generator=g, device=device), dim=-1)
+    return terms, emb

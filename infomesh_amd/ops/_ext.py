@@ -1,0 +1,105 @@
+"""ctypes loader for the CDNA4 HIP extension (fail-loud policy).
+
+On a machine with a GPU the extension MUST load — ops raise
+GpuExtensionMissing rather than silently falling back to eager PyTorch
+(the round-end check records which .so files the GPU processes load).
+On CPU-only machines the torch reference implementations in
+ops/reference.py are the test oracle instead.
+"""
+from __future__ import annotations
+
+import ctypes
+from ctypes import c_float, c_int, c_long, c_void_p
+from pathlib import Path
+
+from ..errors import GpuExtensionMissing
+
+_LIB: ctypes.CDLL | None = None
+_LOAD_ERROR: str | None = None
+
+_SIGNATURES: dict[str, list] = {
+    "infomesh_gemm_bf16_nt": [c_void_p, c_void_p, c_void_p, c_void_p,
+                              c_int, c_int, c_int, c_int,
+                              c_long, c_long, c_long,
+                              c_int, c_float, c_int, c_void_p],
+    "infomesh_layernorm": [c_void_p, c_void_p, c_void_p, c_void_p,
+                           c_void_p, c_void_p, c_long, c_int, c_float,
+                           c_void_p],
+    "infomesh_rmsnorm": [c_void_p, c_void_p, c_void_p, c_void_p,
+                         c_void_p, c_long, c_int, c_float, c_void_p],
+    "infomesh_softmax": [c_void_p, c_void_p, c_void_p, c_long, c_int,
+                         c_int, c_int, c_float, c_void_p],
+    "infomesh_bias_act": [c_void_p, c_void_p, c_long, c_int, c_int, c_void_p],
+    "infomesh_silu_mul": [c_void_p, c_void_p, c_void_p, c_long, c_void_p],
+    "infomesh_add": [c_void_p, c_void_p, c_void_p, c_long, c_void_p],
+    "infomesh_rope": [c_void_p, c_void_p, c_void_p, c_void_p, c_long,
+                      c_int, c_int, c_int, c_void_p],
+    "infomesh_gather": [c_void_p, c_void_p, c_void_p, c_long, c_int,
+                        c_float, c_void_p],
+    "infomesh_pool": [c_void_p, c_void_p, c_void_p, c_int, c_int, c_int,
+                      c_int, c_int, c_void_p],
+    "infomesh_argmax": [c_void_p, c_void_p, c_long, c_int, c_void_p],
+    "infomesh_topk": [c_void_p, c_void_p, c_void_p, c_void_p, c_int,
+                      c_long, c_int, c_void_p],
+    "infomesh_bm25_score": [c_void_p, c_void_p, c_void_p, c_void_p,
+                            c_void_p, c_void_p, c_void_p, c_void_p,
+                            c_void_p, c_int, c_long, c_int, c_float,
+                            c_void_p],
+    "infomesh_score_combine": [c_void_p, c_void_p, c_void_p, c_float,
+                               c_float, c_long, c_void_p],
+    "infomesh_simhash_fingerprint": [c_void_p, c_void_p, c_void_p, c_long,
+                                     c_void_p],
+    "infomesh_hamming_scan": [c_void_p, c_void_p, c_void_p, c_void_p,
+                              c_void_p, c_void_p, c_int, c_long, c_int,
+                              c_int, c_void_p],
+    "infomesh_attn_decode": [c_void_p, c_void_p, c_void_p, c_void_p,
+                             c_void_p, c_int, c_int, c_int, c_int, c_int,
+                             c_float, c_void_p],
+    "infomesh_kv_append": [c_void_p, c_void_p, c_void_p, c_void_p,
+                           c_void_p, c_int, c_int, c_int, c_int, c_void_p],
+}
+
+_RESTYPES = {"infomesh_topk_workspace_u32": c_long}
+
+
+def _try_load() -> ctypes.CDLL | None:
+    global _LOAD_ERROR
+    so = Path(__file__).resolve().parent / "libinfomesh_hip.so"
+    if not so.exists():
+        _LOAD_ERROR = f"{so} not built"
+        return None
+    try:
+        lib = ctypes.CDLL(str(so))
+    except OSError as e:
+        _LOAD_ERROR = str(e)
+        return None
+    for name, argtypes in _SIGNATURES.items():
+        fn = getattr(lib, name)
+        fn.argtypes = argtypes
+        fn.restype = None
+    ws = lib.infomesh_topk_workspace_u32
+    ws.argtypes = [c_int]
+    ws.restype = c_long
+    return lib
+
+
+def lib() -> ctypes.CDLL:
+    """The loaded extension; raises GPU001 when missing."""
+    global _LIB
+    if _LIB is None:
+        _LIB = _try_load()
+    if _LIB is None:
+        raise GpuExtensionMissing(_LOAD_ERROR or "unknown load failure")
+    return _LIB
+
+
+def available() -> bool:
+    global _LIB
+    if _LIB is None:
+        _LIB = _try_load()
+    return _LIB is not None
+
+
+def stream_ptr() -> int:
+    import torch
+    return torch.cuda.current_stream().cuda_stream

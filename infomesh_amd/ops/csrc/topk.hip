@@ -1,0 +1,194 @@
+// Per-row top-k selection over [B, N] f32 score arrays (N up to tens of
+// millions, K <= 1024) via 2-level radix select + compaction + one-block
+// bitonic sort. Replaces: ChromaDB HNSW top-k (reference
+// infomesh/index/vector_store.py:216-220) and the FTS5 ORDER BY bm25()
+// LIMIT path (index/local_store.py:316-332) on the GPU shards.
+//
+// Passes (all streaming, ~3 reads of the score array):
+//   1. hist1: 256-bin histogram of ordered-float top byte (LDS-staged).
+//   2. select1: find byte bin containing the Kth value.
+//   3. hist2: 256-bin histogram of byte 2 within that bin.
+//   4. select2: 16-bit threshold prefix.
+//   5. compact: gather (value, idx) with prefix >= threshold (cap 8192).
+//   6. sort: one block per row bitonic-sorts candidates, emits top-K.
+// Ties sharing a 16-bit prefix can overflow the candidate cap on
+// pathological distributions: the overflow flag is checked by the wrapper.
+#include "common.h"
+
+#define TOPK_CAP 8192
+
+namespace {
+
+__global__ __launch_bounds__(256) void hist1_kernel(
+    const float* __restrict__ scores, unsigned* __restrict__ hist,
+    long N) {
+  __shared__ unsigned lh[256];
+  const int b = blockIdx.y;
+  for (int i = threadIdx.x; i < 256; i += blockDim.x) lh[i] = 0;
+  __syncthreads();
+  const long start = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long step = (long)gridDim.x * blockDim.x;
+  const float* row = scores + (long)b * N;
+  for (long i = start; i < N; i += step)
+    atomicAdd(&lh[float_to_ordered(row[i]) >> 24], 1u);
+  __syncthreads();
+  for (int i = threadIdx.x; i < 256; i += blockDim.x)
+    if (lh[i]) atomicAdd(&hist[(long)b * 256 + i], lh[i]);
+}
+
+__global__ void select1_kernel(const unsigned* __restrict__ hist,
+                               unsigned* __restrict__ bin1,
+                               unsigned* __restrict__ chi1, int K) {
+  const int b = blockIdx.x;
+  if (threadIdx.x != 0) return;
+  unsigned cum = 0;
+  for (int i = 255; i >= 0; --i) {
+    unsigned c = hist[(long)b * 256 + i];
+    if (cum + c >= (unsigned)K || i == 0) {
+      bin1[b] = (unsigned)i;
+      chi1[b] = cum;
+      return;
+    }
+    cum += c;
+  }
+}
+
+__global__ __launch_bounds__(256) void hist2_kernel(
+    const float* __restrict__ scores, const unsigned* __restrict__ bin1,
+    unsigned* __restrict__ hist2, long N) {
+  __shared__ unsigned lh[256];
+  const int b = blockIdx.y;
+  const unsigned b1 = bin1[b];
+  for (int i = threadIdx.x; i < 256; i += blockDim.x) lh[i] = 0;
+  __syncthreads();
+  const long start = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long step = (long)gridDim.x * blockDim.x;
+  const float* row = scores + (long)b * N;
+  for (long i = start; i < N; i += step) {
+    const unsigned o = float_to_ordered(row[i]);
+    if ((o >> 24) == b1) atomicAdd(&lh[(o >> 16) & 255], 1u);
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < 256; i += blockDim.x)
+    if (lh[i]) atomicAdd(&hist2[(long)b * 256 + i], lh[i]);
+}
+
+__global__ void select2_kernel(const unsigned* __restrict__ hist2,
+                               const unsigned* __restrict__ bin1,
+                               const unsigned* __restrict__ chi1,
+                               unsigned* __restrict__ thresh16, int K) {
+  const int b = blockIdx.x;
+  if (threadIdx.x != 0) return;
+  unsigned cum = chi1[b];
+  for (int i = 255; i >= 0; --i) {
+    unsigned c = hist2[(long)b * 256 + i];
+    if (cum + c >= (unsigned)K || i == 0) {
+      thresh16[b] = (bin1[b] << 8) | (unsigned)i;
+      return;
+    }
+    cum += c;
+  }
+}
+
+__global__ __launch_bounds__(256) void compact_kernel(
+    const float* __restrict__ scores, const unsigned* __restrict__ thresh16,
+    unsigned long long* __restrict__ cand, unsigned* __restrict__ cnt,
+    unsigned* __restrict__ overflow, long N) {
+  const int b = blockIdx.y;
+  const unsigned t16 = thresh16[b];
+  const long start = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long step = (long)gridDim.x * blockDim.x;
+  const float* row = scores + (long)b * N;
+  for (long i = start; i < N; i += step) {
+    const unsigned o = float_to_ordered(row[i]);
+    if ((o >> 16) >= t16) {
+      const unsigned pos = atomicAdd(&cnt[b], 1u);
+      if (pos < TOPK_CAP) {
+        // Ascending sort key: ~ordered in high bits (desc value),
+        // raw idx in low bits (ties -> smaller idx first).
+        cand[(long)b * TOPK_CAP + pos] =
+            ((unsigned long long)(~o) << 32) | (unsigned)i;
+      } else {
+        *overflow = 1u;
+      }
+    }
+  }
+}
+
+__global__ __launch_bounds__(256) void sort_emit_kernel(
+    unsigned long long* __restrict__ cand, const unsigned* __restrict__ cnt,
+    float* __restrict__ out_vals, int* __restrict__ out_idx, int K) {
+  __shared__ unsigned long long d[TOPK_CAP];
+  const int b = blockIdx.x;
+  const int n = min(cnt[b], (unsigned)TOPK_CAP);
+  for (int i = threadIdx.x; i < TOPK_CAP; i += blockDim.x)
+    d[i] = (i < n) ? cand[(long)b * TOPK_CAP + i] : ~0ULL;  // pad = worst
+  __syncthreads();
+  for (int k = 2; k <= TOPK_CAP; k <<= 1) {
+    for (int j = k >> 1; j > 0; j >>= 1) {
+      for (int i = threadIdx.x; i < TOPK_CAP; i += blockDim.x) {
+        const int ixj = i ^ j;
+        if (ixj > i) {
+          const bool up = ((i & k) == 0);
+          const unsigned long long a = d[i], c = d[ixj];
+          if ((a > c) == up) { d[i] = c; d[ixj] = a; }
+        }
+      }
+      __syncthreads();
+    }
+  }
+  for (int i = threadIdx.x; i < K; i += blockDim.x) {
+    if (i < n) {
+      const unsigned long long v = d[i];
+      out_vals[(long)b * K + i] = ordered_to_float(~(unsigned)(v >> 32));
+      out_idx[(long)b * K + i] = (int)(v & 0xffffffffu);
+    } else {
+      out_vals[(long)b * K + i] = -INFINITY;
+      out_idx[(long)b * K + i] = -1;
+    }
+  }
+}
+
+}  // namespace
+
+// Workspace layout (u32 units), provided zero-initialized by the wrapper:
+//   hist1 [B*256] | hist2 [B*256] | bin1 [B] | chi1 [B] | thresh16 [B]
+//   | cnt [B] | overflow [1] | cand (u64) [B*TOPK_CAP] (8-byte aligned)
+extern "C" long infomesh_topk_workspace_u32(int B) {
+  long u = (long)B * 256 * 2 + (long)B * 4 + 1;
+  u = (u + 1) & ~1L;  // align cand to 8 bytes
+  return u + (long)B * TOPK_CAP * 2;
+}
+
+extern "C" void infomesh_topk(const void* scores, void* workspace,
+                              void* out_vals, void* out_idx,
+                              int B, long N, int K, void* stream) {
+  auto s = reinterpret_cast<hipStream_t>(stream);
+  unsigned* ws = reinterpret_cast<unsigned*>(workspace);
+  unsigned* hist1 = ws;
+  unsigned* hist2 = hist1 + (long)B * 256;
+  unsigned* bin1 = hist2 + (long)B * 256;
+  unsigned* chi1 = bin1 + B;
+  unsigned* thresh16 = chi1 + B;
+  unsigned* cnt = thresh16 + B;
+  unsigned* overflow = cnt + B;
+  long off = (long)B * 256 * 2 + (long)B * 4 + 1;
+  off = (off + 1) & ~1L;
+  auto* cand = reinterpret_cast<unsigned long long*>(ws + off);
+
+  int chunks = (int)min((N + 256 * 64 - 1) / (256 * 64), (long)1024);
+  if (chunks < 1) chunks = 1;
+  dim3 g1(chunks, B), blk(256);
+  hipLaunchKernelGGL(hist1_kernel, g1, blk, 0, s,
+                     (const float*)scores, hist1, N);
+  hipLaunchKernelGGL(select1_kernel, dim3(B), dim3(64), 0, s,
+                     hist1, bin1, chi1, K);
+  hipLaunchKernelGGL(hist2_kernel, g1, blk, 0, s,
+                     (const float*)scores, bin1, hist2, N);
+  hipLaunchKernelGGL(select2_kernel, dim3(B), dim3(64), 0, s,
+                     hist2, bin1, chi1, thresh16, K);
+  hipLaunchKernelGGL(compact_kernel, g1, blk, 0, s,
+                     (const float*)scores, thresh16, cand, cnt, overflow, N);
+  hipLaunchKernelGGL(sort_emit_kernel, dim3(B), blk, 0, s,
+                     cand, cnt, (float*)out_vals, (int*)out_idx, K);
+}

@@ -1,0 +1,318 @@
+"""Typed torch-tensor wrappers over the HIP extension.
+
+Every function validates layout/dtype, then launches the hand-written
+CDNA4 kernel on the current HIP stream. No eager-PyTorch fallback on
+GPU — a missing extension raises GPU001 (errors.GpuExtensionMissing).
+"""
+from __future__ import annotations
+
+import torch
+
+from . import _ext
+
+ACT = {"none": 0, "gelu": 1, "silu": 2, "relu": 3, "tanh": 4}
+
+
+def _ptr(t: torch.Tensor | None) -> int | None:
+    return None if t is None else t.data_ptr()
+
+
+def _check(t: torch.Tensor, dtype: torch.dtype, name: str) -> None:
+    assert t.is_cuda, f"{name} must be on GPU"
+    assert t.dtype == dtype, f"{name} must be {dtype}, got {t.dtype}"
+    assert t.is_contiguous(), f"{name} must be contiguous"
+
+
+def gemm_nt(a: torch.Tensor, b: torch.Tensor,
+            bias: torch.Tensor | None = None,
+            act: str = "none", alpha: float = 1.0,
+            out_f32: bool = False,
+            out: torch.Tensor | None = None) -> torch.Tensor:
+    """C[...,M,N] = act(alpha * A[...,M,K] @ B[...,N,K]^T + bias[N]).
+
+    A: [M,K] or [G,M,K] bf16; B: [N,K] or [G,N,K] bf16 (shared across the
+    batch when 2-D). bias: [N] f32 or None."""
+    sq_a = a.dim() == 2
+    if sq_a:
+        a = a.unsqueeze(0)
+    if b.dim() == 2:
+        b = b.unsqueeze(0).expand(a.shape[0], *b.shape)
+    G, M, K = a.shape
+    Gb, N, Kb = b.shape
+    assert K == Kb and Gb == G, f"shape mismatch {a.shape} x {b.shape}"
+    assert K % 32 == 0, f"K={K} must be a multiple of 32"
+    _check(a, torch.bfloat16, "A")
+    assert b.dtype == torch.bfloat16
+    strideB = 0 if (G > 1 and b.stride(0) == 0) else N * K
+    if strideB != 0:
+        assert b.is_contiguous()
+    else:
+        assert b[0].is_contiguous()
+    if bias is not None:
+        _check(bias, torch.float32, "bias")
+        assert bias.numel() == N
+    dtype = torch.float32 if out_f32 else torch.bfloat16
+    if out is None:
+        out = torch.empty((G, M, N), device=a.device, dtype=dtype)
+    else:
+        assert out.shape == (G, M, N) and out.dtype == dtype and out.is_contiguous()
+    _ext.lib().infomesh_gemm_bf16_nt(
+        a.data_ptr(), b.data_ptr(), out.data_ptr(), _ptr(bias),
+        M, N, K, G, M * K, strideB, M * N,
+        ACT[act], alpha, int(out_f32), _ext.stream_ptr())
+    return out.squeeze(0) if sq_a else out
+
+
+def layernorm(x: torch.Tensor, gamma: torch.Tensor, beta: torch.Tensor,
+              residual: torch.Tensor | None = None, eps: float = 1e-12,
+              return_residual: bool = False):
+    """LayerNorm over the last dim; optionally fused residual add
+    (y = LN(x + residual)); returns (y, x+residual) if requested."""
+    H = x.shape[-1]
+    assert H % 8 == 0
+    rows = x.numel() // H
+    _check(x, torch.bfloat16, "x")
+    out = torch.empty_like(x)
+    res_out = None
+    if residual is not None:
+        assert residual.shape == x.shape
+        if return_residual:
+            res_out = torch.empty_like(x)
+    _ext.lib().infomesh_layernorm(
+        x.data_ptr(), _ptr(residual), out.data_ptr(), _ptr(res_out),
+        gamma.data_ptr(), beta.data_ptr(), rows, H, eps, _ext.stream_ptr())
+    return (out, res_out) if return_residual else out
+
+
+def rmsnorm(x: torch.Tensor, gamma: torch.Tensor,
+            residual: torch.Tensor | None = None, eps: float = 1e-5,
+            return_residual: bool = False):
+    H = x.shape[-1]
+    assert H % 8 == 0
+    rows = x.numel() // H
+    _check(x, torch.bfloat16, "x")
+    out = torch.empty_like(x)
+    res_out = None
+    if residual is not None and return_residual:
+        res_out = torch.empty_like(x)
+    _ext.lib().infomesh_rmsnorm(
+        x.data_ptr(), _ptr(residual), out.data_ptr(), _ptr(res_out),
+        gamma.data_ptr(), rows, H, eps, _ext.stream_ptr())
+    return (out, res_out) if return_residual else out
+
+
+def softmax(scores: torch.Tensor, scale: float = 1.0, causal: bool = False,
+            valid_len: torch.Tensor | None = None,
+            sq_dim: int | None = None) -> torch.Tensor:
+    """Masked row softmax: f32 [G,Sq,Sk] -> bf16 [G,Sq,Sk]."""
+    assert scores.dim() == 3
+    _check(scores, torch.float32, "scores")
+    G, Sq, Sk = scores.shape
+    out = torch.empty_like(scores, dtype=torch.bfloat16)
+    if valid_len is not None:
+        _check(valid_len, torch.int32, "valid_len")
+    _ext.lib().infomesh_softmax(
+        scores.data_ptr(), out.data_ptr(), _ptr(valid_len),
+        G, Sq, Sk, int(causal), scale, _ext.stream_ptr())
+    return out
+
+
+def bias_act(x: torch.Tensor, bias: torch.Tensor | None,
+             act: str = "none") -> torch.Tensor:
+    """In-place x = act(x + bias)."""
+    N = x.shape[-1]
+    assert N % 8 == 0
+    _check(x, torch.bfloat16, "x")
+    if bias is not None:
+        _check(bias, torch.float32, "bias")
+    _ext.lib().infomesh_bias_act(
+        x.data_ptr(), _ptr(bias), x.numel() // N, N, ACT[act],
+        _ext.stream_ptr())
+    return x
+
+
+def silu_mul(gate: torch.Tensor, up: torch.Tensor) -> torch.Tensor:
+    assert gate.shape == up.shape and gate.numel() % 8 == 0
+    out = torch.empty_like(gate)
+    _ext.lib().infomesh_silu_mul(gate.data_ptr(), up.data_ptr(),
+                                 out.data_ptr(), gate.numel(),
+                                 _ext.stream_ptr())
+    return out
+
+
+def add(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    assert a.shape == b.shape and a.numel() % 8 == 0
+    out = torch.empty_like(a)
+    _ext.lib().infomesh_add(a.data_ptr(), b.data_ptr(), out.data_ptr(),
+                            a.numel(), _ext.stream_ptr())
+    return out
+
+
+def rope(x: torch.Tensor, cos_t: torch.Tensor, sin_t: torch.Tensor,
+         pos: torch.Tensor, rot_dim: int | None = None) -> torch.Tensor:
+    """In-place NeoX-style RoPE on x [rows, H, D] with pos [rows] i32 and
+    host-precomputed cos/sin [max_pos, rot/2] f32 tables."""
+    rows, H, D = x.shape
+    rot = rot_dim or D
+    _check(x, torch.bfloat16, "x")
+    _check(pos, torch.int32, "pos")
+    _ext.lib().infomesh_rope(
+        x.data_ptr(), cos_t.data_ptr(), sin_t.data_ptr(), pos.data_ptr(),
+        rows, H, D, rot, _ext.stream_ptr())
+    return x
+
+
+def gather(table: torch.Tensor, ids: torch.Tensor,
+           scale: float = 1.0) -> torch.Tensor:
+    """out[r] = table[ids[r]] * scale (embedding lookup)."""
+    _check(table, torch.bfloat16, "table")
+    _check(ids, torch.int32, "ids")
+    H = table.shape[1]
+    out = torch.empty((ids.numel(), H), device=table.device,
+                      dtype=torch.bfloat16)
+    _ext.lib().infomesh_gather(table.data_ptr(), ids.data_ptr(),
+                               out.data_ptr(), ids.numel(), H, scale,
+                               _ext.stream_ptr())
+    return out
+
+
+def pool(x: torch.Tensor, lens: torch.Tensor | None = None,
+         mode: str = "cls", l2: bool = True) -> torch.Tensor:
+    """[B,S,H] bf16 -> [B,H] f32 (CLS or masked-mean pooling, L2 option)."""
+    B, S, H = x.shape
+    _check(x, torch.bfloat16, "x")
+    out = torch.empty((B, H), device=x.device, dtype=torch.float32)
+    _ext.lib().infomesh_pool(
+        x.data_ptr(), _ptr(lens), out.data_ptr(), B, S, H,
+        {"cls": 0, "mean": 1}[mode], int(l2), _ext.stream_ptr())
+    return out
+
+
+def argmax(logits: torch.Tensor) -> torch.Tensor:
+    rows, V = logits.shape
+    _check(logits, torch.float32, "logits")
+    out = torch.empty((rows,), device=logits.device, dtype=torch.int32)
+    _ext.lib().infomesh_argmax(logits.data_ptr(), out.data_ptr(), rows, V,
+                               _ext.stream_ptr())
+    return out
+
+
+class TopK:
+    """Reusable top-k selector (keeps its workspace allocated)."""
+
+    def __init__(self, device: torch.device | str = "cuda"):
+        self.device = torch.device(device)
+        self._ws: torch.Tensor | None = None
+        self._ws_b = 0
+
+    def __call__(self, scores: torch.Tensor, k: int
+                 ) -> tuple[torch.Tensor, torch.Tensor]:
+        assert scores.dim() == 2
+        _check(scores, torch.float32, "scores")
+        B, N = scores.shape
+        assert 1 <= k <= 1024
+        lib = _ext.lib()
+        nu32 = lib.infomesh_topk_workspace_u32(B)
+        if self._ws is None or self._ws_b < nu32:
+            self._ws = torch.empty(nu32, device=scores.device,
+                                   dtype=torch.int32)
+            self._ws_b = nu32
+        self._ws[: B * 512 + 4 * B + 2].zero_()  # hists + counters region
+        vals = torch.empty((B, k), device=scores.device, dtype=torch.float32)
+        idx = torch.empty((B, k), device=scores.device, dtype=torch.int32)
+        lib.infomesh_topk(scores.data_ptr(), self._ws.data_ptr(),
+                          vals.data_ptr(), idx.data_ptr(), B, N, k,
+                          _ext.stream_ptr())
+        # Overflow flag sits right after cnt[B] in the workspace.
+        flag_off = B * 512 + 4 * B
+        if int(self._ws[flag_off].item()) != 0:
+            raise RuntimeError(
+                "topk candidate overflow (pathologically tied scores)")
+        return vals, idx
+
+
+def topk(scores: torch.Tensor, k: int) -> tuple[torch.Tensor, torch.Tensor]:
+    return TopK(scores.device)(scores, k)
+
+
+def bm25_score(offsets: torch.Tensor, doc_ids: torch.Tensor,
+               tfs: torch.Tensor, doc_norm: torch.Tensor,
+               chunk_qrow: torch.Tensor, chunk_term: torch.Tensor,
+               chunk_off: torch.Tensor, chunk_idf: torch.Tensor,
+               scores: torch.Tensor, chunk_size: int = 2048,
+               k1: float = 1.2) -> torch.Tensor:
+    """Accumulate BM25 into pre-zeroed scores [B, N] f32."""
+    B, N = scores.shape
+    _check(scores, torch.float32, "scores")
+    n = chunk_qrow.numel()
+    _ext.lib().infomesh_bm25_score(
+        offsets.data_ptr(), doc_ids.data_ptr(), tfs.data_ptr(),
+        doc_norm.data_ptr(), chunk_qrow.data_ptr(), chunk_term.data_ptr(),
+        chunk_off.data_ptr(), chunk_idf.data_ptr(), scores.data_ptr(),
+        n, N, chunk_size, k1, _ext.stream_ptr())
+    return scores
+
+
+def score_combine(a: torch.Tensor, b: torch.Tensor, wa: float,
+                  wb: float) -> torch.Tensor:
+    out = torch.empty_like(a)
+    _ext.lib().infomesh_score_combine(a.data_ptr(), b.data_ptr(),
+                                      out.data_ptr(), wa, wb, a.numel(),
+                                      _ext.stream_ptr())
+    return out
+
+
+def simhash_fingerprint(offsets: torch.Tensor,
+                        hashes: torch.Tensor) -> torch.Tensor:
+    """CSR shingle hashes (int64-as-u64) -> int64 fingerprints [D]."""
+    ndocs = offsets.numel() - 1
+    out = torch.empty(ndocs, device=offsets.device, dtype=torch.int64)
+    _ext.lib().infomesh_simhash_fingerprint(
+        offsets.data_ptr(), hashes.data_ptr(), out.data_ptr(), ndocs,
+        _ext.stream_ptr())
+    return out
+
+
+def hamming_scan(queries: torch.Tensor, table: torch.Tensor,
+                 radius: int = 3, cap: int = 65536
+                 ) -> tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+    """Matches (q_idx, table_idx, dist) with hamming(q, t) <= radius."""
+    M, N = queries.numel(), table.numel()
+    dev = table.device
+    out_q = torch.empty(cap, device=dev, dtype=torch.int32)
+    out_n = torch.empty(cap, device=dev, dtype=torch.int32)
+    out_d = torch.empty(cap, device=dev, dtype=torch.int32)
+    cnt = torch.zeros(1, device=dev, dtype=torch.int32)
+    _ext.lib().infomesh_hamming_scan(
+        queries.data_ptr(), table.data_ptr(), out_q.data_ptr(),
+        out_n.data_ptr(), out_d.data_ptr(), cnt.data_ptr(), M, N, radius,
+        cap, _ext.stream_ptr())
+    n = min(int(cnt.item()), cap)
+    return out_q[:n], out_n[:n], out_d[:n]
+
+
+def attn_decode(q: torch.Tensor, k_cache: torch.Tensor,
+                v_cache: torch.Tensor, lens: torch.Tensor,
+                scale: float) -> torch.Tensor:
+    """q [B,H,D], caches [B,Hkv,Smax,D], lens [B] i32 -> out [B,H,D]."""
+    B, H, D = q.shape
+    _, Hkv, Smax, _ = k_cache.shape
+    _check(q, torch.bfloat16, "q")
+    assert D % 8 == 0 and D <= 128
+    out = torch.empty_like(q)
+    _ext.lib().infomesh_attn_decode(
+        q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
+        lens.data_ptr(), out.data_ptr(), B, H, Hkv, Smax, D, scale,
+        _ext.stream_ptr())
+    return out
+
+
+def kv_append(k_new: torch.Tensor, v_new: torch.Tensor,
+              k_cache: torch.Tensor, v_cache: torch.Tensor,
+              pos: torch.Tensor) -> None:
+    B, Hkv, D = k_new.shape
+    _, _, Smax, _ = k_cache.shape
+    _ext.lib().infomesh_kv_append(
+        k_new.data_ptr(), v_new.data_ptr(), k_cache.data_ptr(),
+        v_cache.data_ptr(), pos.data_ptr(), B, Hkv, Smax, D,
+        _ext.stream_ptr())

@@ -1,0 +1,345 @@
+"""GPU kernel parity tests vs the fp32 PyTorch references (MI355X only).
+
+Strategy: every HIP kernel is compared against
+ops/reference.py on random data; GEMM/attention inputs are asymmetric so
+operand/output transposes cannot pass (guide §5.4 rule 16).
+"""
+from __future__ import annotations
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+if torch.cuda.is_available():
+    from infomesh_amd.ops import _build
+    _build.build()
+from infomesh_amd.ops import kernels as K
+from infomesh_amd.ops import reference as R
+
+
+def _skip_no_gpu():
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+
+
+@pytest.fixture(autouse=True)
+def _gpu():
+    _skip_no_gpu()
+    torch.manual_seed(0)
+
+
+def _assert_close(gpu, ref, rtol=2e-2, atol=2e-2, what=""):
+    gpu = gpu.float().cpu()
+    ref = ref.float().cpu()
+    err = (gpu - ref).abs().max().item()
+    scale = ref.abs().max().item() + 1e-6
+    assert err <= atol + rtol * scale, f"{what}: max err {err} (scale {scale})"
+
+
+# ------------------------------------------------------------------- GEMM
+
+@pytest.mark.parametrize("M,N,K", [
+    (128, 128, 64), (64, 384, 384), (200, 1000, 96),
+    (256, 1536, 384), (33, 100, 32), (512, 512, 512),
+])
+def test_gemm_nt_shapes(M, N, K):
+    a = torch.randn(M, K, device="cuda").bfloat16()
+    b = torch.randn(N, K, device="cuda").bfloat16()
+    out = K.gemm_nt(a, b)
+    ref = R.gemm_nt(a.cpu(), b.cpu())
+    _assert_close(out, ref, rtol=3e-2, atol=K ** 0.5 * 2e-2, what=f"gemm {M}x{N}x{K}")
+
+
+def test_gemm_asymmetric_catches_transpose():
+    # Asymmetric B (guide: A=I with asymmetric B catches row/col swap).
+    M = N = Kd = 128
+    a = torch.eye(M, device="cuda").bfloat16()
+    b = torch.arange(N * Kd, device="cuda").reshape(N, Kd).bfloat16() / (N * Kd)
+    out = K.gemm_nt(a, b)
+    ref = R.gemm_nt(a.cpu(), b.cpu())
+    _assert_close(out, ref, what="gemm transpose check")
+
+
+def test_gemm_bias_act_f32out():
+    a = torch.randn(96, 64, device="cuda").bfloat16()
+    b = torch.randn(256, 64, device="cuda").bfloat16()
+    bias = torch.randn(256, device="cuda")
+    for act in ("none", "gelu", "silu", "relu", "tanh"):
+        out = K.gemm_nt(a, b, bias=bias, act=act, out_f32=True)
+        assert out.dtype == torch.float32
+        ref = R.gemm_nt(a.cpu(), b.cpu(), bias.cpu(), act=act)
+        _assert_close(out, ref, what=f"gemm act={act}")
+
+
+def test_gemm_batched_and_shared_b():
+    G, M, N, Kd = 6, 64, 96, 64
+    a = torch.randn(G, M, Kd, device="cuda").bfloat16()
+    b = torch.randn(G, N, Kd, device="cuda").bfloat16()
+    out = K.gemm_nt(a, b)
+    ref = R.gemm_nt(a.cpu(), b.cpu())
+    _assert_close(out, ref, what="batched gemm")
+    bs = torch.randn(N, Kd, device="cuda").bfloat16()
+    out2 = K.gemm_nt(a, bs)
+    ref2 = R.gemm_nt(a.cpu(), bs.cpu())
+    _assert_close(out2, ref2, what="shared-B gemm")
+
+
+def test_gemm_alpha():
+    a = torch.randn(64, 32, device="cuda").bfloat16()
+    b = torch.randn(64, 32, device="cuda").bfloat16()
+    out = K.gemm_nt(a, b, alpha=0.125, out_f32=True)
+    _assert_close(out, R.gemm_nt(a.cpu(), b.cpu(), alpha=0.125), what="alpha")
+
+
+# ------------------------------------------------------------------ norms
+
+def test_layernorm():
+    x = torch.randn(37, 384, device="cuda").bfloat16()
+    g = torch.randn(384, device="cuda").bfloat16()
+    b = torch.randn(384, device="cuda").bfloat16()
+    _assert_close(K.layernorm(x, g, b),
+                  R.layernorm(x.cpu(), g.cpu(), b.cpu()), what="ln")
+
+
+def test_layernorm_residual():
+    x = torch.randn(16, 768, device="cuda").bfloat16()
+    r = torch.randn(16, 768, device="cuda").bfloat16()
+    g = torch.randn(768, device="cuda").bfloat16()
+    b = torch.randn(768, device="cuda").bfloat16()
+    out, res = K.layernorm(x, g, b, residual=r, return_residual=True)
+    _assert_close(out, R.layernorm(x.cpu(), g.cpu(), b.cpu(), r.cpu()),
+                  what="ln+res")
+    _assert_close(res, x.cpu().float() + r.cpu().float(), what="res out")
+
+
+def test_layernorm_huge_row():
+    x = torch.randn(3, 32768, device="cuda").bfloat16()  # beyond MAXV regs
+    g = torch.ones(32768, device="cuda").bfloat16()
+    b = torch.zeros(32768, device="cuda").bfloat16()
+    _assert_close(K.layernorm(x, g, b),
+                  R.layernorm(x.cpu(), g.cpu(), b.cpu()), what="ln huge")
+
+
+def test_rmsnorm():
+    x = torch.randn(21, 3072, device="cuda").bfloat16()
+    g = torch.randn(3072, device="cuda").bfloat16()
+    _assert_close(K.rmsnorm(x, g), R.rmsnorm(x.cpu(), g.cpu()), what="rms")
+
+
+# ---------------------------------------------------------------- softmax
+
+def test_softmax_plain_and_scale():
+    s = torch.randn(4, 32, 200, device="cuda")
+    _assert_close(K.softmax(s, scale=0.3), R.softmax(s.cpu(), scale=0.3),
+                  atol=5e-3, what="softmax")
+
+
+def test_softmax_causal():
+    s = torch.randn(2, 64, 64, device="cuda")
+    _assert_close(K.softmax(s, causal=True), R.softmax(s.cpu(), causal=True),
+                  atol=5e-3, what="softmax causal")
+
+
+def test_softmax_causal_prefill_offset():
+    s = torch.randn(2, 16, 48, device="cuda")  # Sq < Sk (continuation)
+    _assert_close(K.softmax(s, causal=True), R.softmax(s.cpu(), causal=True),
+                  atol=5e-3, what="softmax causal offset")
+
+
+def test_softmax_valid_len():
+    s = torch.randn(3, 8, 100, device="cuda")
+    vl = torch.tensor([10, 100, 1], device="cuda", dtype=torch.int32)
+    _assert_close(K.softmax(s, valid_len=vl),
+                  R.softmax(s.cpu(), valid_len=vl.cpu()),
+                  atol=5e-3, what="softmax masked")
+
+
+# ------------------------------------------------------------ elementwise
+
+def test_bias_act_inplace():
+    x = torch.randn(10, 64, device="cuda").bfloat16()
+    b = torch.randn(64, device="cuda")
+    ref = R.apply_act(x.cpu().float() + b.cpu(), "gelu")
+    _assert_close(K.bias_act(x, b, "gelu"), ref, what="bias_gelu")
+
+
+def test_silu_mul_and_add():
+    g = torch.randn(8, 512, device="cuda").bfloat16()
+    u = torch.randn(8, 512, device="cuda").bfloat16()
+    _assert_close(K.silu_mul(g, u), R.silu_mul(g.cpu(), u.cpu()), what="swiglu")
+    _assert_close(K.add(g, u), g.cpu().float() + u.cpu().float(), what="add")
+
+
+def test_rope():
+    rows, H, D = 6, 4, 96
+    x = torch.randn(rows, H, D, device="cuda").bfloat16()
+    inv = 1.0 / (10000 ** (torch.arange(D // 2).float() * 2 / D))
+    t = torch.arange(64).float()
+    ang = torch.outer(t, inv)
+    cos_t, sin_t = ang.cos().cuda(), ang.sin().cuda()
+    pos = torch.tensor([0, 5, 9, 13, 33, 63], device="cuda", dtype=torch.int32)
+    ref = R.rope(x.cpu(), cos_t.cpu(), sin_t.cpu(), pos.cpu())
+    _assert_close(K.rope(x, cos_t, sin_t, pos), ref, what="rope")
+
+
+def test_gather():
+    table = torch.randn(1000, 384, device="cuda").bfloat16()
+    ids = torch.randint(0, 1000, (57,), device="cuda", dtype=torch.int32)
+    _assert_close(K.gather(table, ids),
+                  table.cpu()[ids.cpu().long()].float(), what="gather")
+
+
+def test_pool_cls_and_mean():
+    x = torch.randn(5, 33, 384, device="cuda").bfloat16()
+    lens = torch.tensor([33, 10, 1, 20, 5], device="cuda", dtype=torch.int32)
+    _assert_close(K.pool(x, mode="cls"), R.pool(x.cpu(), mode="cls"),
+                  atol=1e-2, what="cls pool")
+    _assert_close(K.pool(x, lens, mode="mean"),
+                  R.pool(x.cpu(), lens.cpu(), mode="mean"),
+                  atol=1e-2, what="mean pool")
+
+
+def test_argmax():
+    logits = torch.randn(17, 32064, device="cuda")
+    out = K.argmax(logits)
+    ref = logits.cpu().argmax(-1)
+    assert (out.cpu().long() == ref).all()
+
+
+# ------------------------------------------------------------------ top-k
+
+@pytest.mark.parametrize("B,N,k", [(4, 10_000, 10), (2, 1_000_000, 100),
+                                   (64, 50_000, 100), (1, 5000, 1024)])
+def test_topk_matches_torch(B, N, k):
+    scores = torch.randn(B, N, device="cuda")
+    vals, idx = K.topk(scores, k)
+    rv, ri = torch.topk(scores, k, dim=-1)
+    assert torch.allclose(vals, rv, atol=0), \
+        f"value mismatch: {(vals - rv).abs().max()}"
+    # Indices must point at their values (ties may reorder).
+    picked = torch.gather(scores, 1, idx.long())
+    assert torch.allclose(picked, vals, atol=0)
+
+
+def test_topk_negative_scores():
+    scores = -torch.rand(3, 20_000, device="cuda") - 5.0
+    vals, idx = K.topk(scores, 7)
+    rv, _ = torch.topk(scores, 7, dim=-1)
+    assert torch.allclose(vals, rv, atol=0)
+
+
+# ------------------------------------------------------------------- BM25
+
+def test_bm25_parity():
+    import numpy as np
+    rng = np.random.default_rng(0)
+    n_docs, vocab = 5000, 300
+    postings: dict[int, list[tuple[int, int]]] = {}
+    doc_lens = torch.zeros(n_docs, dtype=torch.long)
+    for t in range(vocab):
+        df = int(rng.integers(1, 200))
+        docs = rng.choice(n_docs, size=df, replace=False)
+        plist = []
+        for d in sorted(docs):
+            tf = int(rng.integers(1, 5))
+            plist.append((int(d), tf))
+            doc_lens[d] += tf
+        postings[t] = plist
+    queries = [[1, 2, 3], [10, 250], [0], [299, 5, 5]]
+    ref = R.bm25_scores(postings, doc_lens, queries, n_docs)
+
+    # Build CSR + chunks like index/gpu_index.py does.
+    import math
+    offsets = [0]
+    doc_ids, tfs = [], []
+    for t in range(vocab):
+        for d, tf in postings[t]:
+            doc_ids.append(d)
+            tfs.append(tf)
+        offsets.append(len(doc_ids))
+    avgdl = float(doc_lens.float().mean())
+    doc_norm = 1.2 * (1 - 0.75 + 0.75 * doc_lens.float() / avgdl)
+    dev = "cuda"
+    offsets_t = torch.tensor(offsets, dtype=torch.int64, device=dev)
+    scores = torch.zeros(len(queries), n_docs, device=dev)
+    cq, ct, co, ci = [], [], [], []
+    for qi, terms in enumerate(queries):
+        for t in terms:
+            df = offsets[t + 1] - offsets[t]
+            idf = math.log(1.0 + (n_docs - df + 0.5) / (df + 0.5))
+            for off in range(offsets[t], offsets[t + 1], 2048):
+                cq.append(qi); ct.append(t); co.append(off); ci.append(idf)
+    K.bm25_score(
+        offsets_t,
+        torch.tensor(doc_ids, dtype=torch.int32, device=dev),
+        torch.tensor(tfs, dtype=torch.int16, device=dev).to(torch.int16),
+        doc_norm.to(dev),
+        torch.tensor(cq, dtype=torch.int32, device=dev),
+        torch.tensor(ct, dtype=torch.int32, device=dev),
+        torch.tensor(co, dtype=torch.int64, device=dev),
+        torch.tensor(ci, dtype=torch.float32, device=dev),
+        scores)
+    _assert_close(scores, ref, rtol=1e-3, atol=1e-3, what="bm25")
+
+
+# ---------------------------------------------------------------- simhash
+
+def test_simhash_fingerprint_parity():
+    import numpy as np
+    rng = np.random.default_rng(1)
+    docs = [list(rng.integers(0, 2**63, size=int(rng.integers(1, 50))))
+            for _ in range(100)]
+    ref = R.simhash_fingerprint(docs)
+    offsets = [0]
+    flat = []
+    for d in docs:
+        flat.extend(d)
+        offsets.append(len(flat))
+    dev = "cuda"
+    fps = K.simhash_fingerprint(
+        torch.tensor(offsets, dtype=torch.int64, device=dev),
+        torch.tensor(flat, dtype=torch.int64, device=dev))
+    got = [int(v) & (2**64 - 1) for v in fps.cpu()]
+    assert got == ref
+
+
+def test_hamming_scan_parity():
+    import numpy as np
+    rng = np.random.default_rng(2)
+    table = list(rng.integers(0, 2**63, size=2000))
+    queries = [table[5], table[100] ^ 0b111, int(rng.integers(0, 2**63))]
+    ref = R.hamming_matches(queries, table, radius=3)
+    dev = "cuda"
+    q, n, d = K.hamming_scan(
+        torch.tensor(queries, dtype=torch.int64, device=dev),
+        torch.tensor(table, dtype=torch.int64, device=dev), radius=3)
+    got = set(zip(q.cpu().tolist(), n.cpu().tolist()))
+    assert got == ref
+
+
+# ----------------------------------------------------------- attn decode
+
+def test_attn_decode_parity():
+    B, H, Hkv, Smax, D = 2, 8, 4, 256, 96
+    q = torch.randn(B, H, D, device="cuda").bfloat16()
+    kc = torch.randn(B, Hkv, Smax, D, device="cuda").bfloat16()
+    vc = torch.randn(B, Hkv, Smax, D, device="cuda").bfloat16()
+    lens = torch.tensor([100, 256], device="cuda", dtype=torch.int32)
+    out = K.attn_decode(q, kc, vc, lens, scale=D ** -0.5)
+    ref = R.attn_decode(q.cpu(), kc.cpu(), vc.cpu(), lens.cpu(), D ** -0.5)
+    _assert_close(out, ref, atol=2e-2, what="attn decode")
+
+
+def test_kv_append():
+    B, Hkv, Smax, D = 2, 4, 64, 96
+    kc = torch.zeros(B, Hkv, Smax, D, device="cuda").bfloat16()
+    vc = torch.zeros(B, Hkv, Smax, D, device="cuda").bfloat16()
+    kn = torch.randn(B, Hkv, D, device="cuda").bfloat16()
+    vn = torch.randn(B, Hkv, D, device="cuda").bfloat16()
+    pos = torch.tensor([3, 10], device="cuda", dtype=torch.int32)
+    K.kv_append(kn, vn, kc, vc, pos)
+    torch.cuda.synchronize()
+    assert torch.equal(kc[0, :, 3], kn[0])
+    assert torch.equal(vc[1, :, 10], vn[1])
+    assert kc[0, :, 4].abs().sum() == 0
