@@ -83,6 +83,12 @@ class BertEncoder:
         """ids [B,S] int32 (padded), lens [B] int32 -> [B,S,H] bf16."""
         cfg, w = self.cfg, self.w
         B, S = ids.shape
+        if S % 32 != 0:
+            # PV GEMM contracts over S — pad to the MFMA K granularity;
+            # padded keys are zeroed by the valid_len softmax mask.
+            pad = 32 - S % 32
+            ids = torch.nn.functional.pad(ids, (0, pad))
+            S += pad
         H, nh, d = cfg.hidden, cfg.heads, cfg.head_dim
         pos_ids = torch.arange(S, device=ids.device, dtype=torch.int32)\
             .repeat(B)

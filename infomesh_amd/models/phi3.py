@@ -105,6 +105,12 @@ class Phi3Decoder:
         """ids [B,S] i32 (no padding: same length rows) -> logits [B,V] f32
         for the last position. Fills the KV cache positions [0, S)."""
         cfg = self.cfg
+        B, S_true = ids.shape
+        if S_true % 32 != 0:
+            # Pad to the MFMA K granularity; causal masking keeps padded
+            # (future) keys invisible to real queries, padded cache rows
+            # are overwritten by the first decode_step.
+            ids = torch.nn.functional.pad(ids, (0, 32 - S_true % 32))
         B, S = ids.shape
         assert B <= self.max_batch and S <= self.max_seq
         H, nh, nkv, d = cfg.hidden, cfg.heads, cfg.kv_heads, cfg.head_dim
@@ -145,8 +151,8 @@ class Phi3Decoder:
             gate, up = gu[:, :cfg.ffn].contiguous(), gu[:, cfg.ffn:].contiguous()
             mlp = K.gemm_nt(K.silu_mul(gate, up), self.w[p + "down.w"])
             x = K.add(x, mlp)
-        self.lens[:B] = S
-        x_last = x.view(B, S, H)[:, -1, :].contiguous()
+        self.lens[:B] = S_true
+        x_last = x.view(B, S, H)[:, S_true - 1, :].contiguous()
         h = K.rmsnorm(x_last, self.w["final_ln.g"], eps=cfg.eps)
         return K.gemm_nt(h, self.w["lm_head"], out_f32=True)
 

@@ -39,16 +39,17 @@ def _assert_close(gpu, ref, rtol=2e-2, atol=2e-2, what=""):
 
 # ------------------------------------------------------------------- GEMM
 
-@pytest.mark.parametrize("M,N,K", [
+@pytest.mark.parametrize("M,N,Kd", [
     (128, 128, 64), (64, 384, 384), (200, 1000, 96),
     (256, 1536, 384), (33, 100, 32), (512, 512, 512),
 ])
-def test_gemm_nt_shapes(M, N, K):
-    a = torch.randn(M, K, device="cuda").bfloat16()
-    b = torch.randn(N, K, device="cuda").bfloat16()
+def test_gemm_nt_shapes(M, N, Kd):
+    a = torch.randn(M, Kd, device="cuda").bfloat16()
+    b = torch.randn(N, Kd, device="cuda").bfloat16()
     out = K.gemm_nt(a, b)
     ref = R.gemm_nt(a.cpu(), b.cpu())
-    _assert_close(out, ref, rtol=3e-2, atol=K ** 0.5 * 2e-2, what=f"gemm {M}x{N}x{K}")
+    _assert_close(out, ref, rtol=3e-2, atol=Kd ** 0.5 * 2e-2,
+                  what=f"gemm {M}x{N}x{Kd}")
 
 
 def test_gemm_asymmetric_catches_transpose():
