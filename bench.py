@@ -50,6 +50,8 @@ def parse_args():
     p.add_argument("--rerank-candidates", type=int, default=100)
     p.add_argument("--rag-new-tokens", type=int, default=32)
     p.add_argument("--seed", type=int, default=0)
+    p.add_argument("--phase-timers", action="store_true",
+                   help="print per-phase time breakdown (adds syncs)")
     return p.parse_args()
 
 
@@ -157,23 +159,39 @@ def main():
             torch.randn(B, 384, generator=g), dim=-1)
             for _ in range(n_batches)]
 
+    phase_t: dict[str, float] = {}
+
+    def _mark(name: str, t0: float) -> float:
+        if args.phase_timers and use_gpu:
+            torch.cuda.synchronize()
+        t1 = time.perf_counter()
+        if args.phase_timers:
+            phase_t[name] = phase_t.get(name, 0.0) + (t1 - t0)
+        return t1
+
     def step(i: int) -> None:
         j = i % n_batches
         emb = None
+        tp = time.perf_counter()
         if rank == 0 and encoder is not None:
             emb = encoder.encode_ids(qids_all[j], qlens_all[j])
         elif rank == 0 and cpu_emb is not None:
             emb = cpu_emb[j]
+        tp = _mark("encode", tp)
         fused = plane.search_batch(
             qterms_all[j] if rank == 0 else None,
             emb, B, dim=384, n_results=max(
                 args.n_results,
                 args.rerank_candidates if args.rerank else 0),
-            use_dense=not args.bm25_only)
+            use_dense=not args.bm25_only, phase_t=phase_t
+            if args.phase_timers else None)
+        tp = _mark("search+fuse", tp)
         if rank == 0 and reranker is not None:
             run_rerank(fused)
+            tp = _mark("rerank", tp)
         if rank == 0 and summarizer is not None:
             run_rag()
+            tp = _mark("rag", tp)
 
     def sync():
         if use_gpu:
@@ -241,6 +259,10 @@ def main():
                 "setup_s": round(setup_s, 1),
             },
         }
+        if args.phase_timers:
+            result["config"]["phase_ms"] = {
+                k: round(v / (args.steps + args.warmup) * 1e3, 3)
+                for k, v in phase_t.items()}
         print(json.dumps(result))
     fabric.destroy()
 

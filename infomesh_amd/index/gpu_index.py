@@ -183,11 +183,22 @@ class GpuShard:
 
     def search(self, queries_terms: list[np.ndarray],
                query_emb: torch.Tensor | None, k: int = 100,
-               scores_buf: torch.Tensor | None = None) -> ShardHits:
+               scores_buf: torch.Tensor | None = None,
+               phase_t: dict | None = None) -> ShardHits:
         """Score all queries against this shard; returns fixed-size
         top-k with global ids (the per-peer result cap analogue,
         reference p2p/routing.py:48)."""
+        import time as _time
         from ..ops import kernels as K
+
+        def mark(name, t0):
+            if phase_t is None:
+                return t0
+            torch.cuda.synchronize()
+            t1 = _time.perf_counter()
+            phase_t[name] = phase_t.get(name, 0.0) + (t1 - t0)
+            return t1
+
         B = len(queries_terms)
         dev = self.device
         N = self.n_docs
@@ -195,6 +206,7 @@ class GpuShard:
         k = min(k, N)
         topk = self._get_topk()
 
+        tp = _time.perf_counter()
         # --- BM25 plane ---
         if scores_buf is not None and scores_buf.shape == (B, N):
             scores = scores_buf
@@ -202,6 +214,7 @@ class GpuShard:
         else:
             scores = torch.zeros(B, N, device=dev, dtype=torch.float32)
         cq, ct, co, ci = self.bm25_chunks(queries_terms)
+        tp = mark("shard.chunks", tp)
         if cq:
             K.bm25_score(
                 self.offsets, self.doc_ids, self.tfs, self.doc_norm,
@@ -210,7 +223,9 @@ class GpuShard:
                 torch.tensor(co, dtype=torch.int64, device=dev),
                 torch.tensor(ci, dtype=torch.float32, device=dev),
                 scores, k1=BM25_K1)
+        tp = mark("shard.bm25", tp)
         bm_vals, bm_idx = topk(scores, k)
+        tp = mark("shard.bm25topk", tp)
 
         # --- dense plane ---
         if query_emb is not None and self.embeddings is not None:
@@ -219,7 +234,9 @@ class GpuShard:
             d_scores = K.gemm_nt(query_emb.bfloat16(), self.embeddings,
                                  out_f32=True,
                                  out=scores.unsqueeze(0)).squeeze(0)
+            tp = mark("shard.dense", tp)
             dn_vals, dn_idx = topk(d_scores, k)
+            tp = mark("shard.densetopk", tp)
         else:
             dn_vals = torch.full((B, k), -float("inf"), device=dev)
             dn_idx = torch.full((B, k), -1, device=dev, dtype=torch.int32)
@@ -244,7 +261,7 @@ class CpuShard(GpuShard):
         super().__init__(device="cpu", vocab=vocab)
 
     def search(self, queries_terms, query_emb, k: int = 100,
-               scores_buf=None) -> ShardHits:
+               scores_buf=None, phase_t=None) -> ShardHits:
         B = len(queries_terms)
         N = self.n_docs
         assert N > 0, "shard is empty"

@@ -124,18 +124,34 @@ class DistributedQueryPlane:
     def search_batch(self, queries_terms: list[np.ndarray] | None,
                      query_emb: torch.Tensor | None,
                      B: int, dim: int = 384, n_results: int = 10,
-                     use_dense: bool = True) -> FusedHits | None:
+                     use_dense: bool = True,
+                     phase_t: dict | None = None) -> FusedHits | None:
         """Collective search. Rank 0 passes real queries and gets the
         FusedHits; other ranks pass None and get None."""
+        import time as _time
+
+        def mark(name, t0):
+            if phase_t is None:
+                return t0
+            if torch.cuda.is_available():
+                torch.cuda.synchronize()
+            t1 = _time.perf_counter()
+            phase_t[name] = phase_t.get(name, 0.0) + (t1 - t0)
+            return t1
+
+        tp = _time.perf_counter()
         terms, emb = self._pack_queries(queries_terms, query_emb, B, dim)
+        tp = mark("plane.pack", tp)
         hits: ShardHits = self.shard.search(
             terms, emb if use_dense else None, k=self.k,
-            scores_buf=self._get_scores_buf(B))
+            scores_buf=self._get_scores_buf(B), phase_t=phase_t)
+        tp = mark("plane.shard", tp)
         # all-gather fixed [B,k] blocks -> [W, B, k]
         bm_s = self.fabric.all_gather(hits.bm25_scores)
         bm_i = self.fabric.all_gather(hits.bm25_ids)
         dn_s = self.fabric.all_gather(hits.dense_scores)
         dn_i = self.fabric.all_gather(hits.dense_ids)
+        tp = mark("plane.gather", tp)
         if self.fabric.rank != 0:
             return None
         W, _, k = bm_s.shape
@@ -150,6 +166,7 @@ class DistributedQueryPlane:
             order = torch.argsort(bm_s, dim=1, descending=True)
             ids = torch.gather(bm_i, 1, order)[:, :n_results]
             scores = torch.gather(bm_s, 1, order)[:, :n_results]
+        mark("plane.fuse", tp)
         return FusedHits(ids=ids, scores=scores, bm25_ids=bm_i,
                          bm25_scores=bm_s, dense_ids=dn_i,
                          dense_scores=dn_s)
