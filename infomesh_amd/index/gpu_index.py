@@ -162,23 +162,47 @@ class GpuShard:
             return 0.0
         return math.log(1.0 + (self.n_docs - df + 0.5) / (df + 0.5))
 
+    def _host_tables(self):
+        """Cached host copies of offsets/df + idf table for chunking."""
+        if not hasattr(self, "_h_offs") or self._h_offs is None:
+            self._h_offs = self.offsets.cpu().numpy()
+            df = self.df.astype(np.float64)
+            with np.errstate(divide="ignore"):
+                idf = np.log(1.0 + (self.n_docs - df + 0.5) / (df + 0.5))
+            self._h_idf = np.where(df > 0, idf, 0.0).astype(np.float32)
+        return self._h_offs, self._h_idf
+
     def bm25_chunks(self, queries_terms: list[np.ndarray],
                     chunk_size: int = 2048):
-        """Host-side work chunking: (qrow, term, offset, idf) arrays."""
-        offs = self.offsets.cpu().numpy() if self.offsets is not None else None
-        cq, ct, co, ci = [], [], [], []
-        for qi, terms in enumerate(queries_terms):
-            for t in np.unique(terms):
-                t = int(t)
-                begin, end = int(offs[t]), int(offs[t + 1])
-                if begin == end:
-                    continue
-                idf = self._idf(t)
-                for off in range(begin, end, chunk_size):
-                    cq.append(qi)
-                    ct.append(t)
-                    co.append(off)
-                    ci.append(idf)
+        """Vectorized host-side work chunking into (qrow, term, offset,
+        idf) arrays — one work chunk per <=chunk_size posting slice."""
+        offs, idf_t = self._host_tables()
+        qrows = np.concatenate([
+            np.full(len(np.unique(t)), qi, dtype=np.int64)
+            for qi, t in enumerate(queries_terms)]) if queries_terms else \
+            np.zeros(0, np.int64)
+        terms = np.concatenate([np.unique(t) for t in queries_terms]) \
+            if queries_terms else np.zeros(0, np.int64)
+        begins = offs[terms]
+        ends = offs[terms + 1]
+        nchunks = np.maximum((ends - begins + chunk_size - 1) // chunk_size, 0)
+        keep = nchunks > 0
+        qrows, terms, begins, nchunks = (qrows[keep], terms[keep],
+                                         begins[keep], nchunks[keep])
+        if len(terms) == 0:
+            return (np.zeros(0, np.int32), np.zeros(0, np.int32),
+                    np.zeros(0, np.int64), np.zeros(0, np.float32))
+        reps = nchunks.astype(np.int64)
+        cq = np.repeat(qrows, reps).astype(np.int32)
+        ct = np.repeat(terms, reps).astype(np.int32)
+        base = np.repeat(begins, reps)
+        # intra-term chunk index: arange within each repeated group
+        total = int(reps.sum())
+        grp_end = np.cumsum(reps)
+        grp_start = grp_end - reps
+        intra = np.arange(total, dtype=np.int64) - np.repeat(grp_start, reps)
+        co = base + intra * chunk_size
+        ci = idf_t[ct]
         return cq, ct, co, ci
 
     def search(self, queries_terms: list[np.ndarray],
@@ -215,13 +239,13 @@ class GpuShard:
             scores = torch.zeros(B, N, device=dev, dtype=torch.float32)
         cq, ct, co, ci = self.bm25_chunks(queries_terms)
         tp = mark("shard.chunks", tp)
-        if cq:
+        if len(cq):
             K.bm25_score(
                 self.offsets, self.doc_ids, self.tfs, self.doc_norm,
-                torch.tensor(cq, dtype=torch.int32, device=dev),
-                torch.tensor(ct, dtype=torch.int32, device=dev),
-                torch.tensor(co, dtype=torch.int64, device=dev),
-                torch.tensor(ci, dtype=torch.float32, device=dev),
+                torch.from_numpy(np.ascontiguousarray(cq)).to(dev, non_blocking=True),
+                torch.from_numpy(np.ascontiguousarray(ct)).to(dev, non_blocking=True),
+                torch.from_numpy(np.ascontiguousarray(co)).to(dev, non_blocking=True),
+                torch.from_numpy(np.ascontiguousarray(ci)).to(dev, non_blocking=True),
                 scores, k1=BM25_K1)
         tp = mark("shard.bm25", tp)
         bm_vals, bm_idx = topk(scores, k)

@@ -60,11 +60,15 @@ def build_synth_shard(n_docs: int, shard_rank: int = 0, world: int = 1,
 
 def synth_queries(n_queries: int, n_terms: int = 4, dim: int = 384,
                   vocab: int = BM25_VOCAB, seed: int = 1,
-                  device: str = "cuda", zipf_a: float = 1.15):
-    """Random queries: term-id arrays (Zipf, slightly flatter than docs)
-    + unit embeddings. Returns (list[np.ndarray], tensor [B, dim] f32)."""
+                  device: str = "cuda", zipf_a: float = 1.15,
+                  skip_head: int = 50):
+    """Random queries: term-id arrays (Zipf, slightly flatter than docs,
+    skipping the `skip_head` most common terms — the stop-word-removal
+    analogue of search/nlp.remove_stop_words) + unit embeddings.
+    Returns (list[np.ndarray], tensor [B, dim] f32)."""
     rng = np.random.default_rng(seed)
-    terms = [((rng.zipf(zipf_a, size=n_terms) - 1) % vocab).astype(np.int64)
+    terms = [(skip_head + (rng.zipf(zipf_a, size=n_terms) - 1)
+              % (vocab - skip_head)).astype(np.int64)
              for _ in range(n_queries)]
     g = torch.Generator(device=device).manual_seed(seed)
     emb = torch.nn.functional.normalize(

@@ -24,17 +24,27 @@ class EmbeddingEncoder:
     """CLS-pooled, L2-normalized text embeddings on MI355X kernels."""
 
     def __init__(self, device: str = "cuda", seed: int = 1234,
-                 cfg: BertConfig = BGE_SMALL, max_len: int = 128):
+                 cfg: BertConfig = BGE_SMALL, max_len: int = 128,
+                 use_graph: bool = True):
         self.cfg = cfg
         self.device = device
         self.max_len = max_len
         self.tokenizer = HashTokenizer(cfg.vocab_size)
         self.bert = BertEncoder(cfg, init_bert_weights(cfg, seed, device))
+        from ..ops.graphs import GraphedCallable
+        self._graphed = GraphedCallable(self._encode_impl, enabled=use_graph)
 
-    def encode_ids(self, ids: torch.Tensor, lens: torch.Tensor) -> torch.Tensor:
-        """[B,S] i32 -> [B, H] f32 L2-normalized embeddings."""
+    def _encode_impl(self, ids: torch.Tensor,
+                     lens: torch.Tensor) -> torch.Tensor:
         hidden = self.bert.forward(ids, lens)
         return K.pool(hidden, lens, mode="cls", l2=True)
+
+    def encode_ids(self, ids: torch.Tensor, lens: torch.Tensor) -> torch.Tensor:
+        """[B,S] i32 -> [B, H] f32 L2-normalized embeddings.
+
+        Fixed-shape batches replay a captured hipGraph (launch-bound:
+        ~150 kernel launches per forward otherwise)."""
+        return self._graphed(ids, lens)
 
     def encode_texts(self, texts: list[str]) -> torch.Tensor:
         ids_l, lens = self.tokenizer.encode_batch(texts, self.max_len)
