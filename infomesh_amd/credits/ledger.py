@@ -1,0 +1,190 @@
+"""Credit ledger: C = Σ W·Q·M over contribution actions.
+
+Reference parity: infomesh/credits/ledger.py + credits/types.py —
+action weights (crawl 1.0/page, query 0.5, hosting 0.1/hr, uptime
+0.5/hr, LLM 1.5–2.0), off-peak 1.5× multiplier for LLM actions, SQLite
+entries hash-chained + Ed25519-signed, tiers 1–3 with search costs
+0.100/0.050/0.033, 72 h grace then 2× debt mode.
+"""
+from __future__ import annotations
+
+import enum
+import json
+import time
+from dataclasses import dataclass
+from pathlib import Path
+
+from ..db import SQLiteStore
+from ..hashing import content_hash
+from ..trust.keys import KeyPair
+
+
+class Action(enum.Enum):
+    CRAWL = "crawl"
+    QUERY_SERVED = "query_served"
+    HOSTING = "hosting"
+    UPTIME = "uptime"
+    LLM_SUMMARIZE = "llm_summarize"
+    LLM_RERANK = "llm_rerank"
+    SEARCH_SPEND = "search_spend"
+
+
+ACTION_WEIGHTS: dict[Action, float] = {
+    Action.CRAWL: 1.0,            # per page
+    Action.QUERY_SERVED: 0.5,     # per served query
+    Action.HOSTING: 0.1,          # per hour
+    Action.UPTIME: 0.5,           # per hour
+    Action.LLM_SUMMARIZE: 2.0,    # per summary
+    Action.LLM_RERANK: 1.5,       # per rerank batch
+    Action.SEARCH_SPEND: -1.0,    # quantity = cost
+}
+
+LLM_ACTIONS = frozenset({Action.LLM_SUMMARIZE, Action.LLM_RERANK})
+OFF_PEAK_MULTIPLIER = 1.5
+
+TIER_THRESHOLDS = (0.0, 100.0, 1000.0)          # tier 1 / 2 / 3 balances
+TIER_SEARCH_COST = (0.100, 0.050, 0.033)
+GRACE_HOURS = 72.0
+DEBT_MULTIPLIER = 2.0
+
+
+@dataclass
+class CreditEntry:
+    action: str
+    quantity: float
+    multiplier: float
+    credits: float
+    ts: float
+    entry_hash: str
+    prev_hash: str
+    signature: str = ""
+
+
+class CreditLedger(SQLiteStore):
+    SCHEMA = """
+    CREATE TABLE IF NOT EXISTS credit_entries (
+        id INTEGER PRIMARY KEY,
+        action TEXT NOT NULL,
+        quantity REAL NOT NULL,
+        multiplier REAL NOT NULL,
+        credits REAL NOT NULL,
+        ts REAL NOT NULL,
+        entry_hash TEXT NOT NULL,
+        prev_hash TEXT NOT NULL,
+        signature TEXT NOT NULL DEFAULT ''
+    );
+    """
+
+    def __init__(self, path: str | Path = ":memory:",
+                 kp: KeyPair | None = None,
+                 off_peak_fn=None):
+        super().__init__(path)
+        self.kp = kp
+        self._off_peak_fn = off_peak_fn
+        self._debt_since: float | None = None
+
+    # ------------------------------------------------------------ record
+    def _last_hash(self) -> str:
+        row = self.execute(
+            "SELECT entry_hash FROM credit_entries ORDER BY id DESC LIMIT 1"
+        ).fetchone()
+        return row["entry_hash"] if row else "genesis"
+
+    def record_action(self, action: Action, quantity: float = 1.0,
+                      ts: float | None = None) -> CreditEntry:
+        ts = float(ts if ts is not None else time.time())
+        quantity = float(quantity)
+        mult = 1.0
+        if action in LLM_ACTIONS and self._is_off_peak(ts):
+            mult = OFF_PEAK_MULTIPLIER
+        credits = ACTION_WEIGHTS[action] * quantity * mult
+        prev = self._last_hash()
+        payload = json.dumps(
+            {"action": action.value, "q": quantity, "m": mult,
+             "c": credits, "ts": ts, "prev": prev}, sort_keys=True)
+        entry_hash = content_hash(payload)
+        sig = self.kp.sign(payload.encode()).hex() if self.kp else ""
+        self.execute(
+            "INSERT INTO credit_entries (action, quantity, multiplier,"
+            " credits, ts, entry_hash, prev_hash, signature)"
+            " VALUES (?,?,?,?,?,?,?,?)",
+            (action.value, quantity, mult, credits, ts, entry_hash, prev,
+             sig))
+        self.commit()
+        return CreditEntry(action.value, quantity, mult, credits, ts,
+                           entry_hash, prev, sig)
+
+    def _is_off_peak(self, ts: float) -> bool:
+        if self._off_peak_fn is not None:
+            return bool(self._off_peak_fn(ts))
+        from .scheduling import is_off_peak
+        return is_off_peak(ts)
+
+    # ----------------------------------------------------------- balance
+    def balance(self) -> float:
+        row = self.execute(
+            "SELECT COALESCE(SUM(credits), 0) AS b FROM credit_entries"
+        ).fetchone()
+        return float(row["b"])
+
+    def tier(self) -> int:
+        b = self.balance()
+        t = 1
+        for i, thr in enumerate(TIER_THRESHOLDS, start=1):
+            if b >= thr:
+                t = i
+        return t
+
+    def search_cost(self) -> float:
+        cost = TIER_SEARCH_COST[self.tier() - 1]
+        if self.in_debt_mode():
+            cost *= DEBT_MULTIPLIER
+        return cost
+
+    def in_debt_mode(self, now: float | None = None) -> bool:
+        """Negative balance beyond the 72 h grace window ⇒ 2× costs."""
+        now = now or time.time()
+        if self.balance() >= 0:
+            self._debt_since = None
+            return False
+        if self._debt_since is None:
+            self._debt_since = now
+            return False
+        return (now - self._debt_since) > GRACE_HOURS * 3600.0
+
+    def deduct_search_cost(self) -> float:
+        cost = self.search_cost()
+        self.record_action(Action.SEARCH_SPEND, cost)
+        return cost
+
+    # ------------------------------------------------------ verification
+    def verify_chain(self) -> bool:
+        """Recompute the hash chain (and signatures when present)."""
+        prev = "genesis"
+        for row in self.execute(
+                "SELECT * FROM credit_entries ORDER BY id").fetchall():
+            payload = json.dumps(
+                {"action": row["action"], "q": row["quantity"],
+                 "m": row["multiplier"], "c": row["credits"],
+                 "ts": row["ts"], "prev": prev}, sort_keys=True)
+            if content_hash(payload) != row["entry_hash"]:
+                return False
+            if row["signature"] and self.kp is not None:
+                if not KeyPair.verify(self.kp.public, payload.encode(),
+                                      bytes.fromhex(row["signature"])):
+                    return False
+            prev = row["entry_hash"]
+        return True
+
+    def entries(self, limit: int = 100) -> list[dict]:
+        return [dict(r) for r in self.execute(
+            "SELECT * FROM credit_entries ORDER BY id DESC LIMIT ?",
+            (limit,)).fetchall()]
+
+    def stats(self) -> dict:
+        by_action = {r["action"]: r["total"] for r in self.execute(
+            "SELECT action, SUM(credits) AS total FROM credit_entries"
+            " GROUP BY action")}
+        return {"balance": self.balance(), "tier": self.tier(),
+                "search_cost": self.search_cost(), "by_action": by_action,
+                "debt_mode": self.in_debt_mode()}
