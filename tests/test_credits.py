@@ -108,8 +108,8 @@ def test_energy_scheduler_defers():
 def test_farming_detector():
     d = FarmingDetector()
     t0 = 1_000_000.0
-    for i in range(40):
-        d.record("llm_summarize", ts=t0 + i * 0.01)  # 40 in 0.4s = burst
+    for i in range(150):
+        d.record("llm_summarize", ts=t0 + i * 0.01)  # 150 in 1.5 s = burst
     assert d.on_probation(now=t0 + 1)
     assert d.multiplier(now=t0 + 1) == 0.0
     assert d.multiplier(now=t0 + 25 * 3600) == 1.0
