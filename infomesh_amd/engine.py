@@ -1,0 +1,142 @@
+"""HybridEngine: the GPU-side search engine for REAL documents.
+
+Bridges the durable LocalStore ground truth to GPU shards: documents are
+tokenized (BM25 term ids) + encoded (bge-small MFMA encoder), staged in
+pending buffers, and flipped into the scored index on flush() — the GPU
+analogue of FTS5's WAL+optimize cycle (SURVEY.md §7 "hard parts").
+
+On CPU-only machines the same engine runs with CpuShard + no encoder
+(BM25-only or externally-supplied embeddings), so the whole orchestration
+is testable without a GPU and the GPU path is identical code.
+"""
+from __future__ import annotations
+
+import logging
+
+import numpy as np
+import torch
+
+from .index.gpu_index import CpuShard, GpuShard, bm25_term_ids
+from .index.local_store import Document, LocalStore, SearchHit
+from .parallel.fabric import Fabric
+from .parallel.query_plane import DistributedQueryPlane
+
+log = logging.getLogger("infomesh.engine")
+
+
+class HybridEngine:
+    def __init__(self, device: str | None = None, k_per_shard: int = 100,
+                 use_encoder: bool = True, encoder_max_len: int = 128,
+                 fabric: Fabric | None = None):
+        self.gpu = torch.cuda.is_available() if device is None \
+            else device.startswith("cuda")
+        self.device = device or ("cuda" if self.gpu else "cpu")
+        self.fabric = fabric or Fabric()
+        self.shard: GpuShard = (GpuShard(self.device) if self.gpu
+                                else CpuShard())
+        self.plane = DistributedQueryPlane(self.shard, self.fabric,
+                                           k_per_shard=k_per_shard)
+        self.encoder = None
+        if self.gpu and use_encoder:
+            from .models.encoder import EmbeddingEncoder
+            self.encoder = EmbeddingEncoder(device=self.device,
+                                            max_len=encoder_max_len)
+        # pending (not yet searchable) docs
+        self._pending_tokens: list[np.ndarray] = []
+        self._pending_texts: list[str] = []
+        self._pending_ids: list[int] = []
+        self._built_tokens: list[np.ndarray] = []
+        self._built_texts: list[str] = []
+        self._built_ids: list[int] = []
+
+    # ------------------------------------------------------------ ingest
+    def add_document(self, doc: Document) -> None:
+        assert doc.doc_id is not None
+        text = f"{doc.title}\n{doc.text}"[:4000]
+        self._pending_tokens.append(bm25_term_ids(text))
+        self._pending_texts.append(text[:2000])   # embed truncation
+        self._pending_ids.append(doc.doc_id)
+
+    @property
+    def pending_count(self) -> int:
+        return len(self._pending_ids)
+
+    @property
+    def doc_count(self) -> int:
+        return self.shard.n_docs + self.pending_count
+
+    def flush(self, embed_batch: int = 256) -> int:
+        """Make pending docs searchable (epoch flip). Rebuilds the shard
+        from all accumulated docs — segment-merge analogue; incremental
+        segment builds are a planned optimization."""
+        if not self._pending_ids:
+            return 0
+        self._built_tokens.extend(self._pending_tokens)
+        self._built_texts.extend(self._pending_texts)
+        self._built_ids.extend(self._pending_ids)
+        n_new = len(self._pending_ids)
+        self._pending_tokens, self._pending_texts, self._pending_ids = \
+            [], [], []
+
+        emb = None
+        if self.encoder is not None:
+            chunks = []
+            for i in range(0, len(self._built_texts), embed_batch):
+                chunks.append(self.encoder.encode_texts(
+                    self._built_texts[i:i + embed_batch]).bfloat16())
+            emb = torch.cat(chunks, 0)
+
+        lens = np.array([max(len(t), 1) for t in self._built_tokens],
+                        dtype=np.int64)
+        flat_terms = (np.concatenate(self._built_tokens)
+                      if any(len(t) for t in self._built_tokens)
+                      else np.zeros(0, np.int64))
+        flat_docs = np.repeat(np.arange(len(self._built_tokens),
+                                        dtype=np.int64),
+                              [len(t) for t in self._built_tokens])
+        shard = GpuShard(self.device) if self.gpu else CpuShard()
+        shard.build_from_arrays(
+            flat_terms, flat_docs, lens,
+            np.asarray(self._built_ids, dtype=np.int64), emb)
+        self.shard = shard
+        self.plane.shard = shard
+        log.info("engine flush: %d new docs, %d total, %.1f MB HBM",
+                 n_new, shard.n_docs, shard.hbm_bytes() / 1e6)
+        return n_new
+
+    # ------------------------------------------------------------ search
+    def search(self, query: str, limit: int = 10,
+               use_dense: bool | None = None) -> list[SearchHit]:
+        """Single-query search against the engine (shard-fused)."""
+        if self.shard.n_docs == 0:
+            return []
+        terms = [bm25_term_ids(query)]
+        emb = None
+        if use_dense is None:
+            use_dense = self.encoder is not None
+        if use_dense and self.encoder is not None:
+            emb = self.encoder.encode_texts([query])
+        fused = self.plane.search_batch(
+            terms, emb, B=1, dim=emb.shape[1] if emb is not None else 384,
+            n_results=limit, use_dense=use_dense and emb is not None)
+        if fused is None:
+            return []
+        hits: list[SearchHit] = []
+        for gid, score in zip(fused.ids[0].tolist(),
+                              fused.scores[0].tolist()):
+            if gid < 0:
+                continue
+            hits.append(SearchHit(doc_id=int(gid), url="", title="",
+                                  snippet="", bm25=0.0, score=float(score),
+                                  source="gpu-hybrid"))
+        return hits
+
+    def stats(self) -> dict:
+        return {
+            "device": self.device,
+            "docs_indexed": self.shard.n_docs,
+            "docs_pending": self.pending_count,
+            "hbm_bytes": self.shard.hbm_bytes(),
+            "world_size": self.fabric.world,
+            "encoder": self.encoder is not None,
+        }

@@ -1,0 +1,271 @@
+"""Service orchestration: AppContext wiring + the crawl->index->publish
+and search entry paths.
+
+Reference parity: infomesh/services.py (AppContext role-conditional
+construction, index_document single source of truth, crawl_and_index,
+fetch_page cache-first, create_local_search_fn). The network-publish
+path becomes a GPU-engine ingest (SURVEY.md §2.9 last row).
+"""
+from __future__ import annotations
+
+import logging
+import time
+from dataclasses import dataclass, field
+from pathlib import Path
+from typing import Any
+
+from .config import Config, load_config
+from .credits.farming import FarmingDetector
+from .credits.ledger import Action, CreditLedger
+from .crawler.dedup import DeduplicatorDB
+from .crawler.robots import RobotsChecker
+from .crawler.rss import FeedMonitor
+from .crawler.scheduler import Scheduler
+from .crawler.worker import CrawlResult, CrawlWorker
+from .engine import HybridEngine
+from .errors import InfoMeshError
+from .index.link_graph import LinkGraph
+from .index.local_store import Document, LocalStore
+from .search.cache import QueryCache
+from .search.nlp import RelatedSearchTracker
+from .search.query import (SearchResponse, search_distributed, search_hybrid,
+                           search_local)
+from .trust.attestation import create_attestation
+from .trust.dmca import TakedownManager
+from .trust.gdpr import DeletionManager
+from .trust.keys import KeyPair, ensure_keys
+from .trust.scoring import TrustStore
+
+log = logging.getLogger("infomesh.services")
+
+
+@dataclass
+class AppContext:
+    """All wired subsystems for one node process.
+
+    Role-conditional (reference services.py:485-567): 'crawler' skips
+    the query cache + engine; 'search' skips the crawl worker."""
+
+    config: Config
+    store: LocalStore
+    link_graph: LinkGraph
+    dedup: DeduplicatorDB
+    keys: KeyPair
+    ledger: CreditLedger
+    trust: TrustStore
+    takedowns: TakedownManager
+    deletions: DeletionManager
+    cache: QueryCache
+    feeds: FeedMonitor
+    related: RelatedSearchTracker
+    worker: CrawlWorker | None = None
+    engine: HybridEngine | None = None
+    attestations: list = field(default_factory=list)
+    farming: FarmingDetector = field(default_factory=FarmingDetector)
+    started_at: float = field(default_factory=time.time)
+
+    # ------------------------------------------------------------ build
+    @classmethod
+    def create(cls, config: Config | None = None,
+               with_engine: bool | None = None,
+               with_worker: bool | None = None,
+               in_memory: bool = False) -> "AppContext":
+        cfg = config or load_config()
+        role = cfg.node.role
+        data = cfg.data_dir
+        if not in_memory:
+            data.mkdir(parents=True, exist_ok=True)
+
+        def p(name: str):
+            return ":memory:" if in_memory else data / name
+
+        store = LocalStore(p("index.db"), tokenizer=cfg.index.fts_tokenizer)
+        keys = KeyPair.generate() if in_memory else ensure_keys(data)
+        ledger = CreditLedger(p("ledger.db"), kp=keys)
+        trust = TrustStore(p("trust.db"))
+        ctx = cls(
+            config=cfg,
+            store=store,
+            link_graph=LinkGraph(p("links.db")),
+            dedup=DeduplicatorDB(p("dedup.db")),
+            keys=keys,
+            ledger=ledger,
+            trust=trust,
+            takedowns=TakedownManager(store, keys, p("dmca.db")),
+            deletions=DeletionManager(store, keys, p("gdpr.db")),
+            cache=QueryCache(cfg.search.cache_entries, cfg.search.cache_ttl_s),
+            feeds=FeedMonitor(),
+            related=RelatedSearchTracker(),
+        )
+        if with_worker if with_worker is not None else role in ("full", "crawler"):
+            ctx.worker = CrawlWorker(
+                cfg.crawl,
+                scheduler=Scheduler(cfg.crawl.politeness_delay_s,
+                                    cfg.crawl.max_urls_per_hour,
+                                    cfg.crawl.max_depth),
+                dedup=ctx.dedup,
+                robots=RobotsChecker(cfg.crawl.user_agent))
+        want_engine = with_engine if with_engine is not None \
+            else role in ("full", "search")
+        if want_engine:
+            try:
+                ctx.engine = HybridEngine(k_per_shard=cfg.search.max_results_per_shard)
+            except Exception as e:
+                log.warning("engine unavailable: %s", e)
+        return ctx
+
+    # ------------------------------------------------------------ ingest
+    def index_document(self, doc: Document, attest: bool = True,
+                       credit: bool = True) -> int | None:
+        """THE single crawl->index source of truth
+        (reference: services.py:68-110)."""
+        if self.deletions.is_forgotten(doc.url):
+            raise InfoMeshError("SEC001", "url under GDPR deletion record")
+        if self.takedowns.is_blocked(doc.url):
+            raise InfoMeshError("SEC001", "url under DMCA takedown")
+        rowid = self.store.add_document(doc)
+        if rowid is None:
+            return None
+        doc.doc_id = rowid
+        if self.engine is not None:
+            self.engine.add_document(doc)
+        if attest:
+            self.attestations.append(create_attestation(
+                self.keys, doc.url, doc.raw_hash, doc.text_hash))
+            if len(self.attestations) > 10_000:
+                del self.attestations[:5000]
+        if credit:
+            if self.farming.multiplier() > 0:
+                self.ledger.record_action(Action.CRAWL, 1.0)
+            self.farming.record("crawl")
+        self.cache.invalidate()
+        return rowid
+
+    async def crawl_and_index(self, url: str, depth: int = 0,
+                              force: bool = False) -> dict[str, Any]:
+        """Crawl one URL and index it (reference: services.py:354-423)."""
+        if self.worker is None:
+            raise InfoMeshError("RT001", "no crawl worker in this role")
+        doc_meta = self.store.get_document_by_url(url)
+        res: CrawlResult = await self.worker.crawl_url(
+            url, depth=depth, force=force,
+            etag=doc_meta.etag if doc_meta else "",
+            last_modified=doc_meta.last_modified if doc_meta else "")
+        out: dict[str, Any] = {"url": url, "status": res.status,
+                               "reason": res.reason}
+        if res.not_modified:
+            self.store.update_recrawl(url, changed=False)
+            return out
+        if res.status != "ok" or res.page is None:
+            return out
+        page = res.page
+        self.link_graph.add_links(url, page.links)
+        for f in res.feeds:
+            self.feeds.add(f)
+        doc = Document(url=url, title=page.title, text=page.text,
+                       language=page.language, text_hash=page.text_hash,
+                       raw_hash=page.raw_html_hash, etag=res.etag,
+                       last_modified=res.last_modified)
+        rowid = self.index_document(doc)
+        if doc_meta is not None:
+            self.store.update_recrawl(url, changed=rowid is not None,
+                                      etag=res.etag,
+                                      last_modified=res.last_modified)
+        out.update({"doc_id": rowid, "title": page.title,
+                    "links_scheduled": res.links_scheduled,
+                    "indexed": rowid is not None})
+        return out
+
+    # ------------------------------------------------------------ search
+    def search(self, query: str, limit: int | None = None,
+               mode: str = "auto", use_cache: bool = True,
+               deduct: bool = True) -> SearchResponse:
+        """Unified search entry (reference mcp/handlers.py:382 flow):
+        cache -> credits -> local/hybrid/distributed -> cache."""
+        limit = limit or self.config.search.max_results
+        key = QueryCache.make_key(query, limit=limit, mode=mode)
+        if use_cache:
+            cached = self.cache.get(key)
+            if cached is not None:
+                return cached
+        if deduct and self.config.credits.enabled:
+            self.ledger.deduct_search_cost()
+        self.related.record(query)
+        authority = self.link_graph.url_authority
+        trust_fn = self.trust.trust_fn()
+        engine_ready = (self.engine is not None
+                        and self.engine.shard.n_docs > 0)
+        if mode == "local" or (mode == "auto" and not engine_ready):
+            resp = search_local(self.store, query, limit=limit,
+                                authority_fn=authority, trust_fn=trust_fn)
+        elif mode in ("auto", "hybrid", "distributed"):
+            resp = search_hybrid(
+                self.store, _EngineDense(self, limit), query, limit=limit,
+                authority_fn=authority, trust_fn=trust_fn,
+                rrf_k=self.config.search.rrf_k)
+            if mode == "distributed":
+                resp.mode = "distributed"
+        else:
+            raise InfoMeshError("SRCH001", f"unknown mode {mode!r}")
+        if use_cache:
+            self.cache.put(key, resp)
+        self.ledger.record_action(Action.QUERY_SERVED, 1.0)
+        return resp
+
+    def fetch_page(self, url: str) -> Document | None:
+        """Cache-first page fetch (reference: services.py:220-335);
+        network fetch is the async crawl path."""
+        return self.store.get_document_by_url(url)
+
+    def flush_engine(self) -> int:
+        if self.engine is None:
+            return 0
+        return self.engine.flush()
+
+    # ------------------------------------------------------------- stats
+    def status(self) -> dict[str, Any]:
+        return {
+            "node_id": self.keys.node_id,
+            "role": self.config.node.role,
+            "uptime_s": round(time.time() - self.started_at, 1),
+            "index": self.store.stats(),
+            "engine": self.engine.stats() if self.engine else None,
+            "credits": self.ledger.stats(),
+            "cache": self.cache.stats(),
+            "crawler": self.worker.stats if self.worker else None,
+            "link_edges": self.link_graph.edge_count(),
+        }
+
+    def close(self) -> None:
+        for c in (self.store, self.link_graph, self.dedup, self.ledger,
+                  self.trust, self.takedowns, self.deletions):
+            try:
+                c.close()
+            except Exception:
+                pass
+
+
+class _EngineDense:
+    """Adapter: HybridEngine as the DenseSearcher for search_hybrid,
+    resolving engine global ids back to LocalStore documents."""
+
+    def __init__(self, ctx: AppContext, limit: int):
+        self.ctx = ctx
+        self.limit = limit
+
+    def search(self, query: str, limit: int = 10):
+        engine = self.ctx.engine
+        if engine is None or engine.shard.n_docs == 0:
+            return []
+        hits = engine.search(query, limit=limit)
+        out = []
+        for h in hits:
+            doc = self.ctx.store.get_document(h.doc_id)
+            if doc is None:
+                continue
+            h.url = doc.url
+            h.title = doc.title
+            h.domain = doc.domain
+            h.crawled_at = doc.crawled_at
+            out.append(h)
+        return out

@@ -1,0 +1,191 @@
+"""Service-layer tests: AppContext wiring, index/crawl paths, engine
+orchestration (CPU shard), runtime coordination, governor."""
+from __future__ import annotations
+
+import asyncio
+
+import httpx
+import pytest
+
+from infomesh_amd.config import Config
+from infomesh_amd.engine import HybridEngine
+from infomesh_amd.errors import InfoMeshError
+from infomesh_amd.index.local_store import Document
+from infomesh_amd.runtime import (GracefulShutdown, PidFile, RuntimeStatus,
+                                  StartupLock)
+from infomesh_amd.services import AppContext
+from infomesh_amd.utils.governor import (DegradeLevel, ResourceGovernor,
+                                         run_preflight_checks)
+
+
+@pytest.fixture
+def ctx():
+    c = AppContext.create(config=Config(), with_engine=False,
+                          with_worker=True, in_memory=True)
+    yield c
+    c.close()
+
+
+@pytest.fixture
+def ctx_engine():
+    c = AppContext.create(config=Config(), with_engine=True,
+                          with_worker=False, in_memory=True)
+    # CPU builds use the CpuShard-backed engine without encoder
+    c.engine = HybridEngine(device="cpu", use_encoder=False)
+    yield c
+    c.close()
+
+
+def _doc(i=1, text=None):
+    return Document(url=f"https://a.com/{i}", title=f"Doc {i}",
+                    text=text or f"content body number {i} about gpu kernels")
+
+
+def test_index_document_flow(ctx):
+    rid = ctx.index_document(_doc(1))
+    assert rid is not None
+    assert ctx.store.count() == 1
+    assert len(ctx.attestations) == 1
+    assert ctx.ledger.balance() > 0
+
+
+def test_index_respects_gdpr_and_dmca(ctx):
+    ctx.deletions.request_deletion("https://a.com/gone")
+    with pytest.raises(InfoMeshError):
+        ctx.index_document(Document(url="https://a.com/gone", text="x" * 60))
+    ctx.takedowns.file_notice("domain:evil.com", "c", "claimant")
+    with pytest.raises(InfoMeshError):
+        ctx.index_document(Document(url="https://evil.com/p", text="y" * 60))
+
+
+def test_search_modes_and_cache(ctx):
+    for i in range(5):
+        ctx.index_document(_doc(i, text=f"python tutorial part {i} "
+                                        "teaches functions"))
+    r1 = ctx.search("python tutorial")
+    assert r1.results
+    bal = ctx.ledger.balance()
+    r2 = ctx.search("python tutorial")   # cached -> no extra deduction
+    assert ctx.ledger.balance() == bal
+    assert r2 is r1
+
+
+def test_search_explicit_local_mode(ctx):
+    ctx.index_document(_doc(7, text="asyncio event loop coroutines"))
+    resp = ctx.search("asyncio coroutines", mode="local", use_cache=False)
+    assert resp.mode == "local"
+
+
+def test_crawl_and_index(ctx):
+    html = ("<html><head><title>T</title></head><body><p>" +
+            "Crawled content paragraph that is long enough to index. " * 3 +
+            "</p></body></html>")
+
+    def handler(request):
+        return httpx.Response(200, text=html,
+                              headers={"content-type": "text/html"})
+    ctx.worker._client = httpx.AsyncClient(
+        transport=httpx.MockTransport(handler))
+    ctx.worker._own_client = True
+    ctx.worker.resolve_dns = False
+    ctx.worker.cfg = ctx.worker.cfg.__class__(respect_robots=False)
+
+    async def run():
+        out = await ctx.crawl_and_index("https://site.com/page")
+        assert out["status"] == "ok"
+        assert out["indexed"]
+    asyncio.run(run())
+    assert ctx.store.count() == 1
+
+
+def test_engine_ingest_and_search(ctx_engine):
+    for i in range(20):
+        ctx_engine.index_document(Document(
+            url=f"https://b.com/{i}", title=f"GPU doc {i}",
+            text=f"document about rocm hip kernels iteration {i} "
+                 f"with matrix cores and lds tiling"))
+    assert ctx_engine.engine.pending_count == 20
+    n = ctx_engine.flush_engine()
+    assert n == 20
+    assert ctx_engine.engine.shard.n_docs == 20
+    resp = ctx_engine.search("rocm hip kernels", mode="hybrid",
+                             use_cache=False)
+    assert resp.mode in ("hybrid",)
+    assert resp.results
+    hits = ctx_engine.engine.search("matrix cores lds", limit=5)
+    assert hits and all(h.doc_id >= 1 for h in hits)
+
+
+def test_status_surface(ctx):
+    st = ctx.status()
+    assert st["node_id"]
+    assert st["index"]["documents"] == 0
+    assert "credits" in st and "cache" in st
+
+
+# ------------------------------------------------------------- runtime
+
+def test_pidfile(tmp_path):
+    pf = PidFile(tmp_path, marker="python")
+    pf.acquire()
+    assert pf.read_running_pid() is not None
+    pf.acquire()  # same-process re-acquire is fine
+    pf.release()
+    assert pf.read_running_pid() is None
+    # stale pid (dead process) is cleaned up on read
+    pf.path.write_text("999999")
+    assert pf.read_running_pid() is None
+    pf.acquire()
+    pf.release()
+    # a live UNRELATED process (pid 1) fails the cmdline marker check
+    pf.path.write_text("1")
+    assert pf.read_running_pid() is None
+
+
+def test_startup_lock(tmp_path):
+    with StartupLock(tmp_path):
+        l2 = StartupLock(tmp_path)
+        assert not l2.acquire()
+    l3 = StartupLock(tmp_path)
+    assert l3.acquire()
+    l3.release()
+
+
+def test_runtime_status_heartbeat(tmp_path):
+    rs = RuntimeStatus(tmp_path)
+    rs.write("running", docs=5)
+    data = rs.read()
+    assert data["state"] == "running" and data["docs"] == 5
+    import json, time
+    stale = json.loads(rs.path.read_text())
+    stale["ts"] = time.time() - 100
+    rs.path.write_text(json.dumps(stale))
+    assert rs.read()["state"] == "stopped"
+
+
+def test_graceful_shutdown_callbacks():
+    gs = GracefulShutdown()
+    hit = []
+    gs.on_shutdown(lambda: hit.append(1))
+    gs._handler(15, None)
+    assert gs.requested and hit == [1]
+
+
+# ------------------------------------------------------------- governor
+
+def test_governor_levels():
+    g = ResourceGovernor(max_rss_gb=10_000, min_mem_available_gb=0.001,
+                         max_load_per_cpu=10_000)
+    g.sample(force=True)
+    assert g.level == DegradeLevel.NORMAL
+    assert g.crawl_allowed() and g.writes_allowed()
+    g2 = ResourceGovernor(max_load_per_cpu=-1.0, max_rss_gb=10_000,
+                          min_mem_available_gb=0.001)
+    g2.sample(force=True)
+    assert g2.level == DegradeLevel.THROTTLE_CRAWL
+    assert g2.throttle_factor() == 3.0
+
+
+def test_preflight(tmp_path):
+    problems = run_preflight_checks(tmp_path)
+    assert problems == [] or all("disk" not in p for p in problems)
