@@ -1,0 +1,346 @@
+"""CLI entry point (click group).
+
+Reference parity: infomesh/cli/ (start/stop/_serve/status, crawl, mcp,
+search/feedback, index export/import/stats, config, keys, peer->shard,
+doctor, bench). Console entry: `python -m infomesh_amd` or the
+`infomesh-amd` script from setup.py.
+"""
+from __future__ import annotations
+
+import asyncio
+import dataclasses
+import json
+import os
+import signal
+import subprocess
+import sys
+import time
+from pathlib import Path
+
+import click
+
+from ..config import load_config, save_config
+from ..runtime import GracefulShutdown, PidFile, RuntimeStatus, StartupLock
+
+
+@click.group()
+def cli():
+    """infomesh-amd — MI355X-native hybrid search/RAG engine."""
+
+
+def _ctx(with_engine: bool | None = None, with_worker: bool | None = None):
+    from ..services import AppContext
+    return AppContext.create(with_engine=with_engine,
+                             with_worker=with_worker)
+
+
+# ----------------------------------------------------------------- serve
+
+@cli.command()
+@click.option("--foreground", is_flag=True, help="run in this process")
+@click.option("--seed-category", default="quickstart")
+def start(foreground: bool, seed_category: str):
+    """Start the node (crawl loop + heartbeat)."""
+    cfg = load_config()
+    data = cfg.data_dir
+    data.mkdir(parents=True, exist_ok=True)
+    if foreground:
+        _serve_impl(seed_category)
+        return
+    pid = PidFile(data).read_running_pid()
+    if pid:
+        click.echo(f"already running (pid {pid})")
+        return
+    proc = subprocess.Popen(
+        [sys.executable, "-m", "infomesh_amd", "_serve",
+         "--seed-category", seed_category],
+        stdout=subprocess.DEVNULL, stderr=open(data / "serve.log", "ab"),
+        start_new_session=True)
+    click.echo(f"started (pid {proc.pid}); logs: {data / 'serve.log'}")
+
+
+@cli.command(name="_serve", hidden=True)
+@click.option("--seed-category", default="quickstart")
+def _serve(seed_category: str):
+    _serve_impl(seed_category)
+
+
+def _serve_impl(seed_category: str):
+    from ..crawler.crawl_loop import seed_and_crawl_loop
+    cfg = load_config()
+    data = cfg.data_dir
+    lock = StartupLock(data)
+    if not lock.acquire():
+        click.echo("another instance is starting", err=True)
+        sys.exit(1)
+    pidfile = PidFile(data)
+    pidfile.acquire()
+    status = RuntimeStatus(data)
+    shutdown = GracefulShutdown()
+    shutdown.install()
+    ctx = _ctx()
+    try:
+        async def run():
+            async def heartbeat():
+                while not shutdown.requested:
+                    status.write("running", **{
+                        "docs": ctx.store.count(),
+                        "engine_docs": ctx.engine.shard.n_docs
+                        if ctx.engine else 0})
+                    await asyncio.sleep(10)
+            hb = asyncio.ensure_future(heartbeat())
+            await seed_and_crawl_loop(
+                ctx, seed_category=seed_category,
+                stop_check=lambda: shutdown.requested)
+            hb.cancel()
+        asyncio.run(run())
+    finally:
+        status.write("stopped")
+        pidfile.release()
+        lock.release()
+        ctx.close()
+
+
+@cli.command()
+def stop():
+    """Stop a running node (SIGTERM)."""
+    cfg = load_config()
+    pid = PidFile(cfg.data_dir).read_running_pid()
+    if not pid:
+        click.echo("not running")
+        return
+    os.kill(pid, signal.SIGTERM)
+    for _ in range(50):
+        if PidFile(cfg.data_dir).read_running_pid() is None:
+            click.echo("stopped")
+            return
+        time.sleep(0.2)
+    click.echo("still stopping…")
+
+
+@cli.command()
+def status():
+    """Show node status."""
+    cfg = load_config()
+    rs = RuntimeStatus(cfg.data_dir).read()
+    click.echo(json.dumps(rs, indent=2))
+
+
+# ---------------------------------------------------------------- search
+
+@cli.command()
+@click.argument("query", nargs=-1, required=True)
+@click.option("--limit", default=10)
+@click.option("--mode", default="auto")
+@click.option("--json", "as_json", is_flag=True)
+@click.option("--explain", is_flag=True)
+def search(query, limit, mode, as_json, explain):
+    """Search the local index."""
+    from ..search.explain import explain_search, render_explanation
+    from ..search.formatter import format_json, format_text
+    q = " ".join(query)
+    ctx = _ctx(with_worker=False)
+    try:
+        if explain:
+            click.echo(render_explanation(explain_search(ctx.store, q, limit)))
+            return
+        resp = ctx.search(q, limit=limit, mode=mode)
+        click.echo(format_json(resp) if as_json else format_text(resp))
+    finally:
+        ctx.close()
+
+
+@cli.command()
+@click.argument("url")
+@click.option("--signal", "signal_", default="fetch",
+              type=click.Choice(["fetch", "cite", "click", "skip"]))
+def feedback(url, signal_):
+    """Record implicit feedback for a URL."""
+    from ..search.feedback import FeedbackStore
+    cfg = load_config()
+    fs = FeedbackStore(cfg.data_dir / "feedback.db")
+    fs.record(url, signal_)
+    fs.close()
+    click.echo("recorded")
+
+
+# ----------------------------------------------------------------- crawl
+
+@cli.command()
+@click.argument("url")
+@click.option("--force", is_flag=True)
+def crawl(url, force):
+    """Crawl and index one URL."""
+    ctx = _ctx(with_engine=False)
+    try:
+        out = asyncio.run(ctx.crawl_and_index(url, force=force))
+        click.echo(json.dumps(out, indent=2, default=str))
+    finally:
+        ctx.close()
+
+
+@cli.command()
+@click.option("--http", is_flag=True, help="streamable-HTTP instead of stdio")
+@click.option("--port", default=8765)
+@click.option("--api-key", default="")
+def mcp(http, port, api_key):
+    """Run the MCP server (stdio by default)."""
+    from ..mcp.server import run_mcp_http_server, run_mcp_server
+    if http:
+        run_mcp_http_server(host="127.0.0.1", port=port, api_key=api_key)
+    else:
+        run_mcp_server(api_key=api_key)
+
+
+@cli.command()
+@click.option("--port", default=8080)
+@click.option("--api-key", default="")
+def api(port, api_key):
+    """Run the local admin HTTP API."""
+    from ..api.local_api import run_api
+    run_api(port=port, api_key=api_key)
+
+
+@cli.command()
+def dashboard():
+    """Print the text dashboard report."""
+    from ..utils.text_report import render_report
+    ctx = _ctx(with_worker=False, with_engine=False)
+    try:
+        click.echo(render_report(ctx))
+    finally:
+        ctx.close()
+
+
+# ----------------------------------------------------------------- index
+
+@cli.group()
+def index():
+    """Index snapshot/stat operations."""
+
+
+@index.command()
+@click.argument("path", type=click.Path())
+@click.option("--max-docs", default=100_000)
+def export(path, max_docs):
+    """Export the index to a .infomesh-snapshot file."""
+    from ..index.snapshot import export_snapshot
+    ctx = _ctx(with_worker=False, with_engine=False)
+    try:
+        header = export_snapshot(ctx.store, path, max_docs=max_docs,
+                                 node_name=ctx.keys.node_id[:12])
+        click.echo(json.dumps(header, indent=2))
+    finally:
+        ctx.close()
+
+
+@index.command(name="import")
+@click.argument("path", type=click.Path(exists=True))
+def import_(path):
+    """Import a .infomesh-snapshot file."""
+    from ..index.snapshot import import_snapshot
+    ctx = _ctx(with_worker=False)
+    try:
+        hook = (ctx.engine.add_document if ctx.engine is not None else None)
+        res = import_snapshot(ctx.store, path, on_document=hook)
+        if ctx.engine is not None and ctx.engine.pending_count:
+            ctx.flush_engine()
+        click.echo(json.dumps({"imported": res["imported"],
+                               "skipped": res["skipped"]}, indent=2))
+    finally:
+        ctx.close()
+
+
+@index.command()
+def stats():
+    """Index statistics."""
+    ctx = _ctx(with_worker=False, with_engine=False)
+    try:
+        click.echo(json.dumps(ctx.store.stats(), indent=2))
+    finally:
+        ctx.close()
+
+
+# ---------------------------------------------------------------- config
+
+@cli.group()
+def config():
+    """Configuration operations."""
+
+
+@config.command(name="show")
+def config_show():
+    click.echo(json.dumps(dataclasses.asdict(load_config()), indent=2))
+
+
+@config.command(name="set")
+@click.argument("key")     # section.key
+@click.argument("value")
+def config_set(key, value):
+    cfg = load_config()
+    try:
+        section_name, field_name = key.split(".", 1)
+        section = getattr(cfg, section_name)
+        cur = getattr(section, field_name)
+    except (ValueError, AttributeError):
+        raise click.ClickException(f"unknown key {key!r}")
+    from ..config import _coerce
+    new_section = dataclasses.replace(section,
+                                      **{field_name: _coerce(cur, value)})
+    cfg = dataclasses.replace(cfg, **{section_name: new_section})
+    path = save_config(cfg)
+    click.echo(f"wrote {path}")
+
+
+# ------------------------------------------------------------------ keys
+
+@cli.group()
+def keys():
+    """Node identity keys."""
+
+
+@keys.command(name="show")
+def keys_show():
+    from ..trust.keys import ensure_keys
+    kp = ensure_keys(load_config().data_dir)
+    click.echo(json.dumps({"node_id": kp.node_id,
+                           "public_key": kp.public.hex()}, indent=2))
+
+
+@keys.command(name="rotate")
+def keys_rotate():
+    from ..trust.keys import ensure_keys, rotate_keys
+    data = load_config().data_dir
+    old = ensure_keys(data)
+    new, record = rotate_keys(data, old)
+    click.echo(json.dumps({"new_node_id": new.node_id,
+                           "rotation": record}, indent=2))
+
+
+# ---------------------------------------------------------------- doctor
+
+@cli.command()
+def doctor():
+    """Run diagnostics checks."""
+    from ..utils.diagnostics import run_doctor
+    report = run_doctor()
+    for check in report["checks"]:
+        mark = "✓" if check["ok"] else "✗"
+        click.echo(f"{mark} {check['name']}: {check['detail']}")
+    sys.exit(0 if report["ok"] else 1)
+
+
+@cli.command()
+@click.option("--iterations", default=200)
+def bench(iterations):
+    """Run the micro-benchmark suite (CPU pipeline pieces)."""
+    from ..utils.benchmarks import run_micro_suite
+    click.echo(json.dumps(run_micro_suite(iterations), indent=2))
+
+
+def main():
+    cli()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,107 @@
+"""The background crawl loop: seeds -> frontier -> crawl -> index.
+
+Reference parity: infomesh/crawler/crawl_loop.py (seed_and_crawl_loop,
+feed polling, priority recrawl batches, governor backpressure, idle
+re-seeding, hourly FTS optimize).
+"""
+from __future__ import annotations
+
+import asyncio
+import logging
+import time
+
+from ..services import AppContext
+from ..utils.governor import ResourceGovernor
+from .rss import parse_feed
+from .seeds import load_seeds
+
+log = logging.getLogger("infomesh.crawl_loop")
+
+FEED_POLL_EVERY_S = 60.0
+RECRAWL_EVERY_S = 120.0
+OPTIMIZE_EVERY_S = 3600.0
+ENGINE_FLUSH_EVERY_S = 60.0
+IDLE_RESEED_AFTER_S = 600.0
+
+
+async def seed_and_crawl_loop(ctx: AppContext,
+                              governor: ResourceGovernor | None = None,
+                              seed_category: str = "quickstart",
+                              max_iterations: int | None = None,
+                              stop_check=None) -> dict:
+    """Main crawl loop. `max_iterations`/`stop_check` bound it for tests
+    and foreground runs."""
+    assert ctx.worker is not None, "crawl loop needs a worker role"
+    governor = governor or ResourceGovernor()
+    sched = ctx.worker.scheduler
+    for url in load_seeds(seed_category):
+        sched.add_url(url, depth=0)
+    last_feed = last_recrawl = last_flush = 0.0
+    last_optimize = time.time()
+    last_activity = time.time()
+    iterations = 0
+    stats = {"crawled": 0, "indexed": 0, "errors": 0}
+    while True:
+        if stop_check is not None and stop_check():
+            break
+        if max_iterations is not None and iterations >= max_iterations:
+            break
+        iterations += 1
+        now = time.time()
+
+        # governor backpressure (crawl_loop.py:183-221)
+        if not governor.crawl_allowed():
+            await asyncio.sleep(1.0)
+            continue
+
+        # feed polling (crawl_loop.py:38-104)
+        if now - last_feed > FEED_POLL_EVERY_S and ctx.feeds.feeds:
+            last_feed = now
+            for mf in ctx.feeds.due(now)[:5]:
+                try:
+                    client = await ctx.worker._get_client()
+                    resp = await client.get(mf.url)
+                    feed = parse_feed(mf.url, resp.text) \
+                        if resp.status_code == 200 else None
+                except Exception:
+                    feed = None
+                for item in ctx.feeds.record_poll(mf.url, feed, now):
+                    sched.add_url(item.url, depth=0, priority=2)
+
+        # priority recrawl batch (crawl_loop.py:110-180)
+        if now - last_recrawl > RECRAWL_EVERY_S:
+            last_recrawl = now
+            for doc in ctx.store.due_for_recrawl(limit=10):
+                sched.add_url(doc.url, depth=0, priority=3)
+
+        item = await sched.get_url(timeout=2.0)
+        if item is None:
+            if now - last_activity > IDLE_RESEED_AFTER_S:
+                for url in load_seeds(seed_category):
+                    sched.add_url(url, depth=0)
+                last_activity = now
+            await asyncio.sleep(0.2)
+            continue
+        url, depth = item
+        last_activity = now
+        try:
+            out = await ctx.crawl_and_index(url, depth=depth,
+                                            force=depth == -1)
+            stats["crawled"] += 1
+            if out.get("indexed"):
+                stats["indexed"] += 1
+        except Exception as e:
+            stats["errors"] += 1
+            log.warning("crawl %s failed: %s", url, e)
+
+        # GPU ingest flush + hourly FTS optimize (crawl_loop.py:485-496)
+        if ctx.engine is not None and now - last_flush > ENGINE_FLUSH_EVERY_S \
+                and ctx.engine.pending_count:
+            last_flush = now
+            ctx.flush_engine()
+        if now - last_optimize > OPTIMIZE_EVERY_S:
+            last_optimize = now
+            ctx.store.optimize()
+    if ctx.engine is not None and ctx.engine.pending_count:
+        ctx.flush_engine()
+    return stats

@@ -117,9 +117,10 @@ def extract_domain(url: str) -> str:
 _TOKEN_RE = re.compile(r"[^\s\"'()*:^]+")
 
 
-def sanitize_fts_query(query: str) -> str:
+def sanitize_fts_query(query: str, match_any: bool = False) -> str:
     tokens = _TOKEN_RE.findall(query)
-    return " ".join('"' + t.replace('"', "") + '"' for t in tokens if t.strip('"'))
+    quoted = ['"' + t.replace('"', "") + '"' for t in tokens if t.strip('"')]
+    return (" OR " if match_any else " ").join(quoted)
 
 
 class LocalStore:
@@ -224,10 +225,12 @@ class LocalStore:
                language: str | None = None,
                domain: str | None = None,
                after: float | None = None,
-               before: float | None = None) -> list[SearchHit]:
+               before: float | None = None,
+               match_any: bool = False) -> list[SearchHit]:
         """FTS5 MATCH + bm25() ordering + snippet() with filters
-        (reference: local_store.py:253-352)."""
-        fts_query = sanitize_fts_query(query)
+        (reference: local_store.py:253-352). match_any=True ORs the
+        terms (fact-check / recall-first queries)."""
+        fts_query = sanitize_fts_query(query, match_any=match_any)
         if not fts_query:
             return []
         sql = [

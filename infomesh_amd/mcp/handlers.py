@@ -1,0 +1,165 @@
+"""MCP tool handlers over an AppContext.
+
+Reference parity: infomesh/mcp/handlers.py (handle_web_search mode
+routing -> explain / RAG / answer / plain search; fetch/crawl/fact_check/
+status; crawl rate limit 60/hr; query preprocessing + did-you-mean
+post-processing).
+"""
+from __future__ import annotations
+
+import asyncio
+import time
+from typing import Any
+
+from ..errors import InfoMeshError, format_error
+from ..search.explain import explain_search
+from ..search.formatter import result_to_dict
+from ..search.nlp import did_you_mean
+from ..search.rag import format_rag_output, extract_answer
+from ..services import AppContext
+
+CRAWL_RATE_PER_HOUR = 60
+
+
+class Handlers:
+    def __init__(self, ctx: AppContext, reranker=None, summarizer=None):
+        self.ctx = ctx
+        self.reranker = reranker
+        self.summarizer = summarizer
+        self._crawl_times: list[float] = []
+
+    # ------------------------------------------------------------ search
+    def web_search(self, query: str, limit: int = 10, mode: str = "auto",
+                   explain: bool = False, chunk_size: int = 0,
+                   answer_mode: bool = False, rerank: bool = False,
+                   summarize: bool = False, **_) -> dict[str, Any]:
+        if not query or not query.strip():
+            raise InfoMeshError("SRCH001", "empty query")
+        limit = max(1, min(int(limit), 50))
+        if explain:
+            return {"mode": "explain",
+                    "results": explain_search(
+                        self.ctx.store, query, limit,
+                        authority_fn=self.ctx.link_graph.url_authority,
+                        trust_fn=self.ctx.trust.trust_fn())}
+        resp = self.ctx.search(query, limit=limit, mode=mode)
+        results = [result_to_dict(r) for r in resp.results]
+        # attach full text for RAG modes
+        if chunk_size or answer_mode or summarize:
+            for r in results:
+                doc = self.ctx.store.get_document_by_url(r["url"])
+                if doc:
+                    r["text"] = doc.text[:20_000]
+        if rerank and self.reranker is not None:
+            from ..search.reranker import rerank_results
+            results = rerank_results(
+                query, results, self.reranker, keep=limit,
+                text_of=lambda r: f"{r.get('title','')} {r.get('snippet','')}")
+        out: dict[str, Any] = {
+            "query": query,
+            "effective_query": resp.effective_query,
+            "mode": resp.mode,
+            "elapsed_ms": round(resp.elapsed_ms, 2),
+            "degraded": resp.degraded,
+            "results": results,
+        }
+        if chunk_size:
+            rag = format_rag_output(query, results, chunk_size=chunk_size,
+                                    answer_mode=answer_mode,
+                                    summarizer=self.summarizer
+                                    if summarize else None)
+            out["chunks"] = [{"text": c.text, "url": c.url,
+                              "title": c.title, "score": round(c.score, 3)}
+                             for c in rag.chunks]
+            out["entities"] = rag.entities
+            if answer_mode:
+                out["answer"] = rag.answer
+                out["confidence"] = rag.confidence
+            if summarize:
+                out["summary"] = rag.summary
+        elif answer_mode:
+            out["answer"], out["confidence"] = extract_answer(query, results)
+        elif summarize and self.summarizer is not None:
+            out["summary"] = self.summarizer.summarize_results(
+                results, query).summary
+        if not results:
+            vocab = {t.lower(): 1
+                     for s in self.ctx.store.suggest(query.split()[0], 20)
+                     for t in s.split()}
+            suggestion = did_you_mean(query, vocab) if vocab else None
+            if suggestion:
+                out["did_you_mean"] = suggestion
+        return out
+
+    # ------------------------------------------------------------- fetch
+    def fetch_page(self, url: str, live: bool = False,
+                   max_chars: int = 20_000, **_) -> dict[str, Any]:
+        doc = self.ctx.fetch_page(url)
+        if doc is None and live and self.ctx.worker is not None:
+            res = asyncio.run(self.ctx.crawl_and_index(url))
+            if res.get("indexed"):
+                doc = self.ctx.fetch_page(url)
+        if doc is None:
+            return {"url": url, "found": False}
+        return {"url": doc.url, "found": True, "title": doc.title,
+                "language": doc.language, "crawled_at": doc.crawled_at,
+                "text": doc.text[:max_chars],
+                "truncated": len(doc.text) > max_chars}
+
+    # ------------------------------------------------------------- crawl
+    def crawl_url(self, url: str, force: bool = False, **_) -> dict[str, Any]:
+        now = time.time()
+        self._crawl_times = [t for t in self._crawl_times if now - t < 3600]
+        if len(self._crawl_times) >= CRAWL_RATE_PER_HOUR:
+            raise InfoMeshError("SEC001",
+                                f"crawl rate limit {CRAWL_RATE_PER_HOUR}/hr")
+        self._crawl_times.append(now)
+        return asyncio.run(self.ctx.crawl_and_index(url, force=force))
+
+    # -------------------------------------------------------- fact check
+    def fact_check(self, claim: str, limit: int = 5, **_) -> dict[str, Any]:
+        from ..summarizer.verify import _fact_support  # same scorer
+        hits = self.ctx.store.search(claim, limit=limit)
+        if not hits:  # recall-first retry: any-term match
+            hits = self.ctx.store.search(claim, limit=limit, match_any=True)
+        evidence = []
+        for r in hits:
+            url = getattr(r, "url", "")
+            doc = self.ctx.store.get_document_by_url(url)
+            if doc is None:
+                continue
+            score, sent = _fact_support(claim, doc.text[:20_000])
+            evidence.append({"url": url, "title": doc.title,
+                             "support": round(score, 3), "evidence": sent})
+        evidence.sort(key=lambda e: -e["support"])
+        supported = bool(evidence) and evidence[0]["support"] >= 0.5
+        return {"claim": claim, "supported": supported,
+                "confidence": evidence[0]["support"] if evidence else 0.0,
+                "evidence": evidence[:limit]}
+
+    # ------------------------------------------------------------ status
+    def status(self, **_) -> dict[str, Any]:
+        return self.ctx.status()
+
+    # ---------------------------------------------------------- dispatch
+    def call(self, tool: str, args: dict[str, Any]) -> dict[str, Any]:
+        from .tools import resolve_tool
+        resolved = resolve_tool(tool)
+        if resolved is None:
+            raise InfoMeshError("SRCH001", f"unknown tool {tool!r}")
+        # legacy arg adaptation
+        if tool == "explain":
+            args = {**args, "explain": True}
+        elif tool == "search_rag":
+            args.setdefault("chunk_size", 512)
+        elif tool == "extract_answer":
+            args = {**args, "answer_mode": True}
+        elif tool in ("search_local", "search_hybrid", "search_distributed"):
+            args = {**args, "mode": tool.split("_", 1)[1]}
+        elif tool == "verify":
+            args.setdefault("claim", args.pop("query", ""))
+        fn = getattr(self, resolved)
+        try:
+            return fn(**args)
+        except InfoMeshError as e:
+            return {"error": format_error(e), "code": e.code}

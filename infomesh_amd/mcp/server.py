@@ -1,0 +1,162 @@
+"""MCP server: JSON-RPC 2.0 over stdio + streamable-HTTP transport.
+
+Reference parity: infomesh/mcp/server.py (stdio + Streamable-HTTP
+transports, tool dispatch with legacy aliases, query cache wiring,
+API-key check, analytics). The protocol surface implements MCP
+`initialize`, `tools/list`, `tools/call`, `ping`.
+"""
+from __future__ import annotations
+
+import json
+import sys
+import time
+from typing import Any, TextIO
+
+from ..services import AppContext
+from .handlers import Handlers
+from .session import AnalyticsTracker, SessionStore
+from .tools import TOOLS
+
+PROTOCOL_VERSION = "2024-11-05"
+SERVER_INFO = {"name": "infomesh-amd", "version": "0.1.0"}
+
+
+class McpServer:
+    def __init__(self, ctx: AppContext, api_key: str = "",
+                 reranker=None, summarizer=None):
+        self.ctx = ctx
+        self.api_key = api_key
+        self.handlers = Handlers(ctx, reranker=reranker,
+                                 summarizer=summarizer)
+        self.sessions = SessionStore()
+        self.analytics = AnalyticsTracker()
+
+    # ------------------------------------------------------- JSON-RPC
+    def handle_message(self, msg: dict[str, Any]) -> dict[str, Any] | None:
+        mid = msg.get("id")
+        method = msg.get("method", "")
+        params = msg.get("params") or {}
+        try:
+            if method == "initialize":
+                result = {
+                    "protocolVersion": PROTOCOL_VERSION,
+                    "serverInfo": SERVER_INFO,
+                    "capabilities": {"tools": {}},
+                }
+            elif method == "notifications/initialized":
+                return None
+            elif method == "ping":
+                result = {}
+            elif method == "tools/list":
+                result = {"tools": TOOLS}
+            elif method == "tools/call":
+                result = self._call_tool(params)
+            else:
+                return self._error(mid, -32601, f"method {method!r} not found")
+        except Exception as e:  # tool errors -> JSON-RPC error
+            return self._error(mid, -32000, str(e))
+        if mid is None:
+            return None
+        return {"jsonrpc": "2.0", "id": mid, "result": result}
+
+    def _call_tool(self, params: dict[str, Any]) -> dict[str, Any]:
+        name = params.get("name", "")
+        args = params.get("arguments") or {}
+        t0 = time.perf_counter()
+        error = False
+        try:
+            out = self.handlers.call(name, args)
+            error = "error" in out
+        except Exception:
+            error = True
+            raise
+        finally:
+            self.analytics.record(name, (time.perf_counter() - t0) * 1e3,
+                                  error)
+        return {
+            "content": [{"type": "text",
+                         "text": json.dumps(out, ensure_ascii=False,
+                                            default=str)}],
+            "isError": error,
+        }
+
+    @staticmethod
+    def _error(mid, code: int, message: str) -> dict[str, Any]:
+        return {"jsonrpc": "2.0", "id": mid,
+                "error": {"code": code, "message": message}}
+
+    # ---------------------------------------------------------- stdio
+    def run_stdio(self, stdin: TextIO | None = None,
+                  stdout: TextIO | None = None) -> None:
+        stdin = stdin or sys.stdin
+        stdout = stdout or sys.stdout
+        for line in stdin:
+            line = line.strip()
+            if not line:
+                continue
+            try:
+                msg = json.loads(line)
+            except json.JSONDecodeError:
+                continue
+            resp = self.handle_message(msg)
+            if resp is not None:
+                stdout.write(json.dumps(resp) + "\n")
+                stdout.flush()
+
+    # ----------------------------------------------------------- HTTP
+    def asgi_app(self):
+        """Streamable-HTTP transport: POST /mcp with a JSON-RPC body."""
+        server = self
+
+        async def app(scope, receive, send):
+            if scope["type"] != "http":
+                return
+            headers = {k.decode(): v.decode()
+                       for k, v in scope.get("headers", [])}
+            if server.api_key and \
+                    headers.get("x-api-key") != server.api_key:
+                await _respond(send, 401, {"error": "bad api key"})
+                return
+            if scope["method"] == "POST" and scope["path"] in ("/mcp", "/"):
+                body = b""
+                while True:
+                    ev = await receive()
+                    body += ev.get("body", b"")
+                    if not ev.get("more_body"):
+                        break
+                try:
+                    msg = json.loads(body)
+                except json.JSONDecodeError:
+                    await _respond(send, 400, {"error": "bad json"})
+                    return
+                resp = server.handle_message(msg)
+                await _respond(send, 200, resp or {})
+            elif scope["method"] == "GET" and scope["path"] == "/health":
+                await _respond(send, 200, {"ok": True})
+            else:
+                await _respond(send, 404, {"error": "not found"})
+        return app
+
+
+async def _respond(send, status: int, payload: dict) -> None:
+    body = json.dumps(payload, default=str).encode()
+    await send({"type": "http.response.start", "status": status,
+                "headers": [(b"content-type", b"application/json"),
+                            (b"content-length",
+                             str(len(body)).encode()),
+                            (b"access-control-allow-origin", b"*")]})
+    await send({"type": "http.response.body", "body": body})
+
+
+def run_mcp_server(ctx: AppContext | None = None, api_key: str = "") -> None:
+    """stdio entry (blocking)."""
+    ctx = ctx or AppContext.create()
+    McpServer(ctx, api_key=api_key).run_stdio()
+
+
+def run_mcp_http_server(ctx: AppContext | None = None, host: str = "127.0.0.1",
+                        port: int = 8765, api_key: str = "") -> None:
+    import uvicorn
+    ctx = ctx or AppContext.create()
+    server = McpServer(ctx, api_key=api_key)
+    uvicorn.run(server.asgi_app(), host=host, port=port, log_level="warning")

@@ -1,0 +1,102 @@
+"""MCP tool schemas: the 5 consolidated tools + legacy alias map.
+
+Reference parity: infomesh/mcp/tools.py:41-245 (web_search, fetch_page,
+crawl_url, fact_check, status) and the ~15 legacy aliases dispatched in
+mcp/server.py:205-457.
+"""
+from __future__ import annotations
+
+TOOLS: list[dict] = [
+    {
+        "name": "web_search",
+        "description": (
+            "Search the local + GPU-sharded hybrid index (BM25 + dense). "
+            "Modes: plain results, explain (score breakdown), RAG chunks "
+            "(chunk_size), answer extraction (answer_mode)."),
+        "inputSchema": {
+            "type": "object",
+            "properties": {
+                "query": {"type": "string"},
+                "limit": {"type": "integer", "minimum": 1, "maximum": 50},
+                "mode": {"type": "string",
+                         "enum": ["auto", "local", "hybrid", "distributed"]},
+                "explain": {"type": "boolean"},
+                "chunk_size": {"type": "integer"},
+                "answer_mode": {"type": "boolean"},
+                "rerank": {"type": "boolean"},
+                "summarize": {"type": "boolean"},
+            },
+            "required": ["query"],
+        },
+    },
+    {
+        "name": "fetch_page",
+        "description": "Fetch a page's indexed content (cache-first); "
+                       "optionally crawl it live when absent.",
+        "inputSchema": {
+            "type": "object",
+            "properties": {
+                "url": {"type": "string"},
+                "live": {"type": "boolean"},
+                "max_chars": {"type": "integer"},
+            },
+            "required": ["url"],
+        },
+    },
+    {
+        "name": "crawl_url",
+        "description": "Crawl and index a URL (rate-limited).",
+        "inputSchema": {
+            "type": "object",
+            "properties": {
+                "url": {"type": "string"},
+                "force": {"type": "boolean"},
+            },
+            "required": ["url"],
+        },
+    },
+    {
+        "name": "fact_check",
+        "description": "Check a claim against indexed sources: searches, "
+                       "extracts supporting/contradicting passages.",
+        "inputSchema": {
+            "type": "object",
+            "properties": {
+                "claim": {"type": "string"},
+                "limit": {"type": "integer"},
+            },
+            "required": ["claim"],
+        },
+    },
+    {
+        "name": "status",
+        "description": "Node status: index size, GPU shards, credits, "
+                       "cache, crawler stats.",
+        "inputSchema": {"type": "object", "properties": {}},
+    },
+]
+
+# Legacy tool names -> (tool, arg-transform hints) kept for compatibility.
+LEGACY_ALIASES: dict[str, str] = {
+    "search": "web_search",
+    "search_local": "web_search",
+    "search_hybrid": "web_search",
+    "search_distributed": "web_search",
+    "explain": "web_search",
+    "search_rag": "web_search",
+    "extract_answer": "web_search",
+    "suggest": "web_search",
+    "get_page": "fetch_page",
+    "fetch": "fetch_page",
+    "crawl": "crawl_url",
+    "index_url": "crawl_url",
+    "verify": "fact_check",
+    "node_status": "status",
+    "stats": "status",
+}
+
+
+def resolve_tool(name: str) -> str | None:
+    if name in {t["name"] for t in TOOLS}:
+        return name
+    return LEGACY_ALIASES.get(name)

@@ -1,0 +1,290 @@
+"""Entry-point tests: MCP server (direct JSON-RPC), admin API
+(TestClient), CLI (CliRunner), SDK (MockTransport), integrations.
+Reference parity: tests/test_mcp*.py, test_local_api.py,
+test_cli_search.py.
+"""
+from __future__ import annotations
+
+import json
+
+import httpx
+import pytest
+from click.testing import CliRunner
+from fastapi.testclient import TestClient
+
+from infomesh_amd.api.local_api import create_app
+from infomesh_amd.cli import cli
+from infomesh_amd.config import Config
+from infomesh_amd.index.local_store import Document
+from infomesh_amd.integrations import (InfoMeshDocumentStore,
+                                       InfoMeshReader, InfoMeshRetriever)
+from infomesh_amd.mcp.server import McpServer
+from infomesh_amd.mcp.tools import TOOLS, resolve_tool
+from infomesh_amd.sdk.client import InfoMeshClient
+from infomesh_amd.services import AppContext
+
+
+@pytest.fixture
+def ctx():
+    c = AppContext.create(config=Config(), with_engine=False,
+                          with_worker=False, in_memory=True)
+    for i in range(6):
+        c.index_document(Document(
+            url=f"https://a.com/{i}", title=f"Python doc {i}",
+            text=f"python tutorial part {i} covering functions and classes "
+                 f"with plenty of example code to read"),
+            attest=False, credit=False)
+    yield c
+    c.close()
+
+
+# ------------------------------------------------------------------- MCP
+
+def _rpc(server, method, params=None, mid=1):
+    return server.handle_message({"jsonrpc": "2.0", "id": mid,
+                                  "method": method, "params": params or {}})
+
+
+def test_mcp_initialize_and_list(ctx):
+    s = McpServer(ctx)
+    init = _rpc(s, "initialize")
+    assert init["result"]["serverInfo"]["name"] == "infomesh-amd"
+    tools = _rpc(s, "tools/list")["result"]["tools"]
+    assert {t["name"] for t in tools} == \
+        {"web_search", "fetch_page", "crawl_url", "fact_check", "status"}
+
+
+def test_mcp_web_search_call(ctx):
+    s = McpServer(ctx)
+    resp = _rpc(s, "tools/call", {"name": "web_search",
+                                  "arguments": {"query": "python tutorial"}})
+    payload = json.loads(resp["result"]["content"][0]["text"])
+    assert payload["results"]
+    assert payload["results"][0]["url"].startswith("https://a.com/")
+    assert not resp["result"]["isError"]
+
+
+def test_mcp_legacy_aliases(ctx):
+    s = McpServer(ctx)
+    assert resolve_tool("search") == "web_search"
+    assert resolve_tool("nope") is None
+    resp = _rpc(s, "tools/call", {"name": "explain",
+                                  "arguments": {"query": "python"}})
+    payload = json.loads(resp["result"]["content"][0]["text"])
+    assert payload["mode"] == "explain"
+
+
+def test_mcp_rag_modes(ctx):
+    s = McpServer(ctx)
+    resp = _rpc(s, "tools/call", {
+        "name": "web_search",
+        "arguments": {"query": "python functions", "chunk_size": 256,
+                      "answer_mode": True}})
+    payload = json.loads(resp["result"]["content"][0]["text"])
+    assert "chunks" in payload and "answer" in payload
+
+
+def test_mcp_fetch_and_status(ctx):
+    s = McpServer(ctx)
+    resp = _rpc(s, "tools/call", {"name": "fetch_page",
+                                  "arguments": {"url": "https://a.com/1"}})
+    payload = json.loads(resp["result"]["content"][0]["text"])
+    assert payload["found"] and "python tutorial" in payload["text"]
+    st = json.loads(_rpc(s, "tools/call", {"name": "status"})
+                    ["result"]["content"][0]["text"])
+    assert st["index"]["documents"] == 6
+
+
+def test_mcp_fact_check(ctx):
+    s = McpServer(ctx)
+    resp = _rpc(s, "tools/call", {
+        "name": "fact_check",
+        "arguments": {"claim": "python tutorial covers functions"}})
+    payload = json.loads(resp["result"]["content"][0]["text"])
+    assert payload["supported"]
+
+
+def test_mcp_unknown_method(ctx):
+    s = McpServer(ctx)
+    resp = _rpc(s, "bogus/method")
+    assert "error" in resp
+
+
+def test_mcp_stdio_roundtrip(ctx):
+    import io
+    s = McpServer(ctx)
+    stdin = io.StringIO(json.dumps(
+        {"jsonrpc": "2.0", "id": 5, "method": "ping"}) + "\n")
+    stdout = io.StringIO()
+    s.run_stdio(stdin, stdout)
+    out = json.loads(stdout.getvalue())
+    assert out["id"] == 5
+
+
+def test_mcp_analytics(ctx):
+    s = McpServer(ctx)
+    _rpc(s, "tools/call", {"name": "status"})
+    rep = s.analytics.report()
+    assert rep["status"]["calls"] == 1
+
+
+# ------------------------------------------------------------- admin API
+
+@pytest.fixture
+def api_client(ctx):
+    app = create_app(ctx)
+    return TestClient(app)
+
+
+def test_api_health_and_search(api_client):
+    assert api_client.get("/health").json()["ok"]
+    r = api_client.get("/search", params={"q": "python tutorial"})
+    assert r.status_code == 200
+    assert r.json()["results"]
+    assert r.headers["X-Frame-Options"] == "DENY"
+
+
+def test_api_status_config_stats(api_client):
+    assert api_client.get("/status").json()["index"]["documents"] == 6
+    assert "crawl" in api_client.get("/config").json()
+    assert api_client.get("/index/stats").json()["documents"] == 6
+    assert "balance" in api_client.get("/credits/balance").json()
+    assert api_client.get("/network/peers").json()["world_size"] >= 1
+
+
+def test_api_metrics_prometheus(api_client):
+    api_client.get("/search", params={"q": "python"})
+    text = api_client.get("/metrics").text
+    assert "api_requests_total" in text
+
+
+def test_api_dashboard_text(api_client):
+    text = api_client.get("/dashboard").text
+    assert "infomesh-amd node report" in text
+
+
+def test_api_key_enforced(ctx):
+    app = create_app(ctx, api_key="sekrit")
+    c = TestClient(app)
+    assert c.get("/status").status_code == 401
+    assert c.get("/status", headers={"x-api-key": "sekrit"}).status_code == 200
+    assert c.get("/health").status_code == 200  # health exempt
+
+
+def test_api_empty_query_400(api_client):
+    assert api_client.get("/search", params={"q": "  "}).status_code == 400
+
+
+# ------------------------------------------------------------------- SDK
+
+def test_sdk_client_roundtrip(ctx):
+    app = create_app(ctx)
+    tc = TestClient(app)
+
+    def forward(request: httpx.Request) -> httpx.Response:
+        r = tc.request(request.method, request.url.raw_path.decode(),
+                       headers=dict(request.headers))
+        return httpx.Response(r.status_code, content=r.content,
+                              headers=r.headers)
+
+    client = InfoMeshClient(base_url="http://testserver",
+                            transport=httpx.MockTransport(forward))
+    results = client.search("python tutorial")
+    assert results and results[0]["url"].startswith("https://a.com/")
+    assert client.status()["index"]["documents"] == 6
+    assert client.health()
+    client.close()
+
+
+def test_sdk_async_client(ctx):
+    import asyncio
+    from infomesh_amd.sdk.client import AsyncInfoMeshClient
+    app = create_app(ctx)
+
+    async def run():
+        client = AsyncInfoMeshClient(
+            base_url="http://testserver",
+            transport=httpx.ASGITransport(app=app))
+        results = await client.search("python tutorial")
+        assert results
+        st = await client.status()
+        assert st["index"]["documents"] == 6
+        await client.close()
+    asyncio.run(run())
+
+
+# ------------------------------------------------------------------- CLI
+
+def test_cli_help_lists_commands():
+    r = CliRunner().invoke(cli, ["--help"])
+    assert r.exit_code == 0
+    for cmd in ("start", "stop", "status", "search", "crawl", "mcp",
+                "index", "config", "keys", "doctor", "bench"):
+        assert cmd in r.output
+
+
+def test_cli_search_and_index(tmp_data_dir):
+    runner = CliRunner()
+    # build an index via snapshot import
+    from infomesh_amd.index.local_store import LocalStore
+    from infomesh_amd.index.snapshot import export_snapshot
+    src = LocalStore(":memory:")
+    src.add_document(Document(url="https://x.com/1", title="CLI Doc",
+                              text="command line interface search test "
+                                   "body with enough words"))
+    snap = tmp_data_dir / "x.infomesh-snapshot"
+    export_snapshot(src, snap)
+    src.close()
+    r = runner.invoke(cli, ["index", "import", str(snap)])
+    assert r.exit_code == 0, r.output
+    assert json.loads(r.output)["imported"] == 1
+    r2 = runner.invoke(cli, ["search", "command", "line", "--json"])
+    assert r2.exit_code == 0, r2.output
+    assert "x.com" in r2.output
+    r3 = runner.invoke(cli, ["index", "stats"])
+    assert json.loads(r3.output)["documents"] == 1
+
+
+def test_cli_config_set(tmp_data_dir):
+    runner = CliRunner()
+    r = runner.invoke(cli, ["config", "set", "crawl.max_concurrent", "9"])
+    assert r.exit_code == 0, r.output
+    r2 = runner.invoke(cli, ["config", "show"])
+    assert json.loads(r2.output)["crawl"]["max_concurrent"] == 9
+
+
+def test_cli_keys_and_doctor(tmp_data_dir):
+    runner = CliRunner()
+    r = runner.invoke(cli, ["keys", "show"])
+    assert r.exit_code == 0
+    assert json.loads(r.output)["node_id"]
+    r2 = runner.invoke(cli, ["doctor"])
+    assert "python" in r2.output
+
+
+def test_cli_dashboard(tmp_data_dir):
+    r = CliRunner().invoke(cli, ["dashboard"])
+    assert r.exit_code == 0
+    assert "node report" in r.output
+
+
+# ----------------------------------------------------------- integrations
+
+def test_retriever_and_reader_inprocess(ctx):
+    retr = InfoMeshRetriever(ctx=ctx, k=3)
+    docs = retr.get_relevant_documents("python tutorial")
+    assert docs and docs[0].metadata["url"].startswith("https://a.com/")
+    assert retr.invoke("python")  # langchain-style alias
+    reader = InfoMeshReader(ctx=ctx)
+    assert reader.load_data("python", limit=2)
+
+
+def test_document_store(ctx):
+    ds = InfoMeshDocumentStore(ctx)
+    n = ds.write_documents([
+        {"id": "https://h.com/1", "content": "haystack style document body "
+                                             "with sufficient length here",
+         "meta": {"title": "H1"}}])
+    assert n == 1
+    assert ds.count_documents() == 7
+    assert ds.filter_documents("haystack")

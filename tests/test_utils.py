@@ -1,0 +1,177 @@
+"""Cross-cutting utils tests: observability, slo, plugins, scalability,
+data_quality, security_ops, diagnostics, benchmarks, crawl loop."""
+from __future__ import annotations
+
+import asyncio
+
+import httpx
+import pytest
+
+from infomesh_amd.utils.benchmarks import run_micro_suite, time_fn
+from infomesh_amd.utils.data_quality import (cross_reference,
+                                             format_citation,
+                                             grade_document)
+from infomesh_amd.utils.diagnostics import run_doctor
+from infomesh_amd.utils.observability import (MetricsRegistry, QueryTrace,
+                                              grafana_dashboard_json)
+from infomesh_amd.utils.plugins import GLOBAL_PLUGINS, PluginManager
+from infomesh_amd.utils.scalability import (BloomFilter, RoundRobinPool,
+                                            batch_ingest)
+from infomesh_amd.utils.security_ops import ApiKeyManager, AuditLogger
+from infomesh_amd.utils.slo import SLOTracker
+
+
+def test_metrics_registry_render():
+    m = MetricsRegistry()
+    m.inc("requests_total", labels={"path": "/x"})
+    m.inc("requests_total", labels={"path": "/x"})
+    m.set_gauge("docs", 42)
+    m.observe("latency_seconds", 0.02)
+    text = m.render()
+    assert 'requests_total{path="/x"} 2.0' in text
+    assert "docs 42" in text
+    assert "latency_seconds_count 1" in text
+    assert 'le="0.05"' in text
+
+
+def test_query_trace_spans():
+    t = QueryTrace("q")
+    with t.span("encode"):
+        pass
+    with t.span("score"):
+        pass
+    rep = t.report()
+    assert set(rep["spans"]) == {"encode", "score"}
+    assert grafana_dashboard_json()["panels"]
+
+
+def test_slo_tracker():
+    s = SLOTracker()
+    s.define("search", target_p95_ms=100, target_success_rate=0.9)
+    for _ in range(20):
+        s.record("search", 10.0, ok=True)
+    assert s.report()["search"]["met"]
+    for _ in range(80):
+        s.record("search", 500.0, ok=False)
+    assert not s.report()["search"]["met"]
+
+
+def test_plugins():
+    pm = PluginManager()
+
+    @pm.hook("pre_search")
+    def lower(q):
+        return q.lower()
+
+    @pm.hook("pre_search")
+    def strip_x(q):
+        return q.replace("x", "")
+
+    assert pm.run("pre_search", "XQueryX") == "query"
+    with pytest.raises(ValueError):
+        pm.register("bogus", lambda v: v)
+    # failing plugin is isolated
+    pm.register("post_search", lambda v: 1 / 0)
+    assert pm.run("post_search", "ok") == "ok"
+    assert GLOBAL_PLUGINS.count() >= 0
+
+
+def test_bloom_filter():
+    bf = BloomFilter(capacity=1000, error_rate=0.01)
+    for i in range(500):
+        bf.add(f"url-{i}")
+    assert all(f"url-{i}" in bf for i in range(500))
+    false_pos = sum(1 for i in range(1000, 3000) if f"url-{i}" in bf)
+    assert false_pos < 60  # ~1% target
+    assert 0 < bf.fill_ratio() < 1
+
+
+def test_batch_ingest_and_pool():
+    batches = []
+    total = batch_ingest(range(25), lambda b: batches.append(len(b)) or len(b),
+                         batch_size=10)
+    assert total == 25 and batches == [10, 10, 5]
+    pool = RoundRobinPool(lambda: object(), size=2)
+    a, b, c = pool.get(), pool.get(), pool.get()
+    assert a is c and a is not b
+
+
+def test_data_quality():
+    import time
+    g = grade_document(time.time(), trust=0.9, text_len=3000, has_title=True)
+    assert g.grade in ("A", "B")
+    g2 = grade_document(time.time() - 90 * 86400, trust=0.1, text_len=50,
+                        has_title=False)
+    assert g2.grade in ("D", "F")
+    assert "Retrieved" in format_citation("https://a.com", "T", time.time())
+    ref = cross_reference({"a.com": 0.8, "b.com": 0.7, "c.com": 0.1})
+    assert ref["verdict"] == "corroborated"
+
+
+def test_api_key_manager():
+    km = ApiKeyManager()
+    key = km.create_key("ci")
+    assert km.verify(key)
+    assert not km.verify("imk_wrong")
+    assert not km.verify("")
+    km.revoke("ci")
+    assert not km.verify(key)
+    assert km.list_keys()[0]["revoked"] == 1
+    km.close()
+
+
+def test_audit_logger(tmp_path):
+    log = AuditLogger(tmp_path / "audit.jsonl")
+    log.log("search", q="x")
+    log.log("delete", url="https://a.com")
+    assert log.verify()
+    # tamper
+    lines = (tmp_path / "audit.jsonl").read_text().splitlines()
+    lines[0] = lines[0].replace('"search"', '"SEARCH"')
+    (tmp_path / "audit.jsonl").write_text("\n".join(lines) + "\n")
+    log2 = AuditLogger(tmp_path / "audit.jsonl")
+    assert not log2.verify()
+
+
+def test_doctor_runs(tmp_data_dir):
+    report = run_doctor()
+    names = {c["name"] for c in report["checks"]}
+    assert {"python", "torch", "sqlite fts5", "zstd"} <= names
+    assert all(c["ok"] for c in report["checks"]
+               if c["name"] in ("python", "sqlite fts5", "zstd"))
+
+
+def test_micro_bench_suite():
+    out = run_micro_suite(iterations=5)
+    assert "query_expansion" in out and out["query_expansion"]["ops_per_sec"] > 0
+    t = time_fn(lambda: None, iterations=10, warmup=1)
+    assert t["avg_ms"] >= 0
+
+
+def test_crawl_loop_bounded(tmp_data_dir):
+    """Crawl loop with a mock transport: seeds -> crawl -> index."""
+    from infomesh_amd.config import Config, CrawlConfig
+    from infomesh_amd.crawler.crawl_loop import seed_and_crawl_loop
+    from infomesh_amd.services import AppContext
+    import dataclasses
+
+    html = ("<html><head><title>Seed</title></head><body><p>" +
+            "Seed page content that is long enough to index properly. " * 4 +
+            "</p></body></html>")
+
+    def handler(request):
+        return httpx.Response(200, text=html,
+                              headers={"content-type": "text/html"})
+
+    cfg = dataclasses.replace(Config(), crawl=CrawlConfig(
+        politeness_delay_s=0, respect_robots=False, max_urls_per_hour=1000))
+    ctx = AppContext.create(config=cfg, with_engine=False, with_worker=True,
+                            in_memory=True)
+    ctx.worker._client = httpx.AsyncClient(
+        transport=httpx.MockTransport(handler))
+    ctx.worker.resolve_dns = False
+
+    stats = asyncio.run(seed_and_crawl_loop(ctx, max_iterations=6))
+    assert stats["crawled"] >= 1
+    assert ctx.store.count() >= 1  # dedup collapses identical bodies
+    ctx.close()
