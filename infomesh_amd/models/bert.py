@@ -102,16 +102,12 @@ class BertEncoder:
         for i in range(cfg.layers):
             p = f"layer.{i}."
             qkv = K.gemm_nt(x, w[p + "qkv.w"], bias=w[p + "qkv.b"])
-            qkv = qkv.view(B, S, 3, nh, d)
-            q = qkv[:, :, 0].permute(0, 2, 1, 3).reshape(B * nh, S, d).contiguous()
-            k = qkv[:, :, 1].permute(0, 2, 1, 3).reshape(B * nh, S, d).contiguous()
-            v = qkv[:, :, 2].permute(0, 2, 1, 3).reshape(B * nh, S, d).contiguous()
+            # fused head split producing the PV-ready transposed V image
+            q, k, vt = K.qkv_split(qkv, B, S, nh, nh, d)
             scores = K.gemm_nt(q, k, out_f32=True, alpha=scale)
             probs = K.softmax(scores, valid_len=vl)
-            vt = v.transpose(1, 2).contiguous()          # [G, d, S]
             ctx = K.gemm_nt(probs, vt)                   # [G, S, d]
-            merged = ctx.view(B, nh, S, d).permute(0, 2, 1, 3)\
-                .reshape(B * S, H).contiguous()
+            merged = K.merge_heads(ctx, B, S, nh, d)
             attn = K.gemm_nt(merged, w[p + "attn_out.w"],
                              bias=w[p + "attn_out.b"])
             x = K.layernorm(attn, w[p + "ln1.g"], w[p + "ln1.b"],

@@ -73,7 +73,7 @@ def rope_tables(cfg: Phi3Config, device: str) -> tuple[torch.Tensor, torch.Tenso
 class Phi3Decoder:
     def __init__(self, cfg: Phi3Config = PHI3_MINI, device: str = "cuda",
                  seed: int = 777, max_batch: int = 8,
-                 max_seq: int = 2304):
+                 max_seq: int = 2304, use_graph: bool = True):
         self.cfg = cfg
         self.device = device
         self.max_seq = min(max_seq, cfg.max_pos)
@@ -87,6 +87,8 @@ class Phi3Decoder:
         self.v_cache = [torch.zeros_like(self.k_cache[0])
                         for _ in range(cfg.layers)]
         self.lens = torch.zeros(max_batch, dtype=torch.int32, device=device)
+        self.use_graph = use_graph and device.startswith("cuda")
+        self._graphs: dict[int, tuple] = {}
 
     def reset(self) -> None:
         self.lens.zero_()
@@ -121,35 +123,28 @@ class Phi3Decoder:
             p = f"layer.{i}."
             h = K.rmsnorm(x, self.w[p + "ln1.g"], eps=cfg.eps)
             qkv = K.gemm_nt(h, self.w[p + "qkv.w"])
-            q, k, v = self._split_qkv(qkv, B, S)
-            q = q.reshape(B * S, nh, d).contiguous()
-            k = k.reshape(B * S, nkv, d).contiguous()
-            K.rope(q, self.cos, self.sin, pos)
-            K.rope(k, self.cos, self.sin, pos)
-            qh = q.view(B, S, nh, d).permute(0, 2, 1, 3)\
-                .reshape(B * nh, S, d).contiguous()
-            kh = k.view(B, S, nkv, d).permute(0, 2, 1, 3)\
-                .reshape(B * nkv, S, d).contiguous()
-            vh = v.permute(0, 2, 1, 3).reshape(B * nkv, S, d).contiguous()
+            qh, kh, vt = K.qkv_split(qkv, B, S, nh, nkv, d,
+                                     cos_t=self.cos, sin_t=self.sin,
+                                     pos=pos)
             self.k_cache[i][:B, :, :S] = kh.view(B, nkv, S, d)
-            self.v_cache[i][:B, :, :S] = vh.view(B, nkv, S, d)
+            self.v_cache[i][:B, :, :S] = \
+                vt.view(B, nkv, d, S).permute(0, 1, 3, 2)
             if nh != nkv:
                 rep = nh // nkv
                 kh = kh.view(B, nkv, 1, S, d).expand(B, nkv, rep, S, d)\
                     .reshape(B * nh, S, d).contiguous()
-                vh = vh.view(B, nkv, 1, S, d).expand(B, nkv, rep, S, d)\
-                    .reshape(B * nh, S, d).contiguous()
+                vt = vt.view(B, nkv, 1, d, S).expand(B, nkv, rep, d, S)\
+                    .reshape(B * nh, d, S).contiguous()
             scores = K.gemm_nt(qh, kh, out_f32=True, alpha=scale)
             probs = K.softmax(scores, causal=True)
-            ctx = K.gemm_nt(probs, vh.transpose(1, 2).contiguous())
-            merged = ctx.view(B, nh, S, d).permute(0, 2, 1, 3)\
-                .reshape(B * S, nh * d).contiguous()
+            ctx = K.gemm_nt(probs, vt)
+            merged = K.merge_heads(ctx, B, S, nh, d)
             attn = K.gemm_nt(merged, self.w[p + "o.w"])
             x = K.add(x, attn)
             h2 = K.rmsnorm(x, self.w[p + "ln2.g"], eps=cfg.eps)
             gu = K.gemm_nt(h2, self.w[p + "gate_up.w"])
-            gate, up = gu[:, :cfg.ffn].contiguous(), gu[:, cfg.ffn:].contiguous()
-            mlp = K.gemm_nt(K.silu_mul(gate, up), self.w[p + "down.w"])
+            mlp = K.gemm_nt(K.silu_mul_fused(gu, cfg.ffn),
+                            self.w[p + "down.w"])
             x = K.add(x, mlp)
         self.lens[:B] = S_true
         x_last = x.view(B, S, H)[:, S_true - 1, :].contiguous()
@@ -158,7 +153,41 @@ class Phi3Decoder:
 
     def decode_step(self, ids: torch.Tensor) -> torch.Tensor:
         """ids [B] i32 (one token per row) -> logits [B,V] f32.
-        Appends to the KV cache at self.lens and increments it."""
+        Appends to the KV cache at self.lens and increments it.
+
+        Decode is launch-bound (~400 launches per token at L=32), so
+        fixed-batch steps replay a captured hipGraph: all mutable state
+        (lens, caches) lives in device tensors the graph reads/writes."""
+        B = ids.shape[0]
+        if not self.use_graph:
+            return self._decode_impl(ids)
+        entry = self._graphs.get(B)
+        if entry is None:
+            entry = self._capture_decode(ids)
+            self._graphs[B] = entry
+        graph, static_ids, static_out = entry
+        static_ids.copy_(ids, non_blocking=True)
+        graph.replay()
+        return static_out
+
+    def _capture_decode(self, ids: torch.Tensor):
+        assert int(self.lens[: ids.shape[0]].max().item()) + 4 <= self.max_seq, \
+            "no headroom to capture the decode graph"
+        lens_save = self.lens.clone()
+        static_ids = ids.clone()
+        stream = torch.cuda.Stream()
+        stream.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(stream):
+            for _ in range(2):
+                self._decode_impl(static_ids)   # warmup (state restored below)
+        torch.cuda.current_stream().wait_stream(stream)
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph):
+            static_out = self._decode_impl(static_ids)
+        self.lens.copy_(lens_save)  # undo warmup/capture side effects
+        return graph, static_ids, static_out
+
+    def _decode_impl(self, ids: torch.Tensor) -> torch.Tensor:
         cfg = self.cfg
         B = ids.shape[0]
         H, nh, nkv, d = cfg.hidden, cfg.heads, cfg.kv_heads, cfg.head_dim
@@ -183,8 +212,8 @@ class Phi3Decoder:
             x = K.add(x, attn)
             h2 = K.rmsnorm(x, self.w[p + "ln2.g"], eps=cfg.eps)
             gu = K.gemm_nt(h2, self.w[p + "gate_up.w"])
-            gate, up = gu[:, :cfg.ffn].contiguous(), gu[:, cfg.ffn:].contiguous()
-            mlp = K.gemm_nt(K.silu_mul(gate, up), self.w[p + "down.w"])
+            mlp = K.gemm_nt(K.silu_mul_fused(gu, cfg.ffn),
+                            self.w[p + "down.w"])
             x = K.add(x, mlp)
         self.lens[:B] += 1
         h = K.rmsnorm(x, self.w["final_ln.g"], eps=cfg.eps)

@@ -57,6 +57,14 @@ _SIGNATURES: dict[str, list] = {
                              c_float, c_void_p],
     "infomesh_kv_append": [c_void_p, c_void_p, c_void_p, c_void_p,
                            c_void_p, c_int, c_int, c_int, c_int, c_void_p],
+    "infomesh_qkv_split": [c_void_p, c_void_p, c_void_p, c_void_p,
+                           c_void_p, c_void_p, c_void_p,
+                           c_int, c_int, c_int, c_int, c_int, c_int,
+                           c_void_p],
+    "infomesh_merge_heads": [c_void_p, c_void_p, c_int, c_int, c_int,
+                             c_int, c_void_p],
+    "infomesh_silu_mul_fused": [c_void_p, c_void_p, c_long, c_int,
+                                c_void_p],
 }
 
 _RESTYPES = {"infomesh_topk_workspace_u32": c_long}

@@ -307,6 +307,47 @@ def attn_decode(q: torch.Tensor, k_cache: torch.Tensor,
     return out
 
 
+def qkv_split(qkv: torch.Tensor, B: int, S: int, nh: int, nkv: int,
+              d: int, cos_t: torch.Tensor | None = None,
+              sin_t: torch.Tensor | None = None,
+              pos: torch.Tensor | None = None,
+              rot: int | None = None
+              ) -> tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+    """Fused head split (+RoPE on q/k when tables given):
+    qkv [B*S, (nh+2nkv)*d] -> q [B*nh,S,d], k [B*nkv,S,d], vt [B*nkv,d,S]."""
+    _check(qkv, torch.bfloat16, "qkv")
+    dev = qkv.device
+    q = torch.empty(B * nh, S, d, device=dev, dtype=torch.bfloat16)
+    k = torch.empty(B * nkv, S, d, device=dev, dtype=torch.bfloat16)
+    vt = torch.empty(B * nkv, d, S, device=dev, dtype=torch.bfloat16)
+    _ext.lib().infomesh_qkv_split(
+        qkv.data_ptr(), q.data_ptr(), k.data_ptr(), vt.data_ptr(),
+        _ptr(cos_t), _ptr(sin_t), _ptr(pos),
+        B, S, nh, nkv, d, rot or d, _ext.stream_ptr())
+    return q, k, vt
+
+
+def merge_heads(ctx: torch.Tensor, B: int, S: int, nh: int,
+                d: int) -> torch.Tensor:
+    """[B*nh, S, d] -> [B*S, nh*d]."""
+    _check(ctx, torch.bfloat16, "ctx")
+    out = torch.empty(B * S, nh * d, device=ctx.device, dtype=torch.bfloat16)
+    _ext.lib().infomesh_merge_heads(ctx.data_ptr(), out.data_ptr(),
+                                    B, S, nh, d, _ext.stream_ptr())
+    return out
+
+
+def silu_mul_fused(gu: torch.Tensor, F: int) -> torch.Tensor:
+    """gu [rows, 2F] -> silu(gu[:, :F]) * gu[:, F:] without slicing."""
+    _check(gu, torch.bfloat16, "gu")
+    rows = gu.shape[0]
+    assert gu.shape[1] == 2 * F and F % 8 == 0
+    out = torch.empty(rows, F, device=gu.device, dtype=torch.bfloat16)
+    _ext.lib().infomesh_silu_mul_fused(gu.data_ptr(), out.data_ptr(),
+                                       rows, F, _ext.stream_ptr())
+    return out
+
+
 def kv_append(k_new: torch.Tensor, v_new: torch.Tensor,
               k_cache: torch.Tensor, v_cache: torch.Tensor,
               pos: torch.Tensor) -> None:

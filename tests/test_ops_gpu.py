@@ -319,6 +319,56 @@ def test_hamming_scan_parity():
     assert got == ref
 
 
+# ---------------------------------------------------------- reshape ops
+
+def test_qkv_split_no_rope():
+    B, S, nh, nkv, d = 2, 5, 3, 3, 32
+    qkv = torch.randn(B * S, (nh + 2 * nkv) * d, device="cuda").bfloat16()
+    q, k, vt = K.qkv_split(qkv, B, S, nh, nkv, d)
+    ref = qkv.view(B, S, nh + 2 * nkv, d).float()
+    rq = ref[:, :, :nh].permute(0, 2, 1, 3).reshape(B * nh, S, d)
+    rk = ref[:, :, nh:nh + nkv].permute(0, 2, 1, 3).reshape(B * nkv, S, d)
+    rv = ref[:, :, nh + nkv:].permute(0, 2, 1, 3).reshape(B * nkv, S, d)
+    assert torch.equal(q.float().cpu(), rq.cpu())
+    assert torch.equal(k.float().cpu(), rk.cpu())
+    assert torch.equal(vt.float().cpu(), rv.transpose(1, 2).contiguous().cpu())
+
+
+def test_qkv_split_with_rope():
+    B, S, nh, nkv, d = 1, 4, 2, 2, 64
+    qkv = torch.randn(B * S, (nh + 2 * nkv) * d, device="cuda").bfloat16()
+    inv = 1.0 / (10000 ** (torch.arange(d // 2).float() * 2 / d))
+    ang = torch.outer(torch.arange(16).float(), inv)
+    cos_t, sin_t = ang.cos().cuda(), ang.sin().cuda()
+    pos = torch.arange(S, dtype=torch.int32, device="cuda").repeat(B)
+    q, k, vt = K.qkv_split(qkv, B, S, nh, nkv, d, cos_t, sin_t, pos)
+    # reference: split then rope rows
+    ref = qkv.view(B * S, nh + 2 * nkv, d).clone()
+    qk_ref = R.rope(ref[:, :nh + nkv].contiguous(), cos_t.cpu(),
+                    sin_t.cpu(), pos.cpu())
+    rq = qk_ref[:, :nh].view(B, S, nh, d).permute(0, 2, 1, 3)\
+        .reshape(B * nh, S, d)
+    _assert_close(q, rq, atol=1e-2, what="qkv_split rope q")
+    rv = ref[:, nh + nkv:].float().view(B, S, nkv, d)\
+        .permute(0, 2, 3, 1).reshape(B * nkv, d, S)
+    assert torch.equal(vt.float().cpu(), rv.cpu())
+
+
+def test_merge_heads_roundtrip():
+    B, S, nh, d = 2, 7, 4, 32
+    ctx = torch.randn(B * nh, S, d, device="cuda").bfloat16()
+    out = K.merge_heads(ctx, B, S, nh, d)
+    ref = ctx.view(B, nh, S, d).permute(0, 2, 1, 3).reshape(B * S, nh * d)
+    assert torch.equal(out.cpu(), ref.cpu())
+
+
+def test_silu_mul_fused():
+    gu = torch.randn(6, 128, device="cuda").bfloat16()
+    out = K.silu_mul_fused(gu, 64)
+    ref = R.silu_mul(gu[:, :64].cpu(), gu[:, 64:].cpu())
+    _assert_close(out, ref, what="silu_mul_fused")
+
+
 # ----------------------------------------------------------- attn decode
 
 def test_attn_decode_parity():
