@@ -18,19 +18,19 @@ __global__ void qkv_split_kernel(
     const int* __restrict__ pos,
     int B, int S, int nh, int nkv, int d, int rot) {
   const int half = rot / 2;
-  const long total = (long)B * S * (nh + 2 * nkv) * (d / 2);
+  const long total = (long)B * S * (nh + nkv) * (d / 2);
   const long idx = (long)blockIdx.x * blockDim.x + threadIdx.x;
   const int dqkv = (nh + 2 * nkv) * d;
   for (long v = idx; v < total; v += (long)gridDim.x * blockDim.x) {
     // each work item handles a (row, head, d-pair): pair j = (j, j+half)
-    // for rope heads, or two adjacent elements (2*j, 2*j+1) for V.
+    // for roped heads, plain (2j, 2j+1) copies otherwise.
     const int j = (int)(v % (d / 2));
     long rest = v / (d / 2);
-    const int h = (int)(rest % (nh + 2 * nkv));
-    const long row = rest / (nh + 2 * nkv);       // b*S + s
+    const int h = (int)(rest % (nh + nkv));
+    const long row = rest / (nh + nkv);           // b*S + s
     const long b = row / S, s = row % S;
     const bf16* src = qkv + row * dqkv + (long)h * d;
-    if (h < nh + nkv) {
+    {
       // Q or K head: apply rope on the (j, j+half) pair
       float x0, x1;
       if (cos_t != nullptr && j < half) {
@@ -56,13 +56,41 @@ __global__ void qkv_split_kernel(
         dst[j0] = f2bf(x0);
         dst[j1] = f2bf(x1);
       }
-    } else {
-      // V head -> transposed [g, d, S] image for the PV NT-GEMM
-      const int hv = h - nh - nkv;
-      bf16* dst = vt + ((b * nkv + hv) * (long)d) * S + s;
-      dst[(long)(2 * j) * S] = src[2 * j];
-      dst[(long)(2 * j + 1) * S] = src[2 * j + 1];
     }
+    // V heads are handled by the LDS-tiled vt_transpose_kernel below
+    // (scatter stores here would be fully uncoalesced).
+  }
+}
+
+// V slice of qkv [B,S,(nh+2nkv)*d] -> vt [B*nkv, d, S] via a classic
+// 32x32 LDS-tiled transpose (coalesced along d on load, along s on
+// store). Grid: (ceil(S/32), ceil(d/32), B*nkv); block 32x8.
+__global__ __launch_bounds__(256) void vt_transpose_kernel(
+    const bf16* __restrict__ qkv, bf16* __restrict__ vt,
+    int B, int S, int nh, int nkv, int d) {
+  __shared__ bf16 tile[32][33];
+  const int g = blockIdx.z;                  // b*nkv + hv
+  const int b = g / nkv, hv = g % nkv;
+  const int s0 = blockIdx.x * 32, d0 = blockIdx.y * 32;
+  const int dqkv = (nh + 2 * nkv) * d;
+  const long voff = (long)(nh + nkv + hv) * d;
+  const int tx = threadIdx.x & 31, ty = threadIdx.x >> 5;   // 32x8
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int s = s0 + ty + 8 * i;
+    const int dd = d0 + tx;
+    if (s < S && dd < d)
+      tile[ty + 8 * i][tx] =
+          qkv[((long)b * S + s) * dqkv + voff + dd];
+  }
+  __syncthreads();
+  bf16* out = vt + (long)g * d * S;
+#pragma unroll
+  for (int i = 0; i < 4; ++i) {
+    const int dd = d0 + ty + 8 * i;
+    const int s = s0 + tx;
+    if (s < S && dd < d)
+      out[(long)dd * S + s] = tile[tx][ty + 8 * i];
   }
 }
 
@@ -121,11 +149,14 @@ extern "C" void infomesh_qkv_split(
     const void* cos_t, const void* sin_t, const void* pos,
     int B, int S, int nh, int nkv, int d, int rot, void* stream) {
   auto s = reinterpret_cast<hipStream_t>(stream);
-  const long total = (long)B * S * (nh + 2 * nkv) * (d / 2);
+  const long total = (long)B * S * (nh + nkv) * (d / 2);
   hipLaunchKernelGGL(qkv_split_kernel, gs(total), dim3(256), 0, s,
                      (const bf16*)qkv, (bf16*)q, (bf16*)k, (bf16*)vt,
                      (const float*)cos_t, (const float*)sin_t,
                      (const int*)pos, B, S, nh, nkv, d, rot);
+  dim3 grid((S + 31) / 32, (d + 31) / 32, B * nkv);
+  hipLaunchKernelGGL(vt_transpose_kernel, grid, dim3(256), 0, s,
+                     (const bf16*)qkv, (bf16*)vt, B, S, nh, nkv, d);
 }
 
 extern "C" void infomesh_merge_heads(const void* ctx, void* out,

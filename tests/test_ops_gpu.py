@@ -343,7 +343,7 @@ def test_qkv_split_with_rope():
     pos = torch.arange(S, dtype=torch.int32, device="cuda").repeat(B)
     q, k, vt = K.qkv_split(qkv, B, S, nh, nkv, d, cos_t, sin_t, pos)
     # reference: split then rope rows
-    ref = qkv.view(B * S, nh + 2 * nkv, d).clone()
+    ref = qkv.view(B * S, nh + 2 * nkv, d).clone().cpu()
     qk_ref = R.rope(ref[:, :nh + nkv].contiguous(), cos_t.cpu(),
                     sin_t.cpu(), pos.cpu())
     rq = qk_ref[:, :nh].view(B, S, nh, d).permute(0, 2, 1, 3)\
