@@ -6,8 +6,11 @@ Usage: python scripts/gpu_probe.py [encoder|rerank|decode|all]
 """
 from __future__ import annotations
 
+import pathlib
 import sys
 import time
+
+sys.path.insert(0, str(pathlib.Path(__file__).resolve().parents[1]))
 
 import torch
 
