@@ -56,10 +56,17 @@ def gemm_nt(a: torch.Tensor, b: torch.Tensor,
         out = torch.empty((G, M, N), device=a.device, dtype=dtype)
     else:
         assert out.shape == (G, M, N) and out.dtype == dtype and out.is_contiguous()
-    _ext.lib().infomesh_gemm_bf16_nt(
-        a.data_ptr(), b.data_ptr(), out.data_ptr(), _ptr(bias),
-        M, N, K, G, M * K, strideB, M * N,
-        ACT[act], alpha, int(out_f32), _ext.stream_ptr())
+    if M <= 16:
+        # skinny-M decode path: wave-per-column GEMV (csrc/gemv.hip)
+        _ext.lib().infomesh_gemv_bf16_nt(
+            a.data_ptr(), b.data_ptr(), out.data_ptr(), _ptr(bias),
+            M, N, K, G, M * K, strideB, M * N,
+            ACT[act], alpha, int(out_f32), _ext.stream_ptr())
+    else:
+        _ext.lib().infomesh_gemm_bf16_nt(
+            a.data_ptr(), b.data_ptr(), out.data_ptr(), _ptr(bias),
+            M, N, K, G, M * K, strideB, M * N,
+            ACT[act], alpha, int(out_f32), _ext.stream_ptr())
     return out.squeeze(0) if sq_a else out
 
 

@@ -52,6 +52,27 @@ def test_gemm_nt_shapes(M, N, Kd):
                   what=f"gemm {M}x{N}x{Kd}")
 
 
+@pytest.mark.parametrize("M,N,Kd", [(1, 9216, 3072), (4, 100, 64),
+                                    (16, 32064, 3072), (8, 3072, 8192)])
+def test_gemv_skinny_m(M, N, Kd):
+    """M<=16 dispatches to the wave-per-column GEMV kernel."""
+    a = torch.randn(M, Kd, device="cuda").bfloat16()
+    b = torch.randn(N, Kd, device="cuda").bfloat16()
+    bias = torch.randn(N, device="cuda")
+    out = K.gemm_nt(a, b, bias=bias, act="silu", out_f32=True)
+    ref = R.gemm_nt(a.cpu(), b.cpu(), bias.cpu(), act="silu")
+    _assert_close(out, ref, rtol=3e-2, atol=Kd ** 0.5 * 2e-2,
+                  what=f"gemv {M}x{N}x{Kd}")
+
+
+def test_gemv_batched():
+    G, M, N, Kd = 5, 2, 64, 96
+    a = torch.randn(G, M, Kd, device="cuda").bfloat16()
+    b = torch.randn(G, N, Kd, device="cuda").bfloat16()
+    out = K.gemm_nt(a, b)
+    _assert_close(out, R.gemm_nt(a.cpu(), b.cpu()), what="gemv batched")
+
+
 def test_gemm_asymmetric_catches_transpose():
     # Asymmetric B (guide: A=I with asymmetric B catches row/col swap).
     M = N = Kd = 128
