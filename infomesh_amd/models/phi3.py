@@ -194,29 +194,28 @@ class Phi3Decoder:
         pos = self.lens[:B].contiguous()
         x = K.gather(self.w["embed"], ids)
         scale = d ** -0.5
+        new_lens = (pos + 1).contiguous()
+        # residual carried through fused rmsnorm(+add) — no standalone adds
+        delta, res = x, None
         for i in range(cfg.layers):
             p = f"layer.{i}."
-            h = K.rmsnorm(x, self.w[p + "ln1.g"], eps=cfg.eps)
+            h, res = K.rmsnorm(delta, self.w[p + "ln1.g"], residual=res,
+                               eps=cfg.eps, return_residual=True)
             qkv = K.gemm_nt(h, self.w[p + "qkv.w"])
-            q, k, v = self._split_qkv(qkv, B, 1)
-            q = q.reshape(B, nh, d).contiguous()
-            k = k.reshape(B, nkv, d).contiguous()
-            v = v.reshape(B, nkv, d).contiguous()
-            K.rope(q, self.cos, self.sin, pos)
-            K.rope(k, self.cos, self.sin, pos)
-            K.kv_append(k, v, self.k_cache[i], self.v_cache[i], pos)
-            new_lens = pos + 1
-            ctx = K.attn_decode(q, self.k_cache[i][:B], self.v_cache[i][:B],
-                                new_lens.contiguous(), scale)
+            q, k, vt = K.qkv_split(qkv, B, 1, nh, nkv, d,
+                                   cos_t=self.cos, sin_t=self.sin, pos=pos)
+            K.kv_append(k.view(B, nkv, d), vt.view(B, nkv, d),
+                        self.k_cache[i], self.v_cache[i], pos)
+            ctx = K.attn_decode(q.view(B, nh, d), self.k_cache[i][:B],
+                                self.v_cache[i][:B], new_lens, scale)
             attn = K.gemm_nt(ctx.view(B, nh * d), self.w[p + "o.w"])
-            x = K.add(x, attn)
-            h2 = K.rmsnorm(x, self.w[p + "ln2.g"], eps=cfg.eps)
+            h2, res = K.rmsnorm(attn, self.w[p + "ln2.g"], residual=res,
+                                eps=cfg.eps, return_residual=True)
             gu = K.gemm_nt(h2, self.w[p + "gate_up.w"])
-            mlp = K.gemm_nt(K.silu_mul_fused(gu, cfg.ffn),
-                            self.w[p + "down.w"])
-            x = K.add(x, mlp)
+            delta = K.gemm_nt(K.silu_mul_fused(gu, cfg.ffn),
+                              self.w[p + "down.w"])
         self.lens[:B] += 1
-        h = K.rmsnorm(x, self.w["final_ln.g"], eps=cfg.eps)
+        h = K.rmsnorm(delta, self.w["final_ln.g"], residual=res, eps=cfg.eps)
         return K.gemm_nt(h, self.w["lm_head"], out_f32=True)
 
     def generate_greedy(self, prompt_ids: torch.Tensor,

@@ -55,6 +55,9 @@ _SIGNATURES: dict[str, list] = {
     "infomesh_attn_decode": [c_void_p, c_void_p, c_void_p, c_void_p,
                              c_void_p, c_int, c_int, c_int, c_int, c_int,
                              c_float, c_void_p],
+    "infomesh_attn_decode_split": [c_void_p, c_void_p, c_void_p, c_void_p,
+                                   c_void_p, c_void_p, c_int, c_int, c_int,
+                                   c_int, c_int, c_int, c_float, c_void_p],
     "infomesh_kv_append": [c_void_p, c_void_p, c_void_p, c_void_p,
                            c_void_p, c_int, c_int, c_int, c_int, c_void_p],
     "infomesh_gemv_bf16_nt": [c_void_p, c_void_p, c_void_p, c_void_p,

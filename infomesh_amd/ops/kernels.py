@@ -105,7 +105,10 @@ def rmsnorm(x: torch.Tensor, gamma: torch.Tensor,
     _ext.lib().infomesh_rmsnorm(
         x.data_ptr(), _ptr(residual), out.data_ptr(), _ptr(res_out),
         gamma.data_ptr(), rows, H, eps, _ext.stream_ptr())
-    return (out, res_out) if return_residual else out
+    if return_residual:
+        # with no residual input, the running residual IS the input
+        return out, (res_out if res_out is not None else x)
+    return out
 
 
 def softmax(scores: torch.Tensor, scale: float = 1.0, causal: bool = False,
@@ -301,16 +304,30 @@ def hamming_scan(queries: torch.Tensor, table: torch.Tensor,
 def attn_decode(q: torch.Tensor, k_cache: torch.Tensor,
                 v_cache: torch.Tensor, lens: torch.Tensor,
                 scale: float) -> torch.Tensor:
-    """q [B,H,D], caches [B,Hkv,Smax,D], lens [B] i32 -> out [B,H,D]."""
+    """q [B,H,D], caches [B,Hkv,Smax,D], lens [B] i32 -> out [B,H,D].
+
+    Uses flash-decoding sequence splitting when B*H alone cannot fill
+    the chip (partials per chunk + combine kernel)."""
+    import math
     B, H, D = q.shape
     _, Hkv, Smax, _ = k_cache.shape
     _check(q, torch.bfloat16, "q")
     assert D % 8 == 0 and D <= 128
     out = torch.empty_like(q)
-    _ext.lib().infomesh_attn_decode(
-        q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
-        lens.data_ptr(), out.data_ptr(), B, H, Hkv, Smax, D, scale,
-        _ext.stream_ptr())
+    nc_min = max(1, math.ceil(Smax / 2048))
+    nc = max(nc_min, min(64, math.ceil(512 / max(1, B * H))))
+    if nc <= 1:
+        _ext.lib().infomesh_attn_decode(
+            q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
+            lens.data_ptr(), out.data_ptr(), B, H, Hkv, Smax, D, scale,
+            _ext.stream_ptr())
+    else:
+        part = torch.empty(B, H, nc, D + 2, device=q.device,
+                           dtype=torch.float32)
+        _ext.lib().infomesh_attn_decode_split(
+            q.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
+            lens.data_ptr(), part.data_ptr(), out.data_ptr(),
+            B, H, Hkv, Smax, D, nc, scale, _ext.stream_ptr())
     return out
 
 
