@@ -74,9 +74,13 @@ def init_bert_weights(cfg: BertConfig, seed: int = 1234,
 
 
 class BertEncoder:
-    def __init__(self, cfg: BertConfig, weights: dict[str, torch.Tensor]):
+    def __init__(self, cfg: BertConfig, weights: dict[str, torch.Tensor],
+                 use_fused_attn: bool = True):
         self.cfg = cfg
         self.w = weights
+        # fused flash kernel supports d in {32,64,96,128}
+        self.use_fused_attn = use_fused_attn and cfg.head_dim in (32, 64,
+                                                                  96, 128)
 
     # ------------------------------------------------------------- GPU
     def forward(self, ids: torch.Tensor, lens: torch.Tensor) -> torch.Tensor:
@@ -99,14 +103,22 @@ class BertEncoder:
         # per-(batch*head) valid key lengths for the softmax mask
         vl = lens.repeat_interleave(nh).contiguous()
         scale = d ** -0.5
+        vl_b = lens.contiguous()
         for i in range(cfg.layers):
             p = f"layer.{i}."
             qkv = K.gemm_nt(x, w[p + "qkv.w"], bias=w[p + "qkv.b"])
-            # fused head split producing the PV-ready transposed V image
-            q, k, vt = K.qkv_split(qkv, B, S, nh, nh, d)
-            scores = K.gemm_nt(q, k, out_f32=True, alpha=scale)
-            probs = K.softmax(scores, valid_len=vl)
-            ctx = K.gemm_nt(probs, vt)                   # [G, S, d]
+            if self.use_fused_attn:
+                # zero-copy strided head views straight out of qkv
+                qkv5 = qkv.view(B, S, 3, nh, d)
+                qv = qkv5[:, :, 0].permute(0, 2, 1, 3)
+                kv = qkv5[:, :, 1].permute(0, 2, 1, 3)
+                vv = qkv5[:, :, 2].permute(0, 2, 1, 3)
+                ctx = K.attn_fused(qv, kv, vv, valid_len=vl_b, scale=scale)
+            else:
+                q, k, vt = K.qkv_split(qkv, B, S, nh, nh, d)
+                scores = K.gemm_nt(q, k, out_f32=True, alpha=scale)
+                probs = K.softmax(scores, valid_len=vl)
+                ctx = K.gemm_nt(probs, vt)               # [G, S, d]
             merged = K.merge_heads(ctx, B, S, nh, d)
             attn = K.gemm_nt(merged, w[p + "attn_out.w"],
                              bias=w[p + "attn_out.b"])

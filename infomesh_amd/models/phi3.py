@@ -119,25 +119,34 @@ class Phi3Decoder:
         pos = torch.arange(S, device=self.device, dtype=torch.int32).repeat(B)
         x = K.gather(self.w["embed"], ids.reshape(-1))       # [B*S, H]
         scale = d ** -0.5
+        fused = d in (32, 64, 96, 128)
         for i in range(cfg.layers):
             p = f"layer.{i}."
             h = K.rmsnorm(x, self.w[p + "ln1.g"], eps=cfg.eps)
             qkv = K.gemm_nt(h, self.w[p + "qkv.w"])
             qh, kh, vt = K.qkv_split(qkv, B, S, nh, nkv, d,
                                      cos_t=self.cos, sin_t=self.sin,
-                                     pos=pos)
+                                     pos=pos, want_vt=not fused)
+            v_view = qkv.view(B, S, nh + 2 * nkv, d)[:, :, nh + nkv:]\
+                .permute(0, 2, 1, 3)                    # [B, nkv, S, d]
             self.k_cache[i][:B, :, :S] = kh.view(B, nkv, S, d)
-            self.v_cache[i][:B, :, :S] = \
-                vt.view(B, nkv, d, S).permute(0, 1, 3, 2)
-            if nh != nkv:
-                rep = nh // nkv
-                kh = kh.view(B, nkv, 1, S, d).expand(B, nkv, rep, S, d)\
-                    .reshape(B * nh, S, d).contiguous()
-                vt = vt.view(B, nkv, 1, d, S).expand(B, nkv, rep, d, S)\
-                    .reshape(B * nh, d, S).contiguous()
-            scores = K.gemm_nt(qh, kh, out_f32=True, alpha=scale)
-            probs = K.softmax(scores, causal=True)
-            ctx = K.gemm_nt(probs, vt)
+            self.v_cache[i][:B, :, :S] = v_view
+            if fused:
+                ctx = K.attn_fused(qh.view(B, nh, S, d),
+                                   kh.view(B, nkv, S, d), v_view,
+                                   causal=True, scale=scale)
+            else:
+                if nh != nkv:
+                    rep = nh // nkv
+                    kh = kh.view(B, nkv, 1, S, d)\
+                        .expand(B, nkv, rep, S, d)\
+                        .reshape(B * nh, S, d).contiguous()
+                    vt = vt.view(B, nkv, 1, d, S)\
+                        .expand(B, nkv, rep, d, S)\
+                        .reshape(B * nh, d, S).contiguous()
+                scores = K.gemm_nt(qh, kh, out_f32=True, alpha=scale)
+                probs = K.softmax(scores, causal=True)
+                ctx = K.gemm_nt(probs, vt)
             merged = K.merge_heads(ctx, B, S, nh, d)
             attn = K.gemm_nt(merged, self.w[p + "o.w"])
             x = K.add(x, attn)

@@ -58,6 +58,11 @@ _SIGNATURES: dict[str, list] = {
     "infomesh_attn_decode_split": [c_void_p, c_void_p, c_void_p, c_void_p,
                                    c_void_p, c_void_p, c_int, c_int, c_int,
                                    c_int, c_int, c_int, c_float, c_void_p],
+    "infomesh_attn_fused": [c_void_p, c_void_p, c_void_p, c_void_p,
+                            c_void_p, c_int, c_int, c_int, c_int, c_int,
+                            c_int, c_long, c_long, c_long, c_long, c_long,
+                            c_long, c_long, c_long, c_long,
+                            c_int, c_float, c_void_p],
     "infomesh_kv_append": [c_void_p, c_void_p, c_void_p, c_void_p,
                            c_void_p, c_int, c_int, c_int, c_int, c_void_p],
     "infomesh_gemv_bf16_nt": [c_void_p, c_void_p, c_void_p, c_void_p,

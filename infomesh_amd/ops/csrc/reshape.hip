@@ -154,9 +154,11 @@ extern "C" void infomesh_qkv_split(
                      (const bf16*)qkv, (bf16*)q, (bf16*)k, (bf16*)vt,
                      (const float*)cos_t, (const float*)sin_t,
                      (const int*)pos, B, S, nh, nkv, d, rot);
-  dim3 grid((S + 31) / 32, (d + 31) / 32, B * nkv);
-  hipLaunchKernelGGL(vt_transpose_kernel, grid, dim3(256), 0, s,
-                     (const bf16*)qkv, (bf16*)vt, B, S, nh, nkv, d);
+  if (vt != nullptr) {
+    dim3 grid((S + 31) / 32, (d + 31) / 32, B * nkv);
+    hipLaunchKernelGGL(vt_transpose_kernel, grid, dim3(256), 0, s,
+                       (const bf16*)qkv, (bf16*)vt, B, S, nh, nkv, d);
+  }
 }
 
 extern "C" void infomesh_merge_heads(const void* ctx, void* out,

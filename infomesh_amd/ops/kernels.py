@@ -301,6 +301,36 @@ def hamming_scan(queries: torch.Tensor, table: torch.Tensor,
     return out_q[:n], out_n[:n], out_d[:n]
 
 
+def attn_fused(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+               valid_len: torch.Tensor | None = None,
+               causal: bool = False, scale: float | None = None
+               ) -> torch.Tensor:
+    """Fused flash attention: q [B,nh,Sq,d], k/v [B,nhk,Sk,d] — any
+    batch/head/row strides (element stride must be 1, d in {32,64,96,
+    128}). GQA via nh % nhk == 0. valid_len: [B] i32 key limits.
+    Returns O [B*nh, Sq, d] bf16 contiguous."""
+    assert q.dim() == 4 and k.dim() == 4 and v.dim() == 4
+    B, nh, Sq, d = q.shape
+    _, nhk, Sk, dk = k.shape
+    assert d == dk and d in (32, 64, 96, 128) and nh % nhk == 0
+    for t in (q, k, v):
+        assert t.dtype == torch.bfloat16 and t.stride(3) == 1
+    if scale is None:
+        scale = d ** -0.5
+    out = torch.empty(B * nh, Sq, d, device=q.device, dtype=torch.bfloat16)
+    if valid_len is not None:
+        _check(valid_len, torch.int32, "valid_len")
+        assert valid_len.numel() == B
+    _ext.lib().infomesh_attn_fused(
+        q.data_ptr(), k.data_ptr(), v.data_ptr(), _ptr(valid_len),
+        out.data_ptr(), B, nh, nhk, Sq, Sk, d,
+        q.stride(0), q.stride(1), q.stride(2),
+        k.stride(0), k.stride(1), k.stride(2),
+        v.stride(0), v.stride(1), v.stride(2),
+        int(causal), scale, _ext.stream_ptr())
+    return out
+
+
 def attn_decode(q: torch.Tensor, k_cache: torch.Tensor,
                 v_cache: torch.Tensor, lens: torch.Tensor,
                 scale: float) -> torch.Tensor:
@@ -335,17 +365,20 @@ def qkv_split(qkv: torch.Tensor, B: int, S: int, nh: int, nkv: int,
               d: int, cos_t: torch.Tensor | None = None,
               sin_t: torch.Tensor | None = None,
               pos: torch.Tensor | None = None,
-              rot: int | None = None
-              ) -> tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+              rot: int | None = None, want_vt: bool = True
+              ) -> tuple[torch.Tensor, torch.Tensor, torch.Tensor | None]:
     """Fused head split (+RoPE on q/k when tables given):
-    qkv [B*S, (nh+2nkv)*d] -> q [B*nh,S,d], k [B*nkv,S,d], vt [B*nkv,d,S]."""
+    qkv [B*S, (nh+2nkv)*d] -> q [B*nh,S,d], k [B*nkv,S,d], and (when
+    want_vt) the PV-GEMM-ready transposed vt [B*nkv,d,S]; the fused
+    attention path reads V strided from qkv instead (want_vt=False)."""
     _check(qkv, torch.bfloat16, "qkv")
     dev = qkv.device
     q = torch.empty(B * nh, S, d, device=dev, dtype=torch.bfloat16)
     k = torch.empty(B * nkv, S, d, device=dev, dtype=torch.bfloat16)
-    vt = torch.empty(B * nkv, d, S, device=dev, dtype=torch.bfloat16)
+    vt = torch.empty(B * nkv, d, S, device=dev, dtype=torch.bfloat16) \
+        if want_vt else None
     _ext.lib().infomesh_qkv_split(
-        qkv.data_ptr(), q.data_ptr(), k.data_ptr(), vt.data_ptr(),
+        qkv.data_ptr(), q.data_ptr(), k.data_ptr(), _ptr(vt),
         _ptr(cos_t), _ptr(sin_t), _ptr(pos),
         B, S, nh, nkv, d, rot or d, _ext.stream_ptr())
     return q, k, vt

@@ -340,6 +340,83 @@ def test_hamming_scan_parity():
     assert got == ref
 
 
+# ------------------------------------------------------- fused attention
+
+def _attn_ref(q, k, v, valid=None, causal=False, scale=None):
+    B, nh, Sq, d = q.shape
+    _, nhk, Sk, _ = k.shape
+    rep = nh // nhk
+    kf = k.float().repeat_interleave(rep, dim=1)
+    vf = v.float().repeat_interleave(rep, dim=1)
+    s = torch.einsum("bhqd,bhkd->bhqk", q.float(), kf) * (scale or d ** -0.5)
+    mask = torch.zeros(B, 1, Sq, Sk, dtype=torch.bool)
+    if causal:
+        i = torch.arange(Sq).view(Sq, 1)
+        j = torch.arange(Sk).view(1, Sk)
+        mask = mask | (j > i + (Sk - Sq)).view(1, 1, Sq, Sk)
+    if valid is not None:
+        mask = mask | (torch.arange(Sk).view(1, 1, 1, Sk)
+                       >= valid.view(B, 1, 1, 1))
+    s = s.masked_fill(mask, float("-inf"))
+    p = torch.nan_to_num(torch.softmax(s, -1), nan=0.0)
+    return torch.einsum("bhqk,bhkd->bhqd", p, vf).reshape(B * nh, Sq, d)
+
+
+@pytest.mark.parametrize("D", [32, 64, 96, 128])
+def test_attn_fused_basic(D):
+    B, nh, S = 2, 3, 80
+    q = torch.randn(B, nh, S, D, device="cuda").bfloat16()
+    k = torch.randn(B, nh, S, D, device="cuda").bfloat16()
+    v = torch.randn(B, nh, S, D, device="cuda").bfloat16()
+    out = K.attn_fused(q, k, v)
+    ref = _attn_ref(q.cpu(), k.cpu(), v.cpu())
+    _assert_close(out, ref, atol=3e-2, what=f"attn_fused D={D}")
+
+
+def test_attn_fused_valid_len():
+    B, nh, S, D = 3, 2, 100, 64
+    q = torch.randn(B, nh, S, D, device="cuda").bfloat16()
+    k = torch.randn(B, nh, S, D, device="cuda").bfloat16()
+    v = torch.randn(B, nh, S, D, device="cuda").bfloat16()
+    vl = torch.tensor([100, 17, 1], device="cuda", dtype=torch.int32)
+    out = K.attn_fused(q, k, v, valid_len=vl)
+    ref = _attn_ref(q.cpu(), k.cpu(), v.cpu(), valid=vl.cpu())
+    _assert_close(out, ref, atol=3e-2, what="attn_fused masked")
+
+
+def test_attn_fused_causal_and_gqa():
+    B, nh, nhk, S, D = 2, 8, 2, 130, 96
+    q = torch.randn(B, nh, S, D, device="cuda").bfloat16()
+    k = torch.randn(B, nhk, S, D, device="cuda").bfloat16()
+    v = torch.randn(B, nhk, S, D, device="cuda").bfloat16()
+    out = K.attn_fused(q, k, v, causal=True)
+    ref = _attn_ref(q.cpu(), k.cpu(), v.cpu(), causal=True)
+    _assert_close(out, ref, atol=3e-2, what="attn_fused causal gqa")
+
+
+def test_attn_fused_strided_views():
+    """Q/K/V as strided views of a fused qkv buffer (the model path)."""
+    B, S, nh, D = 2, 33, 4, 32
+    qkv = torch.randn(B, S, 3, nh, D, device="cuda").bfloat16()
+    qv = qkv[:, :, 0].permute(0, 2, 1, 3)
+    kv = qkv[:, :, 1].permute(0, 2, 1, 3)
+    vv = qkv[:, :, 2].permute(0, 2, 1, 3)
+    out = K.attn_fused(qv, kv, vv)
+    ref = _attn_ref(qv.contiguous().cpu(), kv.contiguous().cpu(),
+                    vv.contiguous().cpu())
+    _assert_close(out, ref, atol=3e-2, what="attn_fused strided")
+
+
+def test_attn_fused_long_seq_causal():
+    B, nh, S, D = 1, 2, 555, 128
+    q = torch.randn(B, nh, S, D, device="cuda").bfloat16()
+    k = torch.randn(B, nh, S, D, device="cuda").bfloat16()
+    v = torch.randn(B, nh, S, D, device="cuda").bfloat16()
+    out = K.attn_fused(q, k, v, causal=True)
+    ref = _attn_ref(q.cpu(), k.cpu(), v.cpu(), causal=True)
+    _assert_close(out, ref, atol=3e-2, what="attn_fused long causal")
+
+
 # ---------------------------------------------------------- reshape ops
 
 def test_qkv_split_no_rope():
