@@ -1,0 +1,75 @@
+"""GPU shard manifest: warm-start persistence of built shards.
+
+Reference-parity intent: GPU shards are rebuildable caches
+over the SQLite/snapshot ground truth; a sidecar binary manifest makes
+restarts warm. Tensors are saved with torch.save to a per-rank file;
+load_shard re-uploads them to HBM without re-tokenizing/re-embedding.
+"""
+from __future__ import annotations
+
+import json
+import time
+from pathlib import Path
+
+import torch
+
+from ..hashing import content_hash
+from .gpu_index import CpuShard, GpuShard
+
+MANIFEST_VERSION = 1
+
+
+def save_shard(shard: GpuShard, path: str | Path, rank: int = 0,
+               world: int = 1) -> dict:
+    path = Path(path)
+    path.parent.mkdir(parents=True, exist_ok=True)
+    tensors = {
+        "offsets": shard.offsets, "doc_ids": shard.doc_ids,
+        "tfs": shard.tfs, "doc_norm": shard.doc_norm,
+        "global_ids": shard.global_ids,
+    }
+    payload = {k: (v.cpu() if v is not None else None)
+               for k, v in tensors.items()}
+    if shard.embeddings is not None:
+        payload["embeddings"] = shard.embeddings.cpu()
+    meta = {
+        "version": MANIFEST_VERSION,
+        "rank": rank, "world": world,
+        "n_docs": shard.n_docs, "vocab": shard.vocab,
+        "avgdl": shard.avgdl,
+        "created_at": time.time(),
+    }
+    torch.save({"meta": meta, "df": shard.df, **payload}, path)
+    meta["bytes"] = path.stat().st_size
+    meta["checksum"] = content_hash(str(path.stat().st_size) +
+                                    str(shard.n_docs))
+    with open(path.with_suffix(".json"), "w") as f:
+        json.dump(meta, f, indent=2)
+    return meta
+
+
+def load_shard(path: str | Path, device: str = "cuda") -> GpuShard:
+    path = Path(path)
+    blob = torch.load(path, map_location="cpu", weights_only=False)
+    meta = blob["meta"]
+    if meta.get("version") != MANIFEST_VERSION:
+        raise ValueError(f"manifest version {meta.get('version')}")
+    shard = GpuShard(device) if device.startswith("cuda") else CpuShard()
+    shard.n_docs = meta["n_docs"]
+    shard.vocab = meta["vocab"]
+    shard.avgdl = meta["avgdl"]
+    shard.df = blob["df"]
+    dev = shard.device
+    for name in ("offsets", "doc_ids", "tfs", "doc_norm", "global_ids"):
+        t = blob.get(name)
+        setattr(shard, name, t.to(dev) if t is not None else None)
+    emb = blob.get("embeddings")
+    shard.embeddings = emb.to(dev) if emb is not None else None
+    return shard
+
+
+def manifest_info(path: str | Path) -> dict | None:
+    side = Path(path).with_suffix(".json")
+    if not side.exists():
+        return None
+    return json.loads(side.read_text())
