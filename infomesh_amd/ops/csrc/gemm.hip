@@ -12,12 +12,15 @@
 // (infomesh/summarizer/engine.py:111-318) — see SURVEY.md §2.9.
 #include "common.h"
 
-#define BM 128
 #define BN 128
 
 namespace {
 
-template <int BK, bool OUT_F32>
+// BM=128: 4 waves as 2x2, each owning a 64x64 sub-tile (4x4 fragments).
+// BM=64: 4 waves as 1x4 over N, each owning 64x32 -> 4x2 fragments —
+// used for skinny-M shapes (e.g. query-batch x doc-embedding scoring)
+// where a 128-row A tile would be half padding.
+template <int BM, int BK, bool OUT_F32>
 __global__ __launch_bounds__(256) void gemm_bf16_nt_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ B,
     void* __restrict__ C, const float* __restrict__ bias,
@@ -40,14 +43,20 @@ __global__ __launch_bounds__(256) void gemm_bf16_nt_kernel(
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wid = tid >> 6;          // 4 waves
-  const int wm = wid >> 1, wn = wid & 1;
+  constexpr int WM = (BM == 128) ? 2 : 1;   // wave grid M
+  constexpr int WN = 4 / WM;                // wave grid N
+  constexpr int FM = (BM / WM) / 16;        // m-frags per wave (4 or 4)
+  constexpr int FN = (BN / WN) / 16;        // n-frags per wave (4 or 2)
+  const int wm = wid / WN, wn = wid % WN;
 
   // ---- glds staging geometry -------------------------------------------
   // One glds instruction: 64 lanes x 16 B = 1 KiB = 8 rows of BK=64 bf16
   // (or 4 rows-pairs when BK=32: 1 KiB = 16 rows of 64 B).
   // A tile is BM*BK*2 bytes = BM*BK/512 KiB -> chunks of 1 KiB each.
   constexpr int ROWS_PER_CHUNK = 1024 / (BK * 2);      // 8 (BK=64) / 16 (BK=32)
-  constexpr int CHUNKS = BM / ROWS_PER_CHUNK;          // 16 (BK=64) / 8 (BK=32)
+  // Chunk count follows the larger (B) tile; A stages only its first
+  // BM*BK*2/1024 chunks (guarded below) when BM < BN.
+  constexpr int CHUNKS = (BN * BK * 2) / 1024;
   constexpr int CHUNKS_PER_WAVE = CHUNKS / 4;
   const int lanes_per_row = BK / 8;                     // 8 elements per lane
   const int lrow = lane / lanes_per_row;
@@ -55,33 +64,35 @@ __global__ __launch_bounds__(256) void gemm_bf16_nt_kernel(
 
   const int n_ksteps = K / BK;
 
+  constexpr int BCHUNKS = (BN * BK * 2) / 1024;      // B half count
   auto stage = [&](int buf, int kstep) {
     const long k0 = (long)kstep * BK;
 #pragma unroll
     for (int c = 0; c < CHUNKS_PER_WAVE; ++c) {
       const int chunk = wid * CHUNKS_PER_WAVE + c;
       const int row = chunk * ROWS_PER_CHUNK + lrow;
-      // A rows clamp to M-1 (harmless: those C rows are never stored).
-      int arow = m0 + row; arow = arow < M ? arow : M - 1;
       int brow = n0 + row; brow = brow < N ? brow : N - 1;
-      const bf16* ga = Ag + (long)arow * K + k0 + lcol;
       const bf16* gb = Bg + (long)brow * K + k0 + lcol;
-      auto* la = (__attribute__((address_space(3))) unsigned int*)
-          &smem[buf][0][chunk * 512];
       auto* lb = (__attribute__((address_space(3))) unsigned int*)
           &smem[buf][1][chunk * 512];
       __builtin_amdgcn_global_load_lds(
-          (const __attribute__((address_space(1))) unsigned int*)ga, la, 16, 0, 0);
-      __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) unsigned int*)gb, lb, 16, 0, 0);
+      if (chunk < (BM * BK * 2) / 1024) {   // A tile is smaller at BM=64
+        int arow = m0 + row; arow = arow < M ? arow : M - 1;
+        const bf16* ga = Ag + (long)arow * K + k0 + lcol;
+        auto* la = (__attribute__((address_space(3))) unsigned int*)
+            &smem[buf][0][chunk * 512];
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) unsigned int*)ga, la, 16, 0, 0);
+      }
     }
   };
 
-  f32x4 acc[4][4];
+  f32x4 acc[FM][FN];
 #pragma unroll
-  for (int i = 0; i < 4; ++i)
+  for (int i = 0; i < FM; ++i)
 #pragma unroll
-    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+    for (int j = 0; j < FN; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   stage(0, 0);
   __syncthreads();
@@ -94,23 +105,23 @@ __global__ __launch_bounds__(256) void gemm_bf16_nt_kernel(
     if (t + 1 < n_ksteps) stage(cur ^ 1, t + 1);
 #pragma unroll
     for (int ks = 0; ks < BK / 32; ++ks) {
-      bf16x8 a[4], b[4];
+      bf16x8 a[FM], b[FN];
 #pragma unroll
-      for (int i = 0; i < 4; ++i) {
-        const int row = wm * 64 + i * 16 + fr;
+      for (int i = 0; i < FM; ++i) {
+        const int row = wm * (BM / WM) + i * 16 + fr;
         a[i] = *reinterpret_cast<const bf16x8*>(
             &smem[cur][0][row * BK + ks * 32 + fk]);
       }
 #pragma unroll
-      for (int j = 0; j < 4; ++j) {
-        const int row = wn * 64 + j * 16 + fr;
+      for (int j = 0; j < FN; ++j) {
+        const int row = wn * (BN / WN) + j * 16 + fr;
         b[j] = *reinterpret_cast<const bf16x8*>(
             &smem[cur][1][row * BK + ks * 32 + fk]);
       }
 #pragma unroll
-      for (int i = 0; i < 4; ++i)
+      for (int i = 0; i < FM; ++i)
 #pragma unroll
-        for (int j = 0; j < 4; ++j)
+        for (int j = 0; j < FN; ++j)
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               a[i], b[j], acc[i][j], 0, 0, 0);
     }
@@ -122,15 +133,15 @@ __global__ __launch_bounds__(256) void gemm_bf16_nt_kernel(
   const int crow_base = (lane >> 4) * 4;
   const int ccol = lane & 15;
 #pragma unroll
-  for (int i = 0; i < 4; ++i) {
+  for (int i = 0; i < FM; ++i) {
 #pragma unroll
-    for (int j = 0; j < 4; ++j) {
-      const int n = n0 + wn * 64 + j * 16 + ccol;
+    for (int j = 0; j < FN; ++j) {
+      const int n = n0 + wn * (BN / WN) + j * 16 + ccol;
       if (n >= N) continue;
       const float bv = bias ? bias[n] : 0.0f;
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const int m = m0 + wm * 64 + i * 16 + crow_base + r;
+        const int m = m0 + wm * (BM / WM) + i * 16 + crow_base + r;
         if (m >= M) continue;
         float v = apply_act(alpha * acc[i][j][r] + bv, act);
         if (OUT_F32)
@@ -149,16 +160,22 @@ extern "C" void infomesh_gemm_bf16_nt(
     int M, int N, int K, int batch,
     long strideA, long strideB, long strideC,
     int act, float alpha, int out_f32, void* stream) {
-  const int tiles = ((M + BM - 1) / BM) * ((N + BN - 1) / BN);
+  const int bm = (M <= 64) ? 64 : 128;
+  const int tiles = ((M + bm - 1) / bm) * ((N + BN - 1) / BN);
   dim3 grid(tiles, batch);
   dim3 block(256);
   auto s = reinterpret_cast<hipStream_t>(stream);
   const bool bk64 = (K % 64 == 0);
-#define LAUNCH(BKV, OF)                                                      \
-  hipLaunchKernelGGL((gemm_bf16_nt_kernel<BKV, OF>), grid, block, 0, s,      \
+#define LAUNCH(BMV, BKV, OF)                                                 \
+  hipLaunchKernelGGL((gemm_bf16_nt_kernel<BMV, BKV, OF>), grid, block, 0, s, \
                      (const bf16*)A, (const bf16*)B, C, (const float*)bias,  \
                      M, N, K, strideA, strideB, strideC, act, alpha)
-  if (bk64) { if (out_f32) LAUNCH(64, true); else LAUNCH(64, false); }
-  else      { if (out_f32) LAUNCH(32, true); else LAUNCH(32, false); }
+  if (bm == 64) {
+    if (bk64) { if (out_f32) LAUNCH(64, 64, true); else LAUNCH(64, 64, false); }
+    else      { if (out_f32) LAUNCH(64, 32, true); else LAUNCH(64, 32, false); }
+  } else {
+    if (bk64) { if (out_f32) LAUNCH(128, 64, true); else LAUNCH(128, 64, false); }
+    else      { if (out_f32) LAUNCH(128, 32, true); else LAUNCH(128, 32, false); }
+  }
 #undef LAUNCH
 }
