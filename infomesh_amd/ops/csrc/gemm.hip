@@ -28,7 +28,8 @@ __global__ __launch_bounds__(256) void gemm_bf16_nt_kernel(
     long strideA, long strideB, long strideC,
     int act, float alpha) {
   // One __shared__ object only (glds pipeline rule, guide §5 item 4a).
-  __shared__ bf16 smem[2][2][BM * BK];
+  // Layout per buffer: A tile [BM*BK] then B tile [BN*BK].
+  __shared__ bf16 smem[2][(BM + BN) * BK];
 
   const int tiles_n = (N + BN - 1) / BN;
   const int tiles_m = (M + BM - 1) / BM;
@@ -74,14 +75,14 @@ __global__ __launch_bounds__(256) void gemm_bf16_nt_kernel(
       int brow = n0 + row; brow = brow < N ? brow : N - 1;
       const bf16* gb = Bg + (long)brow * K + k0 + lcol;
       auto* lb = (__attribute__((address_space(3))) unsigned int*)
-          &smem[buf][1][chunk * 512];
+          &smem[buf][BM * BK + chunk * 512];
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) unsigned int*)gb, lb, 16, 0, 0);
       if (chunk < (BM * BK * 2) / 1024) {   // A tile is smaller at BM=64
         int arow = m0 + row; arow = arow < M ? arow : M - 1;
         const bf16* ga = Ag + (long)arow * K + k0 + lcol;
         auto* la = (__attribute__((address_space(3))) unsigned int*)
-            &smem[buf][0][chunk * 512];
+            &smem[buf][chunk * 512];
         __builtin_amdgcn_global_load_lds(
             (const __attribute__((address_space(1))) unsigned int*)ga, la, 16, 0, 0);
       }
@@ -110,13 +111,13 @@ __global__ __launch_bounds__(256) void gemm_bf16_nt_kernel(
       for (int i = 0; i < FM; ++i) {
         const int row = wm * (BM / WM) + i * 16 + fr;
         a[i] = *reinterpret_cast<const bf16x8*>(
-            &smem[cur][0][row * BK + ks * 32 + fk]);
+            &smem[cur][row * BK + ks * 32 + fk]);
       }
 #pragma unroll
       for (int j = 0; j < FN; ++j) {
         const int row = wn * (BN / WN) + j * 16 + fr;
         b[j] = *reinterpret_cast<const bf16x8*>(
-            &smem[cur][1][row * BK + ks * 32 + fk]);
+            &smem[cur][BM * BK + row * BK + ks * 32 + fk]);
       }
 #pragma unroll
       for (int i = 0; i < FM; ++i)
