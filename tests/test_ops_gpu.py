@@ -42,6 +42,7 @@ def _assert_close(gpu, ref, rtol=2e-2, atol=2e-2, what=""):
 @pytest.mark.parametrize("M,N,Kd", [
     (128, 128, 64), (64, 384, 384), (200, 1000, 96),
     (256, 1536, 384), (33, 100, 32), (512, 512, 512),
+    (64, 1000, 96), (17, 384, 64), (64, 200, 32), (32, 4096, 384),
 ])
 def test_gemm_nt_shapes(M, N, Kd):
     a = torch.randn(M, Kd, device="cuda").bfloat16()
