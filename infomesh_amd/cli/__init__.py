@@ -202,14 +202,20 @@ def api(port, api_key):
 
 
 @cli.command()
-def dashboard():
-    """Print the text dashboard report."""
-    from ..utils.text_report import render_report
-    ctx = _ctx(with_worker=False, with_engine=False)
-    try:
-        click.echo(render_report(ctx))
-    finally:
-        ctx.close()
+@click.option("--text", is_flag=True, help="one-shot text report")
+@click.option("--refresh", default=2.0, help="live refresh seconds")
+def dashboard(text, refresh):
+    """Live terminal dashboard (--text for a one-shot report)."""
+    if text:
+        from ..utils.text_report import render_report
+        ctx = _ctx(with_worker=False, with_engine=False)
+        try:
+            click.echo(render_report(ctx))
+        finally:
+            ctx.close()
+        return
+    from ..dashboard.app import run_dashboard
+    run_dashboard(refresh_s=refresh)
 
 
 # ----------------------------------------------------------------- index

@@ -263,7 +263,7 @@ def test_cli_keys_and_doctor(tmp_data_dir):
 
 
 def test_cli_dashboard(tmp_data_dir):
-    r = CliRunner().invoke(cli, ["dashboard"])
+    r = CliRunner().invoke(cli, ["dashboard", "--text"])
     assert r.exit_code == 0
     assert "node report" in r.output
 

@@ -175,3 +175,30 @@ def test_crawl_loop_bounded(tmp_data_dir):
     assert stats["crawled"] >= 1
     assert ctx.store.count() >= 1  # dedup collapses identical bodies
     ctx.close()
+
+
+def test_dashboard_renders(tmp_data_dir):
+    """Live dashboard renders against a populated data dir (WAL reads)."""
+    from infomesh_amd.config import Config
+    from infomesh_amd.dashboard.app import DashboardData, render_dashboard
+    from infomesh_amd.runtime import RuntimeStatus
+    from infomesh_amd.services import AppContext
+    from infomesh_amd.index.local_store import Document
+    from rich.console import Console
+    import io
+
+    ctx = AppContext.create(config=Config(), with_engine=False,
+                            with_worker=False)
+    ctx.index_document(Document(url="https://a.com/1",
+                                text="dashboard test doc body " * 5))
+    RuntimeStatus(ctx.config.data_dir).write("running", engine_docs=1)
+    ctx.close()
+
+    data = DashboardData()
+    snap = data.snapshot()
+    assert snap["docs"] == 1
+    assert snap["runtime"]["state"] == "running"
+    console = Console(file=io.StringIO(), width=100)
+    console.print(render_dashboard(data))
+    out = console.file.getvalue()
+    assert "infomesh-amd" in out and "documents" in out
