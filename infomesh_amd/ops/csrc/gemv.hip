@@ -29,6 +29,7 @@ __global__ __launch_bounds__(256) void gemv_bf16_nt_kernel(
 #pragma unroll
   for (int m = 0; m < 16; ++m) acc[m] = 0.f;
 
+#pragma unroll 2
   for (int k0 = lane * 8; k0 < K; k0 += 64 * 8) {
     const bf16x8 b8 = *reinterpret_cast<const bf16x8*>(Brow + k0);
     float bf[8];
