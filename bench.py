@@ -37,7 +37,7 @@ def parse_args():
     p.add_argument("--steps", type=int, default=30)
     p.add_argument("--warmup", type=int, default=5)
     p.add_argument("--docs-per-gpu", type=int, default=1_250_000)
-    p.add_argument("--batch", type=int, default=64)
+    p.add_argument("--batch", type=int, default=128)
     p.add_argument("--k-per-shard", type=int, default=100)
     p.add_argument("--n-results", type=int, default=10)
     p.add_argument("--avg-doc-len", type=int, default=120)
