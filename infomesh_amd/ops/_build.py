@@ -40,24 +40,41 @@ def needs_rebuild() -> bool:
 
 
 def build(force: bool = False, verbose: bool = True) -> Path:
+    """Build the extension (idempotent, multi-process safe).
+
+    At N-rank bench start every rank calls build(); an flock serializes
+    them and late arrivals see a fresh .so after re-checking."""
     if not force and not needs_rebuild():
         return SO_PATH
-    cmd = [
-        _hipcc(), f"--offload-arch={ARCH}", "-O3", "-std=c++17",
-        "-fPIC", "-shared", "-Wall",
-        *[str(s) for s in SOURCES],
-        "-o", str(SO_PATH),
-    ]
-    if verbose:
-        print(f"[infomesh-amd] building HIP extension ({len(SOURCES)} sources, "
-              f"{ARCH}) …", file=sys.stderr)
-    res = subprocess.run(cmd, capture_output=True, text=True)
-    if res.returncode != 0:
-        raise RuntimeError(
-            f"hipcc build failed (exit {res.returncode}):\n{res.stderr[-4000:]}")
-    if verbose and res.stderr.strip():
-        print(res.stderr[-2000:], file=sys.stderr)
-    return SO_PATH
+    import fcntl
+    lock_path = OPS_DIR / ".build.lock"
+    with open(lock_path, "w") as lock:
+        fcntl.flock(lock, fcntl.LOCK_EX)
+        try:
+            if not force and not needs_rebuild():
+                return SO_PATH  # another rank built it while we waited
+            tmp = OPS_DIR / f".libinfomesh_hip.tmp{subprocess.os.getpid()}.so"
+            cmd = [
+                _hipcc(), f"--offload-arch={ARCH}", "-O3", "-std=c++17",
+                "-fPIC", "-shared", "-Wall",
+                *[str(s) for s in SOURCES],
+                "-o", str(tmp),
+            ]
+            if verbose:
+                print(f"[infomesh-amd] building HIP extension "
+                      f"({len(SOURCES)} sources, {ARCH}) …", file=sys.stderr)
+            res = subprocess.run(cmd, capture_output=True, text=True)
+            if res.returncode != 0:
+                tmp.unlink(missing_ok=True)
+                raise RuntimeError(
+                    f"hipcc build failed (exit {res.returncode}):\n"
+                    f"{res.stderr[-4000:]}")
+            tmp.replace(SO_PATH)
+            if verbose and res.stderr.strip():
+                print(res.stderr[-2000:], file=sys.stderr)
+            return SO_PATH
+        finally:
+            fcntl.flock(lock, fcntl.LOCK_UN)
 
 
 if __name__ == "__main__":
