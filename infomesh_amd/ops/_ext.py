@@ -65,6 +65,10 @@ _SIGNATURES: dict[str, list] = {
                             c_int, c_float, c_void_p],
     "infomesh_kv_append": [c_void_p, c_void_p, c_void_p, c_void_p,
                            c_void_p, c_int, c_int, c_int, c_int, c_void_p],
+    "infomesh_gemm8_bf16_nt": [c_void_p, c_void_p, c_void_p, c_void_p,
+                               c_int, c_int, c_int, c_int,
+                               c_long, c_long, c_long,
+                               c_int, c_float, c_int, c_void_p],
     "infomesh_gemv_bf16_nt": [c_void_p, c_void_p, c_void_p, c_void_p,
                               c_int, c_int, c_int, c_int,
                               c_long, c_long, c_long,

@@ -1,0 +1,266 @@
+// Deep-pipelined 256x256 bf16 MFMA GEMM (gfx950) for large projection
+// shapes — the guide's "256² 8-phase" class structure (§5 template):
+// 8 waves (2M x 4N), BK=64, double-buffered LDS staged by
+// global_load_lds, COUNTED vmcnt across tile boundaries (loads stay in
+// flight through barriers), st_16x32 LDS swizzle on the staging SOURCE
+// address + ds_read offsets (rule 21), setprio around MFMA clusters.
+//
+// Half-tile schedule (2 glds per wave per half; halves h0,h1 = A rows
+// 0-127 / 128-255; h2,h3 = B rows 0-127 / 128-255; all four halves of
+// tile T land in buf[T&1]):
+//   prologue: stage t0.h0..h3, t1.h0,h1  -> vmcnt(4), barrier
+//   per tile t (reading buf[cur]):
+//     P1: ds_read frag set 1 | stage t+1.h2 -> buf^1 | bar | MFMA Q0,Q1
+//     P2: ds_read frag set 2 | stage t+1.h3 -> buf^1 | bar | MFMA Q2,Q3
+//         (after P2's barrier every wave has consumed buf[cur])
+//     P3: stage t+2.h0 -> buf[cur] | MFMA Q4,Q5   (frags in registers)
+//     P4: stage t+2.h1 -> buf[cur] | MFMA Q6,Q7
+//     vmcnt(4) + barrier   (own t+1 halves landed; cross-wave via bar)
+// Out-of-range prefetch steps clamp to the last K-tile (harmless
+// redundant loads) so the vmcnt counts stay static.
+#include "common.h"
+
+#define G8_BM 256
+#define G8_BN 256
+#define G8_BK 64
+
+namespace {
+
+// st_16x32 swizzle on a [rows][64 bf16] row-major image:
+// byte addr = row*128 + col2; flip bit5 (32B group) when row bit2 set.
+DEVINL int swz(int byte_off) {
+  return byte_off ^ (((byte_off >> 9) & 1) << 5);
+}
+
+template <bool OUT_F32>
+__global__ __launch_bounds__(512, 1) void gemm8_kernel(
+    const bf16* __restrict__ A, const bf16* __restrict__ B,
+    void* __restrict__ C, const float* __restrict__ bias,
+    int M, int N, int K,
+    long strideA, long strideB, long strideC,
+    int act, float alpha) {
+  // one buffer = A tile (256x64) + B tile (256x64) = 64 KiB
+  __shared__ bf16 smem[2][(G8_BM + G8_BN) * G8_BK];
+
+  const int tiles_n = (N + G8_BN - 1) / G8_BN;
+  const int tiles_m = (M + G8_BM - 1) / G8_BM;
+  const int tile_id = xcd_swizzle(blockIdx.x, tiles_m * tiles_n);
+  const int m0 = (tile_id / tiles_n) * G8_BM;
+  const int n0 = (tile_id % tiles_n) * G8_BN;
+  const int g = blockIdx.y;
+  const bf16* Ag = A + (long)g * strideA;
+  const bf16* Bg = B + (long)g * strideB;
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;        // 8 waves
+  const int wm = wid >> 2, wn = wid & 3;   // 2 x 4 wave grid
+
+  const int n_ksteps = K / G8_BK;
+
+  // ---- staging: one half (128 rows x 64 cols bf16 = 16 KiB) is 16
+  // glds instructions; per wave 2. glds writes lane-linear 16 B: lane's
+  // LDS bytes = chunkbase + lane*16. Swizzle goes on the SOURCE column.
+  // half-chunk c (0..15): rows c*8 .. c*8+8; within: lane/8 row,
+  // (lane%8)*16 byte col (8 bf16).
+  auto stage_half = [&](int buf, int kstep, int half) {
+    const int ks = kstep < n_ksteps ? kstep : n_ksteps - 1;  // clamp
+    const long k0 = (long)ks * G8_BK;
+    const bool is_b = half >= 2;
+    const int row_base = (half & 1) * 128;
+    // region base in BYTES: A at 0, B at BM*BK*2; half 1 at +16 KiB
+    const long region = (is_b ? (long)G8_BM * G8_BK * 2 : 0) +
+                        (long)row_base * 128;
+#pragma unroll
+    for (int c2 = 0; c2 < 2; ++c2) {
+      const int chunk = wid * 2 + c2;               // 0..15
+      // lane-linear LDS byte within the half; glds writes base+lane*16
+      const int lin_byte = chunk * 1024 + lane * 16;
+      // source position under the involution (rule 21: swizzle the
+      // per-lane GLOBAL address, keep the LDS destination linear)
+      const int s_byte = swz(lin_byte);
+      const int lrow = s_byte / 128;                // 0..127 in half
+      const int colb = s_byte % 128;                // byte col in row
+      int grow = (is_b ? n0 : m0) + row_base + lrow;
+      const int lim = is_b ? N : M;
+      grow = grow < lim ? grow : lim - 1;
+      const bf16* gsrc = (is_b ? Bg : Ag) + (long)grow * K + k0 + colb / 2;
+      auto* dst = (__attribute__((address_space(3))) unsigned int*)
+          ((char*)&smem[buf][0] + region + (long)chunk * 1024);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)gsrc,
+          dst, 16, 0, 0);
+    }
+  };
+
+  // ---- fragment reads (swizzled): A frag (mi-quadrant row r, ks):
+  // row = wm*128 + fi*16 + (lane&15); byte = row*128 + (ks*32 +
+  // (lane>>4)*8)*2, swizzled.
+  const int fr = lane & 15;
+  const int fkb = (lane >> 4) * 16;        // byte offset of k slice
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  // prologue: t0 all 4 halves + t1 h0,h1
+  stage_half(0, 0, 0); stage_half(0, 0, 1);
+  stage_half(0, 0, 2); stage_half(0, 0, 3);
+  stage_half(1, 1, 0); stage_half(1, 1, 1);
+  asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  int cur = 0;
+  for (int t = 0; t < n_ksteps; ++t) {
+    const char* abase = (const char*)&smem[cur][0];
+    const char* bbase = abase + (long)G8_BM * G8_BK * 2;
+    // ---- P1: read afr (A mi 0..3) + bfr (B ni 0..1); stage t+1.h2;
+    //          MFMA Q0 = afr x bfr
+    bf16x8 afr[4][2], bfr[2][2];
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        const int row = wm * 128 + i * 16 + fr;
+        afr[i][ks] = *reinterpret_cast<const bf16x8*>(
+            abase + swz(row * 128 + ks * 64 + fkb));
+      }
+#pragma unroll
+    for (int j = 0; j < 2; ++j)
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        const int row = wn * 64 + j * 16 + fr;
+        bfr[j][ks] = *reinterpret_cast<const bf16x8*>(
+            bbase + swz(row * 128 + ks * 64 + fkb));
+      }
+    stage_half(cur ^ 1, t + 1, 2);
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_sched_barrier(0);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < 2; ++j)
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afr[i][ks], bfr[j][ks], acc[i][j], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+
+    // ---- P2: read bfr2 (B ni 2..3); stage t+1.h3; MFMA Q1 = afr x bfr2
+    bf16x8 bfr2[2][2];
+#pragma unroll
+    for (int j = 0; j < 2; ++j)
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        const int row = wn * 64 + (j + 2) * 16 + fr;
+        bfr2[j][ks] = *reinterpret_cast<const bf16x8*>(
+            bbase + swz(row * 128 + ks * 64 + fkb));
+      }
+    stage_half(cur ^ 1, t + 1, 3);
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_sched_barrier(0);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < 2; ++j)
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks)
+          acc[i][j + 2] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afr[i][ks], bfr2[j][ks], acc[i][j + 2], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+
+    // ---- P3: read afr2 (A mi 4..7, reusing afr's registers — afr is
+    //          dead); MFMA Q2 = afr2 x bfr
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        const int row = wm * 128 + (i + 4) * 16 + fr;
+        afr[i][ks] = *reinterpret_cast<const bf16x8*>(
+            abase + swz(row * 128 + ks * 64 + fkb));
+      }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_sched_barrier(0);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < 2; ++j)
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks)
+          acc[i + 4][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afr[i][ks], bfr[j][ks], acc[i + 4][j], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+
+    // ---- P4: ALL buf[cur] reads done -> barrier kills the buffer,
+    //          stage t+2.h0+h1 into it; MFMA Q3 = afr2 x bfr2
+    __builtin_amdgcn_s_barrier();
+    stage_half(cur, t + 2, 0);
+    stage_half(cur, t + 2, 1);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < 2; ++j)
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks)
+          acc[i + 4][j + 2] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afr[i][ks], bfr2[j][ks], acc[i + 4][j + 2], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+
+    // boundary: own t+1 halves (all issued before the last 4 glds)
+    // landed; cross-wave visibility via the barrier.
+    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    cur ^= 1;
+  }
+
+  // ---- epilogue
+  const int crow0 = (lane >> 4) * 4;
+  const int ccol = lane & 15;
+#pragma unroll
+  for (int i = 0; i < 8; ++i) {
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const int n = n0 + wn * 64 + j * 16 + ccol;
+      if (n >= N) continue;
+      const float bv = bias ? bias[n] : 0.0f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int m = m0 + wm * 128 + i * 16 + crow0 + r;
+        if (m >= M) continue;
+        float v = apply_act(alpha * acc[i][j][r] + bv, act);
+        if (OUT_F32)
+          reinterpret_cast<float*>(C)[(long)g * strideC + (long)m * N + n] = v;
+        else
+          reinterpret_cast<bf16*>(C)[(long)g * strideC + (long)m * N + n] =
+              f2bf(v);
+      }
+    }
+  }
+}
+
+}  // namespace
+
+extern "C" void infomesh_gemm8_bf16_nt(
+    const void* A, const void* B, void* C, const void* bias,
+    int M, int N, int K, int batch,
+    long strideA, long strideB, long strideC,
+    int act, float alpha, int out_f32, void* stream) {
+  const int tiles = ((M + G8_BM - 1) / G8_BM) * ((N + G8_BN - 1) / G8_BN);
+  dim3 grid(tiles, batch), block(512);
+  auto s = reinterpret_cast<hipStream_t>(stream);
+  if (out_f32)
+    hipLaunchKernelGGL(gemm8_kernel<true>, grid, block, 0, s,
+                       (const bf16*)A, (const bf16*)B, C,
+                       (const float*)bias, M, N, K, strideA, strideB,
+                       strideC, act, alpha);
+  else
+    hipLaunchKernelGGL(gemm8_kernel<false>, grid, block, 0, s,
+                       (const bf16*)A, (const bf16*)B, C,
+                       (const float*)bias, M, N, K, strideA, strideB,
+                       strideC, act, alpha);
+}

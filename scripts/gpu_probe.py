@@ -26,6 +26,37 @@ def _t(fn, n=5, warmup=2):
     return (time.perf_counter() - t0) / n * 1e3
 
 
+def probe_gemm():
+    from infomesh_amd.ops import _ext, kernels as K
+    lib = _ext.lib()
+    for (M, N, Kd) in [(40960, 2304, 768), (40960, 3072, 768),
+                       (8192, 8192, 8192), (4096, 4096, 4096)]:
+        a = torch.randn(M, Kd, device="cuda").bfloat16()
+        b = torch.randn(N, Kd, device="cuda").bfloat16()
+        out = torch.empty(M, N, device="cuda", dtype=torch.bfloat16)
+        flops = 2.0 * M * N * Kd
+
+        def run128():
+            lib.infomesh_gemm_bf16_nt(a.data_ptr(), b.data_ptr(),
+                                      out.data_ptr(), None, M, N, Kd, 1,
+                                      M * Kd, N * Kd, M * N, 0, 1.0, 0,
+                                      _ext.stream_ptr())
+
+        def run256():
+            lib.infomesh_gemm8_bf16_nt(a.data_ptr(), b.data_ptr(),
+                                       out.data_ptr(), None, M, N, Kd, 1,
+                                       M * Kd, N * Kd, M * N, 0, 1.0, 0,
+                                       _ext.stream_ptr())
+        t1 = _t(run128, n=10, warmup=3)
+        t2 = _t(run256, n=10, warmup=3)
+        # correctness spot check vs each other (bf16 rounding identical path)
+        run128(); torch.cuda.synchronize(); o1 = out.clone()
+        run256(); torch.cuda.synchronize(); o2 = out.clone()
+        diff = (o1.float() - o2.float()).abs().max().item()
+        print(f"gemm {M}x{N}x{Kd}: 128tile {flops/t1/1e9:.0f} GF "
+              f"| 256tile {flops/t2/1e9:.0f} GF | maxdiff {diff:.4f}")
+
+
 def probe_encoder():
     from infomesh_amd.models.encoder import EmbeddingEncoder
     enc = EmbeddingEncoder(device="cuda")
@@ -86,6 +117,8 @@ if __name__ == "__main__":
     assert torch.cuda.is_available()
     from infomesh_amd.ops import _build
     _build.build()
+    if which in ("gemm",):
+        probe_gemm()
     if which in ("encoder", "all"):
         probe_encoder()
     if which in ("rerank", "all"):

@@ -493,3 +493,26 @@ def test_kv_append():
     assert torch.equal(kc[0, :, 3], kn[0])
     assert torch.equal(vc[1, :, 10], vn[1])
     assert kc[0, :, 4].abs().sum() == 0
+
+
+@pytest.mark.parametrize("M,N,Kd", [(4096, 512, 384), (4500, 2304, 768),
+                                    (4096, 513, 128)])
+def test_gemm8_large_shapes(M, N, Kd):
+    """M>=4096 dispatches to the deep-pipelined 256x256 kernel."""
+    a = torch.randn(M, Kd, device="cuda").bfloat16()
+    b = torch.randn(N, Kd, device="cuda").bfloat16()
+    bias = torch.randn(N, device="cuda")
+    out = K.gemm_nt(a, b, bias=bias, act="gelu")
+    ref = R.gemm_nt(a.cpu(), b.cpu(), bias.cpu(), act="gelu")
+    _assert_close(out, ref, rtol=3e-2, atol=Kd ** 0.5 * 2e-2,
+                  what=f"gemm8 {M}x{N}x{Kd}")
+
+
+def test_gemm8_transpose_check():
+    M, N, Kd = 4096, 512, 64
+    a = torch.eye(M, Kd, device="cuda").bfloat16()
+    b = (torch.arange(N * Kd, device="cuda").reshape(N, Kd).bfloat16()
+         / (N * Kd))
+    out = K.gemm_nt(a, b)
+    ref = R.gemm_nt(a.cpu(), b.cpu())
+    _assert_close(out, ref, what="gemm8 transpose check")
