@@ -62,8 +62,10 @@ def gemm_nt(a: torch.Tensor, b: torch.Tensor,
             a.data_ptr(), b.data_ptr(), out.data_ptr(), _ptr(bias),
             M, N, K, G, M * K, strideB, M * N,
             ACT[act], alpha, int(out_f32), _ext.stream_ptr())
-    elif M >= 4096 and N >= 512 and K % 64 == 0:
+    elif (M >= 4096 and N >= 512 and K % 64 == 0
+          and ((M + 255) // 256) * ((N + 255) // 256) * max(G, 1) >= 224):
         # big projection shapes: deep-pipelined 256x256 tile (gemm8.hip)
+        # (tile count must fill the chip at 1 block/CU)
         _ext.lib().infomesh_gemm8_bf16_nt(
             a.data_ptr(), b.data_ptr(), out.data_ptr(), _ptr(bias),
             M, N, K, G, M * K, strideB, M * N,
