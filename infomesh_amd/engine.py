@@ -82,8 +82,14 @@ class HybridEngine:
         if self.encoder is not None:
             chunks = []
             for i in range(0, len(self._built_texts), embed_batch):
-                chunks.append(self.encoder.encode_texts(
-                    self._built_texts[i:i + embed_batch]).bfloat16())
+                batch = self._built_texts[i:i + embed_batch]
+                pad = embed_batch - len(batch)
+                if pad:
+                    # fixed batch shape keeps the encoder's hipGraph
+                    # cache bounded (one graph per (B, S-bucket))
+                    batch = batch + [""] * pad
+                enc = self.encoder.encode_texts(batch).bfloat16()
+                chunks.append(enc[:embed_batch - pad] if pad else enc)
             emb = torch.cat(chunks, 0)
 
         lens = np.array([max(len(t), 1) for t in self._built_tokens],

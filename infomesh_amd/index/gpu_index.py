@@ -104,12 +104,24 @@ class GpuShard:
         self.build_from_arrays(flat_terms, flat_docs, lens,
                                np.asarray(gids, dtype=np.int64), emb)
 
+    def _upload(self, arr: np.ndarray, stream) -> torch.Tensor:
+        """Host -> HBM via a pinned staging buffer on the ingest side
+        stream (hipMemcpyAsync under the hood) so uploads overlap any
+        query work on the compute stream — SURVEY.md §5.8 ingest path."""
+        t = torch.from_numpy(np.ascontiguousarray(arr))
+        if stream is None:
+            return t.to(self.device)
+        pinned = t.pin_memory()
+        with torch.cuda.stream(stream):
+            return pinned.to(self.device, non_blocking=True)
+
     def build_from_arrays(self, flat_terms: np.ndarray,
                           flat_docs: np.ndarray, doc_lens: np.ndarray,
                           global_ids: np.ndarray,
                           embeddings: torch.Tensor | None) -> None:
         """Bulk build from flat (term, doc) pairs. Dedups (term, doc)
-        into term frequencies, sorts into CSR by term."""
+        into term frequencies, sorts into CSR by term. Device uploads
+        go through pinned buffers on a dedicated ingest stream."""
         n = len(doc_lens)
         self.n_docs = n
         # Aggregate tf per (term, doc) via a combined key sort.
@@ -124,22 +136,39 @@ class GpuShard:
         offsets = np.zeros(self.vocab + 1, dtype=np.int64)
         np.cumsum(df, out=offsets[1:])
         self.df = df
-        dev = self.device
-        self.offsets = torch.from_numpy(offsets).to(dev)
-        self.doc_ids = torch.from_numpy(docs_u).to(dev)
-        self.tfs = torch.from_numpy(tf_u.astype(np.int16)).to(dev)
+        on_gpu = self.device.type == "cuda"
+        stream = torch.cuda.Stream(self.device) if on_gpu else None
+        self.offsets = self._upload(offsets, stream)
+        self.doc_ids = self._upload(docs_u, stream)
+        self.tfs = self._upload(tf_u.astype(np.int16), stream)
         self.avgdl = float(doc_lens.mean()) if n else 1.0
         norm = BM25_K1 * (1 - BM25_B + BM25_B *
                           doc_lens.astype(np.float32) / self.avgdl)
-        self.doc_norm = torch.from_numpy(norm).to(dev)
-        self.global_ids = torch.from_numpy(
-            global_ids.astype(np.int64)).to(dev)
+        self.doc_norm = self._upload(norm, stream)
+        self.global_ids = self._upload(global_ids.astype(np.int64), stream)
         if embeddings is not None:
             assert embeddings.shape[0] == n
-            e = embeddings.to(dev)
-            if e.dtype != torch.bfloat16:
-                e = torch.nn.functional.normalize(e.float(), dim=-1).bfloat16()
+            if embeddings.device.type == "cpu":
+                e = embeddings
+                if e.dtype != torch.bfloat16:
+                    e = torch.nn.functional.normalize(
+                        e.float(), dim=-1).bfloat16()
+                if stream is not None:
+                    pinned = e.contiguous().pin_memory()
+                    with torch.cuda.stream(stream):
+                        e = pinned.to(self.device, non_blocking=True)
+                else:
+                    e = e.to(self.device)
+            else:
+                e = embeddings.to(self.device)
+                if e.dtype != torch.bfloat16:
+                    e = torch.nn.functional.normalize(
+                        e.float(), dim=-1).bfloat16()
             self.embeddings = e.contiguous()
+        if stream is not None:
+            # epoch flip: the shard becomes visible only after the side
+            # stream's uploads complete on the compute stream.
+            torch.cuda.current_stream(self.device).wait_stream(stream)
 
     def hbm_bytes(self) -> int:
         total = 0

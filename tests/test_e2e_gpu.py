@@ -1,0 +1,72 @@
+"""GPU end-to-end: real documents through the full MI355X pipeline —
+AppContext ingest -> encoder embedding + CSR build -> GpuShard (HIP
+BM25 + MFMA cosine + radix top-k) -> hybrid search -> relevance."""
+from __future__ import annotations
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+if torch.cuda.is_available():
+    from infomesh_amd.ops import _build
+    _build.build()
+
+from infomesh_amd.config import Config
+from infomesh_amd.engine import HybridEngine
+from infomesh_amd.index.local_store import Document
+from infomesh_amd.services import AppContext
+from tests.test_e2e_quality import CORPUS, QUERIES
+
+
+@pytest.fixture(scope="module")
+def gpu_ctx():
+    if not torch.cuda.is_available():
+        pytest.skip("no GPU")
+    ctx = AppContext.create(config=Config(), with_engine=False,
+                            with_worker=False, in_memory=True)
+    ctx.engine = HybridEngine(device="cuda", use_encoder=True)
+    for url, title, text in CORPUS:
+        ctx.index_document(Document(url=url, title=title, text=text),
+                           attest=False, credit=False)
+    n = ctx.flush_engine()
+    assert n == len(CORPUS)
+    yield ctx
+    ctx.close()
+
+
+def test_gpu_engine_stats(gpu_ctx):
+    st = gpu_ctx.engine.stats()
+    assert st["device"].startswith("cuda")
+    assert st["docs_indexed"] == len(CORPUS)
+    assert st["encoder"]
+
+
+@pytest.mark.parametrize("query,expected", QUERIES)
+def test_gpu_expected_url_in_top3(gpu_ctx, query, expected):
+    resp = gpu_ctx.search(query, limit=5, use_cache=False, deduct=False)
+    urls = [getattr(r, "url", "") for r in resp.results][:3]
+    assert expected in urls, f"{query!r} -> {urls}"
+
+
+def test_gpu_engine_direct_search(gpu_ctx):
+    hits = gpu_ctx.engine.search("hip kernels mfma lds", limit=3)
+    assert hits
+    docs = [gpu_ctx.store.get_document(h.doc_id) for h in hits]
+    assert any(d and "hip-kernels" in d.url for d in docs)
+
+
+def test_gpu_incremental_ingest(gpu_ctx):
+    """New docs become searchable after the next epoch flip."""
+    gpu_ctx.index_document(Document(
+        url="https://new.example/mi355x",
+        title="MI355X memory system",
+        text="The MI355X has 288 gigabytes of HBM3E memory with eight "
+             "terabytes per second of bandwidth and an infinity cache."),
+        attest=False, credit=False)
+    assert gpu_ctx.engine.pending_count == 1
+    gpu_ctx.flush_engine()
+    resp = gpu_ctx.search("hbm3e bandwidth infinity cache", limit=3,
+                          use_cache=False, deduct=False)
+    assert any("new.example" in getattr(r, "url", "")
+               for r in resp.results)
