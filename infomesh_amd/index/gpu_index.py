@@ -286,7 +286,7 @@ class GpuShard:
             # puts this GEMM after the BM25 top-k has consumed it.
             d_scores = K.gemm_nt(query_emb.bfloat16(), self.embeddings,
                                  out_f32=True,
-                                 out=scores.unsqueeze(0)).squeeze(0)
+                                 out=scores.unsqueeze(0)).reshape(B, N)
             tp = mark("shard.dense", tp)
             dn_vals, dn_idx = topk(d_scores, k)
             tp = mark("shard.densetopk", tp)
