@@ -172,9 +172,13 @@ def main():
     def step(i: int) -> None:
         j = i % n_batches
         emb = None
+        encode_fn = None
         tp = time.perf_counter()
         if rank == 0 and encoder is not None:
-            emb = encoder.encode_ids(qids_all[j], qlens_all[j])
+            # passed as a callback: the query plane overlaps the BM25
+            # side-stream work with the encoder forward
+            def encode_fn():
+                return encoder.encode_ids(qids_all[j], qlens_all[j])
         elif rank == 0 and cpu_emb is not None:
             emb = cpu_emb[j]
         tp = _mark("encode", tp)
@@ -184,7 +188,7 @@ def main():
                 args.n_results,
                 args.rerank_candidates if args.rerank else 0),
             use_dense=not args.bm25_only, phase_t=phase_t
-            if args.phase_timers else None)
+            if args.phase_timers else None, encode_fn=encode_fn)
         tp = _mark("search+fuse", tp)
         if rank == 0 and reranker is not None:
             run_rerank(fused)

@@ -185,6 +185,14 @@ class GpuShard:
             self._topk = TopK(self.device)
         return self._topk
 
+    def _get_topk_dense(self):
+        """Separate selector (own workspace) so the dense plane can run
+        on a different stream than the BM25 plane without racing."""
+        from ..ops.kernels import TopK
+        if getattr(self, "_topk_dense", None) is None:
+            self._topk_dense = TopK(self.device)
+        return self._topk_dense
+
     def _idf(self, term: int) -> float:
         df = float(self.df[term]) if self.df is not None else 0.0
         if df <= 0:
@@ -234,6 +242,43 @@ class GpuShard:
         ci = idf_t[ct]
         return cq, ct, co, ci
 
+    def search_bm25(self, queries_terms: list[np.ndarray], k: int,
+                    scores_buf: torch.Tensor | None = None,
+                    mark=None) -> tuple[torch.Tensor, torch.Tensor]:
+        """BM25 plane only (needs terms, not embeddings) — callable on a
+        side stream to overlap with query encoding."""
+        import time as _time
+        from ..ops import kernels as K
+        if mark is None:
+            def mark(name, t0):
+                return t0
+        B = len(queries_terms)
+        dev = self.device
+        N = self.n_docs
+        k = min(k, N)
+        topk = self._get_topk()
+        tp = _time.perf_counter()
+        if scores_buf is not None and scores_buf.shape == (B, N):
+            scores = scores_buf
+            scores.zero_()
+        else:
+            scores = torch.zeros(B, N, device=dev, dtype=torch.float32)
+        cq, ct, co, ci = self.bm25_chunks(queries_terms)
+        tp = mark("shard.chunks", tp)
+        if len(cq):
+            K.bm25_score(
+                self.offsets, self.doc_ids, self.tfs, self.doc_norm,
+                torch.from_numpy(np.ascontiguousarray(cq)).to(dev, non_blocking=True),
+                torch.from_numpy(np.ascontiguousarray(ct)).to(dev, non_blocking=True),
+                torch.from_numpy(np.ascontiguousarray(co)).to(dev, non_blocking=True),
+                torch.from_numpy(np.ascontiguousarray(ci)).to(dev, non_blocking=True),
+                scores, k1=BM25_K1)
+        tp = mark("shard.bm25", tp)
+        out = topk(scores, k)
+        mark("shard.bm25topk", tp)
+        self._bm25_scores_buf = scores
+        return out
+
     def search(self, queries_terms: list[np.ndarray],
                query_emb: torch.Tensor | None, k: int = 100,
                scores_buf: torch.Tensor | None = None,
@@ -253,54 +298,52 @@ class GpuShard:
             return t1
 
         B = len(queries_terms)
-        dev = self.device
         N = self.n_docs
         assert N > 0, "shard is empty"
         k = min(k, N)
-        topk = self._get_topk()
-
         tp = _time.perf_counter()
-        # --- BM25 plane ---
-        if scores_buf is not None and scores_buf.shape == (B, N):
-            scores = scores_buf
-            scores.zero_()
-        else:
-            scores = torch.zeros(B, N, device=dev, dtype=torch.float32)
-        cq, ct, co, ci = self.bm25_chunks(queries_terms)
-        tp = mark("shard.chunks", tp)
-        if len(cq):
-            K.bm25_score(
-                self.offsets, self.doc_ids, self.tfs, self.doc_norm,
-                torch.from_numpy(np.ascontiguousarray(cq)).to(dev, non_blocking=True),
-                torch.from_numpy(np.ascontiguousarray(ct)).to(dev, non_blocking=True),
-                torch.from_numpy(np.ascontiguousarray(co)).to(dev, non_blocking=True),
-                torch.from_numpy(np.ascontiguousarray(ci)).to(dev, non_blocking=True),
-                scores, k1=BM25_K1)
-        tp = mark("shard.bm25", tp)
-        bm_vals, bm_idx = topk(scores, k)
-        tp = mark("shard.bm25topk", tp)
+        bm_vals, bm_idx = self.search_bm25(queries_terms, k,
+                                           scores_buf=scores_buf,
+                                           mark=mark)
+        tp = _time.perf_counter()
 
         # --- dense plane ---
         if query_emb is not None and self.embeddings is not None:
-            # Reusing the BM25 score buffer is safe: same stream ordering
-            # puts this GEMM after the BM25 top-k has consumed it.
-            d_scores = K.gemm_nt(query_emb.bfloat16(), self.embeddings,
-                                 out_f32=True,
-                                 out=scores.unsqueeze(0)).reshape(B, N)
-            tp = mark("shard.dense", tp)
-            dn_vals, dn_idx = topk(d_scores, k)
-            tp = mark("shard.densetopk", tp)
+            dn_vals, dn_idx = self.search_dense(query_emb, k, mark=mark)
         else:
             dn_vals = torch.full((B, k), -float("inf"), device=dev)
             dn_idx = torch.full((B, k), -1, device=dev, dtype=torch.int32)
 
-        def to_global(idx: torch.Tensor) -> torch.Tensor:
-            safe = idx.clamp(min=0).long()
-            g = self.global_ids[safe]
-            return torch.where(idx >= 0, g, torch.full_like(g, -1))
+        return ShardHits(bm25_scores=bm_vals,
+                         bm25_ids=self.to_global(bm_idx),
+                         dense_scores=dn_vals,
+                         dense_ids=self.to_global(dn_idx))
 
-        return ShardHits(bm25_scores=bm_vals, bm25_ids=to_global(bm_idx),
-                         dense_scores=dn_vals, dense_ids=to_global(dn_idx))
+    def search_dense(self, query_emb: torch.Tensor, k: int,
+                     mark=None) -> tuple[torch.Tensor, torch.Tensor]:
+        """Dense (cosine) plane; runs after the query embedding exists."""
+        import time as _time
+        from ..ops import kernels as K
+        if mark is None:
+            def mark(name, t0):
+                return t0
+        B = query_emb.shape[0]
+        N = self.n_docs
+        k = min(k, N)
+        tp = _time.perf_counter()
+        # A fresh buffer: the BM25 scores buffer may still be feeding its
+        # top-k on another stream.
+        d_scores = K.gemm_nt(query_emb.bfloat16(), self.embeddings,
+                             out_f32=True).reshape(B, N)
+        tp = mark("shard.dense", tp)
+        out = self._get_topk_dense()(d_scores, k)
+        mark("shard.densetopk", tp)
+        return out
+
+    def to_global(self, idx: torch.Tensor) -> torch.Tensor:
+        safe = idx.clamp(min=0).long()
+        g = self.global_ids[safe]
+        return torch.where(idx >= 0, g, torch.full_like(g, -1))
 
 
 class CpuShard(GpuShard):

@@ -216,12 +216,18 @@ def argmax(logits: torch.Tensor) -> torch.Tensor:
 
 
 class TopK:
-    """Reusable top-k selector (keeps its workspace allocated)."""
+    """Reusable top-k selector (keeps its workspace allocated).
+
+    With defer_check=True the overflow flag is NOT read per call (a
+    host sync that would break stream overlap); the caller invokes
+    check_pending() after its next natural sync point."""
 
     def __init__(self, device: torch.device | str = "cuda"):
         self.device = torch.device(device)
         self._ws: torch.Tensor | None = None
         self._ws_b = 0
+        self.defer_check = False
+        self._pending: list[tuple[torch.Tensor, int]] = []
 
     def __call__(self, scores: torch.Tensor, k: int
                  ) -> tuple[torch.Tensor, torch.Tensor]:
@@ -243,10 +249,19 @@ class TopK:
                           _ext.stream_ptr())
         # Overflow flag sits right after cnt[B] in the workspace.
         flag_off = B * 512 + 4 * B
-        if int(self._ws[flag_off].item()) != 0:
+        if self.defer_check:
+            self._pending.append((self._ws, flag_off))
+        elif int(self._ws[flag_off].item()) != 0:
             raise RuntimeError(
                 "topk candidate overflow (pathologically tied scores)")
         return vals, idx
+
+    def check_pending(self) -> None:
+        pending, self._pending = self._pending, []
+        for ws, off in pending:
+            if int(ws[off].item()) != 0:
+                raise RuntimeError(
+                    "topk candidate overflow (pathologically tied scores)")
 
 
 def topk(scores: torch.Tensor, k: int) -> tuple[torch.Tensor, torch.Tensor]:

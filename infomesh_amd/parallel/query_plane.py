@@ -95,6 +95,7 @@ class DistributedQueryPlane:
         self.fabric = fabric or Fabric()
         self.k = k_per_shard
         self._scores_buf: torch.Tensor | None = None
+        self._bm25_stream = None
 
     @property
     def world_size(self) -> int:
@@ -125,9 +126,16 @@ class DistributedQueryPlane:
                      query_emb: torch.Tensor | None,
                      B: int, dim: int = 384, n_results: int = 10,
                      use_dense: bool = True,
-                     phase_t: dict | None = None) -> FusedHits | None:
+                     phase_t: dict | None = None,
+                     encode_fn=None) -> FusedHits | None:
         """Collective search. Rank 0 passes real queries and gets the
-        FusedHits; other ranks pass None and get None."""
+        FusedHits; other ranks pass None and get None.
+
+        With encode_fn (rank 0), the BM25 plane is launched on a side
+        stream BEFORE the query encoding runs, overlapping the encoder's
+        MFMA work with the BM25 scatter/top-k (they are independent
+        until fusion). Collective order stays identical on all ranks:
+        bcast(terms) -> [local overlap] -> bcast(emb) -> all-gathers."""
         import time as _time
 
         def mark(name, t0):
@@ -140,17 +148,29 @@ class DistributedQueryPlane:
             return t1
 
         tp = _time.perf_counter()
-        terms, emb = self._pack_queries(queries_terms, query_emb, B, dim)
-        tp = mark("plane.pack", tp)
-        hits: ShardHits = self.shard.search(
-            terms, emb if use_dense else None, k=self.k,
-            scores_buf=self._get_scores_buf(B), phase_t=phase_t)
-        tp = mark("plane.shard", tp)
+        if torch.cuda.is_available() and self.shard.device.type == "cuda":
+            hits = self._search_overlapped(queries_terms, query_emb,
+                                           encode_fn, B, dim, use_dense)
+            tp = mark("plane.shard", tp)
+        else:
+            if encode_fn is not None:
+                query_emb = encode_fn()
+            terms, emb = self._pack_queries(queries_terms, query_emb, B,
+                                            dim)
+            tp = mark("plane.pack", tp)
+            hits = self.shard.search(
+                terms, emb if use_dense else None, k=self.k,
+                scores_buf=self._get_scores_buf(B), phase_t=phase_t)
+            tp = mark("plane.shard", tp)
         # all-gather fixed [B,k] blocks -> [W, B, k]
         bm_s = self.fabric.all_gather(hits.bm25_scores)
         bm_i = self.fabric.all_gather(hits.bm25_ids)
         dn_s = self.fabric.all_gather(hits.dense_scores)
         dn_i = self.fabric.all_gather(hits.dense_ids)
+        for tk in (self.shard._get_topk(),
+                   getattr(self.shard, "_topk_dense", None)):
+            if tk is not None and getattr(tk, "defer_check", False):
+                tk.check_pending()
         tp = mark("plane.gather", tp)
         if self.fabric.rank != 0:
             return None
@@ -170,6 +190,55 @@ class DistributedQueryPlane:
         return FusedHits(ids=ids, scores=scores, bm25_ids=bm_i,
                          bm25_scores=bm_s, dense_ids=dn_i,
                          dense_scores=dn_s)
+
+    def _search_overlapped(self, queries_terms, query_emb, encode_fn,
+                           B, dim, use_dense) -> ShardHits:
+        """BM25 on a side stream || query encoding on the main stream.
+
+        Broadcast order (terms, then embeddings) is identical on all
+        ranks; the BM25 launch between the two is rank-local."""
+        dev = self.shard.device
+        self.shard._get_topk().defer_check = True
+        self.shard._get_topk_dense().defer_check = True
+        if self._bm25_stream is None:
+            self._bm25_stream = torch.cuda.Stream(dev)
+        # terms must be broadcast before any rank's shard work
+        terms = self._broadcast_terms(queries_terms, B)
+        main = torch.cuda.current_stream(dev)
+        self._bm25_stream.wait_stream(main)
+        with torch.cuda.stream(self._bm25_stream):
+            bm_vals, bm_idx = self.shard.search_bm25(
+                terms, self.k, scores_buf=self._get_scores_buf(B))
+            bm_ids = self.shard.to_global(bm_idx)
+        if self.fabric.rank == 0:
+            emb = encode_fn() if encode_fn is not None else query_emb
+        else:
+            emb = None
+        emb_t = (emb.to(dev).float() if emb is not None
+                 else torch.zeros(B, dim, device=dev))
+        self.fabric.broadcast(emb_t)
+        if use_dense and self.shard.embeddings is not None:
+            dn_vals, dn_idx = self.shard.search_dense(emb_t, self.k)
+            dn_ids = self.shard.to_global(dn_idx)
+        else:
+            k = min(self.k, self.shard.n_docs)
+            dn_vals = torch.full((B, k), -float("inf"), device=dev)
+            dn_ids = torch.full((B, k), -1, device=dev, dtype=torch.int64)
+        main.wait_stream(self._bm25_stream)
+        return ShardHits(bm25_scores=bm_vals, bm25_ids=bm_ids,
+                         dense_scores=dn_vals, dense_ids=dn_ids)
+
+    def _broadcast_terms(self, queries_terms, B):
+        dev = self.fabric.device
+        terms_t = torch.full((B, MAX_QUERY_TERMS), -1, dtype=torch.int64)
+        if queries_terms is not None:
+            for i, t in enumerate(queries_terms):
+                t = t[:MAX_QUERY_TERMS]
+                terms_t[i, :len(t)] = torch.from_numpy(t.astype(np.int64))
+        terms_t = terms_t.to(dev)
+        self.fabric.broadcast(terms_t)
+        tt = terms_t.cpu().numpy()
+        return [tt[i][tt[i] >= 0] for i in range(B)]
 
     def _get_scores_buf(self, B: int) -> torch.Tensor | None:
         N = self.shard.n_docs
