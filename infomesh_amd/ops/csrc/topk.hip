@@ -29,7 +29,15 @@ __global__ __launch_bounds__(256) void hist1_kernel(
   const long start = (long)blockIdx.x * blockDim.x + threadIdx.x;
   const long step = (long)gridDim.x * blockDim.x;
   const float* row = scores + (long)b * N;
-  for (long i = start; i < N; i += step)
+  const long n4 = N / 4;
+  for (long i = start; i < n4; i += step) {
+    const float4 v = reinterpret_cast<const float4*>(row)[i];
+    atomicAdd(&lh[float_to_ordered(v.x) >> 24], 1u);
+    atomicAdd(&lh[float_to_ordered(v.y) >> 24], 1u);
+    atomicAdd(&lh[float_to_ordered(v.z) >> 24], 1u);
+    atomicAdd(&lh[float_to_ordered(v.w) >> 24], 1u);
+  }
+  for (long i = n4 * 4 + start; i < N; i += step)
     atomicAdd(&lh[float_to_ordered(row[i]) >> 24], 1u);
   __syncthreads();
   for (int i = threadIdx.x; i < 256; i += blockDim.x)
@@ -64,7 +72,17 @@ __global__ __launch_bounds__(256) void hist2_kernel(
   const long start = (long)blockIdx.x * blockDim.x + threadIdx.x;
   const long step = (long)gridDim.x * blockDim.x;
   const float* row = scores + (long)b * N;
-  for (long i = start; i < N; i += step) {
+  const long n4 = N / 4;
+  for (long i = start; i < n4; i += step) {
+    const float4 v = reinterpret_cast<const float4*>(row)[i];
+    const float f[4] = {v.x, v.y, v.z, v.w};
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const unsigned o = float_to_ordered(f[j]);
+      if ((o >> 24) == b1) atomicAdd(&lh[(o >> 16) & 255], 1u);
+    }
+  }
+  for (long i = n4 * 4 + start; i < N; i += step) {
     const unsigned o = float_to_ordered(row[i]);
     if ((o >> 24) == b1) atomicAdd(&lh[(o >> 16) & 255], 1u);
   }
@@ -99,7 +117,24 @@ __global__ __launch_bounds__(256) void compact_kernel(
   const long start = (long)blockIdx.x * blockDim.x + threadIdx.x;
   const long step = (long)gridDim.x * blockDim.x;
   const float* row = scores + (long)b * N;
-  for (long i = start; i < N; i += step) {
+  const long n4 = N / 4;
+  for (long i = start; i < n4; i += step) {
+    const float4 v = reinterpret_cast<const float4*>(row)[i];
+    const float f[4] = {v.x, v.y, v.z, v.w};
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const unsigned o = float_to_ordered(f[j]);
+      if ((o >> 16) >= t16) {
+        const unsigned pos = atomicAdd(&cnt[b], 1u);
+        if (pos < TOPK_CAP)
+          cand[(long)b * TOPK_CAP + pos] =
+              ((unsigned long long)(~o) << 32) | (unsigned)(i * 4 + j);
+        else
+          *overflow = 1u;
+      }
+    }
+  }
+  for (long i = n4 * 4 + start; i < N; i += step) {
     const unsigned o = float_to_ordered(row[i]);
     if ((o >> 16) >= t16) {
       const unsigned pos = atomicAdd(&cnt[b], 1u);
