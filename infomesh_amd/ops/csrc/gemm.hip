@@ -12,15 +12,15 @@
 // (infomesh/summarizer/engine.py:111-318) — see SURVEY.md §2.9.
 #include "common.h"
 
-#define BN 128
-
 namespace {
 
-// BM=128: 4 waves as 2x2, each owning a 64x64 sub-tile (4x4 fragments).
-// BM=64: 4 waves as 1x4 over N, each owning 64x32 -> 4x2 fragments —
-// used for skinny-M shapes (e.g. query-batch x doc-embedding scoring)
-// where a 128-row A tile would be half padding.
-template <int BM, int BK, bool OUT_F32>
+// Tile configurations (4 waves each):
+//   (128,128): 2x2 wave grid, 64x64 per wave (4x4 fragments) — default.
+//   ( 64,128): 1x4 grid, 64x32 per wave — skinny-M (query scoring).
+//   ( 64, 64): 2x2 grid, 32x32 per wave — latency-bound small-K shapes
+//              (encoder projections) where 128^2 tiles underfill the
+//              chip and the 6-step K loop's stage latency dominates.
+template <int BM, int BN, int BK, bool OUT_F32>
 __global__ __launch_bounds__(256) void gemm_bf16_nt_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ B,
     void* __restrict__ C, const float* __restrict__ bias,
@@ -44,10 +44,10 @@ __global__ __launch_bounds__(256) void gemm_bf16_nt_kernel(
   const int tid = threadIdx.x;
   const int lane = tid & 63;
   const int wid = tid >> 6;          // 4 waves
-  constexpr int WM = (BM == 128) ? 2 : 1;   // wave grid M
+  constexpr int WM = (BM == 128 || BN == 64) ? 2 : 1;   // wave grid M
   constexpr int WN = 4 / WM;                // wave grid N
-  constexpr int FM = (BM / WM) / 16;        // m-frags per wave (4 or 4)
-  constexpr int FN = (BN / WN) / 16;        // n-frags per wave (4 or 2)
+  constexpr int FM = (BM / WM) / 16;        // m-frags per wave
+  constexpr int FN = (BN / WN) / 16;        // n-frags per wave
   const int wm = wid / WN, wn = wid % WN;
 
   // ---- glds staging geometry -------------------------------------------
@@ -57,8 +57,10 @@ __global__ __launch_bounds__(256) void gemm_bf16_nt_kernel(
   constexpr int ROWS_PER_CHUNK = 1024 / (BK * 2);      // 8 (BK=64) / 16 (BK=32)
   // Chunk count follows the larger (B) tile; A stages only its first
   // BM*BK*2/1024 chunks (guarded below) when BM < BN.
-  constexpr int CHUNKS = (BN * BK * 2) / 1024;
+  constexpr int CHUNKS = ((BM > BN ? BM : BN) * BK * 2) / 1024;
   constexpr int CHUNKS_PER_WAVE = CHUNKS / 4;
+  constexpr int ACHUNKS = (BM * BK * 2) / 1024;
+  constexpr int BCHUNKS2 = (BN * BK * 2) / 1024;
   const int lanes_per_row = BK / 8;                     // 8 elements per lane
   const int lrow = lane / lanes_per_row;
   const int lcol = (lane % lanes_per_row) * 8;
@@ -72,13 +74,15 @@ __global__ __launch_bounds__(256) void gemm_bf16_nt_kernel(
     for (int c = 0; c < CHUNKS_PER_WAVE; ++c) {
       const int chunk = wid * CHUNKS_PER_WAVE + c;
       const int row = chunk * ROWS_PER_CHUNK + lrow;
-      int brow = n0 + row; brow = brow < N ? brow : N - 1;
-      const bf16* gb = Bg + (long)brow * K + k0 + lcol;
-      auto* lb = (__attribute__((address_space(3))) unsigned int*)
-          &smem[buf][BM * BK + chunk * 512];
-      __builtin_amdgcn_global_load_lds(
-          (const __attribute__((address_space(1))) unsigned int*)gb, lb, 16, 0, 0);
-      if (chunk < (BM * BK * 2) / 1024) {   // A tile is smaller at BM=64
+      if (chunk < BCHUNKS2) {
+        int brow = n0 + row; brow = brow < N ? brow : N - 1;
+        const bf16* gb = Bg + (long)brow * K + k0 + lcol;
+        auto* lb = (__attribute__((address_space(3))) unsigned int*)
+            &smem[buf][BM * BK + chunk * 512];
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) unsigned int*)gb, lb, 16, 0, 0);
+      }
+      if (chunk < ACHUNKS) {
         int arow = m0 + row; arow = arow < M ? arow : M - 1;
         const bf16* ga = Ag + (long)arow * K + k0 + lcol;
         auto* la = (__attribute__((address_space(3))) unsigned int*)
@@ -161,22 +165,32 @@ extern "C" void infomesh_gemm_bf16_nt(
     int M, int N, int K, int batch,
     long strideA, long strideB, long strideC,
     int act, float alpha, int out_f32, void* stream) {
-  const int bm = (M <= 64) ? 64 : 128;
-  const int tiles = ((M + bm - 1) / bm) * ((N + BN - 1) / BN);
-  dim3 grid(tiles, batch);
-  dim3 block(256);
   auto s = reinterpret_cast<hipStream_t>(stream);
   const bool bk64 = (K % 64 == 0);
-#define LAUNCH(BMV, BKV, OF)                                                 \
-  hipLaunchKernelGGL((gemm_bf16_nt_kernel<BMV, BKV, OF>), grid, block, 0, s, \
-                     (const bf16*)A, (const bf16*)B, C, (const float*)bias,  \
-                     M, N, K, strideA, strideB, strideC, act, alpha)
-  if (bm == 64) {
-    if (bk64) { if (out_f32) LAUNCH(64, 64, true); else LAUNCH(64, 64, false); }
-    else      { if (out_f32) LAUNCH(64, 32, true); else LAUNCH(64, 32, false); }
-  } else {
-    if (bk64) { if (out_f32) LAUNCH(128, 64, true); else LAUNCH(128, 64, false); }
-    else      { if (out_f32) LAUNCH(128, 32, true); else LAUNCH(128, 32, false); }
-  }
+  int bm = (M <= 64) ? 64 : 128, bn = 128;
+  // Latency-bound regime: few K-steps and not enough 128^2 tiles to
+  // fill the chip -> quarter tiles for 4x the block-level overlap.
+  const long blocks128 =
+      (long)((M + 127) / 128) * ((N + 127) / 128) * batch;
+  if (bm == 128 && blocks128 < 384 && K <= 1024) bm = bn = 64;
+  const int tiles = ((M + bm - 1) / bm) * ((N + bn - 1) / bn);
+  dim3 grid(tiles, batch);
+  dim3 block(256);
+#define LAUNCH(BMV, BNV, BKV, OF)                                            \
+  hipLaunchKernelGGL((gemm_bf16_nt_kernel<BMV, BNV, BKV, OF>), grid, block,  \
+                     0, s, (const bf16*)A, (const bf16*)B, C,                \
+                     (const float*)bias, M, N, K, strideA, strideB,          \
+                     strideC, act, alpha)
+#define PICK(BMV, BNV)                                                       \
+  do {                                                                       \
+    if (bk64) { if (out_f32) LAUNCH(BMV, BNV, 64, true);                     \
+                else LAUNCH(BMV, BNV, 64, false); }                          \
+    else      { if (out_f32) LAUNCH(BMV, BNV, 32, true);                     \
+                else LAUNCH(BMV, BNV, 32, false); }                          \
+  } while (0)
+  if (bm == 64 && bn == 64) PICK(64, 64);
+  else if (bm == 64) PICK(64, 128);
+  else PICK(128, 128);
+#undef PICK
 #undef LAUNCH
 }
