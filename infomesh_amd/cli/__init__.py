@@ -323,6 +323,68 @@ def keys_rotate():
                            "rotation": record}, indent=2))
 
 
+# ----------------------------------------------------------------- shard
+
+@cli.group()
+def shard():
+    """GPU shard manifests (warm-start persistence)."""
+
+
+@shard.command(name="save")
+@click.argument("path", type=click.Path())
+def shard_save(path):
+    """Build the GPU shard from the local index and save its manifest."""
+    ctx = _ctx(with_worker=False)
+    try:
+        if ctx.engine is None:
+            raise click.ClickException("engine unavailable on this machine")
+        for doc in ctx.store.export_documents():
+            ctx.engine.add_document(doc)
+        ctx.flush_engine()
+        from ..index.manifest import save_shard
+        meta = save_shard(ctx.engine.shard, path)
+        click.echo(json.dumps(meta, indent=2))
+    finally:
+        ctx.close()
+
+
+@shard.command(name="load")
+@click.argument("path", type=click.Path(exists=True))
+@click.option("--query", default="", help="probe query after loading")
+def shard_load(path, query):
+    """Load a shard manifest and optionally probe it."""
+    from ..index.manifest import load_shard
+    import torch
+    device = "cuda" if torch.cuda.is_available() else "cpu"
+    s = load_shard(path, device=device)
+    out = {"n_docs": s.n_docs, "hbm_bytes": s.hbm_bytes(),
+           "device": device}
+    if query:
+        from ..index.gpu_index import bm25_term_ids
+        hits = s.search([bm25_term_ids(query)], None, k=5)
+        out["probe"] = [int(i) for i in hits.bm25_ids[0] if int(i) >= 0]
+    click.echo(json.dumps(out, indent=2))
+
+
+@shard.command(name="info")
+@click.argument("path", type=click.Path(exists=True))
+def shard_info(path):
+    from ..index.manifest import manifest_info
+    info = manifest_info(path)
+    if info is None:
+        raise click.ClickException("no manifest sidecar found")
+    click.echo(json.dumps(info, indent=2))
+
+
+@cli.command(name="update-check")
+def update_check():
+    """Check for a newer release (offline-first)."""
+    from ..utils.version_check import check_for_update
+    cfg = load_config()
+    click.echo(json.dumps(check_for_update(
+        state_path=cfg.data_dir / "version_check.json"), indent=2))
+
+
 # ---------------------------------------------------------------- doctor
 
 @cli.command()
