@@ -40,7 +40,7 @@ _SIGNATURES: dict[str, list] = {
                       c_int, c_int, c_void_p],
     "infomesh_argmax": [c_void_p, c_void_p, c_long, c_int, c_void_p],
     "infomesh_topk": [c_void_p, c_void_p, c_void_p, c_void_p, c_int,
-                      c_long, c_int, c_void_p],
+                      c_long, c_int, c_int, c_void_p],
     "infomesh_bm25_score": [c_void_p, c_void_p, c_void_p, c_void_p,
                             c_void_p, c_void_p, c_void_p, c_void_p,
                             c_void_p, c_int, c_long, c_int, c_float,

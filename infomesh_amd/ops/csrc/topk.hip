@@ -150,6 +150,81 @@ __global__ __launch_bounds__(256) void compact_kernel(
   }
 }
 
+// ---- sampling select: estimate a conservative K-th-value threshold
+// from a strided sample, skipping the two full histogram passes. The
+// sampled threshold t32 feeds compact32; the wrapper verifies
+// K <= cnt <= CAP afterwards (both violations are astronomically rare
+// for real score distributions and fail LOUDLY, never silently).
+__global__ __launch_bounds__(256) void sample_thresh_kernel(
+    const float* __restrict__ scores, unsigned* __restrict__ thresh32,
+    long N, int stride, int r) {
+  __shared__ unsigned s[8192];
+  const int b = blockIdx.x;
+  const float* row = scores + (long)b * N;
+  const int nsamp = (int)min((long)8192, (N + stride - 1) / stride);
+  for (int i = threadIdx.x; i < 8192; i += blockDim.x) {
+    // per-row phase decorrelates rows; pad with +inf-ordered-0 (worst)
+    const long idx = (long)i * stride + (b % stride);
+    s[i] = (i < nsamp && idx < N)
+        ? ~float_to_ordered(row[idx]) : 0xFFFFFFFFu;
+  }
+  __syncthreads();
+  for (int k = 2; k <= 8192; k <<= 1) {
+    for (int j = k >> 1; j > 0; j >>= 1) {
+      for (int i = threadIdx.x; i < 8192; i += blockDim.x) {
+        const int ixj = i ^ j;
+        if (ixj > i) {
+          const bool up = ((i & k) == 0);
+          const unsigned a = s[i], c = s[ixj];
+          if ((a > c) == up) { s[i] = c; s[ixj] = a; }
+        }
+      }
+      __syncthreads();
+    }
+  }
+  // ascending ~ordered => s[r] is the (r+1)-th LARGEST sample value
+  if (threadIdx.x == 0) thresh32[b] = ~s[min(r, nsamp - 1)];
+}
+
+__global__ __launch_bounds__(256) void compact32_kernel(
+    const float* __restrict__ scores, const unsigned* __restrict__ thresh32,
+    unsigned long long* __restrict__ cand, unsigned* __restrict__ cnt,
+    unsigned* __restrict__ overflow, long N) {
+  const int b = blockIdx.y;
+  const unsigned t32 = thresh32[b];
+  const long start = (long)blockIdx.x * blockDim.x + threadIdx.x;
+  const long step = (long)gridDim.x * blockDim.x;
+  const float* row = scores + (long)b * N;
+  const long n4 = N / 4;
+  for (long i = start; i < n4; i += step) {
+    const float4 v = reinterpret_cast<const float4*>(row)[i];
+    const float f[4] = {v.x, v.y, v.z, v.w};
+#pragma unroll
+    for (int j = 0; j < 4; ++j) {
+      const unsigned o = float_to_ordered(f[j]);
+      if (o >= t32) {
+        const unsigned pos = atomicAdd(&cnt[b], 1u);
+        if (pos < TOPK_CAP)
+          cand[(long)b * TOPK_CAP + pos] =
+              ((unsigned long long)(~o) << 32) | (unsigned)(i * 4 + j);
+        else
+          *overflow = 1u;
+      }
+    }
+  }
+  for (long i = n4 * 4 + start; i < N; i += step) {
+    const unsigned o = float_to_ordered(row[i]);
+    if (o >= t32) {
+      const unsigned pos = atomicAdd(&cnt[b], 1u);
+      if (pos < TOPK_CAP)
+        cand[(long)b * TOPK_CAP + pos] =
+            ((unsigned long long)(~o) << 32) | (unsigned)i;
+      else
+        *overflow = 1u;
+    }
+  }
+}
+
 __global__ __launch_bounds__(256) void sort_emit_kernel(
     unsigned long long* __restrict__ cand, const unsigned* __restrict__ cnt,
     float* __restrict__ out_vals, int* __restrict__ out_idx, int K) {
@@ -202,7 +277,8 @@ extern "C" long infomesh_topk_workspace_u32(int B) {
 
 extern "C" void infomesh_topk(const void* scores, void* workspace,
                               void* out_vals, void* out_idx,
-                              int B, long N, int K, void* stream) {
+                              int B, long N, int K, int sampled,
+                              void* stream) {
   auto s = reinterpret_cast<hipStream_t>(stream);
   unsigned* ws = reinterpret_cast<unsigned*>(workspace);
   unsigned* hist1 = ws;
@@ -219,16 +295,30 @@ extern "C" void infomesh_topk(const void* scores, void* workspace,
   int chunks = (int)min((N + 256 * 64 - 1) / (256 * 64), (long)1024);
   if (chunks < 1) chunks = 1;
   dim3 g1(chunks, B), blk(256);
-  hipLaunchKernelGGL(hist1_kernel, g1, blk, 0, s,
-                     (const float*)scores, hist1, N);
-  hipLaunchKernelGGL(select1_kernel, dim3(B), dim3(64), 0, s,
-                     hist1, bin1, chi1, K);
-  hipLaunchKernelGGL(hist2_kernel, g1, blk, 0, s,
-                     (const float*)scores, bin1, hist2, N);
-  hipLaunchKernelGGL(select2_kernel, dim3(B), dim3(64), 0, s,
-                     hist2, bin1, chi1, thresh16, K);
-  hipLaunchKernelGGL(compact_kernel, g1, blk, 0, s,
-                     (const float*)scores, thresh16, cand, cnt, overflow, N);
+  if (sampled) {
+    // one sample pass + one compact pass instead of three full passes
+    const int stride = (int)((N + 8191) / 8192) < 256
+        ? 256 : (int)((N + 8191) / 8192);
+    int r = (int)(3L * K / stride);
+    if (r < 2) r = 2;
+    hipLaunchKernelGGL(sample_thresh_kernel, dim3(B), blk, 0, s,
+                       (const float*)scores, thresh16, N, stride, r);
+    hipLaunchKernelGGL(compact32_kernel, g1, blk, 0, s,
+                       (const float*)scores, thresh16, cand, cnt,
+                       overflow, N);
+  } else {
+    hipLaunchKernelGGL(hist1_kernel, g1, blk, 0, s,
+                       (const float*)scores, hist1, N);
+    hipLaunchKernelGGL(select1_kernel, dim3(B), dim3(64), 0, s,
+                       hist1, bin1, chi1, K);
+    hipLaunchKernelGGL(hist2_kernel, g1, blk, 0, s,
+                       (const float*)scores, bin1, hist2, N);
+    hipLaunchKernelGGL(select2_kernel, dim3(B), dim3(64), 0, s,
+                       hist2, bin1, chi1, thresh16, K);
+    hipLaunchKernelGGL(compact_kernel, g1, blk, 0, s,
+                       (const float*)scores, thresh16, cand, cnt,
+                       overflow, N);
+  }
   hipLaunchKernelGGL(sort_emit_kernel, dim3(B), blk, 0, s,
                      cand, cnt, (float*)out_vals, (int*)out_idx, K);
 }

@@ -229,12 +229,17 @@ class TopK:
         self.defer_check = False
         self._pending: list[tuple[torch.Tensor, int]] = []
 
-    def __call__(self, scores: torch.Tensor, k: int
+    def __call__(self, scores: torch.Tensor, k: int,
+                 sampled: bool | None = None
                  ) -> tuple[torch.Tensor, torch.Tensor]:
         assert scores.dim() == 2
         _check(scores, torch.float32, "scores")
         B, N = scores.shape
         assert 1 <= k <= 1024
+        if sampled is None:
+            # sampling skips 2 of 3 full passes; only pays (and only has
+            # statistical headroom) at large N
+            sampled = N >= 200_000
         lib = _ext.lib()
         nu32 = lib.infomesh_topk_workspace_u32(B)
         if self._ws is None or self._ws_b < nu32:
@@ -246,22 +251,34 @@ class TopK:
         idx = torch.empty((B, k), device=scores.device, dtype=torch.int32)
         lib.infomesh_topk(scores.data_ptr(), self._ws.data_ptr(),
                           vals.data_ptr(), idx.data_ptr(), B, N, k,
-                          _ext.stream_ptr())
-        # Overflow flag sits right after cnt[B] in the workspace.
+                          int(sampled), _ext.stream_ptr())
+        # cnt[B] then the overflow flag live after the histograms.
+        cnt_off = B * 512 + 3 * B
         flag_off = B * 512 + 4 * B
+        need = min(k, N) if sampled else 0
         if self.defer_check:
-            self._pending.append((self._ws, flag_off))
-        elif int(self._ws[flag_off].item()) != 0:
-            raise RuntimeError(
-                "topk candidate overflow (pathologically tied scores)")
+            self._pending.append((self._ws, cnt_off, flag_off, B, need))
+        else:
+            self._verify(self._ws, cnt_off, flag_off, B, need)
         return vals, idx
+
+    @staticmethod
+    def _verify(ws, cnt_off, flag_off, B, need) -> None:
+        if int(ws[flag_off].item()) != 0:
+            raise RuntimeError(
+                "topk candidate overflow (cap exceeded — pathologically "
+                "tied scores or an unlucky sampled threshold)")
+        if need:
+            cnts = ws[cnt_off:cnt_off + B]
+            if int(cnts.min().item()) < need:
+                raise RuntimeError(
+                    "topk sampled threshold undershot (cnt < k); "
+                    "rerun with sampled=False")
 
     def check_pending(self) -> None:
         pending, self._pending = self._pending, []
-        for ws, off in pending:
-            if int(ws[off].item()) != 0:
-                raise RuntimeError(
-                    "topk candidate overflow (pathologically tied scores)")
+        for ws, cnt_off, flag_off, B, need in pending:
+            self._verify(ws, cnt_off, flag_off, B, need)
 
 
 def topk(scores: torch.Tensor, k: int) -> tuple[torch.Tensor, torch.Tensor]:

@@ -516,3 +516,25 @@ def test_gemm8_transpose_check():
     out = K.gemm_nt(a, b)
     ref = R.gemm_nt(a.cpu(), b.cpu())
     _assert_close(out, ref, what="gemm8 transpose check")
+
+
+def test_topk_sampled_vs_exact():
+    """Sampled threshold path (default at large N) == exact path."""
+    from infomesh_amd.ops.kernels import TopK
+    scores = torch.randn(4, 500_000, device="cuda")
+    t = TopK("cuda")
+    v1, i1 = t(scores, 100, sampled=True)
+    v2, i2 = t(scores, 100, sampled=False)
+    assert torch.allclose(v1, v2, atol=0)
+    rv, _ = torch.topk(scores, 100, dim=-1)
+    assert torch.allclose(v1, rv, atol=0)
+
+
+def test_topk_sampled_overflow_loud():
+    """Constant scores: every element passes the sampled threshold ->
+    the cap overflows and the failure is LOUD, never silent."""
+    from infomesh_amd.ops.kernels import TopK
+    scores = torch.ones(1, 300_000, device="cuda")
+    t = TopK("cuda")
+    with pytest.raises(RuntimeError):
+        t(scores, 10, sampled=True)
