@@ -299,8 +299,13 @@ extern "C" void infomesh_topk(const void* scores, void* workspace,
     // one sample pass + one compact pass instead of three full passes
     const int stride = (int)((N + 8191) / 8192) < 256
         ? 256 : (int)((N + 8191) / 8192);
-    int r = (int)(3L * K / stride);
-    if (r < 2) r = 2;
+    // Rank of the sampled threshold: the count of elements above the
+    // (r+1)-th largest of n samples is ~Gamma(r+1) x stride, so r must
+    // be large enough that P(count < K) is negligible. mean = 12
+    // strides (K small) or 4K (K large) keeps both tails tiny while
+    // staying well under the CAP=8192 candidate budget.
+    int r = (int)((4L * K + stride - 1) / stride) - 1;
+    if (r < 11) r = 11;
     hipLaunchKernelGGL(sample_thresh_kernel, dim3(B), blk, 0, s,
                        (const float*)scores, thresh16, N, stride, r);
     hipLaunchKernelGGL(compact32_kernel, g1, blk, 0, s,
