@@ -237,9 +237,11 @@ class TopK:
         B, N = scores.shape
         assert 1 <= k <= 1024
         if sampled is None:
-            # sampling skips 2 of 3 full passes; only pays (and only has
-            # statistical headroom) at large N
-            sampled = N >= 200_000
+            # Default EXACT: sampling assumes a near-continuous score
+            # distribution (massively tied scores — e.g. BM25 tails —
+            # overflow the candidate cap). Continuous planes (cosine)
+            # opt in explicitly.
+            sampled = False
         lib = _ext.lib()
         nu32 = lib.infomesh_topk_workspace_u32(B)
         if self._ws is None or self._ws_b < nu32:
