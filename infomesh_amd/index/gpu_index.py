@@ -336,9 +336,11 @@ class GpuShard:
         d_scores = K.gemm_nt(query_emb.bfloat16(), self.embeddings,
                              out_f32=True).reshape(B, N)
         tp = mark("shard.dense", tp)
-        # cosine scores are near-continuous: the sampled threshold is
-        # statistically safe and skips 2 of the 3 full passes
-        out = self._get_topk_dense()(d_scores, k, sampled=N >= 200_000)
+        # Cosine scores are near-continuous, so the sampled threshold is
+        # statistically safe there. It saves 2 of the 3 full passes but
+        # inflates the final sort by the candidate slack (~Kp*stride):
+        # net win only when the array passes dominate (large N).
+        out = self._get_topk_dense()(d_scores, k, sampled=N >= 4_000_000)
         mark("shard.densetopk", tp)
         return out
 

@@ -150,78 +150,23 @@ __global__ __launch_bounds__(256) void compact_kernel(
   }
 }
 
-// ---- sampling select: estimate a conservative K-th-value threshold
-// from a strided sample, skipping the two full histogram passes. The
-// sampled threshold t32 feeds compact32; the wrapper verifies
-// K <= cnt <= CAP afterwards (both violations are astronomically rare
-// for real score distributions and fail LOUDLY, never silently).
-__global__ __launch_bounds__(256) void sample_thresh_kernel(
-    const float* __restrict__ scores, unsigned* __restrict__ thresh32,
-    long N, int stride, int r) {
-  __shared__ unsigned s[8192];
-  const int b = blockIdx.x;
-  const float* row = scores + (long)b * N;
-  const int nsamp = (int)min((long)8192, (N + stride - 1) / stride);
-  for (int i = threadIdx.x; i < 8192; i += blockDim.x) {
-    // per-row phase decorrelates rows; pad with +inf-ordered-0 (worst)
-    const long idx = (long)i * stride + (b % stride);
-    s[i] = (i < nsamp && idx < N)
-        ? ~float_to_ordered(row[idx]) : 0xFFFFFFFFu;
-  }
-  __syncthreads();
-  for (int k = 2; k <= 8192; k <<= 1) {
-    for (int j = k >> 1; j > 0; j >>= 1) {
-      for (int i = threadIdx.x; i < 8192; i += blockDim.x) {
-        const int ixj = i ^ j;
-        if (ixj > i) {
-          const bool up = ((i & k) == 0);
-          const unsigned a = s[i], c = s[ixj];
-          if ((a > c) == up) { s[i] = c; s[ixj] = a; }
-        }
-      }
-      __syncthreads();
-    }
-  }
-  // ascending ~ordered => s[r] is the (r+1)-th LARGEST sample value
-  if (threadIdx.x == 0) thresh32[b] = ~s[min(r, nsamp - 1)];
-}
-
-__global__ __launch_bounds__(256) void compact32_kernel(
-    const float* __restrict__ scores, const unsigned* __restrict__ thresh32,
-    unsigned long long* __restrict__ cand, unsigned* __restrict__ cnt,
-    unsigned* __restrict__ overflow, long N) {
+// ---- sampling select: gather a strided sample into the (pre-compact)
+// candidate buffer, run the cheap exact 2-level radix select ON THE
+// SAMPLE (3 passes over ~32 KB/row instead of the full array), then
+// threshold the full array with the resulting 16-bit prefix. Saves two
+// full passes; the wrapper verifies K <= cnt <= CAP afterwards (loud,
+// never silent). Assumes a near-continuous distribution — callers keep
+// heavily-tied planes (BM25 tails) on the exact path.
+__global__ void sample_gather_kernel(const float* __restrict__ scores,
+                                     float* __restrict__ sample,
+                                     long N, int stride) {
   const int b = blockIdx.y;
-  const unsigned t32 = thresh32[b];
-  const long start = (long)blockIdx.x * blockDim.x + threadIdx.x;
-  const long step = (long)gridDim.x * blockDim.x;
   const float* row = scores + (long)b * N;
-  const long n4 = N / 4;
-  for (long i = start; i < n4; i += step) {
-    const float4 v = reinterpret_cast<const float4*>(row)[i];
-    const float f[4] = {v.x, v.y, v.z, v.w};
-#pragma unroll
-    for (int j = 0; j < 4; ++j) {
-      const unsigned o = float_to_ordered(f[j]);
-      if (o >= t32) {
-        const unsigned pos = atomicAdd(&cnt[b], 1u);
-        if (pos < TOPK_CAP)
-          cand[(long)b * TOPK_CAP + pos] =
-              ((unsigned long long)(~o) << 32) | (unsigned)(i * 4 + j);
-        else
-          *overflow = 1u;
-      }
-    }
-  }
-  for (long i = n4 * 4 + start; i < N; i += step) {
-    const unsigned o = float_to_ordered(row[i]);
-    if (o >= t32) {
-      const unsigned pos = atomicAdd(&cnt[b], 1u);
-      if (pos < TOPK_CAP)
-        cand[(long)b * TOPK_CAP + pos] =
-            ((unsigned long long)(~o) << 32) | (unsigned)i;
-      else
-        *overflow = 1u;
-    }
+  float* out = sample + (long)b * TOPK_CAP;
+  for (int i = blockIdx.x * blockDim.x + threadIdx.x; i < TOPK_CAP;
+       i += gridDim.x * blockDim.x) {
+    const long idx = (long)i * stride + (b % stride);
+    out[i] = idx < N ? row[idx] : -INFINITY;
   }
 }
 
@@ -296,19 +241,28 @@ extern "C" void infomesh_topk(const void* scores, void* workspace,
   if (chunks < 1) chunks = 1;
   dim3 g1(chunks, B), blk(256);
   if (sampled) {
-    // one sample pass + one compact pass instead of three full passes
-    const int stride = (int)((N + 8191) / 8192) < 256
-        ? 256 : (int)((N + 8191) / 8192);
-    // Rank of the sampled threshold: the count of elements above the
-    // (r+1)-th largest of n samples is ~Gamma(r+1) x stride, so r must
-    // be large enough that P(count < K) is negligible. mean = 12
-    // strides (K small) or 4K (K large) keeps both tails tiny while
-    // staying well under the CAP=8192 candidate budget.
-    int r = (int)((4L * K + stride - 1) / stride) - 1;
-    if (r < 11) r = 11;
-    hipLaunchKernelGGL(sample_thresh_kernel, dim3(B), blk, 0, s,
-                       (const float*)scores, thresh16, N, stride, r);
-    hipLaunchKernelGGL(compact32_kernel, g1, blk, 0, s,
+    const int stride = (int)((N + TOPK_CAP - 1) / TOPK_CAP) < 256
+        ? 256 : (int)((N + TOPK_CAP - 1) / TOPK_CAP);
+    // Sample rank: the count above the Kp-th largest of the sample is
+    // ~Gamma(Kp) x stride. Kp >= 8 (and mean >= 3K) keeps the
+    // undershoot tail (< K) negligible for continuous scores while the
+    // mean stays far from the CAP (the extra candidates inflate the
+    // final bitonic sort, which is why sampling only pays at large N).
+    int Kp = (int)((3L * K + stride - 1) / stride);
+    if (Kp < 8) Kp = 8;
+    float* sample = reinterpret_cast<float*>(cand);  // pre-compact reuse
+    hipLaunchKernelGGL(sample_gather_kernel, dim3(32, B), blk, 0, s,
+                       (const float*)scores, sample, N, stride);
+    dim3 gs(1, B);
+    hipLaunchKernelGGL(hist1_kernel, gs, blk, 0, s,
+                       sample, hist1, (long)TOPK_CAP);
+    hipLaunchKernelGGL(select1_kernel, dim3(B), dim3(64), 0, s,
+                       hist1, bin1, chi1, Kp);
+    hipLaunchKernelGGL(hist2_kernel, gs, blk, 0, s,
+                       sample, bin1, hist2, (long)TOPK_CAP);
+    hipLaunchKernelGGL(select2_kernel, dim3(B), dim3(64), 0, s,
+                       hist2, bin1, chi1, thresh16, Kp);
+    hipLaunchKernelGGL(compact_kernel, g1, blk, 0, s,
                        (const float*)scores, thresh16, cand, cnt,
                        overflow, N);
   } else {
