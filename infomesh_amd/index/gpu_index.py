@@ -336,11 +336,12 @@ class GpuShard:
         d_scores = K.gemm_nt(query_emb.bfloat16(), self.embeddings,
                              out_f32=True).reshape(B, N)
         tp = mark("shard.dense", tp)
-        # Cosine scores are near-continuous, so the sampled threshold is
-        # statistically safe there. It saves 2 of the 3 full passes but
-        # inflates the final sort by the candidate slack (~Kp*stride):
-        # net win only when the array passes dominate (large N).
-        out = self._get_topk_dense()(d_scores, k, sampled=N >= 4_000_000)
+        # Always exact select. The sampled-threshold variant was measured
+        # a net loss at every shard size: the candidate slack (~Kp*stride)
+        # inflates the final bitonic sort by more than the 2 saved passes
+        # (~80us/pass at 1.25M docs), and with an 8192-sample + 8192-cap
+        # it is only statistically sound for N <~ 2.7M anyway.
+        out = self._get_topk_dense()(d_scores, k)
         mark("shard.densetopk", tp)
         return out
 

@@ -237,10 +237,14 @@ class TopK:
         B, N = scores.shape
         assert 1 <= k <= 1024
         if sampled is None:
-            # Default EXACT: sampling assumes a near-continuous score
-            # distribution (massively tied scores — e.g. BM25 tails —
-            # overflow the candidate cap). Continuous planes (cosine)
-            # opt in explicitly.
+            # Default EXACT. The sampled threshold additionally assumes
+            # (a) a near-continuous score distribution (massively tied
+            # scores — e.g. BM25 tails — overflow the candidate cap) and
+            # (b) N <~ 2.7M: beyond that the sample stride makes the
+            # Gamma(Kp) candidate-count tails collide with the cap.
+            # Failures are loud (overflow/undershoot RuntimeError), and
+            # in-engine benchmarks showed no shape where it is a net
+            # win, so nothing auto-enables it.
             sampled = False
         lib = _ext.lib()
         nu32 = lib.infomesh_topk_workspace_u32(B)
