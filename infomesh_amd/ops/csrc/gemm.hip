@@ -11,6 +11,7 @@
 // (infomesh/index/vector_store.py:120-157) and external-LLM HTTP calls
 // (infomesh/summarizer/engine.py:111-318) — see SURVEY.md §2.9.
 #include "common.h"
+#include <cstdlib>
 
 namespace {
 
@@ -172,7 +173,15 @@ extern "C" void infomesh_gemm_bf16_nt(
   // fill the chip -> quarter tiles for 4x the block-level overlap.
   const long blocks128 =
       (long)((M + 127) / 128) * ((N + 127) / 128) * batch;
-  if (bm == 128 && blocks128 < 384 && K <= 1024) bm = bn = 64;
+  if (bm == 128 && blocks128 < 512 && K <= 2048) bm = bn = 64;
+  // Tuning override (read once): INFOMESH_GEMM_TILE=64|128 forces the
+  // tile; scripts/gemm_tile_probe.py uses it to validate the heuristic.
+  static const int ov = [] {
+    const char* e = getenv("INFOMESH_GEMM_TILE");
+    return e ? atoi(e) : -1;
+  }();
+  if (ov == 64 && M > 64) bm = bn = 64;
+  else if (ov == 128) { bm = 128; bn = 128; }
   const int tiles = ((M + bm - 1) / bm) * ((N + bn - 1) / bn);
   dim3 grid(tiles, batch);
   dim3 block(256);
