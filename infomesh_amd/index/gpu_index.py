@@ -214,12 +214,29 @@ class GpuShard:
         """Vectorized host-side work chunking into (qrow, term, offset,
         idf) arrays — one work chunk per <=chunk_size posting slice."""
         offs, idf_t = self._host_tables()
-        qrows = np.concatenate([
-            np.full(len(np.unique(t)), qi, dtype=np.int64)
-            for qi, t in enumerate(queries_terms)]) if queries_terms else \
-            np.zeros(0, np.int64)
-        terms = np.concatenate([np.unique(t) for t in queries_terms]) \
-            if queries_terms else np.zeros(0, np.int64)
+        # Vectorized per-query term dedupe (the obvious per-query
+        # np.unique loop costs ~1 ms of host time at B=128, serializing
+        # the side-stream launch ahead of the encoder).
+        B = len(queries_terms)
+        if B:
+            lens = np.fromiter((len(t) for t in queries_terms),
+                               np.int64, B)
+            T = int(lens.max()) if B else 0
+        if B == 0 or T == 0:
+            qrows = np.zeros(0, np.int64)
+            terms = np.zeros(0, np.int64)
+        else:
+            if (lens == lens[0]).all():
+                mat = np.stack(queries_terms).astype(np.int64, copy=False)
+            else:
+                mat = np.full((B, T), -1, dtype=np.int64)
+                for qi, t in enumerate(queries_terms):
+                    mat[qi, :len(t)] = t
+            srt = np.sort(mat, axis=1)
+            valid = srt >= 0
+            valid[:, 1:] &= srt[:, 1:] != srt[:, :-1]
+            qrows, cols = np.nonzero(valid)
+            terms = srt[qrows, cols]
         begins = offs[terms]
         ends = offs[terms + 1]
         nchunks = np.maximum((ends - begins + chunk_size - 1) // chunk_size, 0)
@@ -241,6 +258,26 @@ class GpuShard:
         co = base + intra * chunk_size
         ci = idf_t[ct]
         return cq, ct, co, ci
+
+    def _h2d(self, name: str, arr: np.ndarray,
+             dtype: torch.dtype) -> torch.Tensor:
+        """Stage a small host array through a persistent pinned buffer —
+        pageable-memory copies block the host and defeat the side-stream
+        overlap; pinned + non_blocking stays truly async."""
+        t = torch.from_numpy(np.ascontiguousarray(arr)).to(dtype)
+        if self.device.type != "cuda":
+            return t
+        pins = getattr(self, "_h2d_pins", None)
+        if pins is None:
+            pins = self._h2d_pins = {}
+        n = t.numel()
+        buf = pins.get(name)
+        if buf is None or buf.numel() < n:
+            buf = torch.empty(max(2 * n, 4096), dtype=dtype,
+                              pin_memory=True)
+            pins[name] = buf
+        buf[:n].copy_(t)
+        return buf[:n].to(self.device, non_blocking=True)
 
     def search_bm25(self, queries_terms: list[np.ndarray], k: int,
                     scores_buf: torch.Tensor | None = None,
@@ -266,13 +303,22 @@ class GpuShard:
         cq, ct, co, ci = self.bm25_chunks(queries_terms)
         tp = mark("shard.chunks", tp)
         if len(cq):
+            # don't overwrite the pinned staging buffers while a prior
+            # step's async H2D copy could still be in flight
+            evt = getattr(self, "_h2d_evt", None)
+            if evt is not None:
+                evt.synchronize()
             K.bm25_score(
                 self.offsets, self.doc_ids, self.tfs, self.doc_norm,
-                torch.from_numpy(np.ascontiguousarray(cq)).to(dev, non_blocking=True),
-                torch.from_numpy(np.ascontiguousarray(ct)).to(dev, non_blocking=True),
-                torch.from_numpy(np.ascontiguousarray(co)).to(dev, non_blocking=True),
-                torch.from_numpy(np.ascontiguousarray(ci)).to(dev, non_blocking=True),
+                self._h2d("cq", cq, torch.int32),
+                self._h2d("ct", ct, torch.int32),
+                self._h2d("co", co, torch.int64),
+                self._h2d("ci", ci, torch.float32),
                 scores, k1=BM25_K1)
+            if dev.type == "cuda":
+                if evt is None:
+                    evt = self._h2d_evt = torch.cuda.Event()
+                evt.record()
         tp = mark("shard.bm25", tp)
         out = topk(scores, k)
         mark("shard.bm25topk", tp)

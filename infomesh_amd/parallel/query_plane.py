@@ -96,6 +96,9 @@ class DistributedQueryPlane:
         self.k = k_per_shard
         self._scores_buf: torch.Tensor | None = None
         self._bm25_stream = None
+        # RRF fusion is ~30 tiny fixed-shape kernels (~0.5 ms of launch
+        # overhead per batch) -> hipGraph-captured per n_results.
+        self._fuse_graphs: dict = {}
 
     @property
     def world_size(self) -> int:
@@ -180,8 +183,22 @@ class DistributedQueryPlane:
         dn_s = dn_s.permute(1, 0, 2).reshape(B, W * k)
         dn_i = dn_i.permute(1, 0, 2).reshape(B, W * k)
         if use_dense:
-            ids, scores = rrf_fuse([bm_i, dn_i], [bm_s, dn_s],
-                                   [1.0, 1.0], n_results)
+            if bm_i.is_cuda:
+                gf = self._fuse_graphs.get(n_results)
+                if gf is None:
+                    from ..ops.graphs import GraphedCallable
+
+                    def _fuse(a, b, c, d, _n=n_results):
+                        return rrf_fuse([a, b], [c, d], [1.0, 1.0], _n)
+                    gf = GraphedCallable(_fuse)
+                    self._fuse_graphs[n_results] = gf
+                ids, scores = gf(bm_i, dn_i, bm_s, dn_s)
+                # detach results from the graph's static output buffers
+                # (the next replay overwrites them)
+                ids, scores = ids.clone(), scores.clone()
+            else:
+                ids, scores = rrf_fuse([bm_i, dn_i], [bm_s, dn_s],
+                                       [1.0, 1.0], n_results)
         else:
             order = torch.argsort(bm_s, dim=1, descending=True)
             ids = torch.gather(bm_i, 1, order)[:, :n_results]
@@ -229,6 +246,10 @@ class DistributedQueryPlane:
                          dense_scores=dn_vals, dense_ids=dn_ids)
 
     def _broadcast_terms(self, queries_terms, B):
+        if self.fabric.world == 1 and queries_terms is not None:
+            # no collective needed: skip the pack + GPU round-trip +
+            # device sync entirely
+            return queries_terms
         dev = self.fabric.device
         terms_t = torch.full((B, MAX_QUERY_TERMS), -1, dtype=torch.int64)
         if queries_terms is not None:
