@@ -65,6 +65,8 @@ _SIGNATURES: dict[str, list] = {
                             c_int, c_float, c_void_p],
     "infomesh_kv_append": [c_void_p, c_void_p, c_void_p, c_void_p,
                            c_void_p, c_int, c_int, c_int, c_int, c_void_p],
+    "infomesh_dense_scores": [c_void_p, c_void_p, c_void_p,
+                              c_int, c_long, c_int, c_float, c_void_p],
     "infomesh_gemm8_bf16_nt": [c_void_p, c_void_p, c_void_p, c_void_p,
                                c_int, c_int, c_int, c_int,
                                c_long, c_long, c_long,
@@ -100,7 +102,8 @@ def _try_load() -> ctypes.CDLL | None:
     for name, argtypes in _SIGNATURES.items():
         fn = getattr(lib, name)
         fn.argtypes = argtypes
-        fn.restype = None
+        # dense_scores reports dispatch eligibility (0 = launched)
+        fn.restype = c_int if name == "infomesh_dense_scores" else None
     ws = lib.infomesh_topk_workspace_u32
     ws.argtypes = [c_int]
     ws.restype = c_long

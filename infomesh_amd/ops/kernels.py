@@ -62,6 +62,15 @@ def gemm_nt(a: torch.Tensor, b: torch.Tensor,
             a.data_ptr(), b.data_ptr(), out.data_ptr(), _ptr(bias),
             M, N, K, G, M * K, strideB, M * N,
             ACT[act], alpha, int(out_f32), _ext.stream_ptr())
+    elif (M <= 128 and G == 1 and N >= 65536 and out_f32
+          and bias is None and act == "none"
+          and _ext.lib().infomesh_dense_scores(
+              a.data_ptr(), b.data_ptr(), out.data_ptr(),
+              M, N, K, alpha, _ext.stream_ptr()) == 0):
+        # streaming score kernel (densescore.hip): whole query block in
+        # LDS, doc embeddings streamed HBM->regs once — the generic
+        # tile pipeline measured 1.9 TB/s on this shape
+        pass
     elif (M >= 4096 and N >= 512 and K % 64 == 0
           and ((M + 255) // 256) * ((N + 255) // 256) * max(G, 1) >= 224):
         # big projection shapes: deep-pipelined 256x256 tile (gemm8.hip)
