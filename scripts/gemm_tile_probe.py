@@ -49,6 +49,23 @@ def main():
         gf = 2.0 * M * N * K / (us * 1e-6) / 1e9
         print(f"tile={tag:>4} {M}x{N}x{K}{'f32' if f32 else ''}: "
               f"{us:8.1f} us  {gf:7.0f} GF/s")
+        if f32 and M <= 128:
+            def run_ds():
+                rc = lib.infomesh_dense_scores(
+                    a.data_ptr(), b.data_ptr(), out.data_ptr(),
+                    M, N, K, 1.0, _ext.stream_ptr())
+                assert rc == 0
+            for _ in range(5):
+                run_ds()
+            torch.cuda.synchronize()
+            t0 = time.perf_counter()
+            for _ in range(n):
+                run_ds()
+            torch.cuda.synchronize()
+            us = (time.perf_counter() - t0) / n * 1e6
+            bw = (M + N) * K * 2 / (us * 1e-6) / 1e12
+            print(f"  densescore      {M}x{N}x{K}f32: {us:8.1f} us "
+                  f"  B-stream {bw:5.2f} TB/s")
 
 
 if __name__ == "__main__":

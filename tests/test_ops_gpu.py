@@ -544,14 +544,14 @@ def test_topk_sampled_overflow_loud():
 @pytest.mark.parametrize("M,N,Kd", [(128, 131072, 384), (100, 70000, 384),
                                     (128, 131072 + 100, 128)])
 def test_dense_scores_streaming(M, N, Kd):
-    """The streaming dense-score kernel (densescore.hip) dispatches for
-    M<=128, huge N, f32 out — parity vs the fp32 oracle, incl. N tails
-    and non-multiple-of-16 M."""
+    """The streaming dense-score kernel (densescore.hip, explicit
+    opt-in) — parity vs the fp32 oracle, incl. N tails and
+    non-multiple-of-16 M."""
     torch.manual_seed(5)
     a = torch.randn(M, Kd, device="cuda").bfloat16()
     b = torch.randn(N, Kd, device="cuda").bfloat16()
-    out = K.gemm_nt(a, b, out_f32=True)
-    assert out.dtype == torch.float32
+    out = K.dense_scores(a, b)
+    assert out is not None and out.dtype == torch.float32
     ref = a.float() @ b.float().T
     _assert_close(out, ref, rtol=3e-2, atol=Kd ** 0.5 * 2e-2,
                   what=f"dense_scores {M}x{N}x{Kd}")
@@ -560,17 +560,12 @@ def test_dense_scores_streaming(M, N, Kd):
 @pytest.mark.gpu
 def test_dense_scores_matches_generic_tile():
     """Same shape through the streaming kernel and the generic tile path
-    must agree bitwise-closely (both bf16 MFMA, f32 accum)."""
+    (same MFMA order, f32 accum) must agree bitwise."""
     torch.manual_seed(6)
     a = torch.randn(64, 384, device="cuda").bfloat16()
     b = torch.randn(70000, 384, device="cuda").bfloat16()
-    stream = K.gemm_nt(a, b, out_f32=True)
-    from infomesh_amd.ops import _ext
-    out2 = torch.empty(70000 * 64, device="cuda",
-                       dtype=torch.float32).view(1, 64, 70000)
-    _ext.lib().infomesh_gemm_bf16_nt(
-        a.data_ptr(), b.data_ptr(), out2.data_ptr(), None,
-        64, 70000, 384, 1, 64 * 384, 70000 * 384, 64 * 70000,
-        0, 1.0, 1, _ext.stream_ptr())
+    stream = K.dense_scores(a, b)
+    assert stream is not None
+    generic = K.gemm_nt(a, b, out_f32=True)
     torch.cuda.synchronize()
-    assert torch.equal(stream, out2.view(64, 70000))
+    assert torch.equal(stream, generic)
