@@ -30,13 +30,22 @@ class SummarizeRequest:
 class PeerSummarizeHandler:
     def __init__(self, engine: SummarizationEngine,
                  ledger: CreditLedger | None = None,
-                 max_queue: int = MAX_QUEUE):
+                 max_queue: int = MAX_QUEUE,
+                 reputation=None):
         self.engine = engine
         self.ledger = ledger
         self.max_queue = max_queue
+        # trust/reputation.SummaryReputation: requesters with failed
+        # summary grades are deprioritized; our own verified summary
+        # quality feeds back through record_quality().
+        self.reputation = reputation
         self._lock = threading.Lock()
         self._active = 0
         self.stats = {"served": 0, "rejected": 0}
+
+    def record_quality(self, node_id: str, quality: float) -> None:
+        if self.reputation is not None:
+            self.reputation.record(node_id, quality)
 
     def handle(self, req: SummarizeRequest) -> SummaryResult | None:
         """Serve or reject (None) a summarize request."""

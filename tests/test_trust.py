@@ -162,3 +162,63 @@ def test_gdpr_deletion(seeded_store):
     assert dm.enforce() == 1
     assert len(dm.export_records()) == 1
     dm.close()
+
+
+# ------------------------------------------------------- detector
+
+
+def test_detector_escalation_and_isolation():
+    from infomesh_amd.trust.detector import (
+        MaliciousNodeDetector, ThreatLevel)
+    t = [1000.0]
+    d = MaliciousNodeDetector(now=lambda: t[0])
+    assert d.assess("n1").level == ThreatLevel.NONE
+    d.record("n1", "spam")
+    assert d.assess("n1").level == ThreatLevel.NONE
+    d.record("n1", "audit_fail")
+    assert d.assess("n1").level >= ThreatLevel.MEDIUM
+    th = d.record("n1", "invalid_signature", count=2)
+    assert th.level >= ThreatLevel.HIGH and th.isolate
+    # decay: far in the future the score drains back to NONE
+    t[0] += 365 * 24 * 3600.0
+    assert d.assess("n1").level == ThreatLevel.NONE
+    assert d.threats() == []
+
+
+def test_detector_unknown_kind_raises():
+    from infomesh_amd.trust.detector import MaliciousNodeDetector
+    import pytest as _pytest
+    with _pytest.raises(ValueError):
+        MaliciousNodeDetector().record("x", "nope")
+
+
+def test_detector_threat_listing_sorted():
+    from infomesh_amd.trust.detector import MaliciousNodeDetector
+    d = MaliciousNodeDetector(now=lambda: 0.0)
+    d.record("a", "spam")
+    d.record("b", "invalid_proof", count=3)
+    ths = d.threats()
+    assert ths and ths[0].node_id == "b"
+
+
+# ----------------------------------------------------- reputation
+
+
+def test_reputation_grades_and_persistence(tmp_path):
+    from infomesh_amd.trust.reputation import SummaryReputation
+    p = tmp_path / "rep.json"
+    r = SummaryReputation(path=p)
+    assert r.get("n").grade == "?" and r.get("n").accept
+    for _ in range(6):
+        r.record("good", 0.95)
+    for _ in range(8):
+        r.record("bad", 0.05)
+    assert r.get("good").grade == "A" and r.get("good").accept
+    bad = r.get("bad")
+    assert bad.grade == "F" and not bad.accept
+    assert 0.0 <= r.summary_quality("bad") < 0.4
+    lb = r.leaderboard()
+    assert lb[0].node_id == "good"
+    # reload from disk
+    r2 = SummaryReputation(path=p)
+    assert r2.get("good").samples == 6
