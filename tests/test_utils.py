@@ -257,3 +257,46 @@ def test_cli_update_check(tmp_data_dir):
     r = CliRunner().invoke(cli, ["update-check"])
     assert r.exit_code == 0
     assert "current" in _json.loads(r.output)
+
+
+# --------------------------------------------------------- security_ext
+
+
+def test_security_ext_rbac_and_ipfilter():
+    from infomesh_amd.utils.security_ext import IpFilter, role_allows
+    assert role_allows("admin", "config")
+    assert role_allows("reader", "search")
+    assert not role_allows("reader", "config")
+    assert not role_allows("ghost", "search")
+    f = IpFilter(allow=["127.0.0.0/8", "10.0.0.0/8"], deny=["10.1.0.0/16"])
+    assert f.permitted("127.0.0.1")
+    assert f.permitted("10.2.3.4")
+    assert not f.permitted("10.1.2.3")     # deny wins
+    assert not f.permitted("8.8.8.8")      # not in allow
+    assert not f.permitted("not-an-ip")
+    assert IpFilter().permitted("8.8.8.8")  # default allow
+
+
+def test_security_ext_tokens():
+    from infomesh_amd.utils.security_ext import issue_token, verify_token
+    sec = b"s3cret"
+    t = [1000.0]
+    tok = issue_token(sec, "alice", "operator", ttl_s=60,
+                      now=lambda: t[0])
+    data = verify_token(sec, tok, now=lambda: t[0])
+    assert data and data["sub"] == "alice" and data["role"] == "operator"
+    assert verify_token(b"wrong", tok, now=lambda: t[0]) is None
+    assert verify_token(sec, tok + "x", now=lambda: t[0]) is None
+    t[0] += 61
+    assert verify_token(sec, tok, now=lambda: t[0]) is None
+
+
+def test_security_ext_webhook_hmac():
+    from infomesh_amd.utils.security_ext import (
+        sign_webhook, verify_webhook)
+    sec, body = b"whsec", b'{"event":"crawl"}'
+    h = sign_webhook(sec, body)
+    assert h.startswith("sha256=")
+    assert verify_webhook(sec, body, h)
+    assert not verify_webhook(sec, b"tampered", h)
+    assert not verify_webhook(sec, body, "")
