@@ -197,17 +197,10 @@ extern "C" void infomesh_gemm_bf16_nt(
     else      { if (out_f32) LAUNCH(BMV, BNV, 32, true);                     \
                 else LAUNCH(BMV, BNV, 32, false); }                          \
   } while (0)
-  if (bm == 64 && bn == 64) {
-    // Latency-bound tiles: BK=128 halves the K-step count (and its
-    // per-step vmcnt(0)+barrier drains) for the encoder's K=384/1536
-    // projections; LDS 2x(64+64)x128x2B = 64 KB still fits 2 blocks/CU.
-    if (K % 128 == 0) {
-      if (out_f32) LAUNCH(64, 64, 128, true);
-      else LAUNCH(64, 64, 128, false);
-    } else {
-      PICK(64, 64);
-    }
-  }
+  // (BK=128 for the 64^2 tile was probe-tested and is ~60% SLOWER:
+  // doubling LDS to 64 KB halves resident blocks per CU, which costs
+  // more latency hiding than the halved K-step drains save.)
+  if (bm == 64 && bn == 64) PICK(64, 64);
   else if (bm == 64) PICK(64, 128);
   else PICK(128, 128);
 #undef PICK
