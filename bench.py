@@ -268,6 +268,7 @@ def main():
             result["config"]["phase_ms"] = {
                 k: round(v / args.steps * 1e3, 3)
                 for k, v in phase_t.items()}
+            result["config"]["lat_ms_all"] = [round(x, 2) for x in lat_ms]
         print(json.dumps(result))
     fabric.destroy()
 

@@ -153,7 +153,8 @@ class DistributedQueryPlane:
         tp = _time.perf_counter()
         if torch.cuda.is_available() and self.shard.device.type == "cuda":
             hits = self._search_overlapped(queries_terms, query_emb,
-                                           encode_fn, B, dim, use_dense)
+                                           encode_fn, B, dim, use_dense,
+                                           phase_t=phase_t)
             tp = mark("plane.shard", tp)
         else:
             if encode_fn is not None:
@@ -209,24 +210,39 @@ class DistributedQueryPlane:
                          dense_scores=dn_s)
 
     def _search_overlapped(self, queries_terms, query_emb, encode_fn,
-                           B, dim, use_dense) -> ShardHits:
+                           B, dim, use_dense, phase_t=None) -> ShardHits:
         """BM25 on a side stream || query encoding on the main stream.
 
         Broadcast order (terms, then embeddings) is identical on all
-        ranks; the BM25 launch between the two is rank-local."""
+        ranks; the BM25 launch between the two is rank-local.
+
+        phase_t (diagnostic only): sub-phase wall times WITH syncs —
+        the syncs break the overlap, so only enable to locate time."""
+        import time as _time
+
+        def mark(name, t0):
+            if phase_t is None:
+                return t0
+            torch.cuda.synchronize()
+            t1 = _time.perf_counter()
+            phase_t[name] = phase_t.get(name, 0.0) + (t1 - t0)
+            return t1
         dev = self.shard.device
         self.shard._get_topk().defer_check = True
         self.shard._get_topk_dense().defer_check = True
         if self._bm25_stream is None:
             self._bm25_stream = torch.cuda.Stream(dev)
         # terms must be broadcast before any rank's shard work
+        tp = _time.perf_counter()
         terms = self._broadcast_terms(queries_terms, B)
+        tp = mark("ov.terms", tp)
         main = torch.cuda.current_stream(dev)
         self._bm25_stream.wait_stream(main)
         with torch.cuda.stream(self._bm25_stream):
             bm_vals, bm_idx = self.shard.search_bm25(
                 terms, self.k, scores_buf=self._get_scores_buf(B))
             bm_ids = self.shard.to_global(bm_idx)
+        tp = mark("ov.bm25", tp)
         if self.fabric.rank == 0:
             emb = encode_fn() if encode_fn is not None else query_emb
         else:
@@ -234,9 +250,11 @@ class DistributedQueryPlane:
         emb_t = (emb.to(dev).float() if emb is not None
                  else torch.zeros(B, dim, device=dev))
         self.fabric.broadcast(emb_t)
+        tp = mark("ov.encode", tp)
         if use_dense and self.shard.embeddings is not None:
             dn_vals, dn_idx = self.shard.search_dense(emb_t, self.k)
             dn_ids = self.shard.to_global(dn_idx)
+            tp = mark("ov.dense", tp)
         else:
             k = min(self.k, self.shard.n_docs)
             dn_vals = torch.full((B, k), -float("inf"), device=dev)
