@@ -377,10 +377,18 @@ class GpuShard:
         N = self.n_docs
         k = min(k, N)
         tp = _time.perf_counter()
-        # A fresh buffer: the BM25 scores buffer may still be feeding its
-        # top-k on another stream.
-        d_scores = K.gemm_nt(query_emb.bfloat16(), self.embeddings,
-                             out_f32=True).reshape(B, N)
+        # Persistent scores buffer, separate from the BM25 one (which may
+        # still be feeding its top-k on another stream). Reuse avoids a
+        # multi-GB alloc/free per batch at large N (allocator churn shows
+        # up as periodic multi-ms stalls).
+        if (getattr(self, "_dense_scores_buf", None) is None
+                or self._dense_scores_buf.shape != (1, B, N)):
+            self._dense_scores_buf = torch.empty(
+                1, B, N, device=self.device, dtype=torch.float32)
+        d_scores = K.gemm_nt(query_emb.bfloat16().unsqueeze(0),
+                             self.embeddings,
+                             out_f32=True,
+                             out=self._dense_scores_buf).reshape(B, N)
         tp = mark("shard.dense", tp)
         # Always exact select. The sampled-threshold variant was measured
         # a net loss at every shard size: the candidate slack (~Kp*stride)
