@@ -15,6 +15,16 @@
 
 namespace {
 
+// st_16x32 swizzle on a [rows][64 bf16] row-major LDS image (byte addr
+// = row*128 + col2): flip bit5 (32B group) when row bit2 is set. Applied
+// to the glds SOURCE address and to the ds_read address (rule 21:
+// destination stays lane-linear), it spreads the b128 fragment reads of
+// a wave across banks. PMC before: 1.59e9 LDS bank conflicts in this
+// kernel. BK=64 only (BK=32 rows are 64 B and keep the linear image).
+__device__ __forceinline__ int swz64(int byte_off) {
+  return byte_off ^ (((byte_off >> 9) & 1) << 5);
+}
+
 // Tile configurations (4 waves each):
 //   (128,128): 2x2 wave grid, 64x64 per wave (4x4 fragments) — default.
 //   ( 64,128): 1x4 grid, 64x32 per wave — skinny-M (query scoring).
@@ -74,10 +84,15 @@ __global__ __launch_bounds__(256) void gemm_bf16_nt_kernel(
 #pragma unroll
     for (int c = 0; c < CHUNKS_PER_WAVE; ++c) {
       const int chunk = wid * CHUNKS_PER_WAVE + c;
-      const int row = chunk * ROWS_PER_CHUNK + lrow;
+      // region-relative source byte under the swizzle involution; the
+      // LDS destination stays lane-linear (glds writes base + lane*16)
+      const int lin = chunk * 1024 + lane * 16;
+      const int src = (BK == 64) ? swz64(lin) : lin;
+      const int row = src / (BK * 2);
+      const int colb = src % (BK * 2);
       if (chunk < BCHUNKS2) {
         int brow = n0 + row; brow = brow < N ? brow : N - 1;
-        const bf16* gb = Bg + (long)brow * K + k0 + lcol;
+        const bf16* gb = Bg + (long)brow * K + k0 + colb / 2;
         auto* lb = (__attribute__((address_space(3))) unsigned int*)
             &smem[buf][BM * BK + chunk * 512];
         __builtin_amdgcn_global_load_lds(
@@ -85,7 +100,7 @@ __global__ __launch_bounds__(256) void gemm_bf16_nt_kernel(
       }
       if (chunk < ACHUNKS) {
         int arow = m0 + row; arow = arow < M ? arow : M - 1;
-        const bf16* ga = Ag + (long)arow * K + k0 + lcol;
+        const bf16* ga = Ag + (long)arow * K + k0 + colb / 2;
         auto* la = (__attribute__((address_space(3))) unsigned int*)
             &smem[buf][chunk * 512];
         __builtin_amdgcn_global_load_lds(
@@ -115,14 +130,18 @@ __global__ __launch_bounds__(256) void gemm_bf16_nt_kernel(
 #pragma unroll
       for (int i = 0; i < FM; ++i) {
         const int row = wm * (BM / WM) + i * 16 + fr;
+        const int off = (row * BK + ks * 32 + fk) * 2;
         a[i] = *reinterpret_cast<const bf16x8*>(
-            &smem[cur][row * BK + ks * 32 + fk]);
+            (const char*)&smem[cur][0]
+            + ((BK == 64) ? swz64(off) : off));
       }
 #pragma unroll
       for (int j = 0; j < FN; ++j) {
         const int row = wn * (BN / WN) + j * 16 + fr;
+        const int off = (row * BK + ks * 32 + fk) * 2;
         b[j] = *reinterpret_cast<const bf16x8*>(
-            &smem[cur][BM * BK + row * BK + ks * 32 + fk]);
+            (const char*)&smem[cur][BM * BK]
+            + ((BK == 64) ? swz64(off) : off));
       }
 #pragma unroll
       for (int i = 0; i < FM; ++i)
