@@ -377,27 +377,71 @@ class GpuShard:
         N = self.n_docs
         k = min(k, N)
         tp = _time.perf_counter()
-        # Persistent scores buffer, separate from the BM25 one (which may
-        # still be feeding its top-k on another stream). Reuse avoids a
-        # multi-GB alloc/free per batch at large N (allocator churn shows
-        # up as periodic multi-ms stalls).
-        if (getattr(self, "_dense_scores_buf", None) is None
-                or self._dense_scores_buf.shape != (1, B, N)):
-            self._dense_scores_buf = torch.empty(
-                1, B, N, device=self.device, dtype=torch.float32)
+        tk = self._get_topk_dense()
+        # hipGraph capture of the whole plane (GEMM + top-k passes, all
+        # fixed shapes) removes ~10 launch gaps per batch. Only when the
+        # caller runs deferred top-k verification (the query plane does):
+        # the eager path reads the overflow flag (D2H), which is illegal
+        # inside a capture.
+        if self.device.type == "cuda" and tk.defer_check:
+            key = (B, k)
+            graphs = getattr(self, "_dense_graphs", None)
+            if graphs is None:
+                graphs = self._dense_graphs = {}
+            entry = graphs.get(key)
+            if entry is None:
+                from ..ops.graphs import GraphedCallable
+
+                def _plane(emb, _B=B, _k=k):
+                    d = K.gemm_nt(emb.unsqueeze(0), self.embeddings,
+                                  out_f32=True,
+                                  out=self._get_dense_buf(_B)
+                                  ).reshape(_B, self.n_docs)
+                    return tk(d, _k)
+                g = GraphedCallable(_plane)
+                vals, idx = g(query_emb.bfloat16())  # captures
+                # pin the capture-time workspace: replays write THIS
+                # tensor even if tk._ws is later reallocated
+                entry = (g, tk._ws)
+                graphs[key] = entry
+            else:
+                g, ws = entry
+                vals, idx = g(query_emb.bfloat16())
+                # Graph replay skips the wrapper's python, so re-arm
+                # the deferred overflow check by hand (same static
+                # workspace/layout as at capture; need=0 -> exact).
+                cnt_off = B * 512 + 3 * B
+                flag_off = B * 512 + 4 * B
+                tk._pending.append((ws, cnt_off, flag_off, B, 0))
+            # detach from the graph's static outputs (next replay
+            # overwrites them)
+            out = (vals.clone(), idx.clone())
+            mark("shard.dense+topk", tp)
+            return out
         d_scores = K.gemm_nt(query_emb.bfloat16().unsqueeze(0),
                              self.embeddings,
                              out_f32=True,
-                             out=self._dense_scores_buf).reshape(B, N)
+                             out=self._get_dense_buf(B)).reshape(B, N)
         tp = mark("shard.dense", tp)
         # Always exact select. The sampled-threshold variant was measured
         # a net loss at every shard size: the candidate slack (~Kp*stride)
         # inflates the final bitonic sort by more than the 2 saved passes
         # (~80us/pass at 1.25M docs), and with an 8192-sample + 8192-cap
         # it is only statistically sound for N <~ 2.7M anyway.
-        out = self._get_topk_dense()(d_scores, k)
+        out = tk(d_scores, k)
         mark("shard.densetopk", tp)
         return out
+
+    def _get_dense_buf(self, B: int) -> torch.Tensor:
+        """Persistent dense scores buffer, separate from the BM25 one
+        (which may still be feeding its top-k on another stream). Reuse
+        avoids a multi-GB alloc/free per batch at large N."""
+        N = self.n_docs
+        if (getattr(self, "_dense_scores_buf", None) is None
+                or self._dense_scores_buf.shape != (1, B, N)):
+            self._dense_scores_buf = torch.empty(
+                1, B, N, device=self.device, dtype=torch.float32)
+        return self._dense_scores_buf
 
     def to_global(self, idx: torch.Tensor) -> torch.Tensor:
         safe = idx.clamp(min=0).long()
