@@ -46,8 +46,12 @@ class HybridEngine:
         self._pending_texts: list[str] = []
         self._pending_ids: list[int] = []
         self._built_tokens: list[np.ndarray] = []
-        self._built_texts: list[str] = []
         self._built_ids: list[int] = []
+        # embeddings of already-built docs: encoding is the expensive
+        # part of a flush, so it is INCREMENTAL — only new docs run
+        # through the encoder; the CSR rebuild itself is a fast numpy
+        # sort (segment-merge analogue).
+        self._built_emb: torch.Tensor | None = None
 
     # ------------------------------------------------------------ ingest
     def add_document(self, doc: Document) -> None:
@@ -66,23 +70,23 @@ class HybridEngine:
         return self.shard.n_docs + self.pending_count
 
     def flush(self, embed_batch: int = 256) -> int:
-        """Make pending docs searchable (epoch flip). Rebuilds the shard
-        from all accumulated docs — segment-merge analogue; incremental
-        segment builds are a planned optimization."""
+        """Make pending docs searchable (epoch flip). Embeds only the
+        NEW docs (incremental), then rebuilds the CSR postings from all
+        accumulated docs — the segment-merge analogue."""
         if not self._pending_ids:
             return 0
         self._built_tokens.extend(self._pending_tokens)
-        self._built_texts.extend(self._pending_texts)
         self._built_ids.extend(self._pending_ids)
         n_new = len(self._pending_ids)
+        new_texts = self._pending_texts
         self._pending_tokens, self._pending_texts, self._pending_ids = \
             [], [], []
 
         emb = None
         if self.encoder is not None:
             chunks = []
-            for i in range(0, len(self._built_texts), embed_batch):
-                batch = self._built_texts[i:i + embed_batch]
+            for i in range(0, len(new_texts), embed_batch):
+                batch = new_texts[i:i + embed_batch]
                 pad = embed_batch - len(batch)
                 if pad:
                     # fixed batch shape keeps the encoder's hipGraph
@@ -90,7 +94,11 @@ class HybridEngine:
                     batch = batch + [""] * pad
                 enc = self.encoder.encode_texts(batch).bfloat16()
                 chunks.append(enc[:embed_batch - pad] if pad else enc)
-            emb = torch.cat(chunks, 0)
+            new_emb = torch.cat(chunks, 0)
+            self._built_emb = (new_emb if self._built_emb is None
+                               else torch.cat([self._built_emb, new_emb],
+                                              0))
+            emb = self._built_emb
 
         lens = np.array([max(len(t), 1) for t in self._built_tokens],
                         dtype=np.int64)

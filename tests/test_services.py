@@ -189,3 +189,41 @@ def test_governor_levels():
 def test_preflight(tmp_path):
     problems = run_preflight_checks(tmp_path)
     assert problems == [] or all("disk" not in p for p in problems)
+
+
+def test_engine_flush_is_incremental_on_encoding():
+    """A second flush must encode ONLY the new docs (embedding reuse),
+    and search must still see docs from both flushes."""
+    import torch
+    from infomesh_amd.engine import HybridEngine
+    from infomesh_amd.index.local_store import Document
+
+    class StubEncoder:
+        def __init__(self):
+            self.rows = 0
+
+        def encode_texts(self, texts):
+            self.rows += len(texts)
+            g = torch.Generator().manual_seed(hash(tuple(texts)) & 0xFFFF)
+            return torch.nn.functional.normalize(
+                torch.randn(len(texts), 384, generator=g), dim=-1)
+
+    eng = HybridEngine(device="cpu", use_encoder=False)
+    eng.encoder = StubEncoder()
+    for i in range(3):
+        eng.add_document(Document(url=f"u{i}", title=f"alpha doc{i}",
+                                  text="alpha beta", doc_id=i))
+    eng.flush(embed_batch=4)
+    first = eng.encoder.rows
+    assert eng.shard.n_docs == 3
+    for i in range(3, 5):
+        eng.add_document(Document(url=f"u{i}", title=f"gamma doc{i}",
+                                  text="gamma delta", doc_id=i))
+    eng.flush(embed_batch=4)
+    # second flush encoded one padded batch (4 rows), NOT all 5 docs
+    assert eng.encoder.rows - first == 4
+    assert eng.shard.n_docs == 5
+    assert eng._built_emb.shape[0] == 5
+    hits_old = eng.search("alpha", limit=5)
+    hits_new = eng.search("gamma", limit=5)
+    assert hits_old and hits_new
