@@ -14,7 +14,6 @@ from collections import defaultdict, deque
 from fastapi import FastAPI, HTTPException, Request
 from fastapi.responses import JSONResponse, PlainTextResponse
 
-from ..config import save_config
 from ..search.formatter import result_to_dict
 from ..services import AppContext
 from ..utils.observability import MetricsRegistry

@@ -6,7 +6,6 @@ peer spot-checks via sampled Merkle proofs).
 """
 from __future__ import annotations
 
-import json
 import random
 
 from ..trust.keys import KeyPair

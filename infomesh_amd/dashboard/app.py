@@ -9,9 +9,7 @@ Overview, Index, GPU engine, Crawl, Credits, Cache/Search.
 from __future__ import annotations
 
 import time
-from pathlib import Path
 
-from rich.console import Group
 from rich.layout import Layout
 from rich.live import Live
 from rich.panel import Panel

@@ -7,8 +7,8 @@ from __future__ import annotations
 
 from typing import Any
 
-from ..index.local_store import LocalStore, SearchHit
-from ..index.ranking import (ScoreBreakdown, W_AUTHORITY, W_BM25,
+from ..index.local_store import LocalStore
+from ..index.ranking import (W_AUTHORITY, W_BM25,
                              W_FRESHNESS, W_TITLE, W_TRUST, W_URL_PATH,
                              rank_local_results)
 

@@ -8,7 +8,6 @@ detection via score deviation + snippet similarity).
 from __future__ import annotations
 
 import statistics
-import time
 from typing import Any, Callable, Sequence
 
 from .cache import QueryCache

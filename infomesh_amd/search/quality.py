@@ -7,7 +7,6 @@ intent lives in passage.classify_intent).
 from __future__ import annotations
 
 import math
-import random
 import re
 from dataclasses import dataclass, field
 from typing import Sequence

@@ -11,7 +11,6 @@ from __future__ import annotations
 import logging
 import time
 from dataclasses import dataclass, field
-from pathlib import Path
 from typing import Any
 
 from .config import Config, load_config
@@ -28,7 +27,7 @@ from .index.link_graph import LinkGraph
 from .index.local_store import Document, LocalStore
 from .search.cache import QueryCache
 from .search.nlp import RelatedSearchTracker
-from .search.query import (SearchResponse, search_distributed, search_hybrid,
+from .search.query import (SearchResponse, search_hybrid,
                            search_local)
 from .trust.attestation import create_attestation
 from .trust.dmca import TakedownManager

@@ -5,7 +5,6 @@ from __future__ import annotations
 import asyncio
 
 import httpx
-import pytest
 
 from infomesh_amd.config import CrawlConfig
 from infomesh_amd.crawler.dedup import DeduplicatorDB, normalize_url

@@ -19,7 +19,7 @@ from infomesh_amd.index.local_store import Document
 from infomesh_amd.integrations import (InfoMeshDocumentStore,
                                        InfoMeshReader, InfoMeshRetriever)
 from infomesh_amd.mcp.server import McpServer
-from infomesh_amd.mcp.tools import TOOLS, resolve_tool
+from infomesh_amd.mcp.tools import resolve_tool
 from infomesh_amd.sdk.client import InfoMeshClient
 from infomesh_amd.services import AppContext
 

@@ -9,7 +9,6 @@ from __future__ import annotations
 import os
 
 import numpy as np
-import pytest
 import torch
 
 from infomesh_amd.index.gpu_index import CpuShard, BM25_VOCAB, bm25_term_ids

@@ -6,7 +6,7 @@ import struct
 import pytest
 
 from infomesh_amd.errors import InfoMeshError
-from infomesh_amd.index.local_store import Document, LocalStore
+from infomesh_amd.index.local_store import LocalStore
 from infomesh_amd.index.snapshot import (export_snapshot, import_snapshot,
                                          read_snapshot_header, FORMAT_VERSION)
 

@@ -231,7 +231,6 @@ def test_cli_shard_roundtrip(tmp_data_dir):
     runner = CliRunner()
     # index a doc, save the (CPU) shard, reload and probe it
     from infomesh_amd.services import AppContext
-    from infomesh_amd.engine import HybridEngine
     from infomesh_amd.index.local_store import Document
     ctx = AppContext.create(with_engine=False, with_worker=False)
     ctx.index_document(Document(url="https://a.com/1",
