@@ -3,7 +3,8 @@
 Reference parity: infomesh/mcp/server.py (stdio + Streamable-HTTP
 transports, tool dispatch with legacy aliases, query cache wiring,
 API-key check, analytics). The protocol surface implements MCP
-`initialize`, `tools/list`, `tools/call`, `ping`.
+`initialize`, `tools/list`, `tools/call`, `resources/list`,
+`resources/read`, `prompts/list`, `prompts/get`, `ping`.
 """
 from __future__ import annotations
 
@@ -15,7 +16,7 @@ from typing import Any, TextIO
 from ..services import AppContext
 from .handlers import Handlers
 from .session import AnalyticsTracker, SessionStore
-from .tools import TOOLS
+from .tools import PROMPTS, RESOURCES, TOOLS
 
 PROTOCOL_VERSION = "2024-11-05"
 SERVER_INFO = {"name": "infomesh-amd", "version": "0.1.0"}
@@ -41,7 +42,8 @@ class McpServer:
                 result = {
                     "protocolVersion": PROTOCOL_VERSION,
                     "serverInfo": SERVER_INFO,
-                    "capabilities": {"tools": {}},
+                    "capabilities": {"tools": {}, "resources": {},
+                                     "prompts": {}},
                 }
             elif method == "notifications/initialized":
                 return None
@@ -51,6 +53,16 @@ class McpServer:
                 result = {"tools": TOOLS}
             elif method == "tools/call":
                 result = self._call_tool(params)
+            elif method == "resources/list":
+                result = {"resources": RESOURCES}
+            elif method == "resources/read":
+                result = self._read_resource(params)
+            elif method == "prompts/list":
+                result = {"prompts": [
+                    {k: v for k, v in p.items() if k != "template"}
+                    for p in PROMPTS]}
+            elif method == "prompts/get":
+                result = self._get_prompt(params)
             else:
                 return self._error(mid, -32601, f"method {method!r} not found")
         except Exception as e:  # tool errors -> JSON-RPC error
@@ -58,6 +70,45 @@ class McpServer:
         if mid is None:
             return None
         return {"jsonrpc": "2.0", "id": mid, "result": result}
+
+    def _read_resource(self, params: dict[str, Any]) -> dict[str, Any]:
+        import json as _json
+        uri = params.get("uri", "")
+        if uri == "infomesh://index/stats":
+            body = _json.dumps(self.ctx.store.stats(), default=str)
+        elif uri == "infomesh://node/status":
+            from ..runtime import RuntimeStatus
+            body = _json.dumps(
+                RuntimeStatus(self.ctx.config.data_dir).read(),
+                default=str)
+        elif uri == "infomesh://credits/balance":
+            body = _json.dumps(
+                self.ctx.ledger.summary() if self.ctx.ledger else {},
+                default=str)
+        elif uri.startswith("infomesh://doc/"):
+            doc = self.ctx.store.get_document_by_url(
+                uri[len("infomesh://doc/"):])
+            if doc is None:
+                raise ValueError(f"no document for {uri!r}")
+            body = _json.dumps({"url": doc.url, "title": doc.title,
+                                "text": doc.text[:20000]}, default=str)
+        else:
+            raise ValueError(f"unknown resource {uri!r}")
+        return {"contents": [{"uri": uri, "mimeType": "application/json",
+                              "text": body}]}
+
+    def _get_prompt(self, params: dict[str, Any]) -> dict[str, Any]:
+        name = params.get("name", "")
+        args = params.get("arguments") or {}
+        for p in PROMPTS:
+            if p["name"] == name:
+                text = p["template"].format(**{
+                    a["name"]: args.get(a["name"], "")
+                    for a in p.get("arguments", [])})
+                return {"description": p["description"],
+                        "messages": [{"role": "user", "content": {
+                            "type": "text", "text": text}}]}
+        raise ValueError(f"unknown prompt {name!r}")
 
     def _call_tool(self, params: dict[str, Any]) -> dict[str, Any]:
         name = params.get("name", "")

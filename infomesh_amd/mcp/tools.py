@@ -100,3 +100,38 @@ def resolve_tool(name: str) -> str | None:
     if name in {t["name"] for t in TOOLS}:
         return name
     return LEGACY_ALIASES.get(name)
+
+
+# MCP resources (read-only context an agent can pull without a tool
+# call) and prompt templates — reference parity for the MCP surface.
+RESOURCES = [
+    {"uri": "infomesh://index/stats", "name": "index-stats",
+     "description": "Local index statistics (doc counts, domains, size)",
+     "mimeType": "application/json"},
+    {"uri": "infomesh://node/status", "name": "node-status",
+     "description": "Runtime status heartbeat of the node",
+     "mimeType": "application/json"},
+    {"uri": "infomesh://credits/balance", "name": "credits-balance",
+     "description": "Credit ledger summary for this node",
+     "mimeType": "application/json"},
+    {"uri": "infomesh://doc/{url}", "name": "document",
+     "description": "An indexed document by URL (title + text)",
+     "mimeType": "application/json"},
+]
+
+PROMPTS = [
+    {"name": "research",
+     "description": "Research a topic using the local index",
+     "arguments": [{"name": "topic", "description": "what to research",
+                    "required": True}],
+     "template": ("Research the topic {topic!s} using web_search over "
+                  "the local index. Cross-check claims with fact_check "
+                  "and cite the source URLs for every statement.")},
+    {"name": "summarize-url",
+     "description": "Fetch and summarize a page from the index/web",
+     "arguments": [{"name": "url", "description": "page URL",
+                    "required": True}],
+     "template": ("Use fetch_page on {url!s} and produce a faithful "
+                  "summary with the key facts; note anything that "
+                  "looks paywalled or truncated.")},
+]

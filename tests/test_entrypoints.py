@@ -288,3 +288,42 @@ def test_document_store(ctx):
     assert n == 1
     assert ds.count_documents() == 7
     assert ds.filter_documents("haystack")
+
+
+def test_mcp_resources_and_prompts(tmp_path, monkeypatch):
+    monkeypatch.setenv("INFOMESH_NODE_DATA_DIR", str(tmp_path))
+    from infomesh_amd.mcp.server import McpServer
+    ctx = AppContext.create(with_engine=False, with_worker=False)
+    try:
+        srv = McpServer(ctx)
+        init = srv.handle_message({"jsonrpc": "2.0", "id": 1,
+                                   "method": "initialize", "params": {}})
+        caps = init["result"]["capabilities"]
+        assert "resources" in caps and "prompts" in caps
+        rl = srv.handle_message({"jsonrpc": "2.0", "id": 2,
+                                 "method": "resources/list"})
+        uris = [r["uri"] for r in rl["result"]["resources"]]
+        assert "infomesh://index/stats" in uris
+        rd = srv.handle_message({"jsonrpc": "2.0", "id": 3,
+                                 "method": "resources/read",
+                                 "params": {"uri": "infomesh://index/stats"}})
+        import json as _json
+        body = _json.loads(rd["result"]["contents"][0]["text"])
+        assert "documents" in body or body  # stats dict present
+        pl = srv.handle_message({"jsonrpc": "2.0", "id": 4,
+                                 "method": "prompts/list"})
+        names = [p["name"] for p in pl["result"]["prompts"]]
+        assert "research" in names
+        assert all("template" not in p for p in pl["result"]["prompts"])
+        pg = srv.handle_message({"jsonrpc": "2.0", "id": 5,
+                                 "method": "prompts/get",
+                                 "params": {"name": "research",
+                                            "arguments": {"topic": "gpus"}}})
+        txt = pg["result"]["messages"][0]["content"]["text"]
+        assert "gpus" in txt
+        err = srv.handle_message({"jsonrpc": "2.0", "id": 6,
+                                  "method": "resources/read",
+                                  "params": {"uri": "infomesh://nope"}})
+        assert "error" in err
+    finally:
+        ctx.close()
