@@ -36,7 +36,8 @@ class AuditResult:
 class AuditScheduler:
     def __init__(self, store: LocalStore, trust: TrustStore,
                  fetch_fn: FetchFn, rate_per_hour: float = AUDITS_PER_HOUR,
-                 auditors: int = N_AUDITORS, rng: random.Random | None = None):
+                 auditors: int = N_AUDITORS, rng: random.Random | None = None,
+                 detector=None):
         self.store = store
         self.trust = trust
         self.fetch_fn = fetch_fn
@@ -45,6 +46,9 @@ class AuditScheduler:
         self.rng = rng or random.Random()
         self.last_audit = 0.0
         self.history: list[AuditResult] = []
+        # trust/detector.MaliciousNodeDetector: failed audits feed the
+        # threat ladder so repeat offenders escalate toward isolation.
+        self.detector = detector
 
     def due(self, now: float | None = None) -> bool:
         now = now or time.time()
@@ -85,7 +89,10 @@ class AuditScheduler:
             # an all-stale result only mildly penalizes (handled by EMA).
             passed = sum(votes) * 2 >= len(votes)
             result = AuditResult(url, passed=passed, votes=votes)
-            self.trust.record_audit(extract_domain(url), passed)
+            domain = extract_domain(url)
+            self.trust.record_audit(domain, passed)
+            if self.detector is not None and not passed:
+                self.detector.record(domain, "audit_fail")
         self.history.append(result)
         return result
 

@@ -222,3 +222,29 @@ def test_reputation_grades_and_persistence(tmp_path):
     # reload from disk
     r2 = SummaryReputation(path=p)
     assert r2.get("good").samples == 6
+
+
+def test_audit_failures_feed_detector(tmp_path):
+    import asyncio
+    from infomesh_amd.index.local_store import Document, LocalStore
+    from infomesh_amd.trust.audit import AuditScheduler
+    from infomesh_amd.trust.detector import MaliciousNodeDetector
+    from infomesh_amd.trust.scoring import TrustStore
+    store = LocalStore(tmp_path / "ls.db")
+    store.add_document(Document(url="https://bad.example/x",
+                                title="t", text="original body text"))
+    trust = TrustStore(tmp_path / "trust.db")
+    det = MaliciousNodeDetector()
+
+    async def tampered_fetch(url):
+        return "tampered content entirely different"
+
+    sched = AuditScheduler(store, trust, tampered_fetch, detector=det)
+    for _ in range(3):
+        res = asyncio.run(sched.run_audit("https://bad.example/x"))
+        assert res is not None and not res.passed
+    threat = det.assess("bad.example")
+    assert threat.events.get("audit_fail", 0) == 3
+    assert threat.score > 0
+    store.close()
+    trust.close()
