@@ -75,16 +75,14 @@ class HybridEngine:
         accumulated docs — the segment-merge analogue."""
         if not self._pending_ids:
             return 0
-        self._built_tokens.extend(self._pending_tokens)
-        self._built_ids.extend(self._pending_ids)
         n_new = len(self._pending_ids)
-        new_texts = self._pending_texts
-        self._pending_tokens, self._pending_texts, self._pending_ids = \
-            [], [], []
-
-        emb = None
+        # Encode BEFORE committing any state: a failed embed (OOM, ...)
+        # leaves the engine exactly as it was — the old epoch keeps
+        # serving and the pending docs stay pending for a retry.
+        new_emb = None
         if self.encoder is not None:
             chunks = []
+            new_texts = self._pending_texts
             for i in range(0, len(new_texts), embed_batch):
                 batch = new_texts[i:i + embed_batch]
                 pad = embed_batch - len(batch)
@@ -95,6 +93,13 @@ class HybridEngine:
                 enc = self.encoder.encode_texts(batch).bfloat16()
                 chunks.append(enc[:embed_batch - pad] if pad else enc)
             new_emb = torch.cat(chunks, 0)
+
+        self._built_tokens.extend(self._pending_tokens)
+        self._built_ids.extend(self._pending_ids)
+        self._pending_tokens, self._pending_texts, self._pending_ids = \
+            [], [], []
+        emb = None
+        if new_emb is not None:
             self._built_emb = (new_emb if self._built_emb is None
                                else torch.cat([self._built_emb, new_emb],
                                               0))
