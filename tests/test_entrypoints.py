@@ -327,3 +327,13 @@ def test_mcp_resources_and_prompts(tmp_path, monkeypatch):
         assert "error" in err
     finally:
         ctx.close()
+
+
+def test_driver_contract_script():
+    import subprocess, sys, pathlib
+    root = pathlib.Path(__file__).resolve().parents[1]
+    out = subprocess.run([sys.executable, "scripts/verify_contract.py"],
+                         cwd=root, capture_output=True, text=True,
+                         timeout=900)
+    assert out.returncode == 0, out.stdout + out.stderr
+    assert "ALL OK" in out.stdout
