@@ -33,17 +33,25 @@ class Fabric:
                  timeout_s: float = 300.0):
         self.rank = env_rank()
         self.world = env_world()
+        # Bind this process to its GPU BEFORE the process group exists:
+        # otherwise every rank's first CUDA touch lands on device 0 and
+        # RCCL communicators bind to the wrong device.
+        use_cuda = torch.cuda.is_available()
+        if use_cuda and self.world > 1:
+            torch.cuda.set_device(env_local_rank())
         if self.world > 1 and not dist.is_initialized():
             if backend is None:
-                backend = "nccl" if torch.cuda.is_available() else "gloo"
+                backend = "nccl" if use_cuda else "gloo"
             os.environ.setdefault("MASTER_ADDR", "127.0.0.1")
             os.environ.setdefault("MASTER_PORT", "29500")
+            kwargs = {}
+            if backend == "nccl" and use_cuda:
+                kwargs["device_id"] = torch.device("cuda",
+                                                   env_local_rank())
             dist.init_process_group(
                 backend=backend, rank=self.rank, world_size=self.world,
-                timeout=datetime.timedelta(seconds=timeout_s))
+                timeout=datetime.timedelta(seconds=timeout_s), **kwargs)
         self.backend = dist.get_backend() if dist.is_initialized() else "none"
-        if torch.cuda.is_available() and self.world > 1:
-            torch.cuda.set_device(env_local_rank())
 
     @property
     def initialized(self) -> bool:
