@@ -39,7 +39,9 @@ def export_snapshot(store: LocalStore, path: str | Path,
     for doc in store.export_documents():
         docs.append({
             "url": doc.url, "title": doc.title, "text": doc.text,
-            "language": doc.language, "text_hash": doc.text_hash,
+            "language": doc.language,
+            "raw_html_hash": doc.raw_hash,  # reference field name
+            "text_hash": doc.text_hash,
             "crawled_at": doc.crawled_at,
         })
         if len(docs) >= max_docs:
@@ -47,6 +49,9 @@ def export_snapshot(store: LocalStore, path: str | Path,
     header = {
         "format_version": FORMAT_VERSION,
         "created_at": time.time(),
+        # the reference reads "document_count" (snapshot.py:85); keep
+        # "doc_count" too for our own earlier snapshots
+        "document_count": len(docs),
         "doc_count": len(docs),
         "node": node_name,
         "generator": "infomesh-amd",
@@ -123,6 +128,7 @@ def import_snapshot(store: LocalStore, path: str | Path,
             text=str(d.get("text", "")),
             language=str(d.get("language", "")),
             text_hash=str(d.get("text_hash", "")) or content_hash(str(d.get("text", ""))),
+            raw_hash=str(d.get("raw_html_hash", d.get("raw_hash", ""))),
             crawled_at=float(d.get("crawled_at", 0.0) or 0.0),
         )
         rowid = store.add_document(doc)

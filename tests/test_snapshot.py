@@ -64,3 +64,35 @@ def test_on_document_hook(seeded_store, tmp_path):
     assert len(seen) == seeded_store.count()
     assert all(d.doc_id is not None for d in seen)
     dst.close()
+
+
+def test_snapshot_reference_field_parity(tmp_path):
+    """Byte-level interop with the reference format: header carries
+    `document_count` and docs carry `raw_html_hash` (reference
+    snapshot.py:82-90, local_store.py:487-498)."""
+    import json
+    import struct
+
+    import msgpack
+
+    from infomesh_amd import compression
+    from infomesh_amd.index.local_store import Document
+    from infomesh_amd.index.snapshot import export_snapshot
+
+    store = LocalStore(tmp_path / "s.db")
+    store.add_document(Document(url="https://p/1", title="t",
+                                text="body " * 20, language="en",
+                                raw_hash="rawhash123"))
+    path = tmp_path / "p.infomesh-snapshot"
+    export_snapshot(store, path, node_name="n")
+    raw = path.read_bytes()
+    hlen = struct.unpack(">I", raw[:4])[0]
+    comp = compression.Compressor()
+    header = json.loads(comp.decompress(raw[4:4 + hlen]))
+    assert header["document_count"] == 1
+    assert header["format_version"] == FORMAT_VERSION
+    docs = msgpack.unpackb(comp.decompress(raw[4 + hlen:]), raw=False)
+    assert docs[0]["raw_html_hash"] == "rawhash123"
+    assert set(docs[0]) >= {"url", "title", "text", "language",
+                            "raw_html_hash", "text_hash", "crawled_at"}
+    store.close()
