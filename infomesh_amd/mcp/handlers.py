@@ -31,10 +31,30 @@ class Handlers:
     # ------------------------------------------------------------ search
     def web_search(self, query: str, limit: int = 10, mode: str = "auto",
                    explain: bool = False, chunk_size: int = 0,
-                   answer_mode: bool = False, rerank: bool = False,
-                   summarize: bool = False, **_) -> dict[str, Any]:
+                   answer_mode: bool | str = False, rerank: bool = False,
+                   summarize: bool = False, top_k: int | None = None,
+                   local_only: bool = False, language: str = "",
+                   recency_days: float = 0,
+                   domain_allowlist: list[str] | None = None,
+                   domain_blocklist: list[str] | None = None,
+                   fetch_full_content: bool = False,
+                   **_) -> dict[str, Any]:
         if not query or not query.strip():
             raise InfoMeshError("SRCH001", "empty query")
+        # Reference-style argument forms (infomesh mcp/tools.py:53-136):
+        # top_k aliases limit; local_only forces the local mode;
+        # answer_mode may be the 'snippets'/'answer'/'summary' enum.
+        if top_k:
+            limit = int(top_k)
+        if local_only:
+            mode = "local"
+        if isinstance(answer_mode, str):
+            summarize = summarize or answer_mode == "summary"
+            answer_mode = answer_mode in ("answer", "summary")
+        if answer_mode and not chunk_size:
+            chunk_size = 800
+        if language:
+            query = f"{query} lang:{language}"
         limit = max(1, min(int(limit), 50))
         if explain:
             return {"mode": "explain",
@@ -42,10 +62,34 @@ class Handlers:
                         self.ctx.store, query, limit,
                         authority_fn=self.ctx.link_graph.url_authority,
                         trust_fn=self.ctx.trust.trust_fn())}
-        resp = self.ctx.search(query, limit=limit, mode=mode)
+        over = limit * 2 if (domain_allowlist or domain_blocklist
+                             or recency_days) else limit
+        resp = self.ctx.search(query, limit=over, mode=mode)
         results = [result_to_dict(r) for r in resp.results]
+        if domain_allowlist or domain_blocklist or recency_days:
+            from urllib.parse import urlparse
+            import time as _t
+            cutoff = _t.time() - float(recency_days) * 86400 \
+                if recency_days else None
+
+            def _keep(r):
+                dom = urlparse(r["url"]).netloc.lower()
+                if domain_allowlist and not any(
+                        dom == a.lower() or dom.endswith("." + a.lower())
+                        for a in domain_allowlist):
+                    return False
+                if domain_blocklist and any(
+                        dom == b.lower() or dom.endswith("." + b.lower())
+                        for b in domain_blocklist):
+                    return False
+                if cutoff is not None:
+                    doc = self.ctx.store.get_document_by_url(r["url"])
+                    if doc is None or doc.crawled_at < cutoff:
+                        return False
+                return True
+            results = [r for r in results if _keep(r)][:limit]
         # attach full text for RAG modes
-        if chunk_size or answer_mode or summarize:
+        if chunk_size or answer_mode or summarize or fetch_full_content:
             for r in results:
                 doc = self.ctx.store.get_document_by_url(r["url"])
                 if doc:

@@ -18,11 +18,22 @@ TOOLS: list[dict] = [
             "properties": {
                 "query": {"type": "string"},
                 "limit": {"type": "integer", "minimum": 1, "maximum": 50},
+                "top_k": {"type": "integer", "minimum": 1, "maximum": 50,
+                          "description": "alias of limit (reference name)"},
                 "mode": {"type": "string",
                          "enum": ["auto", "local", "hybrid", "distributed"]},
+                "local_only": {"type": "boolean"},
+                "language": {"type": "string"},
+                "recency_days": {"type": "number"},
+                "domain_allowlist": {"type": "array",
+                                     "items": {"type": "string"}},
+                "domain_blocklist": {"type": "array",
+                                     "items": {"type": "string"}},
+                "fetch_full_content": {"type": "boolean"},
                 "explain": {"type": "boolean"},
                 "chunk_size": {"type": "integer"},
-                "answer_mode": {"type": "boolean"},
+                "answer_mode": {
+                    "description": "bool, or 'snippets'|'answer'|'summary'"},
                 "rerank": {"type": "boolean"},
                 "summarize": {"type": "boolean"},
             },

@@ -337,3 +337,38 @@ def test_driver_contract_script():
                          timeout=900)
     assert out.returncode == 0, out.stdout + out.stderr
     assert "ALL OK" in out.stdout
+
+
+def test_web_search_reference_style_args(tmp_path, monkeypatch):
+    """Reference agents call web_search with top_k / domain lists /
+    recency_days / answer_mode enum (infomesh mcp/tools.py:53-136) —
+    all must be honored."""
+    monkeypatch.setenv("INFOMESH_NODE_DATA_DIR", str(tmp_path))
+    from infomesh_amd.index.local_store import Document
+    ctx = AppContext.create(with_engine=False, with_worker=False)
+    try:
+        for i, dom in enumerate(["good.org", "good.org", "spam.net"]):
+            ctx.index_document(Document(
+                url=f"https://{dom}/p{i}", title=f"quantum doc {i}",
+                text="quantum computing research " * 10))
+        h = ctx.handlers if hasattr(ctx, "handlers") else None
+        from infomesh_amd.mcp.handlers import Handlers
+        h = Handlers(ctx)
+        out = h.web_search(query="quantum", top_k=5,
+                           domain_allowlist=["good.org"])
+        assert out["results"]
+        assert all("good.org" in r["url"] for r in out["results"])
+        out = h.web_search(query="quantum", top_k=5,
+                           domain_blocklist=["good.org"])
+        assert all("spam.net" in r["url"] for r in out["results"])
+        out = h.web_search(query="quantum", recency_days=30, top_k=5)
+        assert out["results"]  # crawled just now -> within 30 days
+        out = h.web_search(query="quantum", answer_mode="answer", top_k=3)
+        assert "answer" in out and "chunks" in out
+        out = h.web_search(query="quantum", local_only=True, top_k=2)
+        assert len(out["results"]) <= 2
+        out = h.web_search(query="quantum", fetch_full_content=True,
+                           top_k=2)
+        assert all("text" in r for r in out["results"])
+    finally:
+        ctx.close()
