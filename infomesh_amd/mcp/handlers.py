@@ -151,17 +151,24 @@ class Handlers:
                 "truncated": len(doc.text) > max_chars}
 
     # ------------------------------------------------------------- crawl
-    def crawl_url(self, url: str, force: bool = False, **_) -> dict[str, Any]:
+    def crawl_url(self, url: str, force: bool = False, depth: int = 0,
+                  **_) -> dict[str, Any]:
         now = time.time()
         self._crawl_times = [t for t in self._crawl_times if now - t < 3600]
         if len(self._crawl_times) >= CRAWL_RATE_PER_HOUR:
             raise InfoMeshError("SEC001",
                                 f"crawl rate limit {CRAWL_RATE_PER_HOUR}/hr")
         self._crawl_times.append(now)
-        return asyncio.run(self.ctx.crawl_and_index(url, force=force))
+        # depth flows to the crawl worker, which BFS-schedules the
+        # page's links within its own politeness/budget limits
+        return asyncio.run(self.ctx.crawl_and_index(url, depth=depth,
+                                                    force=force))
 
     # -------------------------------------------------------- fact check
-    def fact_check(self, claim: str, limit: int = 5, **_) -> dict[str, Any]:
+    def fact_check(self, claim: str, limit: int = 5,
+                   top_k: int | None = None, **_) -> dict[str, Any]:
+        if top_k:
+            limit = int(top_k)
         from ..summarizer.verify import _fact_support  # same scorer
         hits = self.ctx.store.search(claim, limit=limit)
         if not hits:  # recall-first retry: any-term match

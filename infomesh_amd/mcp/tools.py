@@ -62,6 +62,8 @@ TOOLS: list[dict] = [
             "properties": {
                 "url": {"type": "string"},
                 "force": {"type": "boolean"},
+                "depth": {"type": "integer", "minimum": 0, "maximum": 2,
+                          "default": 0},
             },
             "required": ["url"],
         },
@@ -75,6 +77,8 @@ TOOLS: list[dict] = [
             "properties": {
                 "claim": {"type": "string"},
                 "limit": {"type": "integer"},
+                "top_k": {"type": "integer",
+                          "description": "alias of limit (reference name)"},
             },
             "required": ["claim"],
         },

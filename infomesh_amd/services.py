@@ -171,6 +171,7 @@ class AppContext:
                                       etag=res.etag,
                                       last_modified=res.last_modified)
         out.update({"doc_id": rowid, "title": page.title,
+                    "links": page.links[:50],
                     "links_scheduled": res.links_scheduled,
                     "indexed": rowid is not None})
         return out
