@@ -193,6 +193,90 @@ class Handlers:
         return self.ctx.status()
 
     # ---------------------------------------------------------- dispatch
+    # -------- legacy utility tools (reference mcp/server.py:205-457) ---
+
+    def ping(self, **_) -> dict[str, Any]:
+        return {"pong": True, "ts": time.time()}
+
+    def credit_balance(self, **_) -> dict[str, Any]:
+        if self.ctx.ledger is None:
+            return {"enabled": False}
+        return self.ctx.ledger.stats()
+
+    def index_stats(self, **_) -> dict[str, Any]:
+        return self.ctx.store.stats()
+
+    def network_stats(self, **_) -> dict[str, Any]:
+        """Single-node analogue of the reference's peer stats: the GPU
+        shard fabric replaces the libp2p swarm."""
+        eng = self.ctx.engine
+        out: dict[str, Any] = {"world_size": 1, "shards": []}
+        if eng is not None:
+            out["world_size"] = eng.fabric.world
+            out["shards"] = [{"rank": eng.fabric.rank,
+                              "docs": eng.shard.n_docs,
+                              "hbm_bytes": eng.shard.hbm_bytes(),
+                              "device": str(eng.shard.device)}]
+        return out
+
+    def batch_search(self, queries: list[str] | None = None,
+                     limit: int = 10, **_) -> dict[str, Any]:
+        from ..search.extended import batch_search as _bs
+        queries = list(queries or [])[:20]
+        resps = _bs(lambda q: self.web_search(query=q, limit=limit),
+                    queries)
+        return {"batches": resps}
+
+    def search_history(self, limit: int = 20, query: str = "",
+                       **_) -> dict[str, Any]:
+        tracker = getattr(self.ctx, "related", None)
+        if tracker is None:
+            return {"history": []}
+        if query:
+            return {"related": tracker.related(query, limit=limit)}
+        hist = list(getattr(tracker, "_recent", []))[-limit:]
+        return {"history": hist}
+
+    def analytics(self, **_) -> dict[str, Any]:
+        tracker = getattr(self, "_analytics", None)
+        return tracker.report() if tracker is not None else {}
+
+    def register_webhook(self, event: str = "", url: str = "",
+                         **_) -> dict[str, Any]:
+        reg = getattr(self, "_webhooks", None)
+        if reg is None or not event:
+            raise InfoMeshError("SRCH001", "webhook registry unavailable "
+                                           "or event missing")
+        if url:
+            def _post(payload, _url=url):  # fired by reg.fire(event,...)
+                import json as _json
+                import urllib.request
+                req = urllib.request.Request(
+                    _url, data=_json.dumps(payload).encode(),
+                    headers={"Content-Type": "application/json"})
+                urllib.request.urlopen(req, timeout=5)
+            reg.register(event, _post)
+        return {"registered": event}
+
+    def unregister_webhook(self, event: str = "", **_) -> dict[str, Any]:
+        reg = getattr(self, "_webhooks", None)
+        if reg is None:
+            raise InfoMeshError("SRCH001", "webhook registry unavailable")
+        n = len(reg._hooks.pop(event, [])) if hasattr(reg, "_hooks") else 0
+        return {"unregistered": event, "removed": n}
+
+    def remove_url(self, url: str = "", reason: str = "user-request",
+                   **_) -> dict[str, Any]:
+        """GDPR-backed removal: records a durable deletion so the URL
+        cannot re-enter via crawl/import (reference remove_url)."""
+        if not url:
+            raise InfoMeshError("SRCH001", "url required")
+        self.ctx.deletions.request_deletion(url, reason=reason)
+        self.ctx.deletions.enforce()
+        removed = not self.ctx.store.get_document_by_url(url)
+        return {"url": url, "removed": removed,
+                "deletion_recorded": True}
+
     def call(self, tool: str, args: dict[str, Any]) -> dict[str, Any]:
         from .tools import resolve_tool
         resolved = resolve_tool(tool)

@@ -15,7 +15,7 @@ from typing import Any, TextIO
 
 from ..services import AppContext
 from .handlers import Handlers
-from .session import AnalyticsTracker, SessionStore
+from .session import AnalyticsTracker, SessionStore, WebhookRegistry
 from .tools import PROMPTS, RESOURCES, TOOLS
 
 PROTOCOL_VERSION = "2024-11-05"
@@ -31,6 +31,10 @@ class McpServer:
                                  summarizer=summarizer)
         self.sessions = SessionStore()
         self.analytics = AnalyticsTracker()
+        self.webhooks = WebhookRegistry()
+        # legacy analytics/webhook tools reach these through the handlers
+        self.handlers._analytics = self.analytics
+        self.handlers._webhooks = self.webhooks
 
     # ------------------------------------------------------- JSON-RPC
     def handle_message(self, msg: dict[str, Any]) -> dict[str, Any] | None:

@@ -108,6 +108,17 @@ LEGACY_ALIASES: dict[str, str] = {
     "verify": "fact_check",
     "node_status": "status",
     "stats": "status",
+    # direct legacy tools (dispatch to same-named handler methods)
+    "ping": "ping",
+    "credit_balance": "credit_balance",
+    "index_stats": "index_stats",
+    "network_stats": "network_stats",
+    "batch_search": "batch_search",
+    "search_history": "search_history",
+    "analytics": "analytics",
+    "register_webhook": "register_webhook",
+    "unregister_webhook": "unregister_webhook",
+    "remove_url": "remove_url",
 }
 
 

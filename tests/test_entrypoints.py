@@ -372,3 +372,47 @@ def test_web_search_reference_style_args(tmp_path, monkeypatch):
         assert all("text" in r for r in out["results"])
     finally:
         ctx.close()
+
+
+def test_mcp_legacy_utility_tools(tmp_path, monkeypatch):
+    """The reference dispatches ping/credit_balance/index_stats/
+    network_stats/batch_search/search_history/analytics/webhooks/
+    remove_url (mcp/server.py:205-457) — all must resolve and work."""
+    monkeypatch.setenv("INFOMESH_NODE_DATA_DIR", str(tmp_path))
+    from infomesh_amd.index.local_store import Document
+    from infomesh_amd.mcp.server import McpServer
+    ctx = AppContext.create(with_engine=False, with_worker=False)
+    try:
+        ctx.index_document(Document(url="https://lg.org/1", title="t",
+                                    text="legacy tools body " * 10))
+        srv = McpServer(ctx)
+
+        def call(name, args=None):
+            r = srv.handle_message({"jsonrpc": "2.0", "id": 1,
+                                    "method": "tools/call",
+                                    "params": {"name": name,
+                                               "arguments": args or {}}})
+            import json as _json
+            assert "result" in r, r
+            return _json.loads(r["result"]["content"][0]["text"])
+
+        assert call("ping")["pong"] is True
+        assert "documents" in call("index_stats") or call("index_stats")
+        assert isinstance(call("credit_balance"), dict)
+        assert "world_size" in call("network_stats")
+        out = call("batch_search", {"queries": ["legacy", "tools"],
+                                    "limit": 3})
+        assert len(out["batches"]) == 2
+        call("search", {"query": "legacy"})      # populates history
+        hist = call("search_history")
+        assert any("legacy" in h for h in hist["history"])
+        assert isinstance(call("analytics"), dict)
+        assert call("register_webhook",
+                    {"event": "crawl"})["registered"] == "crawl"
+        assert call("unregister_webhook",
+                    {"event": "crawl"})["unregistered"] == "crawl"
+        rm = call("remove_url", {"url": "https://lg.org/1"})
+        assert rm["removed"] and rm["deletion_recorded"]
+        assert ctx.store.get_document_by_url("https://lg.org/1") is None
+    finally:
+        ctx.close()
