@@ -121,6 +121,34 @@ def create_app(ctx: AppContext, api_key: str = "",
         return {"cache": ctx.cache.stats(),
                 "crawler": ctx.worker.stats if ctx.worker else None}
 
+    @app.get("/analytics/tools")
+    def analytics_tools():
+        """MCP tool usage breakdown (reference local_api.py:396-413)."""
+        stats = getattr(app.state, "tool_counts",
+                        {"web_search": 0, "crawl_url": 0, "fetch_page": 0})
+        total = sum(stats.values())
+        searches = stats.get("web_search", 0)
+        return {"tool_usage": {**stats, "total": total},
+                "search_fetch_rate": round(
+                    stats.get("fetch_page", 0) / searches * 100, 1)
+                if searches else 0.0}
+
+    @app.get("/index/compression")
+    def index_compression():
+        """Index compression statistics (reference local_api.py:417-435)."""
+        st = ctx.store.stats()
+        docs = int(st.get("documents", 0))
+        db_mb = round(float(st.get("db_bytes", 0)) / 1e6, 2)
+        return {"documents": docs, "db_size_mb": db_mb,
+                "avg_doc_kb": round(db_mb * 1024 / docs, 2) if docs else 0.0,
+                "compression_enabled": True,
+                "compression_level": 12}
+
+    @app.get("/openapi-spec")
+    def openapi_spec():
+        """Reference-named alias of FastAPI's /openapi.json."""
+        return app.openapi()
+
     @app.get("/metrics")
     def prom_metrics():
         return PlainTextResponse(metrics.render(),

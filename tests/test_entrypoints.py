@@ -416,3 +416,23 @@ def test_mcp_legacy_utility_tools(tmp_path, monkeypatch):
         assert ctx.store.get_document_by_url("https://lg.org/1") is None
     finally:
         ctx.close()
+
+
+def test_api_reference_routes(tmp_path, monkeypatch):
+    monkeypatch.setenv("INFOMESH_NODE_DATA_DIR", str(tmp_path))
+    from starlette.testclient import TestClient
+    from infomesh_amd.api.local_api import create_app
+    ctx = AppContext.create(with_engine=False, with_worker=False)
+    try:
+        app = create_app(ctx)
+        with TestClient(app) as client:
+            r = client.get("/analytics/tools")
+            assert r.status_code == 200 and "tool_usage" in r.json()
+            r = client.get("/index/compression")
+            body = r.json()
+            assert r.status_code == 200
+            assert {"documents", "db_size_mb", "avg_doc_kb"} <= set(body)
+            r = client.get("/openapi-spec")
+            assert r.status_code == 200 and "openapi" in r.json()
+    finally:
+        ctx.close()
