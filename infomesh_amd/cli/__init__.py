@@ -412,3 +412,57 @@ def main():
 
 if __name__ == "__main__":
     main()
+
+
+# ----------------------------------------------------------------- feeds
+
+@cli.group()
+def feeds():
+    """Manage RSS/Atom feed monitoring (persistent across restarts)."""
+
+
+@feeds.command(name="list")
+def feeds_list():
+    ctx = _ctx(with_worker=False, with_engine=False)
+    try:
+        rows = [{"url": f.url, "tier": f.tier, "failures": f.failures}
+                for f in ctx.feeds.feeds.values()]
+        click.echo(json.dumps(rows, indent=2))
+    finally:
+        ctx.close()
+
+
+@feeds.command(name="add")
+@click.argument("url")
+@click.option("--tier", default=3, help="poll tier 0 (1 min) .. 3 (60 min)")
+def feeds_add(url, tier):
+    ctx = _ctx(with_worker=False, with_engine=False)
+    try:
+        ctx.feeds.add(url, tier=tier)
+        click.echo(f"added {url} (tier {tier})")
+    finally:
+        ctx.close()
+
+
+@feeds.command(name="remove")
+@click.argument("url")
+def feeds_remove(url):
+    ctx = _ctx(with_worker=False, with_engine=False)
+    try:
+        ctx.feeds.remove(url)
+        click.echo(f"removed {url}")
+    finally:
+        ctx.close()
+
+
+@feeds.command(name="import")
+@click.argument("opml_file", type=click.Path(exists=True))
+def feeds_import(opml_file):
+    """Import RSS/Atom feeds from an OPML file."""
+    ctx = _ctx(with_worker=False, with_engine=False)
+    try:
+        n = ctx.feeds.import_opml(Path(opml_file).read_text())
+        ctx.feeds.save()
+        click.echo(f"imported {n} feeds")
+    finally:
+        ctx.close()

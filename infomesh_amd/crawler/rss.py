@@ -74,17 +74,54 @@ class MonitoredFeed:
 
 
 class FeedMonitor:
-    """Tracks feeds and reports which are due + which items are new."""
+    """Tracks feeds and reports which are due + which items are new.
 
-    def __init__(self):
+    Optionally persistent (path=...): discovered feeds survive restarts
+    — the reference's `infomesh feeds list/import` CLI expects that."""
+
+    def __init__(self, path=None):
         self.feeds: dict[str, MonitoredFeed] = {}
+        self._path = path
+        if path is not None:
+            self._load()
+
+    def _load(self) -> None:
+        import json
+        from pathlib import Path
+        p = Path(self._path)
+        if not p.exists():
+            return
+        try:
+            data = json.loads(p.read_text())
+        except (ValueError, OSError):
+            return
+        for d in data.get("feeds", []):
+            self.feeds[d["url"]] = MonitoredFeed(
+                url=d["url"], tier=int(d.get("tier", 3)),
+                last_poll=float(d.get("last_poll", 0.0)),
+                failures=int(d.get("failures", 0)))
+
+    def save(self) -> None:
+        if self._path is None:
+            return
+        import json
+        from pathlib import Path
+        p = Path(self._path)
+        tmp = p.with_suffix(".tmp")
+        tmp.write_text(json.dumps({"feeds": [
+            {"url": f.url, "tier": f.tier, "last_poll": f.last_poll,
+             "failures": f.failures} for f in self.feeds.values()]}))
+        tmp.replace(p)
 
     def add(self, url: str, tier: int = 3) -> None:
         tier = min(max(tier, 0), len(POLL_TIERS_MIN) - 1)
-        self.feeds.setdefault(url, MonitoredFeed(url=url, tier=tier))
+        if url not in self.feeds:
+            self.feeds[url] = MonitoredFeed(url=url, tier=tier)
+            self.save()
 
     def remove(self, url: str) -> None:
-        self.feeds.pop(url, None)
+        if self.feeds.pop(url, None) is not None:
+            self.save()
 
     def due(self, now: float | None = None) -> list[MonitoredFeed]:
         now = now or time.time()

@@ -93,7 +93,7 @@ class AppContext:
             takedowns=TakedownManager(store, keys, p("dmca.db")),
             deletions=DeletionManager(store, keys, p("gdpr.db")),
             cache=QueryCache(cfg.search.cache_entries, cfg.search.cache_ttl_s),
-            feeds=FeedMonitor(),
+            feeds=FeedMonitor(path=None if in_memory else p("feeds.json")),
             related=RelatedSearchTracker(),
         )
         if with_worker if with_worker is not None else role in ("full", "crawler"):

@@ -436,3 +436,29 @@ def test_api_reference_routes(tmp_path, monkeypatch):
             assert r.status_code == 200 and "openapi" in r.json()
     finally:
         ctx.close()
+
+
+def test_feeds_cli_persistence(tmp_path, monkeypatch):
+    monkeypatch.setenv("INFOMESH_NODE_DATA_DIR", str(tmp_path))
+    from click.testing import CliRunner
+    from infomesh_amd.cli import cli as root_cli
+    runner = CliRunner()
+    r = runner.invoke(root_cli, ["feeds", "add",
+                                 "https://ex.org/feed.xml", "--tier", "1"])
+    assert r.exit_code == 0, r.output
+    r = runner.invoke(root_cli, ["feeds", "list"])
+    assert "ex.org/feed.xml" in r.output
+    # persists to a new context (fresh process equivalent)
+    assert (tmp_path / "feeds.json").exists()
+    opml = tmp_path / "feeds.opml"
+    opml.write_text('<opml><body>'
+                    '<outline type="rss" xmlUrl="https://a.io/rss"/>'
+                    '<outline type="rss" xmlUrl="https://b.io/rss"/>'
+                    '</body></opml>')
+    r = runner.invoke(root_cli, ["feeds", "import", str(opml)])
+    assert "imported 2" in r.output
+    r = runner.invoke(root_cli, ["feeds", "remove",
+                                 "https://ex.org/feed.xml"])
+    assert r.exit_code == 0
+    r = runner.invoke(root_cli, ["feeds", "list"])
+    assert "ex.org" not in r.output and "a.io" in r.output
