@@ -197,3 +197,21 @@ class RelatedSearchTracker:
                 scored.append((overlap * cnt, past))
         scored.sort(reverse=True)
         return [p for _, p in scored[:limit]]
+
+
+def extract_keywords(text: str, top_n: int = 50, min_len: int = 2,
+                     language: str = "en") -> list[tuple[str, int]]:
+    """TF-based keyword extraction (reference: index/distributed.py:
+    30-159, where it fed the DHT keyword->pointer publish; here the GPU
+    shard indexes every term exhaustively, so this serves faceting,
+    related-search seeding and summary key-fact hints).
+
+    Returns (keyword, count) pairs, most frequent first, stop-worded
+    and length-filtered, capped at top_n."""
+    from collections import Counter
+    sw = STOP_WORDS.get(language, STOP_WORDS["en"])
+    words = re.findall(r"[\w'-]+", text.lower())
+    counts = Counter(w for w in words
+                     if len(w) >= min_len and w not in sw
+                     and not w.isdigit())
+    return counts.most_common(top_n)

@@ -241,3 +241,16 @@ def test_cross_validate_summaries():
         "gpu kernels run quickly on mi355x hardware",
         "bananas are yellow fruit entirely unrelated"])
     assert sims[0] > sims[2]
+
+
+def test_extract_keywords():
+    from infomesh_amd.search.nlp import extract_keywords
+    text = ("The quantum computer uses quantum gates. Quantum "
+            "error correction protects the computer from noise. "
+            "the and of to in is 42")
+    kws = extract_keywords(text, top_n=5)
+    words = [w for w, _ in kws]
+    assert words[0] == "quantum" and kws[0][1] == 3
+    assert "computer" in words
+    assert "the" not in words and "42" not in words
+    assert len(kws) <= 5
