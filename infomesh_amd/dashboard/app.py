@@ -72,6 +72,53 @@ class DashboardData:
         }
 
 
+SPARK_CHARS = " ▁▂▃▄▅▆▇█"
+
+
+def sparkline(values: list[float], width: int = 24) -> str:
+    """Unicode sparkline (reference dashboard widgets analogue)."""
+    if not values:
+        return ""
+    vals = values[-width:]
+    lo, hi = min(vals), max(vals)
+    span = (hi - lo) or 1.0
+    return "".join(SPARK_CHARS[1 + int((v - lo) / span * 7)]
+                   for v in vals)
+
+
+class History:
+    """In-process sample history for sparklines."""
+
+    def __init__(self, maxlen: int = 60):
+        self.maxlen = maxlen
+        self.series: dict[str, list[float]] = {}
+
+    def push(self, name: str, value: float) -> None:
+        xs = self.series.setdefault(name, [])
+        xs.append(float(value))
+        del xs[:-self.maxlen]
+
+    def get(self, name: str) -> list[float]:
+        return self.series.get(name, [])
+
+
+def _load_and_rss() -> tuple[float, float]:
+    try:
+        load = __import__("os").getloadavg()[0]
+    except OSError:
+        load = 0.0
+    rss = 0.0
+    try:
+        with open("/proc/self/status") as f:
+            for line in f:
+                if line.startswith("VmRSS"):
+                    rss = float(line.split()[1]) / 1e6  # GB
+                    break
+    except OSError:
+        pass
+    return load, rss
+
+
 def _kv_table(rows: list[tuple[str, str]]) -> Table:
     t = Table.grid(padding=(0, 2))
     t.add_column(style="dim")
@@ -81,9 +128,14 @@ def _kv_table(rows: list[tuple[str, str]]) -> Table:
     return t
 
 
-def render_dashboard(data: DashboardData):
+def render_dashboard(data: DashboardData, hist: History | None = None):
     s = data.snapshot()
     rt = s["runtime"]
+    if hist is not None:
+        hist.push("docs", s["docs"])
+        hist.push("recent", s["recent_docs"])
+        load, rss = _load_and_rss()
+        hist.push("load", load)
     state = rt.get("state", "stopped")
     color = "green" if state == "running" else "red"
     layout = Layout()
@@ -98,18 +150,26 @@ def render_dashboard(data: DashboardData):
         Layout(name="mid"),
         Layout(name="bottom"),
     )
+    doc_rows = [
+        ("documents", s["docs"]),
+        ("domains", s["domains"]),
+        ("indexed last hour", s["recent_docs"]),
+        ("link edges", s["link_edges"]),
+    ]
+    if hist is not None and len(hist.get("docs")) > 1:
+        doc_rows.append(("docs trend", sparkline(hist.get("docs"))))
+        doc_rows.append(("crawl rate", sparkline(hist.get("recent"))))
+    eng_rows = [
+        ("engine docs", rt.get("engine_docs", "—")),
+        ("seen URLs", s["seen_urls"]),
+        ("trust subjects", s["trust_subjects"]),
+    ]
+    if hist is not None and hist.get("load"):
+        eng_rows.append(("cpu load", f"{hist.get('load')[-1]:.2f}  "
+                         + sparkline(hist.get("load"))))
     layout["mid"].split_row(
-        Layout(Panel(_kv_table([
-            ("documents", s["docs"]),
-            ("domains", s["domains"]),
-            ("indexed last hour", s["recent_docs"]),
-            ("link edges", s["link_edges"]),
-        ]), title="index")),
-        Layout(Panel(_kv_table([
-            ("engine docs", rt.get("engine_docs", "—")),
-            ("seen URLs", s["seen_urls"]),
-            ("trust subjects", s["trust_subjects"]),
-        ]), title="crawl / engine")),
+        Layout(Panel(_kv_table(doc_rows), title="index")),
+        Layout(Panel(_kv_table(eng_rows), title="crawl / engine")),
     )
     layout["bottom"].split_row(
         Layout(Panel(_kv_table([
@@ -127,10 +187,11 @@ def render_dashboard(data: DashboardData):
 def run_dashboard(refresh_s: float = 2.0,
                   iterations: int | None = None) -> None:
     data = DashboardData()
-    with Live(render_dashboard(data), refresh_per_second=4,
+    hist = History()
+    with Live(render_dashboard(data, hist), refresh_per_second=4,
               screen=iterations is None) as live:
         n = 0
         while iterations is None or n < iterations:
             time.sleep(refresh_s if iterations is None else 0.01)
-            live.update(render_dashboard(data))
+            live.update(render_dashboard(data, hist))
             n += 1

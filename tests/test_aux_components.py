@@ -127,3 +127,22 @@ def test_version_check(tmp_path):
     # cached second call
     out2 = check_for_update(state_path=tmp_path / "vc.json")
     assert out2 == out
+
+
+def test_dashboard_sparkline_and_history(tmp_path, monkeypatch):
+    monkeypatch.setenv("INFOMESH_NODE_DATA_DIR", str(tmp_path))
+    from infomesh_amd.dashboard.app import (
+        DashboardData, History, render_dashboard, sparkline)
+    assert sparkline([]) == ""
+    sp = sparkline([0, 1, 2, 3, 4, 5, 6, 7])
+    assert len(sp) == 8 and sp[0] != sp[-1]
+    assert sparkline([5.0] * 4) == sparkline([5.0] * 4)  # flat ok
+    h = History(maxlen=5)
+    for i in range(10):
+        h.push("x", i)
+    assert h.get("x") == [5.0, 6.0, 7.0, 8.0, 9.0]
+    data = DashboardData()
+    layout = render_dashboard(data, h)
+    assert layout is not None
+    # two renders with history grow sparkline series without error
+    render_dashboard(data, h)
