@@ -58,3 +58,32 @@ class PluginManager:
 
 
 GLOBAL_PLUGINS = PluginManager()
+
+
+def load_plugin_module(module_path: str,
+                       manager: PluginManager | None = None) -> int:
+    """Import a plugin module by dotted path (reference dx.py:84-115).
+
+    The module may either call ``register(point, fn)`` on the GLOBAL
+    manager at import time, or expose a ``setup(plugins)`` function
+    that receives the manager. Returns hooks added."""
+    import importlib
+    mgr = manager or GLOBAL_PLUGINS
+    before = mgr.count()
+    mod = importlib.import_module(module_path)
+    setup = getattr(mod, "setup", None)
+    if callable(setup):
+        setup(mgr)
+    return mgr.count() - before
+
+
+def load_plugins_from_config(paths: list[str],
+                             manager: PluginManager | None = None) -> int:
+    """Best-effort bulk load; a broken plugin never takes the node down."""
+    total = 0
+    for p in paths:
+        try:
+            total += load_plugin_module(p, manager)
+        except Exception as e:  # noqa: BLE001
+            log.warning("plugin module %s failed to load: %s", p, e)
+    return total

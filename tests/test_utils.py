@@ -299,3 +299,26 @@ def test_security_ext_webhook_hmac():
     assert verify_webhook(sec, body, h)
     assert not verify_webhook(sec, b"tampered", h)
     assert not verify_webhook(sec, body, "")
+
+
+def test_plugin_module_loading(tmp_path, monkeypatch):
+    import sys
+    from infomesh_amd.utils.plugins import (
+        PluginManager, load_plugin_module, load_plugins_from_config)
+    plug_dir = tmp_path / "plugs"
+    plug_dir.mkdir()
+    (plug_dir / "myplug.py").write_text(
+        "def setup(plugins):\n"
+        "    plugins.register('pre_search', lambda q, **k: q.upper())\n")
+    (plug_dir / "broken.py").write_text("raise RuntimeError('boom')\n")
+    monkeypatch.syspath_prepend(str(plug_dir))
+    mgr = PluginManager()
+    n = load_plugin_module("myplug", mgr)
+    assert n == 1
+    assert mgr.run("pre_search", "hello") == "HELLO"
+    # broken plugin isolated, good one still counted
+    mgr2 = PluginManager()
+    total = load_plugins_from_config(["myplug", "broken", "ghost"], mgr2)
+    assert total == 1
+    sys.modules.pop("myplug", None)
+    sys.modules.pop("broken", None)
