@@ -37,6 +37,7 @@ class NodeConfig:
     role: str = "full"                 # full | crawler | search
     data_dir: str = ""                 # empty → ~/.infomesh
     name: str = ""
+    plugins: str = ""                  # comma-separated dotted paths
 
 
 @dataclass(frozen=True)

@@ -78,6 +78,11 @@ class AppContext:
         def p(name: str):
             return ":memory:" if in_memory else data / name
 
+        if cfg.node.plugins:
+            from .utils.plugins import load_plugins_from_config
+            load_plugins_from_config(
+                [x.strip() for x in cfg.node.plugins.split(",")
+                 if x.strip()])
         store = LocalStore(p("index.db"), tokenizer=cfg.index.fts_tokenizer)
         keys = KeyPair.generate() if in_memory else ensure_keys(data)
         ledger = CreditLedger(p("ledger.db"), kp=keys)
@@ -118,6 +123,8 @@ class AppContext:
                        credit: bool = True) -> int | None:
         """THE single crawl->index source of truth
         (reference: services.py:68-110)."""
+        from .utils.plugins import GLOBAL_PLUGINS
+        doc = GLOBAL_PLUGINS.run("pre_index", doc)
         if self.deletions.is_forgotten(doc.url):
             raise InfoMeshError("SEC001", "url under GDPR deletion record")
         if self.takedowns.is_blocked(doc.url):
@@ -138,6 +145,7 @@ class AppContext:
                 self.ledger.record_action(Action.CRAWL, 1.0)
             self.farming.record("crawl")
         self.cache.invalidate()
+        GLOBAL_PLUGINS.run("post_index", doc, rowid=rowid)
         return rowid
 
     async def crawl_and_index(self, url: str, depth: int = 0,
@@ -145,6 +153,8 @@ class AppContext:
         """Crawl one URL and index it (reference: services.py:354-423)."""
         if self.worker is None:
             raise InfoMeshError("RT001", "no crawl worker in this role")
+        from .utils.plugins import GLOBAL_PLUGINS
+        url = GLOBAL_PLUGINS.run("pre_crawl", url)
         doc_meta = self.store.get_document_by_url(url)
         res: CrawlResult = await self.worker.crawl_url(
             url, depth=depth, force=force,
@@ -174,7 +184,7 @@ class AppContext:
                     "links": page.links[:50],
                     "links_scheduled": res.links_scheduled,
                     "indexed": rowid is not None})
-        return out
+        return GLOBAL_PLUGINS.run("post_crawl", out)
 
     # ------------------------------------------------------------ search
     def search(self, query: str, limit: int | None = None,
@@ -182,6 +192,8 @@ class AppContext:
                deduct: bool = True) -> SearchResponse:
         """Unified search entry (reference mcp/handlers.py:382 flow):
         cache -> credits -> local/hybrid/distributed -> cache."""
+        from .utils.plugins import GLOBAL_PLUGINS
+        query = GLOBAL_PLUGINS.run("pre_search", query, mode=mode)
         limit = limit or self.config.search.max_results
         key = QueryCache.make_key(query, limit=limit, mode=mode)
         if use_cache:
@@ -207,6 +219,7 @@ class AppContext:
                 resp.mode = "distributed"
         else:
             raise InfoMeshError("SRCH001", f"unknown mode {mode!r}")
+        resp = GLOBAL_PLUGINS.run("post_search", resp, query=query)
         if use_cache:
             self.cache.put(key, resp)
         self.ledger.record_action(Action.QUERY_SERVED, 1.0)

@@ -227,3 +227,49 @@ def test_engine_flush_is_incremental_on_encoding():
     hits_old = eng.search("alpha", limit=5)
     hits_new = eng.search("gamma", limit=5)
     assert hits_old and hits_new
+
+
+def test_plugin_hooks_fire_in_real_paths(tmp_path, monkeypatch):
+    """pre/post index + search hooks run inside AppContext flows."""
+    from infomesh_amd.index.local_store import Document
+    from infomesh_amd.services import AppContext
+    from infomesh_amd.utils.plugins import GLOBAL_PLUGINS
+    monkeypatch.setenv("INFOMESH_NODE_DATA_DIR", str(tmp_path))
+    seen = {"pre_index": 0, "post_index": 0,
+            "pre_search": 0, "post_search": 0}
+
+    def pre_index(doc, **k):
+        seen["pre_index"] += 1
+        doc.title = doc.title + " [plug]"
+        return doc
+
+    def post_index(doc, **k):
+        seen["post_index"] += 1
+
+    def pre_search(q, **k):
+        seen["pre_search"] += 1
+        return q
+
+    def post_search(resp, **k):
+        seen["post_search"] += 1
+        return resp
+
+    GLOBAL_PLUGINS.register("pre_index", pre_index)
+    GLOBAL_PLUGINS.register("post_index", post_index)
+    GLOBAL_PLUGINS.register("pre_search", pre_search)
+    GLOBAL_PLUGINS.register("post_search", post_search)
+    try:
+        ctx = AppContext.create(with_engine=False, with_worker=False)
+        try:
+            ctx.index_document(Document(url="https://pl.org/1",
+                                        title="hooked",
+                                        text="plugin hook body " * 10))
+            doc = ctx.store.get_document_by_url("https://pl.org/1")
+            assert doc.title.endswith("[plug]")   # pre_index transformed
+            ctx.search("plugin")
+            assert seen["pre_index"] == seen["post_index"] == 1
+            assert seen["pre_search"] >= 1 and seen["post_search"] >= 1
+        finally:
+            ctx.close()
+    finally:
+        GLOBAL_PLUGINS._hooks.clear()
