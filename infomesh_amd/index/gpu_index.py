@@ -30,8 +30,13 @@ _WORD_RE = re.compile(r"[a-z0-9_]+")
 
 
 def bm25_term_ids(text: str, vocab: int = BM25_VOCAB) -> np.ndarray:
-    """Tokenize + hash into the BM25 term space."""
-    toks = _WORD_RE.findall(text.lower())
+    """Tokenize + hash into the BM25 term space. A registered
+    'tokenizer' plugin (utils/plugins.py, reference dx.py:153-188)
+    replaces the default word tokenizer for BOTH indexing and queries
+    — the hash space keeps them consistent automatically."""
+    from ..utils.plugins import GLOBAL_PLUGINS
+    tok = GLOBAL_PLUGINS.get_single("tokenizer")
+    toks = tok(text) if tok is not None else _WORD_RE.findall(text.lower())
     if not toks:
         return np.zeros(0, dtype=np.int64)
     return np.fromiter((hash64(t) % vocab for t in toks), dtype=np.int64,

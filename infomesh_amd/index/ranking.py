@@ -40,6 +40,10 @@ class ScoreBreakdown:
 
     @property
     def total(self) -> float:
+        from ..utils.plugins import GLOBAL_PLUGINS
+        custom = GLOBAL_PLUGINS.get_single("scorer")
+        if custom is not None:
+            return float(custom(self))
         return (W_BM25 * self.bm25 + W_FRESHNESS * self.freshness
                 + W_TRUST * self.trust + W_AUTHORITY * self.authority
                 + W_TITLE * self.title_match + W_URL_PATH * self.url_path)

@@ -322,3 +322,23 @@ def test_plugin_module_loading(tmp_path, monkeypatch):
     assert total == 1
     sys.modules.pop("myplug", None)
     sys.modules.pop("broken", None)
+
+
+def test_tokenizer_and_scorer_plugin_slots():
+    import numpy as np
+    from infomesh_amd.index.gpu_index import bm25_term_ids
+    from infomesh_amd.index.ranking import ScoreBreakdown
+    from infomesh_amd.utils.plugins import GLOBAL_PLUGINS
+    try:
+        default = bm25_term_ids("hello world")
+        GLOBAL_PLUGINS.register("tokenizer", lambda t: ["onlytoken"])
+        custom = bm25_term_ids("hello world")
+        assert len(custom) == 1 and not np.array_equal(default, custom)
+        GLOBAL_PLUGINS.register("scorer", lambda sb: 42.0)
+        sb = ScoreBreakdown(bm25=1.0, freshness=1.0, trust=1.0,
+                            authority=1.0, title_match=1.0, url_path=1.0)
+        assert sb.total == 42.0
+    finally:
+        GLOBAL_PLUGINS._hooks.clear()
+    assert ScoreBreakdown(bm25=1.0, freshness=0, trust=0, authority=0,
+                          title_match=0, url_path=0).total > 0
