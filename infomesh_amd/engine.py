@@ -45,13 +45,9 @@ class HybridEngine:
         self._pending_tokens: list[np.ndarray] = []
         self._pending_texts: list[str] = []
         self._pending_ids: list[int] = []
-        self._built_tokens: list[np.ndarray] = []
-        self._built_ids: list[int] = []
-        # embeddings of already-built docs: encoding is the expensive
-        # part of a flush, so it is INCREMENTAL — only new docs run
-        # through the encoder; the CSR rebuild itself is a fast numpy
-        # sort (segment-merge analogue).
-        self._built_emb: torch.Tensor | None = None
+        # (embeddings and postings of already-built docs live in the
+        # shard itself: flush merges rather than re-accumulating, so
+        # engine memory does not grow with corpus size)
 
     # ------------------------------------------------------------ ingest
     def add_document(self, doc: Document) -> None:
@@ -94,29 +90,27 @@ class HybridEngine:
                 chunks.append(enc[:embed_batch - pad] if pad else enc)
             new_emb = torch.cat(chunks, 0)
 
-        self._built_tokens.extend(self._pending_tokens)
-        self._built_ids.extend(self._pending_ids)
+        tokens = self._pending_tokens
+        ids = self._pending_ids
         self._pending_tokens, self._pending_texts, self._pending_ids = \
             [], [], []
-        emb = None
-        if new_emb is not None:
-            self._built_emb = (new_emb if self._built_emb is None
-                               else torch.cat([self._built_emb, new_emb],
-                                              0))
-            emb = self._built_emb
-
-        lens = np.array([max(len(t), 1) for t in self._built_tokens],
-                        dtype=np.int64)
-        flat_terms = (np.concatenate(self._built_tokens)
-                      if any(len(t) for t in self._built_tokens)
-                      else np.zeros(0, np.int64))
-        flat_docs = np.repeat(np.arange(len(self._built_tokens),
-                                        dtype=np.int64),
-                              [len(t) for t in self._built_tokens])
-        shard = GpuShard(self.device) if self.gpu else CpuShard()
-        shard.build_from_arrays(
-            flat_terms, flat_docs, lens,
-            np.asarray(self._built_ids, dtype=np.int64), emb)
+        if self.shard.n_docs == 0:
+            lens = np.array([max(len(t), 1) for t in tokens],
+                            dtype=np.int64)
+            flat_terms = (np.concatenate(tokens)
+                          if any(len(t) for t in tokens)
+                          else np.zeros(0, np.int64))
+            flat_docs = np.repeat(np.arange(len(tokens), dtype=np.int64),
+                                  [len(t) for t in tokens])
+            shard = GpuShard(self.device) if self.gpu else CpuShard()
+            shard.build_from_arrays(
+                flat_terms, flat_docs, lens,
+                np.asarray(ids, dtype=np.int64), new_emb)
+        else:
+            # epoch flip via segment merge: only the NEW docs are
+            # tokenized/encoded; readers keep the old shard until the
+            # merged one is fully installed
+            shard = self.shard.merged_with(tokens, ids, new_emb)
         self.shard = shard
         self.plane.shard = shard
         log.info("engine flush: %d new docs, %d total, %.1f MB HBM",

@@ -87,27 +87,101 @@ class GpuShard:
 
     def build(self) -> None:
         """(Re)build the CSR postings + embedding matrix from pending
-        docs plus any existing index (the FTS5-optimize analogue)."""
-        if not self._pend_tokens and self.n_docs:
+        docs plus any existing index — the FTS5-optimize / segment-merge
+        analogue. With existing docs this MERGES: the old postings are
+        pulled back once, concatenated with the new aggregated postings
+        (new docs get fresh local ids, so (term, doc) pairs never
+        collide), and BM25 stats (avgdl, norms) are recomputed over the
+        whole corpus."""
+        if not self._pend_tokens:
             return
         token_lists = self._pend_tokens
         gids = list(self._pend_gids)
         embs = self._pend_emb
-        if self.n_docs:
-            raise NotImplementedError(
-                "incremental rebuild-with-existing not yet supported; "
-                "use build_from_arrays for bulk loads")
         self._pend_tokens, self._pend_gids, self._pend_emb = [], [], []
-        if not token_lists:
+        lens2 = np.array([max(len(t), 1) for t in token_lists],
+                         dtype=np.int64)
+        n_old = self.n_docs
+        n_new = len(token_lists)
+        flat_terms = (np.concatenate(token_lists)
+                      if any(len(t) for t in token_lists)
+                      else np.zeros(0, np.int64))
+        flat_docs = np.repeat(np.arange(n_new, dtype=np.int64),
+                              [len(t) for t in token_lists])
+        new_emb = torch.cat(embs, 0) if embs else None
+        if n_old == 0:
+            self.build_from_arrays(flat_terms, flat_docs, lens2,
+                                   np.asarray(gids, dtype=np.int64),
+                                   new_emb)
             return
-        lens = np.array([len(t) for t in token_lists], dtype=np.int64)
-        flat_terms = np.concatenate(token_lists) if token_lists else \
-            np.zeros(0, np.int64)
+        self._merge_install(self, flat_terms, flat_docs, lens2, gids,
+                            new_emb)
+
+    def merged_with(self, token_lists: list[np.ndarray],
+                    gids: list[int],
+                    new_emb: torch.Tensor | None) -> "GpuShard":
+        """Return a NEW shard = this shard + the given docs (epoch-flip
+        variant of the in-place incremental build(): readers of the old
+        shard are never exposed to a partially-built index)."""
+        lens2 = np.array([max(len(t), 1) for t in token_lists],
+                         dtype=np.int64)
+        flat_terms = (np.concatenate(
+            [t.astype(np.int64) for t in token_lists])
+            if any(len(t) for t in token_lists)
+            else np.zeros(0, np.int64))
         flat_docs = np.repeat(np.arange(len(token_lists), dtype=np.int64),
-                              lens)
-        emb = torch.cat(embs, 0) if embs else None
-        self.build_from_arrays(flat_terms, flat_docs, lens,
-                               np.asarray(gids, dtype=np.int64), emb)
+                              [len(t) for t in token_lists])
+        out = type(self)() if type(self).__init__ is not GpuShard.__init__ \
+            else GpuShard(str(self.device), vocab=self.vocab)
+        out.vocab = self.vocab
+        self._merge_install(out, flat_terms, flat_docs, lens2,
+                            list(gids), new_emb)
+        return out
+
+    def _merge_install(self, target: "GpuShard", flat_terms, flat_docs,
+                       lens2, gids, new_emb) -> None:
+        """Merge THIS shard's postings with new aggregated postings and
+        install into `target` (which may be self)."""
+        n_old = self.n_docs
+        n_new = len(lens2)
+        t2, d2, tf2 = self._aggregate(flat_terms, flat_docs, n_new)
+        d2 = (d2.astype(np.int64) + n_old).astype(np.int32)
+        # pull the OLD postings back (CSR -> flat aggregated form)
+        t1 = np.repeat(np.arange(self.vocab, dtype=np.int64), self.df)
+        d1 = self.doc_ids.cpu().numpy()
+        tf1 = self.tfs.cpu().numpy().view(np.uint16)
+        terms = np.concatenate([t1, t2])
+        order = np.argsort(terms, kind="stable")
+        terms = terms[order]
+        docs = np.concatenate([d1, d2])[order]
+        tfs = np.concatenate([tf1, tf2])[order]
+        lens = np.concatenate([self._doc_lens, lens2])
+        old_gids = self.global_ids.cpu().numpy()
+        all_gids = np.concatenate(
+            [old_gids, np.asarray(gids, dtype=np.int64)])
+        emb = None
+        if self.embeddings is not None:
+            assert new_emb is not None and new_emb.shape[0] == n_new, \
+                "dense shard requires embeddings for every pending doc"
+            e2 = new_emb
+            if e2.dtype != torch.bfloat16:
+                e2 = torch.nn.functional.normalize(e2.float(),
+                                                   dim=-1).bfloat16()
+            emb = torch.cat([self.embeddings, e2.to(self.device)], 0)
+        target.n_docs = n_old + n_new
+        target._install_postings(terms, docs, tfs, lens, all_gids, emb)
+
+    @staticmethod
+    def _aggregate(flat_terms: np.ndarray, flat_docs: np.ndarray,
+                   n: int) -> tuple[np.ndarray, np.ndarray, np.ndarray]:
+        """(term, doc) pairs -> unique pairs + tf counts, term-sorted."""
+        key = flat_terms * np.int64(max(n, 1)) + flat_docs
+        key.sort(kind="stable")
+        uniq, counts = np.unique(key, return_counts=True)
+        terms_u = (uniq // max(n, 1)).astype(np.int64)
+        docs_u = (uniq % max(n, 1)).astype(np.int32)
+        tf_u = np.minimum(counts, 65535).astype(np.uint16)
+        return terms_u, docs_u, tf_u
 
     def _upload(self, arr: np.ndarray, stream) -> torch.Tensor:
         """Host -> HBM via a pinned staging buffer on the ingest side
@@ -130,12 +204,18 @@ class GpuShard:
         n = len(doc_lens)
         self.n_docs = n
         # Aggregate tf per (term, doc) via a combined key sort.
-        key = flat_terms * np.int64(n) + flat_docs
-        key.sort(kind="stable")
-        uniq, counts = np.unique(key, return_counts=True)
-        terms_u = (uniq // n).astype(np.int64)
-        docs_u = (uniq % n).astype(np.int32)
-        tf_u = np.minimum(counts, 65535).astype(np.uint16)
+        terms_u, docs_u, tf_u = self._aggregate(flat_terms, flat_docs, n)
+        self._install_postings(terms_u, docs_u, tf_u, doc_lens,
+                               global_ids, embeddings)
+
+    def _install_postings(self, terms_u: np.ndarray, docs_u: np.ndarray,
+                          tf_u: np.ndarray, doc_lens: np.ndarray,
+                          global_ids: np.ndarray,
+                          embeddings: torch.Tensor | None) -> None:
+        """Install term-sorted aggregated postings as the CSR index and
+        recompute BM25 stats; uploads via pinned staging."""
+        n = self.n_docs
+        self._doc_lens = np.asarray(doc_lens, dtype=np.int64)
         # CSR offsets per term (terms_u already sorted).
         df = np.bincount(terms_u, minlength=self.vocab).astype(np.int64)
         offsets = np.zeros(self.vocab + 1, dtype=np.int64)

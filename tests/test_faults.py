@@ -132,7 +132,7 @@ def test_engine_flush_failure_keeps_old_epoch():
     eng.encoder.calls = -10
     eng.flush(embed_batch=4)
     assert eng.shard.n_docs == 2
-    assert eng._built_emb.shape[0] == 2
+    assert eng.shard.embeddings.shape[0] == 2
     assert eng.search("alpha", limit=3) and eng.search("beta", limit=3)
 
 

@@ -253,3 +253,48 @@ def test_search_distributed_merges_shards(seeded_store):
     urls = [h.url for h in resp.results]
     assert urls == ["https://a.com/1", "https://b.com/2"]
     assert resp.results[0].score == 0.9
+
+
+def test_shard_incremental_build_matches_bulk():
+    """build() with existing docs (segment merge) must produce the same
+    index as one bulk build of all docs: identical search results AND
+    identical BM25 stats (avgdl/norms recomputed over the union)."""
+    import numpy as np
+    import torch
+    from infomesh_amd.index.gpu_index import CpuShard
+
+    rng = np.random.default_rng(7)
+    docs = [rng.integers(0, 300, size=rng.integers(4, 25)).astype(np.int64)
+            for _ in range(120)]
+    g = torch.Generator().manual_seed(7)
+    emb = torch.nn.functional.normalize(
+        torch.randn(120, 16, generator=g), dim=-1).bfloat16()
+
+    # incremental: 3 batches through add_document + build
+    inc = CpuShard()
+    start = 0
+    for batch in (40, 50, 30):
+        for i in range(start, start + batch):
+            inc.add_document(1000 + i, docs[i], emb[i])
+        inc.build()
+        start += batch
+    assert inc.n_docs == 120
+
+    # bulk oracle
+    bulk = CpuShard()
+    for i in range(120):
+        bulk.add_document(1000 + i, docs[i], emb[i])
+    bulk.build()
+
+    assert abs(inc.avgdl - bulk.avgdl) < 1e-9
+    assert np.array_equal(inc.df, bulk.df)
+    assert torch.equal(inc.doc_norm, bulk.doc_norm)
+    queries = [np.array([5, 17, 40]), np.array([100, 2]),
+               np.array([250])]
+    qe = torch.nn.functional.normalize(
+        torch.randn(3, 16, generator=g), dim=-1)
+    hi = inc.search(queries, qe, k=10)
+    hb = bulk.search(queries, qe, k=10)
+    assert torch.equal(hi.bm25_ids, hb.bm25_ids)
+    assert torch.allclose(hi.bm25_scores, hb.bm25_scores, atol=1e-5)
+    assert torch.equal(hi.dense_ids, hb.dense_ids)

@@ -223,7 +223,7 @@ def test_engine_flush_is_incremental_on_encoding():
     # second flush encoded one padded batch (4 rows), NOT all 5 docs
     assert eng.encoder.rows - first == 4
     assert eng.shard.n_docs == 5
-    assert eng._built_emb.shape[0] == 5
+    assert eng.shard.embeddings.shape[0] == 5
     hits_old = eng.search("alpha", limit=5)
     hits_new = eng.search("gamma", limit=5)
     assert hits_old and hits_new
