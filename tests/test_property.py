@@ -76,3 +76,48 @@ def test_bm25_chunks_matches_naive(queries):
     got = sorted(zip(cq.tolist(), ct.tolist(), co.tolist()))
     assert got == sorted(exp)
     assert np.allclose(ci, idf[ct])
+
+
+@given(st.integers(1, 5), st.integers(2, 12), st.integers(1, 8),
+       st.integers(0, 2 ** 31))
+@settings(max_examples=30, deadline=None)
+def test_rrf_fuse_matches_naive(batch, m_per_source, n_out, seed):
+    """Graph-captured vectorized RRF == naive dict-based RRF for random
+    candidate lists (with -1 padding and duplicate ids across
+    sources)."""
+    import torch
+    from infomesh_amd.parallel.query_plane import RRF_K, rrf_fuse
+    g = torch.Generator().manual_seed(seed)
+    ids1 = torch.randint(-1, 30, (batch, m_per_source), generator=g)
+    ids2 = torch.randint(-1, 30, (batch, m_per_source), generator=g)
+    # scores: distinct to avoid tie-order ambiguity between impls
+    s1 = torch.randperm(1000, generator=g)[:batch * m_per_source] \
+        .float().view(batch, m_per_source)
+    s2 = torch.randperm(1000, generator=g)[batch * m_per_source:
+                                           2 * batch * m_per_source] \
+        .float().view(batch, m_per_source)
+    out_ids, out_scores = rrf_fuse([ids1.long(), ids2.long()], [s1, s2],
+                                   [1.0, 1.0], n_out)
+    for b in range(batch):
+        contrib: dict[int, float] = {}
+        for ids, sc in ((ids1[b], s1[b]), (ids2[b], s2[b])):
+            order = torch.argsort(sc, descending=True)
+            rank = 0
+            for j in order.tolist():
+                rank += 1
+                i = int(ids[j])
+                if i >= 0:
+                    contrib[i] = contrib.get(i, 0.0) + 1.0 / (RRF_K + rank)
+        # Equal RRF contributions (cross-source rank ties) make id
+        # order within a tie group arbitrary: compare the score
+        # sequence exactly and ids as a multiset per tie group.
+        want_all = sorted(contrib.items(), key=lambda kv: -kv[1])
+        got_ids = [int(x) for x in out_ids[b] if int(x) >= 0]
+        got_sc = [float(x) for x in out_scores[b][:len(got_ids)]]
+        want_sc = [v for _, v in want_all[:len(got_ids)]]
+        assert len(got_ids) == min(n_out, len(want_all))
+        for gv, wv in zip(got_sc, want_sc):
+            assert abs(gv - wv) < 1e-5
+        # each returned id's true contribution equals its reported score
+        for gid, gv in zip(got_ids, got_sc):
+            assert abs(contrib[gid] - gv) < 1e-5
