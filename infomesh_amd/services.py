@@ -250,6 +250,10 @@ class AppContext:
         }
 
     def close(self) -> None:
+        try:
+            self.feeds.save()
+        except Exception:
+            pass
         for c in (self.store, self.link_graph, self.dedup, self.ledger,
                   self.trust, self.takedowns, self.deletions):
             try:
