@@ -2,6 +2,10 @@
 // residual add, RoPE apply (host-precomputed cos/sin tables — guide App. B),
 // embedding gather (+scaled add), CLS/mean pooling with L2 norm, greedy
 // argmax. All bf16 I/O vectorized 8-wide.
+// Replaces: torch elementwise glue in the reference's model stacks
+// (sentence-transformers internals behind infomesh/index/vector_store.py
+// :104-118 and the external-LLM serving the reference calls over HTTP,
+// infomesh/summarizer/engine.py:111-318).
 #include "common.h"
 
 namespace {

@@ -18,6 +18,10 @@
 //     vmcnt(4) + barrier   (own t+1 halves landed; cross-wave via bar)
 // Out-of-range prefetch steps clamp to the last K-tile (harmless
 // redundant loads) so the vmcnt counts stay static.
+// Replaces (with gemm.hip): the reference's large matmuls inside
+// sentence-transformers/cross-encoder models (infomesh/index/
+// vector_store.py:120-157) at the shapes where the 128^2 tile is
+// pipeline-bound.
 #include "common.h"
 
 #define G8_BM 256

@@ -7,6 +7,9 @@
 // wave-reduce per m, lane 0 stores. Weights stream once at HBM rate;
 // the 128x128 MFMA tile kernel wastes 127/128 of its A tile here and
 // runs ~10x slower at M=1 (measured in profiles/r01_decode.md).
+// Replaces: the decode-time projection matmuls of the external LLM
+// the reference calls over HTTP (infomesh/summarizer/engine.py:111-318)
+// -- in-process Phi-3-shaped decode runs these at HBM rate.
 #include "common.h"
 
 namespace {

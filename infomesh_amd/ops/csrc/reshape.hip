@@ -3,6 +3,9 @@
 // in the encoder/reranker/decoder forward paths (4-6 HBM round trips
 // per layer become 2 fused passes), plus a strided SwiGLU that reads
 // gate/up halves in place.
+// Replaces: torch permute/contiguous head-reshape glue inside the
+// reference's model stacks (sentence-transformers internals behind
+// infomesh/index/vector_store.py:104-157).
 #include "common.h"
 
 namespace {

@@ -1,6 +1,10 @@
 // Masked row softmax for attention scores: f32 in -> bf16 out (PV input).
 // Supports causal masking and per-group valid-length masking.
 // One block per row; online not needed (row fits in a grid-stride pass).
+// Replaces: torch softmax in the reference's model stacks (attention
+// softmax inside sentence-transformers, infomesh/index/vector_store.py
+// :104-118); the fused attention path subsumes it -- this kernel serves
+// the decomposed fallback and tests.
 #include "common.h"
 
 namespace {
