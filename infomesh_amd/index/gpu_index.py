@@ -33,10 +33,23 @@ def bm25_term_ids(text: str, vocab: int = BM25_VOCAB) -> np.ndarray:
     """Tokenize + hash into the BM25 term space. A registered
     'tokenizer' plugin (utils/plugins.py, reference dx.py:153-188)
     replaces the default word tokenizer for BOTH indexing and queries
-    — the hash space keeps them consistent automatically."""
+    — the hash space keeps them consistent automatically.
+
+    CJK runs are expanded into character bigrams (the GPU analogue of
+    the reference's FTS5 trigram/CJK handling, search/cjk.py): the
+    default latin word regex alone would drop CJK text entirely from
+    the GPU BM25 plane."""
     from ..utils.plugins import GLOBAL_PLUGINS
     tok = GLOBAL_PLUGINS.get_single("tokenizer")
-    toks = tok(text) if tok is not None else _WORD_RE.findall(text.lower())
+    if tok is not None:
+        toks = tok(text)
+    else:
+        toks = _WORD_RE.findall(text.lower())
+        # fast ordinal check before paying the regex: any char >= U+2E80
+        if any(ord(c) >= 0x2E80 for c in text):
+            from ..search.cjk import _CJK_RUN_RE, ngram_expand
+            for m in _CJK_RUN_RE.finditer(text):
+                toks.extend(ngram_expand(m.group(0), 2))
     if not toks:
         return np.zeros(0, dtype=np.int64)
     return np.fromiter((hash64(t) % vocab for t in toks), dtype=np.int64,

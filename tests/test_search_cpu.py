@@ -298,3 +298,21 @@ def test_shard_incremental_build_matches_bulk():
     assert torch.equal(hi.bm25_ids, hb.bm25_ids)
     assert torch.allclose(hi.bm25_scores, hb.bm25_scores, atol=1e-5)
     assert torch.equal(hi.dense_ids, hb.dense_ids)
+
+
+def test_bm25_term_ids_cjk_bigrams():
+    """CJK text must produce GPU BM25 terms (bigrams), and a CJK query
+    must retrieve a CJK doc through the shard."""
+    import numpy as np
+    from infomesh_amd.index.gpu_index import CpuShard, bm25_term_ids
+
+    t = bm25_term_ids("量子计算机")
+    assert len(t) == 4  # 4 bigrams from a 5-char run
+    assert len(bm25_term_ids("mixed 量子 text")) == 2 + 1
+    shard = CpuShard()
+    shard.add_document(1, bm25_term_ids("量子计算机的研究进展"), None)
+    shard.add_document(2, bm25_term_ids("classic english doc"), None)
+    shard.build()
+    hits = shard.search([bm25_term_ids("量子计算")], None, k=2)
+    top = int(hits.bm25_ids[0, 0])
+    assert top == 1
