@@ -26,7 +26,9 @@ BM25_K1 = 1.2
 BM25_B = 0.75
 BM25_VOCAB = 1 << 17
 
-_WORD_RE = re.compile(r"[a-z0-9_]+")
+# \w+ covers all unicode letters/digits (FTS5 unicode61 analogue);
+# diacritics are NFKD-folded below like unicode61 remove_diacritics
+_WORD_RE = re.compile(r"\w+")
 
 
 def bm25_term_ids(text: str, vocab: int = BM25_VOCAB) -> np.ndarray:
@@ -44,6 +46,13 @@ def bm25_term_ids(text: str, vocab: int = BM25_VOCAB) -> np.ndarray:
     if tok is not None:
         toks = tok(text)
     else:
+        if any(ord(c) > 127 for c in text):
+            # diacritic folding (cafe == café), matching the FTS5
+            # unicode61 tokenizer's remove_diacritics so both score
+            # planes see the same terms
+            import unicodedata
+            text = "".join(c for c in unicodedata.normalize("NFKD", text)
+                           if not unicodedata.combining(c))
         toks = _WORD_RE.findall(text.lower())
         # fast ordinal check before paying the regex: any char >= U+2E80
         if any(ord(c) >= 0x2E80 for c in text):

@@ -316,3 +316,22 @@ def test_bm25_term_ids_cjk_bigrams():
     hits = shard.search([bm25_term_ids("量子计算")], None, k=2)
     top = int(hits.bm25_ids[0, 0])
     assert top == 1
+
+
+def test_bm25_term_ids_unicode_folding():
+    """GPU tokenizer matches FTS5 unicode61 semantics: diacritic
+    folding and non-Latin scripts."""
+    import numpy as np
+    from infomesh_amd.index.gpu_index import CpuShard, bm25_term_ids
+
+    assert np.array_equal(bm25_term_ids("café"), bm25_term_ids("cafe"))
+    assert np.array_equal(bm25_term_ids("Müller"), bm25_term_ids("muller"))
+    ru = bm25_term_ids("квантовый компьютер")
+    assert len(ru) == 2  # Cyrillic words are real tokens now
+    shard = CpuShard()
+    shard.add_document(1, bm25_term_ids("квантовый компьютер исследования"),
+                       None)
+    shard.add_document(2, bm25_term_ids("english only text"), None)
+    shard.build()
+    hits = shard.search([bm25_term_ids("квантовый")], None, k=2)
+    assert int(hits.bm25_ids[0, 0]) == 1
