@@ -17,6 +17,9 @@ from ..hashing import content_hash
 from .gpu_index import CpuShard, GpuShard
 
 MANIFEST_VERSION = 1
+# bump when bm25_term_ids tokenization changes — old term ids
+# become incompatible and saved shards must be rebuilt
+TOKENIZER_VERSION = 2  # v2: unicode \w + NFKD folding + CJK bigrams
 
 
 def save_shard(shard: GpuShard, path: str | Path, rank: int = 0,
@@ -34,6 +37,7 @@ def save_shard(shard: GpuShard, path: str | Path, rank: int = 0,
         payload["embeddings"] = shard.embeddings.cpu()
     meta = {
         "version": MANIFEST_VERSION,
+        "tokenizer_version": TOKENIZER_VERSION,
         "rank": rank, "world": world,
         "n_docs": shard.n_docs, "vocab": shard.vocab,
         "avgdl": shard.avgdl,
@@ -54,6 +58,12 @@ def load_shard(path: str | Path, device: str = "cuda") -> GpuShard:
     meta = blob["meta"]
     if meta.get("version") != MANIFEST_VERSION:
         raise ValueError(f"manifest version {meta.get('version')}")
+    tv = meta.get("tokenizer_version", 1)
+    if tv != TOKENIZER_VERSION:
+        raise ValueError(
+            f"shard was built with tokenizer v{tv}, current is "
+            f"v{TOKENIZER_VERSION}: term ids are incompatible — rebuild "
+            "the shard from the LocalStore/snapshot ground truth")
     shard = GpuShard(device) if device.startswith("cuda") else CpuShard()
     shard.n_docs = meta["n_docs"]
     shard.vocab = meta["vocab"]

@@ -462,3 +462,26 @@ def test_feeds_cli_persistence(tmp_path, monkeypatch):
     assert r.exit_code == 0
     r = runner.invoke(root_cli, ["feeds", "list"])
     assert "ex.org" not in r.output and "a.io" in r.output
+
+
+def test_manifest_tokenizer_version_guard(tmp_path):
+    import numpy as np
+    import pytest as _pytest
+    from infomesh_amd.index.gpu_index import CpuShard, bm25_term_ids
+    from infomesh_amd.index import manifest as M
+    shard = CpuShard()
+    shard.add_document(1, bm25_term_ids("hello world"), None)
+    shard.build()
+    path = tmp_path / "s.shard"
+    meta = M.save_shard(shard, path)
+    assert meta["tokenizer_version"] == M.TOKENIZER_VERSION
+    s2 = M.load_shard(path, device="cpu")
+    assert s2.n_docs == 1
+    # a stale tokenizer version refuses to load (term-id mismatch)
+    import json
+    mpath = path.with_suffix(".json")
+    d = json.loads(mpath.read_text())
+    d["tokenizer_version"] = 1
+    mpath.write_text(json.dumps(d))
+    with _pytest.raises(ValueError, match="tokenizer"):
+        M.load_shard(path, device="cpu")
