@@ -477,11 +477,11 @@ def test_manifest_tokenizer_version_guard(tmp_path):
     assert meta["tokenizer_version"] == M.TOKENIZER_VERSION
     s2 = M.load_shard(path, device="cpu")
     assert s2.n_docs == 1
-    # a stale tokenizer version refuses to load (term-id mismatch)
-    import json
-    mpath = path.with_suffix(".json")
-    d = json.loads(mpath.read_text())
-    d["tokenizer_version"] = 1
-    mpath.write_text(json.dumps(d))
+    # a stale tokenizer version refuses to load (term-id mismatch);
+    # the authoritative meta lives in the torch blob, not the sidecar
+    import torch
+    blob = torch.load(path, map_location="cpu", weights_only=False)
+    blob["meta"]["tokenizer_version"] = 1
+    torch.save(blob, path)
     with _pytest.raises(ValueError, match="tokenizer"):
         M.load_shard(path, device="cpu")
