@@ -62,6 +62,10 @@ class AppContext:
     attestations: list = field(default_factory=list)
     farming: FarmingDetector = field(default_factory=FarmingDetector)
     started_at: float = field(default_factory=time.time)
+    # resource ladder: crawl loop consults it for backpressure and
+    # index_document refuses writes at READ_ONLY (reference
+    # governor.py:49-74 semantics)
+    governor: Any = None
 
     # ------------------------------------------------------------ build
     @classmethod
@@ -101,6 +105,8 @@ class AppContext:
             feeds=FeedMonitor(path=None if in_memory else p("feeds.json")),
             related=RelatedSearchTracker(),
         )
+        from .utils.governor import ResourceGovernor
+        ctx.governor = ResourceGovernor()
         if with_worker if with_worker is not None else role in ("full", "crawler"):
             ctx.worker = CrawlWorker(
                 cfg.crawl,
@@ -124,6 +130,9 @@ class AppContext:
         """THE single crawl->index source of truth
         (reference: services.py:68-110)."""
         from .utils.plugins import GLOBAL_PLUGINS
+        if self.governor is not None and not self.governor.writes_allowed():
+            raise InfoMeshError(
+                "RT002", "node degraded to read-only (resource governor)")
         doc = GLOBAL_PLUGINS.run("pre_index", doc)
         if self.deletions.is_forgotten(doc.url):
             raise InfoMeshError("SEC001", "url under GDPR deletion record")

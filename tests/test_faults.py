@@ -169,3 +169,31 @@ def test_localstore_concurrent_readers_during_writes(tmp_path):
     assert not errs
     assert store.count() == 200
     store.close()
+
+
+def test_read_only_degrade_blocks_writes(tmp_path, monkeypatch):
+    """At DegradeLevel.READ_ONLY, index_document refuses loudly and
+    search still works (reference governor ladder)."""
+    monkeypatch.setenv("INFOMESH_NODE_DATA_DIR", str(tmp_path))
+    from infomesh_amd.errors import InfoMeshError
+    from infomesh_amd.services import AppContext
+    ctx = AppContext.create(with_engine=False, with_worker=False)
+    try:
+        ctx.index_document(Document(url="https://g.org/1", title="t",
+                                    text="governor body " * 10))
+
+        class FrozenGov:
+            def writes_allowed(self):
+                return False
+
+            def crawl_allowed(self):
+                return False
+
+        ctx.governor = FrozenGov()
+        with pytest.raises(InfoMeshError):
+            ctx.index_document(Document(url="https://g.org/2", title="t",
+                                        text="more body " * 10))
+        assert ctx.search("governor").results  # reads unaffected
+    finally:
+        ctx.governor = None
+        ctx.close()
