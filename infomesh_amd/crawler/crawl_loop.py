@@ -32,7 +32,8 @@ async def seed_and_crawl_loop(ctx: AppContext,
     """Main crawl loop. `max_iterations`/`stop_check` bound it for tests
     and foreground runs."""
     assert ctx.worker is not None, "crawl loop needs a worker role"
-    governor = governor or ResourceGovernor()
+    governor = governor or getattr(ctx, "governor", None) \
+        or ResourceGovernor()
     sched = ctx.worker.scheduler
     for url in load_seeds(seed_category):
         sched.add_url(url, depth=0)
