@@ -87,11 +87,13 @@ class Phi3Decoder:
         self.v_cache = [torch.zeros_like(self.k_cache[0])
                         for _ in range(cfg.layers)]
         self.lens = torch.zeros(max_batch, dtype=torch.int32, device=device)
+        self._len_host = 0
         self.use_graph = use_graph and device.startswith("cuda")
         self._graphs: dict[int, tuple] = {}
 
     def reset(self) -> None:
         self.lens.zero_()
+        self._len_host = 0  # host shadow of max(lens): bounds decode
 
     # ------------------------------------------------------------ layers
     def _split_qkv(self, qkv: torch.Tensor, B: int, S: int):
@@ -156,6 +158,7 @@ class Phi3Decoder:
                             self.w[p + "down.w"])
             x = K.add(x, mlp)
         self.lens[:B] = S_true
+        self._len_host = max(self._len_host, S_true)
         x_last = x.view(B, S, H)[:, S_true - 1, :].contiguous()
         h = K.rmsnorm(x_last, self.w["final_ln.g"], eps=cfg.eps)
         return K.gemm_nt(h, self.w["lm_head"], out_f32=True)
@@ -168,6 +171,13 @@ class Phi3Decoder:
         fixed-batch steps replay a captured hipGraph: all mutable state
         (lens, caches) lives in device tensors the graph reads/writes."""
         B = ids.shape[0]
+        # host-side bound: kv_append at lens == max_seq would write out
+        # of the cache (device-side, unchecked by design in the kernel)
+        if self._len_host + 1 > self.max_seq:
+            raise RuntimeError(
+                f"KV cache full ({self._len_host}/{self.max_seq}); "
+                "raise max_seq or reset()")
+        self._len_host += 1
         if not self.use_graph:
             return self._decode_impl(ids)
         entry = self._graphs.get(B)
@@ -194,6 +204,8 @@ class Phi3Decoder:
         with torch.cuda.graph(graph):
             static_out = self._decode_impl(static_ids)
         self.lens.copy_(lens_save)  # undo warmup/capture side effects
+        # (_len_host is managed by decode_step; capture ran 3 impls on
+        # restored device lens, so the host shadow needs no change)
         return graph, static_ids, static_out
 
     def _decode_impl(self, ids: torch.Tensor) -> torch.Tensor:

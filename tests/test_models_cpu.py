@@ -57,3 +57,16 @@ def test_phi3_weights_shapes():
     cos, sin = rope_tables(PHI3_TINY, "cpu")
     assert cos.shape == (256, 16)
     assert torch.allclose(cos[0], torch.ones(16))
+
+
+def test_phi3_kv_cache_bound_guard():
+    """decode_step refuses when the KV cache is full instead of letting
+    kv_append write out of bounds on device."""
+    import pytest
+    import torch
+    from infomesh_amd.models.phi3 import PHI3_TINY, Phi3Decoder
+    dec = Phi3Decoder(PHI3_TINY, device="cpu", max_batch=1, max_seq=8,
+                      use_graph=False)
+    dec._len_host = 8  # cache full
+    with pytest.raises(RuntimeError, match="KV cache full"):
+        dec.decode_step(torch.zeros(1, dtype=torch.int32))
