@@ -485,3 +485,34 @@ def test_manifest_tokenizer_version_guard(tmp_path):
     torch.save(blob, path)
     with _pytest.raises(ValueError, match="tokenizer"):
         M.load_shard(path, device="cpu")
+
+
+def test_mcp_http_transport(tmp_path, monkeypatch):
+    """Streamable-HTTP transport: JSON-RPC over POST, api-key auth,
+    health, and error paths."""
+    monkeypatch.setenv("INFOMESH_NODE_DATA_DIR", str(tmp_path))
+    from starlette.testclient import TestClient
+    from infomesh_amd.mcp.server import McpServer
+    ctx = AppContext.create(with_engine=False, with_worker=False)
+    try:
+        srv = McpServer(ctx, api_key="sekrit")
+        client = TestClient(srv.asgi_app())
+        # no key -> 401
+        r = client.post("/mcp", json={"jsonrpc": "2.0", "id": 1,
+                                      "method": "ping"})
+        assert r.status_code == 401
+        h = {"x-api-key": "sekrit"}
+        r = client.post("/mcp", json={"jsonrpc": "2.0", "id": 1,
+                                      "method": "ping"}, headers=h)
+        assert r.status_code == 200 and r.json()["result"] == {}
+        r = client.get("/health", headers=h)
+        assert r.status_code == 200 and r.json()["ok"]
+        r = client.post("/mcp", content=b"not json", headers=h)
+        assert r.status_code == 400
+        r = client.get("/nope", headers=h)
+        assert r.status_code == 404
+        r = client.post("/mcp", json={"jsonrpc": "2.0", "id": 2,
+                                      "method": "tools/list"}, headers=h)
+        assert len(r.json()["result"]["tools"]) == 5
+    finally:
+        ctx.close()
