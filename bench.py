@@ -100,13 +100,16 @@ def main():
     if rank == 0 and not args.bm25_only and use_gpu:
         from infomesh_amd.models.encoder import EmbeddingEncoder
         encoder = EmbeddingEncoder(device=device)
-    if rank == 0 and args.rerank:
+    if rank == 0 and args.rerank and use_gpu:
         from infomesh_amd.models.reranker import CrossEncoderReranker
         reranker = CrossEncoderReranker(device=device)
-    if rank == 0 and args.rag:
+    if rank == 0 and args.rag and use_gpu:
         from infomesh_amd.models.phi3 import PHI3_MINI, Phi3Decoder
         summarizer = Phi3Decoder(PHI3_MINI, device=device, max_batch=1,
                                  max_seq=2304)
+    if (args.rerank or args.rag) and not use_gpu and rank == 0:
+        print("# --rerank/--rag need a GPU; running the hybrid step only",
+              flush=True)
 
     # Pre-generate rotating query batches (token ids for the encoder,
     # term ids for BM25). Generation is setup; encoding is TIMED.
