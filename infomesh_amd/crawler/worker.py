@@ -31,6 +31,16 @@ TEXT_CONTENT_TYPES = ("text/html", "application/xhtml", "text/plain")
 
 
 @dataclass
+class FetchedResponse:
+    """What crawl_url needs from an HTTP response (body already capped
+    by the streaming reader)."""
+    status_code: int
+    headers: dict[str, str]
+    content: bytes
+    encoding: str
+
+
+@dataclass
 class CrawlResult:
     url: str
     status: str                     # ok | skipped | error
@@ -80,8 +90,10 @@ class CrawlWorker:
             self._client = None
 
     async def _fetch(self, url: str, etag: str = "",
-                     last_modified: str = "") -> httpx.Response:
-        """GET with redirect re-validation, 5xx retries, size guard."""
+                     last_modified: str = "") -> "FetchedResponse":
+        """GET with redirect re-validation, 5xx retries, and a STREAMING
+        size guard: at most max_response_bytes are ever read off the
+        wire, so a huge/hostile body cannot spike memory."""
         client = await self._get_client()
         headers = {}
         if etag:
@@ -89,19 +101,36 @@ class CrawlWorker:
         if last_modified:
             headers["If-Modified-Since"] = last_modified
         current = url
+        resp = None
         for attempt in range(self.cfg.retries + 1):
             redirects = 0
             while True:
-                resp = await client.get(current, headers=headers)
-                if resp.status_code in (301, 302, 303, 307, 308):
-                    redirects += 1
-                    if redirects > 5:
-                        raise InfoMeshError("CRWL003", "too many redirects")
-                    location = resp.headers.get("location", "")
-                    current = str(httpx.URL(current).join(location))
-                    validate_url_post_redirect(current, resolve_dns=False)
-                    continue
-                break
+                async with client.stream("GET", current,
+                                         headers=headers) as r:
+                    if r.status_code in (301, 302, 303, 307, 308):
+                        redirects += 1
+                        if redirects > 5:
+                            raise InfoMeshError("CRWL003",
+                                                "too many redirects")
+                        location = r.headers.get("location", "")
+                        current = str(httpx.URL(current).join(location))
+                        validate_url_post_redirect(current,
+                                                   resolve_dns=False)
+                        continue
+                    body = b""
+                    if r.status_code == 200:
+                        cap = self.cfg.max_response_bytes
+                        async for chunk in r.aiter_bytes():
+                            body += chunk
+                            if len(body) >= cap:
+                                body = body[:cap]
+                                break
+                    resp = FetchedResponse(
+                        status_code=r.status_code,
+                        headers=dict(r.headers),
+                        content=body,
+                        encoding=r.encoding or "utf-8")
+                    break
             if resp.status_code >= 500 and attempt < self.cfg.retries:
                 bo = self.retry_backoff_s
                 await asyncio.sleep(bo[min(attempt, len(bo) - 1)])
@@ -159,11 +188,10 @@ class CrawlWorker:
             self.stats["skipped"] += 1
             return CrawlResult(url, "skipped", f"content-type {ctype}",
                                http_status=200)
-        body = resp.content[: self.cfg.max_response_bytes]
         try:
-            html = body.decode(resp.encoding or "utf-8", errors="replace")
+            html = resp.content.decode(resp.encoding, errors="replace")
         except LookupError:
-            html = body.decode("utf-8", errors="replace")
+            html = resp.content.decode("utf-8", errors="replace")
 
         # 5. extract
         page = extract_content(url, html)

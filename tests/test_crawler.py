@@ -326,3 +326,28 @@ def test_worker_5xx_retries():
         assert calls["n"] == 3
         await w.close()
     asyncio.run(run())
+
+
+def test_fetch_streams_capped(tmp_path):
+    """The streaming fetch never holds more than max_response_bytes:
+    a 10x-oversized body is truncated at the cap, not fully buffered."""
+    import asyncio
+
+    import httpx
+
+    from infomesh_amd.config import CrawlConfig
+    from infomesh_amd.crawler.worker import CrawlWorker
+
+    big = "<html><body>" + ("spam " * 40_000) + "</body></html>"
+
+    def handler(request):
+        return httpx.Response(200, text=big,
+                              headers={"content-type": "text/html"})
+
+    cfg = CrawlConfig(max_response_bytes=16_384, respect_robots=False)
+    w = CrawlWorker(cfg, client=httpx.AsyncClient(
+        transport=httpx.MockTransport(handler)), resolve_dns=False)
+    resp = asyncio.run(w._fetch("https://big.example/x"))
+    assert resp.status_code == 200
+    assert len(resp.content) == 16_384
+    asyncio.run(w.close())
