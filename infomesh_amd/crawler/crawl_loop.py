@@ -60,10 +60,19 @@ async def seed_and_crawl_loop(ctx: AppContext,
             last_feed = now
             for mf in ctx.feeds.due(now)[:5]:
                 try:
+                    # capped streaming read (feeds are small; a hostile
+                    # endpoint must not buffer unbounded XML)
                     client = await ctx.worker._get_client()
-                    resp = await client.get(mf.url)
-                    feed = parse_feed(mf.url, resp.text) \
-                        if resp.status_code == 200 else None
+                    async with client.stream("GET", mf.url) as r:
+                        body = b""
+                        if r.status_code == 200:
+                            async for chunk in r.aiter_bytes():
+                                body += chunk
+                                if len(body) > 2_000_000:
+                                    break
+                    feed = parse_feed(
+                        mf.url, body.decode("utf-8", errors="replace")) \
+                        if body else None
                 except Exception:
                     feed = None
                 for item in ctx.feeds.record_poll(mf.url, feed, now):
