@@ -198,6 +198,17 @@ class Handlers:
     def ping(self, **_) -> dict[str, Any]:
         return {"pong": True, "ts": time.time()}
 
+    def suggest(self, query: str = "", prefix: str = "",
+                limit: int = 5, **_) -> dict[str, Any]:
+        """Title-prefix completions (reference suggest tool) — related
+        searches are appended when the tracker knows the query."""
+        p = (prefix or query).strip()
+        out = {"suggestions": self.ctx.store.suggest(p, limit=limit)}
+        rel = getattr(self.ctx, "related", None)
+        if rel is not None and p:
+            out["related"] = rel.related(p, limit=limit)
+        return out
+
     def credit_balance(self, **_) -> dict[str, Any]:
         if self.ctx.ledger is None:
             return {"enabled": False}

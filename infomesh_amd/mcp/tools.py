@@ -100,7 +100,7 @@ LEGACY_ALIASES: dict[str, str] = {
     "explain": "web_search",
     "search_rag": "web_search",
     "extract_answer": "web_search",
-    "suggest": "web_search",
+    "suggest": "suggest",
     "get_page": "fetch_page",
     "fetch": "fetch_page",
     "crawl": "crawl_url",

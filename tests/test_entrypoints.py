@@ -516,3 +516,23 @@ def test_mcp_http_transport(tmp_path, monkeypatch):
         assert len(r.json()["result"]["tools"]) == 5
     finally:
         ctx.close()
+
+
+def test_mcp_suggest_tool(tmp_path, monkeypatch):
+    monkeypatch.setenv("INFOMESH_NODE_DATA_DIR", str(tmp_path))
+    from infomesh_amd.index.local_store import Document
+    from infomesh_amd.mcp.server import McpServer
+    ctx = AppContext.create(with_engine=False, with_worker=False)
+    try:
+        ctx.index_document(Document(url="https://s.org/1",
+                                    title="Quantum computing primer",
+                                    text="intro " * 20))
+        srv = McpServer(ctx)
+        r = srv.handle_message({"jsonrpc": "2.0", "id": 1,
+                                "method": "tools/call",
+                                "params": {"name": "suggest",
+                                           "arguments": {"prefix": "Quan"}}})
+        payload = json.loads(r["result"]["content"][0]["text"])
+        assert payload["suggestions"] == ["Quantum computing primer"]
+    finally:
+        ctx.close()
