@@ -121,3 +121,22 @@ def test_rrf_fuse_matches_naive(batch, m_per_source, n_out, seed):
         # each returned id's true contribution equals its reported score
         for gid, gv in zip(got_ids, got_sc):
             assert abs(contrib[gid] - gv) < 1e-5
+
+
+@given(st.text(alphabet=st.characters(min_codepoint=32, max_codepoint=0x2FFF),
+               min_size=0, max_size=60))
+@settings(max_examples=60, deadline=None)
+def test_search_never_crashes_on_hostile_queries(q):
+    """FTS5 syntax injection (quotes, NEAR, parens, *, unicode) must
+    never raise from the search stack — worst case: empty results."""
+    global _HOSTILE_STORE
+    try:
+        store = _HOSTILE_STORE
+    except NameError:
+        from infomesh_amd.index.local_store import Document, LocalStore
+        store = _HOSTILE_STORE = LocalStore(":memory:")
+        store.add_document(Document(url="https://h.org/1", title="doc",
+                                    text="hostile query fuzzing body"))
+    from infomesh_amd.search.query import search_local
+    resp = search_local(store, q, limit=3)
+    assert isinstance(resp.results, list)
