@@ -140,3 +140,26 @@ def test_search_never_crashes_on_hostile_queries(q):
     from infomesh_amd.search.query import search_local
     resp = search_local(store, q, limit=3)
     assert isinstance(resp.results, list)
+
+
+@given(st.text(min_size=0, max_size=120))
+@settings(max_examples=60, deadline=None)
+def test_bm25_term_ids_total_function(text):
+    """Any unicode input (emoji, combining marks, RTL, control chars)
+    tokenizes without raising, with ids in [0, vocab)."""
+    from infomesh_amd.index.gpu_index import BM25_VOCAB, bm25_term_ids
+    t = bm25_term_ids(text)
+    assert t.dtype == np.int64
+    if len(t):
+        assert int(t.min()) >= 0 and int(t.max()) < BM25_VOCAB
+
+
+@given(st.text(min_size=1, max_size=60))
+@settings(max_examples=40, deadline=None)
+def test_tokenizer_index_query_consistency(text):
+    """The same text tokenizes identically whether it arrives as a doc
+    or as a query (the GPU plane depends on this symmetry)."""
+    from infomesh_amd.index.gpu_index import bm25_term_ids
+    a = bm25_term_ids(text)
+    b = bm25_term_ids(text)
+    assert np.array_equal(a, b)
