@@ -163,3 +163,47 @@ def test_tokenizer_index_query_consistency(text):
     a = bm25_term_ids(text)
     b = bm25_term_ids(text)
     assert np.array_equal(a, b)
+
+
+@given(st.text(min_size=0, max_size=400))
+@settings(max_examples=40, deadline=None)
+def test_html_and_feed_parsers_total(raw):
+    """Hostile/broken markup never raises from the extraction stack."""
+    from infomesh_amd.crawler.parser import (extract_content,
+                                             extract_links,
+                                             looks_like_js_app)
+    from infomesh_amd.crawler.rss import parse_feed
+    from infomesh_amd.crawler.structured import (extract_code_blocks,
+                                                 extract_json_ld,
+                                                 extract_open_graph,
+                                                 extract_tables)
+    page = extract_content("https://f.org/x", raw)
+    assert page.url == "https://f.org/x"
+    extract_links("https://f.org/x", raw)
+    looks_like_js_app(raw, page.text)
+    parse_feed("https://f.org/feed", raw)
+    extract_json_ld(raw)
+    extract_open_graph(raw)
+    extract_code_blocks(raw)
+    extract_tables(raw)
+
+
+@given(st.binary(min_size=0, max_size=200))
+@settings(max_examples=30, deadline=None)
+def test_snapshot_import_arbitrary_bytes_fail_closed(data):
+    """Random bytes as a snapshot: import raises a clean error, never
+    imports anything."""
+    import pathlib
+    import tempfile
+
+    import pytest as _pytest
+    from infomesh_amd.index.local_store import LocalStore
+    from infomesh_amd.index.snapshot import import_snapshot
+    with tempfile.TemporaryDirectory() as d:
+        p = pathlib.Path(d) / "x.infomesh-snapshot"
+        p.write_bytes(data)
+        store = LocalStore(":memory:")
+        with _pytest.raises(Exception):
+            import_snapshot(store, p)
+        assert store.count() == 0
+        store.close()
