@@ -24,6 +24,8 @@ from ..runtime import GracefulShutdown, PidFile, RuntimeStatus, StartupLock
 
 
 @click.group()
+@click.version_option(package_name=None, prog_name="infomesh-amd",
+                      version=__import__("infomesh_amd").__version__)
 def cli():
     """infomesh-amd — MI355X-native hybrid search/RAG engine."""
 
