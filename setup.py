@@ -2,7 +2,7 @@
 CDNA4 HIP extension in-tree (gfx950) via hipcc."""
 from __future__ import annotations
 
-from setuptools import Command, find_packages, setup
+from setuptools import find_packages, setup
 from setuptools.command.build_ext import build_ext as _build_ext
 
 
