@@ -104,3 +104,38 @@ def test_full_lifecycle(node, tmp_path, monkeypatch):
     # 5. re-crawl is conditional: second crawl of same URL is deduped
     out = asyncio.run(ctx.crawl_and_index("https://site-a.org/gpu"))
     assert out["status"] in ("skipped", "ok")
+
+
+def test_lifecycle_compliance_enforcement(node):
+    """DMCA takedown + GDPR deletion remove content and block
+    re-indexing through the real index path."""
+    ctx = node
+    for url in PAGES:
+        asyncio.run(ctx.crawl_and_index(url))
+    assert ctx.store.count() == 3
+
+    # DMCA: takedown site-b -> applied -> gone and blocked
+    ctx.takedowns.file_notice("domain:site-b.net", "copyright claim",
+                              claimant="rights-holder@example.com")
+    removed = ctx.takedowns.apply_pending()
+    assert removed >= 1
+    assert ctx.store.get_document_by_url("https://site-b.net/other") is None
+    from infomesh_amd.errors import InfoMeshError
+    from infomesh_amd.index.local_store import Document
+    with pytest.raises(InfoMeshError):
+        ctx.index_document(Document(url="https://site-b.net/back",
+                                    title="t", text="returns " * 20))
+
+    # GDPR: forget one site-a URL -> enforced -> blocked at re-crawl
+    ctx.deletions.request_deletion("https://site-a.org/lds",
+                                   reason="user request")
+    ctx.deletions.enforce()
+    assert ctx.store.get_document_by_url("https://site-a.org/lds") is None
+    # re-crawling the forgotten URL is refused LOUDLY at index time
+    # (the background crawl loop isolates this per-URL)
+    with pytest.raises(InfoMeshError):
+        asyncio.run(ctx.crawl_and_index("https://site-a.org/lds",
+                                        force=True))
+    assert ctx.store.get_document_by_url("https://site-a.org/lds") is None
+    # the other site-a page is untouched
+    assert ctx.store.get_document_by_url("https://site-a.org/gpu")
