@@ -207,3 +207,39 @@ def test_snapshot_import_arbitrary_bytes_fail_closed(data):
             import_snapshot(store, p)
         assert store.count() == 0
         store.close()
+
+
+@given(st.lists(st.text(min_size=1, max_size=20), min_size=1,
+                max_size=24), st.data())
+@settings(max_examples=30, deadline=None)
+def test_merkle_membership_and_tamper(items, data):
+    from infomesh_amd.trust.merkle import MerkleTree
+    tree = MerkleTree.from_items(items)
+    idx = data.draw(st.integers(0, len(items) - 1))
+    proof = tree.prove(idx)
+    assert MerkleTree.verify_proof(tree.root, proof, items[idx])
+    assert not MerkleTree.verify_proof(tree.root, proof,
+                                       items[idx] + "x")
+    assert not MerkleTree.verify_proof(b"\x01" * 32, proof, items[idx])
+
+
+@given(st.integers(0, 2 ** 63 - 1), st.integers(0, 2 ** 63 - 1))
+@settings(max_examples=50, deadline=None)
+def test_simhash_hamming_metric(a, b):
+    from infomesh_amd.crawler.simhash import hamming_distance as hamming
+    assert hamming(a, a) == 0
+    assert hamming(a, b) == hamming(b, a)
+    assert 0 <= hamming(a, b) <= 64
+
+
+@given(st.floats(0, 1), st.floats(1, 400))
+@settings(max_examples=40, deadline=None)
+def test_freshness_monotonic(bm, age_days):
+    """Fresher documents never score lower, all else equal."""
+    import time as _t
+    from infomesh_amd.index.ranking import freshness_score
+    now = 1_700_000_000.0
+    newer = freshness_score(now - age_days * 43200, now=now)
+    older = freshness_score(now - age_days * 86400, now=now)
+    assert newer >= older >= 0.05 - 1e-9
+    assert freshness_score(now, now=now) <= 1.0
