@@ -20,8 +20,12 @@ TRACKING_PARAMS = frozenset(
 
 
 def normalize_url(url: str) -> str:
+    # urlsplit (NOT urlparse): urlparse splits ';params' out of the
+    # last path segment, which made normalization non-idempotent for
+    # paths containing ';' (found by the hypothesis suite).
+    from urllib.parse import urlsplit, urlunsplit
     try:
-        p = urlparse(url.strip())
+        p = urlsplit(url.strip())
     except ValueError:
         return url
     host = (p.hostname or "").lower()
@@ -32,9 +36,9 @@ def normalize_url(url: str) -> str:
         (k, v) for k, v in parse_qsl(p.query, keep_blank_values=True)
         if k.lower() not in TRACKING_PARAMS))
     path = p.path or "/"
-    if path != "/" and path.endswith("/"):
-        path = path[:-1]
-    return urlunparse((p.scheme.lower(), host, path, "", query, ""))
+    if path != "/":
+        path = path.rstrip("/") or "/"
+    return urlunsplit((p.scheme.lower(), host, path, query, ""))
 
 
 class DeduplicatorDB(SQLiteStore):

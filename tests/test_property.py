@@ -243,3 +243,24 @@ def test_freshness_monotonic(bm, age_days):
     older = freshness_score(now - age_days * 86400, now=now)
     assert newer >= older >= 0.05 - 1e-9
     assert freshness_score(now, now=now) <= 1.0
+
+
+@given(st.text(alphabet=st.characters(min_codepoint=33, max_codepoint=126),
+               min_size=1, max_size=60))
+@settings(max_examples=60, deadline=None)
+def test_url_normalization_idempotent(tail):
+    from infomesh_amd.crawler.dedup import normalize_url
+    u = "https://Example.COM/" + tail
+    n1 = normalize_url(u)
+    assert normalize_url(n1) == n1
+
+
+@given(st.floats(min_value=0.0, max_value=1e6),
+       st.floats(min_value=1e-6, max_value=1e6))
+@settings(max_examples=60, deadline=None)
+def test_bm25_normalization_bounded(raw, batch_max):
+    from infomesh_amd.index.ranking import normalize_bm25
+    v = normalize_bm25(raw, batch_max)
+    assert 0.0 <= v <= 1.0
+    # monotonic in raw
+    assert normalize_bm25(raw * 0.5, batch_max) <= v + 1e-12
