@@ -38,6 +38,8 @@ def contains_cjk(text: str) -> bool:
 
 def ngram_expand(run: str, n: int = 2) -> list[str]:
     """Overlapping n-grams of a contiguous CJK run (bigrams by default)."""
+    if not run:
+        return []
     if len(run) <= n:
         return [run]
     return [run[i:i + n] for i in range(len(run) - n + 1)]

@@ -264,3 +264,39 @@ def test_bm25_normalization_bounded(raw, batch_max):
     assert 0.0 <= v <= 1.0
     # monotonic in raw
     assert normalize_bm25(raw * 0.5, batch_max) <= v + 1e-12
+
+
+@given(st.text(min_size=0, max_size=300), st.text(min_size=0, max_size=40))
+@settings(max_examples=40, deadline=None)
+def test_search_nlp_and_passage_total(doc_text, query):
+    """NLP preprocessing, filter parsing, passage selection and
+    highlighting are total functions over arbitrary input."""
+    from infomesh_amd.search.nlp import (did_you_mean, expand_query,
+                                         parse_query_filters,
+                                         remove_stop_words)
+    from infomesh_amd.search.query import preprocess_query
+    from infomesh_amd.search.passage import (select_best_passage,
+                                             split_passages)
+    pq = parse_query_filters(query)
+    assert isinstance(pq.text, str)
+    preprocess_query(pq.text)
+    remove_stop_words(query)
+    expand_query(query)
+    did_you_mean(query, {"example", "sample", "words"})
+    parts = split_passages(doc_text)
+    assert isinstance(parts, list)
+    select_best_passage(doc_text, query)
+
+
+@given(st.text(alphabet=st.characters(min_codepoint=0x4E00,
+                                      max_codepoint=0x4FFF),
+               min_size=0, max_size=12), st.integers(1, 4))
+@settings(max_examples=40, deadline=None)
+def test_cjk_ngram_properties(run, n):
+    from infomesh_amd.search.cjk import ngram_expand
+    grams = ngram_expand(run, n)
+    if len(run) < n:
+        assert grams == ([run] if run else [])
+    else:
+        assert len(grams) == len(run) - n + 1
+        assert all(len(g) == n and g in run for g in grams)
