@@ -300,3 +300,31 @@ def test_cjk_ngram_properties(run, n):
     else:
         assert len(grams) == len(run) - n + 1
         assert all(len(g) == n and g in run for g in grams)
+
+
+@given(st.text(min_size=0, max_size=300))
+@settings(max_examples=40, deadline=None)
+def test_lang_detect_total(text):
+    from infomesh_amd.crawler.lang_detect import detect_language
+    lang = detect_language(text)
+    assert isinstance(lang, str) and len(lang) <= 5
+
+
+@given(st.text(min_size=0, max_size=200), st.text(min_size=0, max_size=200))
+@settings(max_examples=30, deadline=None)
+def test_content_diff_total(a, b):
+    from infomesh_amd.crawler.diff import diff_content, significant_change
+    d = diff_content(a, b)
+    assert 0.0 <= d.changed_ratio <= 1.0
+    assert diff_content(a, a).changed_ratio == 0.0
+    significant_change(a, b)
+
+
+@given(st.text(min_size=0, max_size=100), st.text(min_size=0, max_size=20))
+@settings(max_examples=30, deadline=None)
+def test_snippet_highlight_total(text, q):
+    from infomesh_amd.search.passage import highlight
+    out = highlight(text, q)
+    # highlighting only adds tags, never loses content
+    import re as _re
+    assert _re.sub(r"</?b>", "", out) == text
