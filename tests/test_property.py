@@ -328,3 +328,40 @@ def test_snippet_highlight_total(text, q):
     # highlighting only adds tags, never loses content
     import re as _re
     assert _re.sub(r"</?b>", "", out) == text
+
+
+@given(st.lists(
+    st.tuples(st.text(alphabet=st.characters(min_codepoint=33,
+                                             max_codepoint=0x2FFF),
+                      min_size=1, max_size=24),
+              st.text(min_size=0, max_size=40),
+              st.text(min_size=1, max_size=120)),
+    min_size=1, max_size=8, unique_by=lambda t: t[0]))
+@settings(max_examples=20, deadline=None)
+def test_snapshot_roundtrip_property(docs):
+    """Export -> import preserves every document's url/title/text for
+    arbitrary unicode content."""
+    import pathlib
+    import tempfile
+
+    from infomesh_amd.index.local_store import Document, LocalStore
+    from infomesh_amd.index.snapshot import export_snapshot, import_snapshot
+
+    src = LocalStore(":memory:")
+    n_unique_texts = len({t for _, _, t in docs})
+    for i, (tail, title, text) in enumerate(docs):
+        src.add_document(Document(url=f"https://rt.org/{i}-{tail}",
+                                  title=title, text=text))
+    with tempfile.TemporaryDirectory() as d:
+        p = pathlib.Path(d) / "rt.infomesh-snapshot"
+        export_snapshot(src, p)
+        dst = LocalStore(":memory:")
+        res = import_snapshot(dst, p)
+        # text-hash dedup may drop duplicate TEXTS (by design)
+        assert res["imported"] == n_unique_texts
+        for i, (tail, title, text) in enumerate(docs):
+            doc = dst.get_document_by_url(f"https://rt.org/{i}-{tail}")
+            if doc is not None:
+                assert doc.title == title and doc.text == text
+        dst.close()
+    src.close()
