@@ -114,8 +114,13 @@ class CrawlWorker:
                                                 "too many redirects")
                         location = r.headers.get("location", "")
                         current = str(httpx.URL(current).join(location))
-                        validate_url_post_redirect(current,
-                                                   resolve_dns=False)
+                        # DNS-resolve the redirect target: a hostname
+                        # resolving to a private/internal IP is the
+                        # classic SSRF/DNS-rebinding hop (reference
+                        # security.py validate_url_post_redirect forces
+                        # resolve_dns=True for exactly this)
+                        validate_url_post_redirect(
+                            current, resolve_dns=self.resolve_dns)
                         continue
                     body = b""
                     if r.status_code == 200:

@@ -1,5 +1,6 @@
-"""GPU hybrid index shard: CSR inverted index (BM25) + dense embeddings
-(cosine) resident in HBM3E, scored by hand-written CDNA4 kernels.
+"""GPU hybrid index shard: segmented CSR inverted index (BM25) + dense
+embeddings (cosine) resident in HBM3E, scored by hand-written CDNA4
+kernels.
 
 Replaces intra-node: SQLite FTS5 MATCH+bm25() (reference
 infomesh/index/local_store.py:316-332) and ChromaDB HNSW
@@ -7,8 +8,19 @@ infomesh/index/local_store.py:316-332) and ChromaDB HNSW
 hash-partitioned across shards; the query plane (parallel/query_plane.py)
 fans out and all-gathers top-k.
 
-Shard sizing: 1.25M docs × (≈120 postings × 6 B + 384 × 2 B embedding)
-≈ 1.9 GB — far under the 288 GB HBM budget, so shards scale to 100M+
+Segmented design (the FTS5 append/optimize analogue, reference
+local_store.py:528-541): a flush appends ONE new posting segment built
+only from the new docs — O(new), no host pull-back of old postings —
+and `optimize()` merges segments GPU-side (torch sort over (term, doc)
+keys). BM25 stays exact across segments because the per-posting doc
+length travels packed with tf and the norm is computed in-kernel from
+the CURRENT global avgdl; idf comes from the global df table summed
+over segments. Segments partition the doc-id axis, so per-segment
+score kernels write disjoint column ranges of the [B, N] score matrix
+(each column exactly once — no zero-fill, no atomics across segments).
+
+Shard sizing: 1.25M docs × (≈120 postings × 8 B + 384 × 2 B embedding)
+≈ 2.2 GB — far under the 288 GB HBM budget, so shards scale to 100M+
 docs per GPU; the bench uses the BASELINE 10M-doc/8-GPU config.
 """
 from __future__ import annotations
@@ -25,6 +37,10 @@ from ..hashing import hash64
 BM25_K1 = 1.2
 BM25_B = 0.75
 BM25_VOCAB = 1 << 17
+
+# auto-merge threshold: beyond this many live segments a flush triggers
+# optimize() (GPU-side merge) — the FTS5 automerge analogue
+MAX_SEGMENTS = 16
 
 # \w+ covers all unicode letters/digits (FTS5 unicode61 analogue);
 # diacritics are NFKD-folded below like unicode61 remove_diacritics
@@ -74,6 +90,22 @@ class ShardHits:
     dense_ids: torch.Tensor     # [B, k] i64
 
 
+@dataclass
+class PostingSegment:
+    """One immutable CSR posting segment covering the doc-id range
+    [doc_base, doc_base + n_docs)."""
+    offsets: torch.Tensor   # [V+1] i64 (device)
+    doc_ids: torch.Tensor   # [P] i32, SEGMENT-local ids, asc per term
+    tfdl: torch.Tensor      # [P] i32: tf | (dl << 16)
+    doc_base: int
+    n_docs: int
+    h_offs: np.ndarray      # host copy of offsets (query-time chunking)
+
+    def hbm_bytes(self) -> int:
+        return sum(t.numel() * t.element_size()
+                   for t in (self.offsets, self.doc_ids, self.tfdl))
+
+
 class GpuShard:
     """One GPU's slice of the hybrid index."""
 
@@ -81,23 +113,41 @@ class GpuShard:
         self.device = torch.device(device)
         self.vocab = vocab
         self.n_docs = 0
-        # CSR postings
-        self.offsets: torch.Tensor | None = None   # [V+1] i64
-        self.doc_ids: torch.Tensor | None = None   # [P] i32 (local ids)
-        self.tfs: torch.Tensor | None = None       # [P] i16
-        self.doc_norm: torch.Tensor | None = None  # [N] f32
-        self.df: np.ndarray | None = None          # [V] i64 (host)
+        self.segments: list[PostingSegment] = []
+        self.df: np.ndarray = np.zeros(vocab, dtype=np.int64)  # global df
+        self._doc_lens = np.zeros(0, dtype=np.int64)           # host [N]
         self.avgdl = 1.0
-        # dense
-        self.embeddings: torch.Tensor | None = None  # [N, D] bf16 (unit)
-        # local idx -> global doc id
-        self.global_ids: torch.Tensor | None = None  # [N] i64
+        # dense plane: capacity-growth device buffers ([:n_docs] live)
+        self._emb_buf: torch.Tensor | None = None   # [cap, D] bf16 (unit)
+        self._gid_buf: torch.Tensor | None = None   # [cap] i64
         self._topk = None
+        self._h_idf: np.ndarray | None = None
         # pending (un-built) batch buffers — the ingest side-buffer;
         # GPU visibility flips at build() (epoch-style).
         self._pend_tokens: list[np.ndarray] = []
         self._pend_emb: list[torch.Tensor] = []
         self._pend_gids: list[int] = []
+
+    # ------------------------------------------------- derived views
+    @property
+    def embeddings(self) -> torch.Tensor | None:
+        if self._emb_buf is None or self.n_docs == 0:
+            return None
+        return self._emb_buf[:self.n_docs]
+
+    @property
+    def global_ids(self) -> torch.Tensor | None:
+        if self._gid_buf is None:
+            return None
+        return self._gid_buf[:self.n_docs]
+
+    @property
+    def doc_norm(self) -> torch.Tensor:
+        """BM25 length norm per doc (host-derived; kernels compute this
+        in-flight from packed dl — kept for the CPU oracle/tests)."""
+        dl = self._doc_lens.astype(np.float32)
+        return torch.from_numpy(
+            BM25_K1 * (1 - BM25_B + BM25_B * dl / self.avgdl))
 
     # ------------------------------------------------------------ build
     def add_document(self, global_id: int, term_ids: np.ndarray,
@@ -108,43 +158,44 @@ class GpuShard:
             self._pend_emb.append(embedding.reshape(1, -1))
 
     def build(self) -> None:
-        """(Re)build the CSR postings + embedding matrix from pending
-        docs plus any existing index — the FTS5-optimize / segment-merge
-        analogue. With existing docs this MERGES: the old postings are
-        pulled back once, concatenated with the new aggregated postings
-        (new docs get fresh local ids, so (term, doc) pairs never
-        collide), and BM25 stats (avgdl, norms) are recomputed over the
-        whole corpus."""
+        """Make pending docs searchable by appending ONE new posting
+        segment — O(new docs), never O(corpus). Triggers a GPU-side
+        optimize() merge when the segment count exceeds MAX_SEGMENTS
+        (the FTS5 automerge analogue)."""
         if not self._pend_tokens:
             return
         token_lists = self._pend_tokens
         gids = list(self._pend_gids)
         embs = self._pend_emb
         self._pend_tokens, self._pend_gids, self._pend_emb = [], [], []
-        lens2 = np.array([max(len(t), 1) for t in token_lists],
-                         dtype=np.int64)
-        n_old = self.n_docs
-        n_new = len(token_lists)
-        flat_terms = (np.concatenate(token_lists)
-                      if any(len(t) for t in token_lists)
-                      else np.zeros(0, np.int64))
-        flat_docs = np.repeat(np.arange(n_new, dtype=np.int64),
-                              [len(t) for t in token_lists])
         new_emb = torch.cat(embs, 0) if embs else None
-        if n_old == 0:
-            self.build_from_arrays(flat_terms, flat_docs, lens2,
-                                   np.asarray(gids, dtype=np.int64),
-                                   new_emb)
-            return
-        self._merge_install(self, flat_terms, flat_docs, lens2, gids,
-                            new_emb)
+        self._append_segment(token_lists, gids, new_emb)
+        if len(self.segments) > MAX_SEGMENTS:
+            self.optimize()
 
     def merged_with(self, token_lists: list[np.ndarray],
                     gids: list[int],
                     new_emb: torch.Tensor | None) -> "GpuShard":
-        """Return a NEW shard = this shard + the given docs (epoch-flip
-        variant of the in-place incremental build(): readers of the old
-        shard are never exposed to a partially-built index)."""
+        """Return a NEW shard = this shard + the given docs, sharing
+        the immutable segments/buffers with this one (epoch-flip:
+        readers of the old shard never see the new docs because every
+        read is bounded by the old n_docs). Linear use only — append to
+        the RETURNED shard, not to this one, afterwards."""
+        out = type(self)() if type(self).__init__ is not GpuShard.__init__ \
+            else GpuShard(str(self.device), vocab=self.vocab)
+        out.vocab = self.vocab
+        out.n_docs = self.n_docs
+        out.segments = list(self.segments)
+        out.df = self.df.copy()
+        out._doc_lens = self._doc_lens
+        out.avgdl = self.avgdl
+        out._emb_buf = self._emb_buf
+        out._gid_buf = self._gid_buf
+        out._append_segment(token_lists, gids, new_emb)
+        return out
+
+    def _append_segment(self, token_lists: list[np.ndarray],
+                        gids: list, new_emb: torch.Tensor | None) -> None:
         lens2 = np.array([max(len(t), 1) for t in token_lists],
                          dtype=np.int64)
         flat_terms = (np.concatenate(
@@ -153,45 +204,19 @@ class GpuShard:
             else np.zeros(0, np.int64))
         flat_docs = np.repeat(np.arange(len(token_lists), dtype=np.int64),
                               [len(t) for t in token_lists])
-        out = type(self)() if type(self).__init__ is not GpuShard.__init__ \
-            else GpuShard(str(self.device), vocab=self.vocab)
-        out.vocab = self.vocab
-        self._merge_install(out, flat_terms, flat_docs, lens2,
-                            list(gids), new_emb)
-        return out
+        self._install_segment(flat_terms, flat_docs, lens2,
+                              np.asarray(gids, dtype=np.int64), new_emb)
 
-    def _merge_install(self, target: "GpuShard", flat_terms, flat_docs,
-                       lens2, gids, new_emb) -> None:
-        """Merge THIS shard's postings with new aggregated postings and
-        install into `target` (which may be self)."""
-        n_old = self.n_docs
-        n_new = len(lens2)
-        t2, d2, tf2 = self._aggregate(flat_terms, flat_docs, n_new)
-        d2 = (d2.astype(np.int64) + n_old).astype(np.int32)
-        # pull the OLD postings back (CSR -> flat aggregated form)
-        t1 = np.repeat(np.arange(self.vocab, dtype=np.int64), self.df)
-        d1 = self.doc_ids.cpu().numpy()
-        tf1 = self.tfs.cpu().numpy().view(np.uint16)
-        terms = np.concatenate([t1, t2])
-        order = np.argsort(terms, kind="stable")
-        terms = terms[order]
-        docs = np.concatenate([d1, d2])[order]
-        tfs = np.concatenate([tf1, tf2])[order]
-        lens = np.concatenate([self._doc_lens, lens2])
-        old_gids = self.global_ids.cpu().numpy()
-        all_gids = np.concatenate(
-            [old_gids, np.asarray(gids, dtype=np.int64)])
-        emb = None
-        if self.embeddings is not None:
-            assert new_emb is not None and new_emb.shape[0] == n_new, \
-                "dense shard requires embeddings for every pending doc"
-            e2 = new_emb
-            if e2.dtype != torch.bfloat16:
-                e2 = torch.nn.functional.normalize(e2.float(),
-                                                   dim=-1).bfloat16()
-            emb = torch.cat([self.embeddings, e2.to(self.device)], 0)
-        target.n_docs = n_old + n_new
-        target._install_postings(terms, docs, tfs, lens, all_gids, emb)
+    def build_from_arrays(self, flat_terms: np.ndarray,
+                          flat_docs: np.ndarray, doc_lens: np.ndarray,
+                          global_ids: np.ndarray,
+                          embeddings: torch.Tensor | None) -> None:
+        """Bulk build from flat (term, doc-local-id) pairs — installs a
+        single segment (doc ids must be 0..len(doc_lens)-1)."""
+        self._install_segment(flat_terms, flat_docs,
+                              np.asarray(doc_lens, dtype=np.int64),
+                              np.asarray(global_ids, dtype=np.int64),
+                              embeddings)
 
     @staticmethod
     def _aggregate(flat_terms: np.ndarray, flat_docs: np.ndarray,
@@ -211,76 +236,139 @@ class GpuShard:
         query work on the compute stream — SURVEY.md §5.8 ingest path."""
         t = torch.from_numpy(np.ascontiguousarray(arr))
         if stream is None:
-            return t.to(self.device)
+            return t.clone()
         pinned = t.pin_memory()
         with torch.cuda.stream(stream):
             return pinned.to(self.device, non_blocking=True)
 
-    def build_from_arrays(self, flat_terms: np.ndarray,
-                          flat_docs: np.ndarray, doc_lens: np.ndarray,
-                          global_ids: np.ndarray,
-                          embeddings: torch.Tensor | None) -> None:
-        """Bulk build from flat (term, doc) pairs. Dedups (term, doc)
-        into term frequencies, sorts into CSR by term. Device uploads
-        go through pinned buffers on a dedicated ingest stream."""
-        n = len(doc_lens)
-        self.n_docs = n
-        # Aggregate tf per (term, doc) via a combined key sort.
-        terms_u, docs_u, tf_u = self._aggregate(flat_terms, flat_docs, n)
-        self._install_postings(terms_u, docs_u, tf_u, doc_lens,
-                               global_ids, embeddings)
+    def _grow(self, buf: torch.Tensor | None, need: int,
+              row_shape: tuple, dtype: torch.dtype) -> torch.Tensor:
+        """Capacity-doubling device buffer growth (amortized O(1) per
+        append; the old tensor stays valid for prior-epoch readers)."""
+        if buf is None:
+            cap = max(need, 256)
+            return torch.empty((cap, *row_shape), device=self.device,
+                               dtype=dtype)
+        if buf.shape[0] >= need:
+            return buf
+        cap = max(need, 2 * buf.shape[0])
+        new = torch.empty((cap, *row_shape), device=self.device,
+                          dtype=dtype)
+        new[:self.n_docs] = buf[:self.n_docs]
+        return new
 
-    def _install_postings(self, terms_u: np.ndarray, docs_u: np.ndarray,
-                          tf_u: np.ndarray, doc_lens: np.ndarray,
-                          global_ids: np.ndarray,
-                          embeddings: torch.Tensor | None) -> None:
-        """Install term-sorted aggregated postings as the CSR index and
-        recompute BM25 stats; uploads via pinned staging."""
-        n = self.n_docs
-        self._doc_lens = np.asarray(doc_lens, dtype=np.int64)
-        # CSR offsets per term (terms_u already sorted).
-        df = np.bincount(terms_u, minlength=self.vocab).astype(np.int64)
+    def _install_segment(self, flat_terms: np.ndarray,
+                         flat_docs: np.ndarray, doc_lens: np.ndarray,
+                         global_ids: np.ndarray,
+                         embeddings: torch.Tensor | None) -> None:
+        n_new = len(doc_lens)
+        if n_new == 0:
+            return
+        if self.n_docs > 0:
+            assert (embeddings is not None) == (self._emb_buf is not None), \
+                "dense/sparse mode must be consistent across appends"
+        terms_u, docs_u, tf_u = self._aggregate(flat_terms, flat_docs,
+                                                n_new)
+        dl_u = np.minimum(doc_lens[docs_u.astype(np.int64)], 65535)
+        tfdl = (tf_u.astype(np.uint32)
+                | (dl_u.astype(np.uint32) << np.uint32(16))).view(np.int32)
+        df_new = np.bincount(terms_u, minlength=self.vocab).astype(np.int64)
         offsets = np.zeros(self.vocab + 1, dtype=np.int64)
-        np.cumsum(df, out=offsets[1:])
-        self.df = df
+        np.cumsum(df_new, out=offsets[1:])
         on_gpu = self.device.type == "cuda"
         stream = torch.cuda.Stream(self.device) if on_gpu else None
-        self.offsets = self._upload(offsets, stream)
-        self.doc_ids = self._upload(docs_u, stream)
-        self.tfs = self._upload(tf_u.astype(np.int16), stream)
-        self.avgdl = float(doc_lens.mean()) if n else 1.0
-        norm = BM25_K1 * (1 - BM25_B + BM25_B *
-                          doc_lens.astype(np.float32) / self.avgdl)
-        self.doc_norm = self._upload(norm, stream)
-        self.global_ids = self._upload(global_ids.astype(np.int64), stream)
+        seg = PostingSegment(
+            offsets=self._upload(offsets, stream),
+            doc_ids=self._upload(docs_u.astype(np.int32), stream),
+            tfdl=self._upload(tfdl, stream),
+            doc_base=self.n_docs, n_docs=n_new, h_offs=offsets)
+        # dense-plane buffers: append rows [n_docs : n_docs+n_new]
+        need = self.n_docs + n_new
+        self._gid_buf = self._grow(self._gid_buf, need, (), torch.int64)
+        gid_t = self._upload(global_ids.astype(np.int64), stream)
         if embeddings is not None:
-            assert embeddings.shape[0] == n
-            if embeddings.device.type == "cpu":
-                e = embeddings
-                if e.dtype != torch.bfloat16:
-                    e = torch.nn.functional.normalize(
-                        e.float(), dim=-1).bfloat16()
-                if stream is not None:
-                    pinned = e.contiguous().pin_memory()
-                    with torch.cuda.stream(stream):
-                        e = pinned.to(self.device, non_blocking=True)
-                else:
-                    e = e.to(self.device)
+            e = embeddings
+            if e.dtype != torch.bfloat16:
+                e = torch.nn.functional.normalize(e.float(),
+                                                  dim=-1).bfloat16()
+            assert e.shape[0] == n_new, \
+                "dense shard requires embeddings for every pending doc"
+            self._emb_buf = self._grow(self._emb_buf, need,
+                                       (e.shape[1],), torch.bfloat16)
+            if e.device != self.device:
+                e = (self._upload_t(e, stream) if stream is not None
+                     else e.clone())
+            if stream is not None:
+                with torch.cuda.stream(stream):
+                    self._emb_buf[self.n_docs:need].copy_(e)
+                    self._gid_buf[self.n_docs:need].copy_(gid_t)
             else:
-                e = embeddings.to(self.device)
-                if e.dtype != torch.bfloat16:
-                    e = torch.nn.functional.normalize(
-                        e.float(), dim=-1).bfloat16()
-            self.embeddings = e.contiguous()
+                self._emb_buf[self.n_docs:need].copy_(e)
+                self._gid_buf[self.n_docs:need].copy_(gid_t)
+        else:
+            if stream is not None:
+                with torch.cuda.stream(stream):
+                    self._gid_buf[self.n_docs:need].copy_(gid_t)
+            else:
+                self._gid_buf[self.n_docs:need].copy_(gid_t)
         if stream is not None:
-            # epoch flip: the shard becomes visible only after the side
-            # stream's uploads complete on the compute stream.
+            # epoch flip: the segment becomes visible only after the
+            # side stream's uploads complete on the compute stream.
             torch.cuda.current_stream(self.device).wait_stream(stream)
+        self.segments.append(seg)
+        self.df = self.df + df_new
+        self._doc_lens = np.concatenate([self._doc_lens, doc_lens])
+        self.avgdl = float(self._doc_lens.mean())
+        self.n_docs = need
+        self._invalidate_query_caches()
+
+    def _upload_t(self, t: torch.Tensor, stream) -> torch.Tensor:
+        pinned = t.contiguous().pin_memory()
+        with torch.cuda.stream(stream):
+            return pinned.to(self.device, non_blocking=True)
+
+    def _invalidate_query_caches(self) -> None:
+        self._h_idf = None
+        # dense hipGraphs captured the old embeddings view/shape
+        self._dense_graphs = {}
+        self._dense_scores_buf = None
+
+    def optimize(self) -> None:
+        """Merge all posting segments into one, entirely on-device:
+        reconstruct (term, global-doc) keys, single torch.sort, gather
+        tfdl — the FTS5 `optimize()` analogue (reference
+        local_store.py:528-541). Docs never repeat across segments, so
+        no tf re-aggregation is needed."""
+        if len(self.segments) <= 1:
+            return
+        dev = self.device
+        V = self.vocab
+        term_parts, doc_parts, tfdl_parts = [], [], []
+        arangeV = torch.arange(V, dtype=torch.int64, device=dev)
+        for seg in self.segments:
+            dfg = seg.offsets.diff()
+            term_parts.append(torch.repeat_interleave(arangeV, dfg))
+            doc_parts.append(seg.doc_ids.to(torch.int64) + seg.doc_base)
+            tfdl_parts.append(seg.tfdl)
+        terms = torch.cat(term_parts)
+        docs = torch.cat(doc_parts)
+        tfdl = torch.cat(tfdl_parts)
+        key = terms * self.n_docs + docs
+        key, order = torch.sort(key)
+        docs_sorted = (key % self.n_docs).to(torch.int32)
+        tfdl_sorted = tfdl[order].contiguous()
+        offsets = torch.zeros(V + 1, dtype=torch.int64, device=dev)
+        torch.cumsum(torch.from_numpy(self.df).to(dev), 0,
+                     out=offsets[1:])
+        self.segments = [PostingSegment(
+            offsets=offsets, doc_ids=docs_sorted, tfdl=tfdl_sorted,
+            doc_base=0, n_docs=self.n_docs,
+            h_offs=offsets.cpu().numpy())]
+        self._invalidate_query_caches()
 
     def hbm_bytes(self) -> int:
-        total = 0
-        for t in (self.offsets, self.doc_ids, self.tfs, self.doc_norm,
-                  self.embeddings, self.global_ids):
+        total = sum(s.hbm_bytes() for s in self.segments)
+        for t in (self._emb_buf, self._gid_buf):
             if t is not None:
                 total += t.numel() * t.element_size()
         return total
@@ -301,70 +389,54 @@ class GpuShard:
         return self._topk_dense
 
     def _idf(self, term: int) -> float:
-        df = float(self.df[term]) if self.df is not None else 0.0
+        df = float(self.df[term])
         if df <= 0:
             return 0.0
         return math.log(1.0 + (self.n_docs - df + 0.5) / (df + 0.5))
 
-    def _host_tables(self):
-        """Cached host copies of offsets/df + idf table for chunking."""
-        if not hasattr(self, "_h_offs") or self._h_offs is None:
-            self._h_offs = self.offsets.cpu().numpy()
+    def _idf_table(self) -> np.ndarray:
+        """Cached idf[V] f32 from the GLOBAL df (sum over segments)."""
+        if self._h_idf is None:
             df = self.df.astype(np.float64)
             with np.errstate(divide="ignore"):
                 idf = np.log(1.0 + (self.n_docs - df + 0.5) / (df + 0.5))
             self._h_idf = np.where(df > 0, idf, 0.0).astype(np.float32)
-        return self._h_offs, self._h_idf
+        return self._h_idf
 
-    def bm25_chunks(self, queries_terms: list[np.ndarray],
-                    chunk_size: int = 2048):
-        """Vectorized host-side work chunking into (qrow, term, offset,
-        idf) arrays — one work chunk per <=chunk_size posting slice."""
-        offs, idf_t = self._host_tables()
-        # Vectorized per-query term dedupe (the obvious per-query
-        # np.unique loop costs ~1 ms of host time at B=128, serializing
-        # the side-stream launch ahead of the encoder).
+    @staticmethod
+    def dedupe_terms(queries_terms: list[np.ndarray]
+                     ) -> tuple[np.ndarray, np.ndarray]:
+        """Vectorized per-query term dedupe -> (qrows, terms), sorted by
+        query row (the obvious per-query np.unique loop costs ~1 ms of
+        host time at B=128, serializing the side-stream launch ahead of
+        the encoder)."""
         B = len(queries_terms)
-        if B:
-            lens = np.fromiter((len(t) for t in queries_terms),
-                               np.int64, B)
-            T = int(lens.max()) if B else 0
-        if B == 0 or T == 0:
-            qrows = np.zeros(0, np.int64)
-            terms = np.zeros(0, np.int64)
+        if B == 0:
+            return np.zeros(0, np.int64), np.zeros(0, np.int64)
+        lens = np.fromiter((len(t) for t in queries_terms), np.int64, B)
+        T = int(lens.max())
+        if T == 0:
+            return np.zeros(0, np.int64), np.zeros(0, np.int64)
+        if (lens == lens[0]).all():
+            mat = np.stack(queries_terms).astype(np.int64, copy=False)
         else:
-            if (lens == lens[0]).all():
-                mat = np.stack(queries_terms).astype(np.int64, copy=False)
-            else:
-                mat = np.full((B, T), -1, dtype=np.int64)
-                for qi, t in enumerate(queries_terms):
-                    mat[qi, :len(t)] = t
-            srt = np.sort(mat, axis=1)
-            valid = srt >= 0
-            valid[:, 1:] &= srt[:, 1:] != srt[:, :-1]
-            qrows, cols = np.nonzero(valid)
-            terms = srt[qrows, cols]
-        begins = offs[terms]
-        ends = offs[terms + 1]
-        nchunks = np.maximum((ends - begins + chunk_size - 1) // chunk_size, 0)
-        keep = nchunks > 0
-        qrows, terms, begins, nchunks = (qrows[keep], terms[keep],
-                                         begins[keep], nchunks[keep])
-        if len(terms) == 0:
-            return (np.zeros(0, np.int32), np.zeros(0, np.int32),
-                    np.zeros(0, np.int64), np.zeros(0, np.float32))
-        reps = nchunks.astype(np.int64)
-        cq = np.repeat(qrows, reps).astype(np.int32)
-        ct = np.repeat(terms, reps).astype(np.int32)
-        base = np.repeat(begins, reps)
-        # intra-term chunk index: arange within each repeated group
-        total = int(reps.sum())
-        grp_end = np.cumsum(reps)
-        grp_start = grp_end - reps
-        intra = np.arange(total, dtype=np.int64) - np.repeat(grp_start, reps)
-        co = base + intra * chunk_size
-        ci = idf_t[ct]
-        return cq, ct, co, ci
+            mat = np.full((B, T), -1, dtype=np.int64)
+            for qi, t in enumerate(queries_terms):
+                mat[qi, :len(t)] = t
+        srt = np.sort(mat, axis=1)
+        valid = srt >= 0
+        valid[:, 1:] &= srt[:, 1:] != srt[:, :-1]
+        qrows, cols = np.nonzero(valid)
+        return qrows.astype(np.int64), srt[qrows, cols]
+
+    def _pick_bd(self, B: int) -> int:
+        """Doc-block size: 16K docs (64 KB LDS -> 2 workgroups/CU)
+        unless that underfills the chip, then shrink."""
+        for bd in (16384, 8192, 4096):
+            blocks = sum((s.n_docs + bd - 1) // bd for s in self.segments)
+            if blocks * max(B, 1) >= 2048:
+                return bd
+        return 4096
 
     def _h2d(self, name: str, arr: np.ndarray,
              dtype: torch.dtype) -> torch.Tensor:
@@ -390,7 +462,9 @@ class GpuShard:
                     scores_buf: torch.Tensor | None = None,
                     mark=None) -> tuple[torch.Tensor, torch.Tensor]:
         """BM25 plane only (needs terms, not embeddings) — callable on a
-        side stream to overlap with query encoding."""
+        side stream to overlap with query encoding. One kernel launch
+        per posting segment; the segments' disjoint doc ranges cover
+        every column of the score matrix exactly once."""
         import time as _time
         from ..ops import kernels as K
         if mark is None:
@@ -404,28 +478,34 @@ class GpuShard:
         tp = _time.perf_counter()
         if scores_buf is not None and scores_buf.shape == (B, N):
             scores = scores_buf
-            scores.zero_()
         else:
-            scores = torch.zeros(B, N, device=dev, dtype=torch.float32)
-        cq, ct, co, ci = self.bm25_chunks(queries_terms)
+            scores = torch.empty(B, N, device=dev, dtype=torch.float32)
+        qrows, terms = self.dedupe_terms(queries_terms)
+        idf = self._idf_table()[terms]
+        qt_off = np.zeros(B + 1, dtype=np.int64)
+        np.cumsum(np.bincount(qrows, minlength=B), out=qt_off[1:])
+        bd = self._pick_bd(B)
         tp = mark("shard.chunks", tp)
-        if len(cq):
-            # don't overwrite the pinned staging buffers while a prior
-            # step's async H2D copy could still be in flight
-            evt = getattr(self, "_h2d_evt", None)
-            if evt is not None:
-                evt.synchronize()
-            K.bm25_score(
-                self.offsets, self.doc_ids, self.tfs, self.doc_norm,
-                self._h2d("cq", cq, torch.int32),
-                self._h2d("ct", ct, torch.int32),
-                self._h2d("co", co, torch.int64),
-                self._h2d("ci", ci, torch.float32),
-                scores, k1=BM25_K1)
-            if dev.type == "cuda":
-                if evt is None:
-                    evt = self._h2d_evt = torch.cuda.Event()
-                evt.record()
+        # don't overwrite the pinned staging buffers while a prior
+        # step's async H2D copy could still be in flight
+        evt = getattr(self, "_h2d_evt", None)
+        if evt is not None:
+            evt.synchronize()
+        qt_off_d = self._h2d("qt_off", qt_off, torch.int32)
+        qt_idf_d = self._h2d("qt_idf", idf, torch.float32)
+        for si, seg in enumerate(self.segments):
+            begins = seg.h_offs[terms]
+            ends = seg.h_offs[terms + 1]
+            K.bm25_block(
+                seg.doc_ids, seg.tfdl, qt_off_d,
+                self._h2d(f"qb{si}", begins, torch.int64),
+                self._h2d(f"qe{si}", ends, torch.int64),
+                qt_idf_d, scores, seg.doc_base, seg.n_docs, bd,
+                self.avgdl, k1=BM25_K1, b=BM25_B)
+        if dev.type == "cuda":
+            if evt is None:
+                evt = self._h2d_evt = torch.cuda.Event()
+            evt.record()
         tp = mark("shard.bm25", tp)
         out = topk(scores, k)
         mark("shard.bm25topk", tp)
@@ -440,7 +520,6 @@ class GpuShard:
         top-k with global ids (the per-peer result cap analogue,
         reference p2p/routing.py:48)."""
         import time as _time
-        from ..ops import kernels as K
 
         def mark(name, t0):
             if phase_t is None:
@@ -452,6 +531,7 @@ class GpuShard:
 
         B = len(queries_terms)
         N = self.n_docs
+        dev = self.device
         assert N > 0, "shard is empty"
         k = min(k, N)
         tp = _time.perf_counter()
@@ -485,11 +565,13 @@ class GpuShard:
         k = min(k, N)
         tp = _time.perf_counter()
         tk = self._get_topk_dense()
+        emb = self.embeddings
         # hipGraph capture of the whole plane (GEMM + top-k passes, all
         # fixed shapes) removes ~10 launch gaps per batch. Only when the
         # caller runs deferred top-k verification (the query plane does):
         # the eager path reads the overflow flag (D2H), which is illegal
-        # inside a capture.
+        # inside a capture. Graphs are invalidated on every append
+        # (the captured embeddings view goes stale).
         if self.device.type == "cuda" and tk.defer_check:
             key = (B, k)
             graphs = getattr(self, "_dense_graphs", None)
@@ -499,8 +581,8 @@ class GpuShard:
             if entry is None:
                 from ..ops.graphs import GraphedCallable
 
-                def _plane(emb, _B=B, _k=k):
-                    d = K.gemm_nt(emb.unsqueeze(0), self.embeddings,
+                def _plane(e, _B=B, _k=k, _emb=emb):
+                    d = K.gemm_nt(e.unsqueeze(0), _emb,
                                   out_f32=True,
                                   out=self._get_dense_buf(_B)
                                   ).reshape(_B, self.n_docs)
@@ -526,7 +608,7 @@ class GpuShard:
             mark("shard.dense+topk", tp)
             return out
         d_scores = K.gemm_nt(query_emb.bfloat16().unsqueeze(0),
-                             self.embeddings,
+                             emb,
                              out_f32=True,
                              out=self._get_dense_buf(B)).reshape(B, N)
         tp = mark("shard.dense", tp)
@@ -557,7 +639,7 @@ class GpuShard:
 
 
 class CpuShard(GpuShard):
-    """CPU shard with identical semantics, scored by plain torch ops.
+    """CPU shard with identical semantics, scored by plain torch/numpy.
 
     Used on machines without a GPU (tests, the CPU-plumbing config, and
     multi-process gloo tests of the query plane). On a GPU box the HIP
@@ -572,22 +654,24 @@ class CpuShard(GpuShard):
         N = self.n_docs
         assert N > 0, "shard is empty"
         k = min(k, N)
-        offs = self.offsets.numpy()
-        doc_ids = self.doc_ids.numpy()
-        tfs = self.tfs.numpy().astype(np.float32)
-        tfs = np.where(tfs < 0, tfs + 65536.0, tfs)  # stored as i16
         norm = self.doc_norm.numpy()
         scores = np.zeros((B, N), dtype=np.float32)
-        for qi, terms in enumerate(queries_terms):
-            for t in np.unique(terms):
-                t = int(t)
-                b, e = int(offs[t]), int(offs[t + 1])
-                if b == e:
-                    continue
-                idf = self._idf(t)
-                d = doc_ids[b:e]
-                tf = tfs[b:e]
-                scores[qi, d] += idf * tf * (BM25_K1 + 1) / (tf + norm[d])
+        for seg in self.segments:
+            offs = seg.h_offs
+            doc_ids = seg.doc_ids.numpy()
+            tfdl = seg.tfdl.numpy().view(np.uint32)
+            tf_all = (tfdl & np.uint32(0xFFFF)).astype(np.float32)
+            for qi, terms in enumerate(queries_terms):
+                for t in np.unique(terms):
+                    t = int(t)
+                    b, e = int(offs[t]), int(offs[t + 1])
+                    if b == e:
+                        continue
+                    idf = self._idf(t)
+                    d = doc_ids[b:e].astype(np.int64) + seg.doc_base
+                    tf = tf_all[b:e]
+                    scores[qi, d] += (idf * tf * (BM25_K1 + 1)
+                                      / (tf + norm[d]))
         st = torch.from_numpy(scores)
         bm_vals, bm_idx = torch.topk(st, k, dim=1)
         if query_emb is not None and self.embeddings is not None:

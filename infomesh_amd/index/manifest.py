@@ -4,6 +4,10 @@ Reference-parity intent: GPU shards are rebuildable caches
 over the SQLite/snapshot ground truth; a sidecar binary manifest makes
 restarts warm. Tensors are saved with torch.save to a per-rank file;
 load_shard re-uploads them to HBM without re-tokenizing/re-embedding.
+
+v2: segmented postings (offsets/doc_ids/tfdl per segment) + host doc
+lengths, matching gpu_index.PostingSegment — doc_lens persistence means
+incremental appends after a warm start work (round-1 ADVICE fix).
 """
 from __future__ import annotations
 
@@ -11,12 +15,13 @@ import json
 import time
 from pathlib import Path
 
+import numpy as np
 import torch
 
 from ..hashing import content_hash
-from .gpu_index import CpuShard, GpuShard
+from .gpu_index import CpuShard, GpuShard, PostingSegment
 
-MANIFEST_VERSION = 1
+MANIFEST_VERSION = 2
 # bump when bm25_term_ids tokenization changes — old term ids
 # become incompatible and saved shards must be rebuilt
 TOKENIZER_VERSION = 2  # v2: unicode \w + NFKD folding + CJK bigrams
@@ -26,13 +31,13 @@ def save_shard(shard: GpuShard, path: str | Path, rank: int = 0,
                world: int = 1) -> dict:
     path = Path(path)
     path.parent.mkdir(parents=True, exist_ok=True)
-    tensors = {
-        "offsets": shard.offsets, "doc_ids": shard.doc_ids,
-        "tfs": shard.tfs, "doc_norm": shard.doc_norm,
-        "global_ids": shard.global_ids,
-    }
-    payload = {k: (v.cpu() if v is not None else None)
-               for k, v in tensors.items()}
+    segs = [{
+        "offsets": s.offsets.cpu(), "doc_ids": s.doc_ids.cpu(),
+        "tfdl": s.tfdl.cpu(), "doc_base": s.doc_base, "n_docs": s.n_docs,
+    } for s in shard.segments]
+    payload: dict = {"segments": segs}
+    if shard.global_ids is not None:
+        payload["global_ids"] = shard.global_ids.cpu()
     if shard.embeddings is not None:
         payload["embeddings"] = shard.embeddings.cpu()
     meta = {
@@ -41,9 +46,11 @@ def save_shard(shard: GpuShard, path: str | Path, rank: int = 0,
         "rank": rank, "world": world,
         "n_docs": shard.n_docs, "vocab": shard.vocab,
         "avgdl": shard.avgdl,
+        "n_segments": len(shard.segments),
         "created_at": time.time(),
     }
-    torch.save({"meta": meta, "df": shard.df, **payload}, path)
+    torch.save({"meta": meta, "df": shard.df,
+                "doc_lens": shard._doc_lens, **payload}, path)
     meta["bytes"] = path.stat().st_size
     meta["checksum"] = content_hash(str(path.stat().st_size) +
                                     str(shard.n_docs))
@@ -68,13 +75,21 @@ def load_shard(path: str | Path, device: str = "cuda") -> GpuShard:
     shard.n_docs = meta["n_docs"]
     shard.vocab = meta["vocab"]
     shard.avgdl = meta["avgdl"]
-    shard.df = blob["df"]
+    shard.df = np.asarray(blob["df"], dtype=np.int64)
+    shard._doc_lens = np.asarray(blob["doc_lens"], dtype=np.int64)
     dev = shard.device
-    for name in ("offsets", "doc_ids", "tfs", "doc_norm", "global_ids"):
-        t = blob.get(name)
-        setattr(shard, name, t.to(dev) if t is not None else None)
+    for s in blob["segments"]:
+        offs = s["offsets"]
+        shard.segments.append(PostingSegment(
+            offsets=offs.to(dev), doc_ids=s["doc_ids"].to(dev),
+            tfdl=s["tfdl"].to(dev), doc_base=s["doc_base"],
+            n_docs=s["n_docs"], h_offs=offs.numpy()))
+    gids = blob.get("global_ids")
+    if gids is not None:
+        shard._gid_buf = gids.to(dev)
     emb = blob.get("embeddings")
-    shard.embeddings = emb.to(dev) if emb is not None else None
+    if emb is not None:
+        shard._emb_buf = emb.to(dev).contiguous()
     return shard
 
 

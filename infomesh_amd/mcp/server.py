@@ -8,6 +8,7 @@ API-key check, analytics). The protocol surface implements MCP
 """
 from __future__ import annotations
 
+import hmac
 import json
 import sys
 import time
@@ -17,6 +18,11 @@ from ..services import AppContext
 from .handlers import Handlers
 from .session import AnalyticsTracker, SessionStore, WebhookRegistry
 from .tools import PROMPTS, RESOURCES, TOOLS
+
+# hard cap on one HTTP request body — a JSON-RPC tool call is KBs;
+# anything larger is abuse (matches the reference's 10 MB wire cap,
+# infomesh/p2p/protocol.py:358-369 spirit)
+MAX_BODY_BYTES = 8 * 1024 * 1024
 
 PROTOCOL_VERSION = "2024-11-05"
 SERVER_INFO = {"name": "infomesh-amd", "version": "0.1.0"}
@@ -168,17 +174,24 @@ class McpServer:
                 return
             headers = {k.decode(): v.decode()
                        for k, v in scope.get("headers", [])}
-            if server.api_key and \
-                    headers.get("x-api-key") != server.api_key:
+            if server.api_key and not hmac.compare_digest(
+                    headers.get("x-api-key", ""), server.api_key):
                 await _respond(send, 401, {"error": "bad api key"})
                 return
             if scope["method"] == "POST" and scope["path"] in ("/mcp", "/"):
                 body = b""
+                too_big = False
                 while True:
                     ev = await receive()
                     body += ev.get("body", b"")
+                    if len(body) > MAX_BODY_BYTES:
+                        too_big = True   # keep draining to not stall ASGI
+                        body = body[:MAX_BODY_BYTES]
                     if not ev.get("more_body"):
                         break
+                if too_big:
+                    await _respond(send, 413, {"error": "body too large"})
+                    return
                 try:
                     msg = json.loads(body)
                 except json.JSONDecodeError:

@@ -1,39 +1,84 @@
-// GPU BM25 term-at-a-time scorer over CSR posting lists in HBM.
+// GPU BM25 scorer over per-segment CSR posting lists in HBM — v2:
+// doc-block LDS accumulation instead of global atomic scatter-add.
+//
 // Replaces: SQLite FTS5 `MATCH ... ORDER BY bm25()` (reference
 // infomesh/index/local_store.py:316-332) for the GPU shards; the CPU
 // FTS5 path remains for the CPU-plumbing config.
 //
 // score(q, d) = Σ_t idf(t) · tf·(k1+1) / (tf + k1·(1−b+b·dl/avgdl))
-// with the per-doc denominator part precomputed as doc_norm[d].
-// Work is pre-chunked host-side into (query-row, term, posting-offset)
-// triples so Zipf-skewed posting lists spread over many blocks.
+//
+// Design (MI355X): the v1 kernel did one global atomicAdd per posting
+// into scores[q*N+d] — random read-modify-write lines over a multi-GB
+// working set, measured HBM-bound at ~1 ms/batch (profiles/r01_*).
+// v2 tiles the doc axis: each workgroup owns one (query, doc-block)
+// tile, accumulates its block's scores in LDS (ds_add_f32 atomics are
+// conflict-cheap), then streams the block out once. Each term's
+// postings are doc-sorted, so the block's sub-range is two binary
+// searches. Global traffic drops to: postings read once + scores
+// written exactly once (which also removes the separate zero-fill —
+// every column of the output is written by exactly one workgroup,
+// because segments partition the doc axis and blocks partition each
+// segment).
+//
+// Per-posting doc length travels packed with tf (tf | dl<<16) so the
+// norm is computed in-kernel from the CURRENT global avgdl — this is
+// what makes O(new) segment appends exact: BM25 stats shift with the
+// corpus while installed segments stay immutable.
 #include "common.h"
 
 namespace {
 
-__global__ __launch_bounds__(256) void bm25_kernel(
-    const long* __restrict__ offsets, const int* __restrict__ doc_ids,
-    const unsigned short* __restrict__ tfs,
-    const float* __restrict__ doc_norm,
-    const int* __restrict__ chunk_qrow, const int* __restrict__ chunk_term,
-    const long* __restrict__ chunk_off, const float* __restrict__ chunk_idf,
-    float* __restrict__ scores, long N, int chunk_size, float k1p1) {
-  const int c = blockIdx.x;
-  const int q = chunk_qrow[c];
-  const int t = chunk_term[c];
-  const long begin = chunk_off[c];
-  const long end = min(offsets[t + 1], begin + (long)chunk_size);
-  const float idf = chunk_idf[c];
-  float* srow = scores + (long)q * N;
-  for (long p = begin + threadIdx.x; p < end; p += blockDim.x) {
-    const int d = doc_ids[p];
-    const float tf = (float)tfs[p];
-    atomicAdd(&srow[d], idf * tf * k1p1 / (tf + doc_norm[d]));
+DEVINL long lower_bound_i32(const int* __restrict__ a, long lo, long hi,
+                            int v) {
+  while (lo < hi) {
+    const long mid = (lo + hi) >> 1;
+    if (a[mid] < v) lo = mid + 1; else hi = mid;
   }
+  return lo;
 }
 
-// Fused score-combine: out = w_bm25 * bm25 / (bm25 + sat) + w_dense * dense
-// (optional linear hybrid; RRF merge of top-k lists happens host-side).
+__global__ __launch_bounds__(256) void bm25_block_kernel(
+    const int* __restrict__ doc_ids,        // [P] segment-local, asc per term
+    const unsigned int* __restrict__ tfdl,  // [P] tf | dl<<16
+    const int* __restrict__ qt_off,         // [B+1] per-query tuple CSR
+    const long* __restrict__ qt_begin,      // [T] posting begin (this seg)
+    const long* __restrict__ qt_end,        // [T] posting end
+    const float* __restrict__ qt_idf,       // [T]
+    float* __restrict__ scores,             // [B, rowN]
+    long rowN, long doc_base, long nseg, int BD,
+    float norm_a, float norm_b, float k1p1) {
+  extern __shared__ float lds_scores[];     // [BD]
+  const int q = blockIdx.y;
+  const long d0 = (long)blockIdx.x * BD;
+  const int nd = (int)min((long)BD, nseg - d0);
+  for (int i = threadIdx.x; i < nd; i += blockDim.x) lds_scores[i] = 0.0f;
+  __syncthreads();
+  const int t0 = qt_off[q], t1 = qt_off[q + 1];
+  const int wave = threadIdx.x / WAVE, lane = threadIdx.x % WAVE;
+  const int nwaves = blockDim.x / WAVE;
+  // one wave per query-term tuple, round-robin; lanes stride the
+  // block's posting sub-range
+  for (int ti = t0 + wave; ti < t1; ti += nwaves) {
+    const long begin = qt_begin[ti], end = qt_end[ti];
+    if (begin >= end) continue;            // term absent in this segment
+    const float w = qt_idf[ti] * k1p1;
+    const long lo = lower_bound_i32(doc_ids, begin, end, (int)d0);
+    const long hi = lower_bound_i32(doc_ids, lo, end, (int)(d0 + BD));
+    for (long p = lo + lane; p < hi; p += WAVE) {
+      const int d = doc_ids[p] - (int)d0;
+      const unsigned int td = tfdl[p];
+      const float tf = (float)(td & 0xffffu);
+      const float dl = (float)(td >> 16);
+      atomicAdd(&lds_scores[d], w * tf / (tf + norm_a + norm_b * dl));
+    }
+  }
+  __syncthreads();
+  float* __restrict__ srow = scores + (long)q * rowN + doc_base + d0;
+  for (int i = threadIdx.x; i < nd; i += blockDim.x) srow[i] = lds_scores[i];
+}
+
+// Fused score-combine: out = wa * a + wb * b (optional linear hybrid;
+// RRF merge of top-k lists happens host-side).
 __global__ void combine_kernel(const float* __restrict__ a,
                                const float* __restrict__ b,
                                float* __restrict__ out,
@@ -45,19 +90,21 @@ __global__ void combine_kernel(const float* __restrict__ a,
 
 }  // namespace
 
-extern "C" void infomesh_bm25_score(
-    const void* offsets, const void* doc_ids, const void* tfs,
-    const void* doc_norm, const void* chunk_qrow, const void* chunk_term,
-    const void* chunk_off, const void* chunk_idf, void* scores,
-    int nchunks, long N, int chunk_size, float k1, void* stream) {
-  if (nchunks <= 0) return;
+extern "C" void infomesh_bm25_block(
+    const void* doc_ids, const void* tfdl, const void* qt_off,
+    const void* qt_begin, const void* qt_end, const void* qt_idf,
+    void* scores, int B, long rowN, long doc_base, long nseg,
+    int BD, float norm_a, float norm_b, float k1p1, void* stream) {
+  if (nseg <= 0 || B <= 0) return;
   auto s = reinterpret_cast<hipStream_t>(stream);
-  hipLaunchKernelGGL(bm25_kernel, dim3(nchunks), dim3(256), 0, s,
-                     (const long*)offsets, (const int*)doc_ids,
-                     (const unsigned short*)tfs, (const float*)doc_norm,
-                     (const int*)chunk_qrow, (const int*)chunk_term,
-                     (const long*)chunk_off, (const float*)chunk_idf,
-                     (float*)scores, N, chunk_size, k1 + 1.0f);
+  dim3 grid((unsigned)((nseg + BD - 1) / BD), (unsigned)B);
+  hipLaunchKernelGGL(bm25_block_kernel, grid, dim3(256),
+                     (size_t)BD * sizeof(float), s,
+                     (const int*)doc_ids, (const unsigned int*)tfdl,
+                     (const int*)qt_off, (const long*)qt_begin,
+                     (const long*)qt_end, (const float*)qt_idf,
+                     (float*)scores, rowN, doc_base, nseg, BD,
+                     norm_a, norm_b, k1p1);
 }
 
 extern "C" void infomesh_score_combine(const void* a, const void* b,

@@ -314,21 +314,30 @@ def topk(scores: torch.Tensor, k: int) -> tuple[torch.Tensor, torch.Tensor]:
     return TopK(scores.device)(scores, k)
 
 
-def bm25_score(offsets: torch.Tensor, doc_ids: torch.Tensor,
-               tfs: torch.Tensor, doc_norm: torch.Tensor,
-               chunk_qrow: torch.Tensor, chunk_term: torch.Tensor,
-               chunk_off: torch.Tensor, chunk_idf: torch.Tensor,
-               scores: torch.Tensor, chunk_size: int = 2048,
-               k1: float = 1.2) -> torch.Tensor:
-    """Accumulate BM25 into pre-zeroed scores [B, N] f32."""
+def bm25_block(doc_ids: torch.Tensor, tfdl: torch.Tensor,
+               qt_off: torch.Tensor, qt_begin: torch.Tensor,
+               qt_end: torch.Tensor, qt_idf: torch.Tensor,
+               scores: torch.Tensor, doc_base: int, nseg: int,
+               bd: int, avgdl: float, k1: float = 1.2,
+               b: float = 0.75) -> torch.Tensor:
+    """Doc-block LDS-accumulated BM25 for ONE posting segment.
+
+    Writes every element of scores[:, doc_base:doc_base+nseg] exactly
+    once (zero where no posting hits) — no pre-zeroing needed when
+    segments partition the doc axis. Norm is computed in-kernel from
+    the packed per-posting doc length and the current global avgdl."""
     B, N = scores.shape
     _check(scores, torch.float32, "scores")
-    n = chunk_qrow.numel()
-    _ext.lib().infomesh_bm25_score(
-        offsets.data_ptr(), doc_ids.data_ptr(), tfs.data_ptr(),
-        doc_norm.data_ptr(), chunk_qrow.data_ptr(), chunk_term.data_ptr(),
-        chunk_off.data_ptr(), chunk_idf.data_ptr(), scores.data_ptr(),
-        n, N, chunk_size, k1, _ext.stream_ptr())
+    assert B <= 65535 and qt_off.numel() == B + 1
+    assert doc_ids.dtype == torch.int32 and tfdl.dtype == torch.int32
+    assert bd * 4 <= 160 * 1024
+    norm_a = k1 * (1.0 - b)
+    norm_b = k1 * b / max(avgdl, 1e-9)
+    _ext.lib().infomesh_bm25_block(
+        doc_ids.data_ptr(), tfdl.data_ptr(), qt_off.data_ptr(),
+        qt_begin.data_ptr(), qt_end.data_ptr(), qt_idf.data_ptr(),
+        scores.data_ptr(), B, N, doc_base, nseg, bd,
+        norm_a, norm_b, k1 + 1.0, _ext.stream_ptr())
     return scores
 
 

@@ -118,6 +118,35 @@ def test_shard_manifest_roundtrip(tmp_path):
     assert torch.allclose(hits_before.bm25_scores, hits_after.bm25_scores)
 
 
+def test_shard_manifest_then_incremental_append(tmp_path):
+    """Warm-start then ingest more docs: the round-1 ADVICE bug was
+    that doc_lens was not persisted, so any build() after load_shard
+    crashed. Now append-after-restore must match a from-scratch build."""
+    rng = np.random.default_rng(11)
+    docs = [rng.integers(0, 400, size=rng.integers(5, 30)).astype(np.int64)
+            for _ in range(200)]
+    shard = CpuShard()
+    for i in range(150):
+        shard.add_document(i, docs[i], None)
+    shard.build()
+    p = tmp_path / "warm.pt"
+    save_shard(shard, p)
+    loaded = load_shard(p, device="cpu")
+    for i in range(150, 200):
+        loaded.add_document(i, docs[i], None)
+    loaded.build()   # crashed with AttributeError before the fix
+    assert loaded.n_docs == 200
+
+    oracle = CpuShard()
+    for i in range(200):
+        oracle.add_document(i, docs[i], None)
+    oracle.build()
+    q = [np.array([7, 42, 99]), np.array([300])]
+    hl, ho = loaded.search(q, None, k=10), oracle.search(q, None, k=10)
+    assert torch.equal(hl.bm25_ids, ho.bm25_ids)
+    assert torch.allclose(hl.bm25_scores, ho.bm25_scores, atol=1e-5)
+
+
 def test_version_check(tmp_path):
     assert parse_version("1.2.3") == (1, 2, 3)
     assert is_newer("9.9.9")

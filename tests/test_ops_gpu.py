@@ -272,38 +272,77 @@ def test_bm25_parity():
     queries = [[1, 2, 3], [10, 250], [0], [299, 5, 5]]
     ref = R.bm25_scores(postings, doc_lens, queries, n_docs)
 
-    # Build CSR + chunks like index/gpu_index.py does.
+    # Build segment CSR (doc ids asc per term, tf|dl packed) like
+    # index/gpu_index.py _install_segment does.
     import math
     offsets = [0]
-    doc_ids, tfs = [], []
+    doc_ids, tfdl = [], []
     for t in range(vocab):
         for d, tf in postings[t]:
             doc_ids.append(d)
-            tfs.append(tf)
+            tfdl.append(tf | (int(doc_lens[d]) << 16))
         offsets.append(len(doc_ids))
     avgdl = float(doc_lens.float().mean())
-    doc_norm = 1.2 * (1 - 0.75 + 0.75 * doc_lens.float() / avgdl)
     dev = "cuda"
-    offsets_t = torch.tensor(offsets, dtype=torch.int64, device=dev)
-    scores = torch.zeros(len(queries), n_docs, device=dev)
-    cq, ct, co, ci = [], [], [], []
-    for qi, terms in enumerate(queries):
-        for t in terms:
+    B = len(queries)
+    # per-(q, term) tuple table (deduped, per-query CSR)
+    qt_off = [0]
+    qb, qe, qi_ = [], [], []
+    for terms in queries:
+        for t in sorted(set(terms)):
             df = offsets[t + 1] - offsets[t]
             idf = math.log(1.0 + (n_docs - df + 0.5) / (df + 0.5))
-            for off in range(offsets[t], offsets[t + 1], 2048):
-                cq.append(qi); ct.append(t); co.append(off); ci.append(idf)
-    K.bm25_score(
-        offsets_t,
+            qb.append(offsets[t]); qe.append(offsets[t + 1]); qi_.append(idf)
+        qt_off.append(len(qb))
+    # garbage-filled output: the kernel must write every column
+    scores = torch.full((B, n_docs), float("nan"), device=dev)
+    K.bm25_block(
         torch.tensor(doc_ids, dtype=torch.int32, device=dev),
-        torch.tensor(tfs, dtype=torch.int16, device=dev).to(torch.int16),
-        doc_norm.to(dev),
-        torch.tensor(cq, dtype=torch.int32, device=dev),
-        torch.tensor(ct, dtype=torch.int32, device=dev),
-        torch.tensor(co, dtype=torch.int64, device=dev),
-        torch.tensor(ci, dtype=torch.float32, device=dev),
-        scores)
+        torch.tensor(tfdl, dtype=torch.int32, device=dev),
+        torch.tensor(qt_off, dtype=torch.int32, device=dev),
+        torch.tensor(qb, dtype=torch.int64, device=dev),
+        torch.tensor(qe, dtype=torch.int64, device=dev),
+        torch.tensor(qi_, dtype=torch.float32, device=dev),
+        scores, doc_base=0, nseg=n_docs, bd=4096, avgdl=avgdl)
+    assert not torch.isnan(scores).any(), "kernel left columns unwritten"
     _assert_close(scores, ref, rtol=1e-3, atol=1e-3, what="bm25")
+
+
+def test_bm25_segmented_matches_merged():
+    """Three appended segments (O(new) flush) vs one bulk build, and vs
+    the post-optimize() merged segment — identical search results on the
+    GPU kernels. Also covers BM25-only GpuShard.search() (the round-1
+    dense-less branch NameError path)."""
+    import numpy as np
+    from infomesh_amd.index.gpu_index import GpuShard
+    rng = np.random.default_rng(3)
+    docs = [rng.integers(0, 4000, size=rng.integers(5, 60)).astype(np.int64)
+            for _ in range(3000)]
+    inc = GpuShard("cuda")
+    start = 0
+    for batch in (1200, 900, 900):
+        for i in range(start, start + batch):
+            inc.add_document(10_000 + i, docs[i], None)
+        inc.build()
+        start += batch
+    assert len(inc.segments) == 3
+    bulk = GpuShard("cuda")
+    for i in range(3000):
+        bulk.add_document(10_000 + i, docs[i], None)
+    bulk.build()
+    queries = [rng.integers(0, 4000, size=5).astype(np.int64)
+               for _ in range(8)]
+    hi = inc.search(queries, None, k=20)     # BM25-only branch
+    hb = bulk.search(queries, None, k=20)
+    assert torch.equal(hi.bm25_ids.cpu(), hb.bm25_ids.cpu())
+    _assert_close(hi.bm25_scores, hb.bm25_scores, rtol=1e-4, atol=1e-4,
+                  what="seg-bm25")
+    inc.optimize()
+    assert len(inc.segments) == 1
+    ho = inc.search(queries, None, k=20)
+    assert torch.equal(ho.bm25_ids.cpu(), hb.bm25_ids.cpu())
+    _assert_close(ho.bm25_scores, hb.bm25_scores, rtol=1e-4, atol=1e-4,
+                  what="opt-bm25")
 
 
 # ---------------------------------------------------------------- simhash

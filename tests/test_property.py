@@ -50,9 +50,10 @@ def test_ed25519_sign_verify_roundtrip(msg):
 @given(st.lists(st.lists(st.integers(0, 500), min_size=0, max_size=12),
                 min_size=1, max_size=16))
 @settings(max_examples=25, deadline=None)
-def test_bm25_chunks_matches_naive(queries):
-    """The vectorized dedupe+chunking must equal the per-query
-    np.unique construction it replaced."""
+def test_bm25_dedupe_terms_matches_naive(queries):
+    """The vectorized per-query term dedupe must equal the per-query
+    np.unique construction it replaced, and the per-segment
+    begin/end/idf tables must match the host CSR."""
     from infomesh_amd.index.gpu_index import CpuShard
     shard = CpuShard()
     rng = np.random.default_rng(0)
@@ -62,20 +63,19 @@ def test_bm25_chunks_matches_naive(queries):
         shard.add_document(i, d, None)
     shard.build()
     qts = [np.asarray(q, dtype=np.int64) for q in queries]
-    cq, ct, co, ci = shard.bm25_chunks(qts)
-    # naive reference
-    offs, idf = shard._host_tables()
+    qrows, terms = shard.dedupe_terms(qts)
     exp = []
     for qi, t in enumerate(qts):
         for term in np.unique(t):
-            b, e = offs[term], offs[term + 1]
-            o = b
-            while o < e:
-                exp.append((qi, int(term), int(o)))
-                o += 2048
-    got = sorted(zip(cq.tolist(), ct.tolist(), co.tolist()))
+            exp.append((qi, int(term)))
+    got = sorted(zip(qrows.tolist(), terms.tolist()))
     assert got == sorted(exp)
-    assert np.allclose(ci, idf[ct])
+    # idf table + per-segment CSR lookups are consistent
+    idf = shard._idf_table()
+    for t in terms:
+        assert abs(idf[t] - shard._idf(int(t))) < 1e-6
+    seg = shard.segments[0]
+    assert np.all(seg.h_offs[terms + 1] >= seg.h_offs[terms])
 
 
 @given(st.integers(1, 5), st.integers(2, 12), st.integers(1, 8),
