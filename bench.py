@@ -34,8 +34,8 @@ import torch
 def parse_args():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
-    p.add_argument("--steps", type=int, default=30)
-    p.add_argument("--warmup", type=int, default=5)
+    p.add_argument("--steps", type=int, default=200)
+    p.add_argument("--warmup", type=int, default=20)
     p.add_argument("--docs-per-gpu", type=int, default=1_250_000)
     p.add_argument("--batch", type=int, default=128)
     p.add_argument("--k-per-shard", type=int, default=100)
@@ -264,6 +264,8 @@ def main():
                 "p50_query_ms": round(p50, 3),
                 "p95_batch_ms": round(
                     sorted(lat_ms)[int(len(lat_ms) * 0.95) - 1], 3),
+                "p99_batch_ms": round(
+                    sorted(lat_ms)[max(int(len(lat_ms) * 0.99) - 1, 0)], 3),
                 "setup_s": round(setup_s, 1),
             },
         }
