@@ -295,9 +295,8 @@ class GpuShard:
                 "dense shard requires embeddings for every pending doc"
             self._emb_buf = self._grow(self._emb_buf, need,
                                        (e.shape[1],), torch.bfloat16)
-            if e.device != self.device:
-                e = (self._upload_t(e, stream) if stream is not None
-                     else e.clone())
+            if e.device.type == "cpu" and self.device.type == "cuda":
+                e = self._upload_t(e, stream)
             if stream is not None:
                 with torch.cuda.stream(stream):
                     self._emb_buf[self.n_docs:need].copy_(e)

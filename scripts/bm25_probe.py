@@ -9,10 +9,14 @@ Usage (GPU box): python scripts/bm25_probe.py [--docs N] [--batch B]
 from __future__ import annotations
 
 import argparse
+import sys
 import time
+from pathlib import Path
 
 import numpy as np
 import torch
+
+sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
 
 
 def main() -> None:
