@@ -277,6 +277,10 @@ class GpuShard:
         np.cumsum(df_new, out=offsets[1:])
         on_gpu = self.device.type == "cuda"
         stream = torch.cuda.Stream(self.device) if on_gpu else None
+        if stream is not None:
+            # new_emb may have been produced on the compute stream
+            # (encoder output): order the side-stream copies after it
+            stream.wait_stream(torch.cuda.current_stream(self.device))
         seg = PostingSegment(
             offsets=self._upload(offsets, stream),
             doc_ids=self._upload(docs_u.astype(np.int32), stream),
@@ -437,6 +441,15 @@ class GpuShard:
                 return bd
         return 4096
 
+    def _bounds_ws(self, n: int) -> torch.Tensor:
+        """Device i32 workspace for the bounds pre-pass (reused across
+        segments/batches; launches are stream-ordered so one suffices)."""
+        ws = getattr(self, "_bounds_buf", None)
+        if ws is None or ws.numel() < n:
+            ws = self._bounds_buf = torch.empty(
+                max(n, 1 << 16), device=self.device, dtype=torch.int32)
+        return ws
+
     def _h2d(self, name: str, arr: np.ndarray,
              dtype: torch.dtype) -> torch.Tensor:
         """Stage a small host array through a persistent pinned buffer —
@@ -480,10 +493,14 @@ class GpuShard:
         else:
             scores = torch.empty(B, N, device=dev, dtype=torch.float32)
         qrows, terms = self.dedupe_terms(queries_terms)
+        # queries share Zipf-common terms: dedupe ACROSS queries so the
+        # bounds pre-pass searches each term's postings once per block
+        uterms, qt_ut = np.unique(terms, return_inverse=True)
         idf = self._idf_table()[terms]
         qt_off = np.zeros(B + 1, dtype=np.int64)
         np.cumsum(np.bincount(qrows, minlength=B), out=qt_off[1:])
         bd = self._pick_bd(B)
+        U = len(uterms)
         tp = mark("shard.chunks", tp)
         # don't overwrite the pinned staging buffers while a prior
         # step's async H2D copy could still be in flight
@@ -491,15 +508,16 @@ class GpuShard:
         if evt is not None:
             evt.synchronize()
         qt_off_d = self._h2d("qt_off", qt_off, torch.int32)
+        qt_ut_d = self._h2d("qt_ut", qt_ut, torch.int32)
         qt_idf_d = self._h2d("qt_idf", idf, torch.float32)
         for si, seg in enumerate(self.segments):
-            begins = seg.h_offs[terms]
-            ends = seg.h_offs[terms + 1]
+            nblocks = (seg.n_docs + bd - 1) // bd
+            bounds = self._bounds_ws(U * nblocks * 2)
             K.bm25_block(
-                seg.doc_ids, seg.tfdl, qt_off_d,
-                self._h2d(f"qb{si}", begins, torch.int64),
-                self._h2d(f"qe{si}", ends, torch.int64),
-                qt_idf_d, scores, seg.doc_base, seg.n_docs, bd,
+                seg.doc_ids, seg.tfdl, qt_off_d, qt_ut_d, qt_idf_d,
+                self._h2d(f"qb{si}", seg.h_offs[uterms], torch.int64),
+                self._h2d(f"qe{si}", seg.h_offs[uterms + 1], torch.int64),
+                bounds, scores, seg.doc_base, seg.n_docs, bd,
                 self.avgdl, k1=BM25_K1, b=BM25_B)
         if dev.type == "cuda":
             if evt is None:

@@ -37,19 +37,48 @@ DEVINL long lower_bound_i32(const int* __restrict__ a, long lo, long hi,
   return lo;
 }
 
+// Pre-pass: for every (unique-term, doc-block) pair, binary-search the
+// posting sub-range once into a bounds table (i32 offsets relative to
+// the term's posting begin). The main kernel then has ZERO serial
+// binary searches — per (q,term) they were ~34 dependent HBM loads,
+// poorly hidden at 2 workgroups/CU; here 1 thread per pair with tens
+// of thousands in flight hides them completely, and queries sharing a
+// term (Zipf-common) reuse the same entry.
+__global__ __launch_bounds__(256) void bm25_bounds_kernel(
+    const int* __restrict__ doc_ids,
+    const long* __restrict__ u_begin,   // [U] per unique term
+    const long* __restrict__ u_end,
+    int* __restrict__ bounds,           // [U * nblocks * 2]
+    int U, int nblocks, int BD) {
+  const int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= U * nblocks) return;
+  const int ut = i / nblocks, blk = i % nblocks;
+  const long b = u_begin[ut], e = u_end[ut];
+  const long d0 = (long)blk * BD;
+  const long lo = lower_bound_i32(doc_ids, b, e, (int)d0);
+  const long hi = lower_bound_i32(doc_ids, lo, e, (int)(d0 + BD));
+  bounds[2 * i] = (int)(lo - b);
+  bounds[2 * i + 1] = (int)(hi - b);
+}
+
 __global__ __launch_bounds__(256) void bm25_block_kernel(
     const int* __restrict__ doc_ids,        // [P] segment-local, asc per term
     const unsigned int* __restrict__ tfdl,  // [P] tf | dl<<16
     const int* __restrict__ qt_off,         // [B+1] per-query tuple CSR
-    const long* __restrict__ qt_begin,      // [T] posting begin (this seg)
-    const long* __restrict__ qt_end,        // [T] posting end
+    const int* __restrict__ qt_ut,          // [T] unique-term index
     const float* __restrict__ qt_idf,       // [T]
+    const long* __restrict__ u_begin,       // [U]
+    const int* __restrict__ bounds,         // [U * nblocks * 2]
     float* __restrict__ scores,             // [B, rowN]
-    long rowN, long doc_base, long nseg, int BD,
+    long rowN, long doc_base, long nseg, int BD, int nblocks,
     float norm_a, float norm_b, float k1p1) {
   extern __shared__ float lds_scores[];     // [BD]
-  const int q = blockIdx.y;
-  const long d0 = (long)blockIdx.x * BD;
+  // grid: x = query (fast), y = doc-block — adjacent workgroups are
+  // the SAME posting sub-range for different queries, so the XCD's L2
+  // serves the repeat reads instead of HBM
+  const int q = blockIdx.x;
+  const int blk = blockIdx.y;
+  const long d0 = (long)blk * BD;
   const int nd = (int)min((long)BD, nseg - d0);
   for (int i = threadIdx.x; i < nd; i += blockDim.x) lds_scores[i] = 0.0f;
   __syncthreads();
@@ -59,11 +88,11 @@ __global__ __launch_bounds__(256) void bm25_block_kernel(
   // one wave per query-term tuple, round-robin; lanes stride the
   // block's posting sub-range
   for (int ti = t0 + wave; ti < t1; ti += nwaves) {
-    const long begin = qt_begin[ti], end = qt_end[ti];
-    if (begin >= end) continue;            // term absent in this segment
+    const int ut = qt_ut[ti];
+    const int* bp = bounds + 2 * ((long)ut * nblocks + blk);
+    const long base = u_begin[ut];
+    const long lo = base + bp[0], hi = base + bp[1];
     const float w = qt_idf[ti] * k1p1;
-    const long lo = lower_bound_i32(doc_ids, begin, end, (int)d0);
-    const long hi = lower_bound_i32(doc_ids, lo, end, (int)(d0 + BD));
     for (long p = lo + lane; p < hi; p += WAVE) {
       const int d = doc_ids[p] - (int)d0;
       const unsigned int td = tfdl[p];
@@ -73,8 +102,12 @@ __global__ __launch_bounds__(256) void bm25_block_kernel(
     }
   }
   __syncthreads();
+  // scores are written once and only re-read by the top-k streaming
+  // pass — nontemporal keeps them out of L2, which the posting reads
+  // (shared across adjacent same-block workgroups) actually want
   float* __restrict__ srow = scores + (long)q * rowN + doc_base + d0;
-  for (int i = threadIdx.x; i < nd; i += blockDim.x) srow[i] = lds_scores[i];
+  for (int i = threadIdx.x; i < nd; i += blockDim.x)
+    __builtin_nontemporal_store(lds_scores[i], srow + i);
 }
 
 // Fused score-combine: out = wa * a + wb * b (optional linear hybrid;
@@ -92,18 +125,28 @@ __global__ void combine_kernel(const float* __restrict__ a,
 
 extern "C" void infomesh_bm25_block(
     const void* doc_ids, const void* tfdl, const void* qt_off,
-    const void* qt_begin, const void* qt_end, const void* qt_idf,
-    void* scores, int B, long rowN, long doc_base, long nseg,
+    const void* qt_ut, const void* qt_idf, const void* u_begin,
+    const void* u_end, void* bounds, void* scores,
+    int B, int U, long rowN, long doc_base, long nseg,
     int BD, float norm_a, float norm_b, float k1p1, void* stream) {
   if (nseg <= 0 || B <= 0) return;
   auto s = reinterpret_cast<hipStream_t>(stream);
-  dim3 grid((unsigned)((nseg + BD - 1) / BD), (unsigned)B);
+  const int nblocks = (int)((nseg + BD - 1) / BD);
+  if (U > 0) {
+    const int total = U * nblocks;
+    hipLaunchKernelGGL(bm25_bounds_kernel,
+                       dim3((unsigned)((total + 255) / 256)), dim3(256), 0, s,
+                       (const int*)doc_ids, (const long*)u_begin,
+                       (const long*)u_end, (int*)bounds, U, nblocks, BD);
+  }
+  dim3 grid((unsigned)B, (unsigned)nblocks);
   hipLaunchKernelGGL(bm25_block_kernel, grid, dim3(256),
                      (size_t)BD * sizeof(float), s,
                      (const int*)doc_ids, (const unsigned int*)tfdl,
-                     (const int*)qt_off, (const long*)qt_begin,
-                     (const long*)qt_end, (const float*)qt_idf,
-                     (float*)scores, rowN, doc_base, nseg, BD,
+                     (const int*)qt_off, (const int*)qt_ut,
+                     (const float*)qt_idf, (const long*)u_begin,
+                     (const int*)bounds, (float*)scores,
+                     rowN, doc_base, nseg, BD, nblocks,
                      norm_a, norm_b, k1p1);
 }
 

@@ -315,28 +315,40 @@ def topk(scores: torch.Tensor, k: int) -> tuple[torch.Tensor, torch.Tensor]:
 
 
 def bm25_block(doc_ids: torch.Tensor, tfdl: torch.Tensor,
-               qt_off: torch.Tensor, qt_begin: torch.Tensor,
-               qt_end: torch.Tensor, qt_idf: torch.Tensor,
+               qt_off: torch.Tensor, qt_ut: torch.Tensor,
+               qt_idf: torch.Tensor, u_begin: torch.Tensor,
+               u_end: torch.Tensor, bounds: torch.Tensor,
                scores: torch.Tensor, doc_base: int, nseg: int,
                bd: int, avgdl: float, k1: float = 1.2,
                b: float = 0.75) -> torch.Tensor:
     """Doc-block LDS-accumulated BM25 for ONE posting segment.
 
-    Writes every element of scores[:, doc_base:doc_base+nseg] exactly
-    once (zero where no posting hits) — no pre-zeroing needed when
-    segments partition the doc axis. Norm is computed in-kernel from
-    the packed per-posting doc length and the current global avgdl."""
+    Two launches: a bounds pre-pass binary-searching each (unique term,
+    doc-block) posting sub-range once, then the block kernel (one
+    workgroup per (query, doc-block), query-major so adjacent
+    workgroups reuse posting reads through L2). Writes every element of
+    scores[:, doc_base:doc_base+nseg] exactly once (zero where no
+    posting hits) — no pre-zeroing needed when segments partition the
+    doc axis. Norm is computed in-kernel from the packed per-posting
+    doc length and the current global avgdl.
+
+    bounds: caller-provided i32 workspace of at least
+    U * ceil(nseg/bd) * 2 elements."""
     B, N = scores.shape
+    U = u_begin.numel()
+    nblocks = (nseg + bd - 1) // bd
     _check(scores, torch.float32, "scores")
-    assert B <= 65535 and qt_off.numel() == B + 1
+    assert nblocks <= 65535 and qt_off.numel() == B + 1
     assert doc_ids.dtype == torch.int32 and tfdl.dtype == torch.int32
+    assert bounds.numel() >= U * nblocks * 2
     assert bd * 4 <= 160 * 1024
     norm_a = k1 * (1.0 - b)
     norm_b = k1 * b / max(avgdl, 1e-9)
     _ext.lib().infomesh_bm25_block(
         doc_ids.data_ptr(), tfdl.data_ptr(), qt_off.data_ptr(),
-        qt_begin.data_ptr(), qt_end.data_ptr(), qt_idf.data_ptr(),
-        scores.data_ptr(), B, N, doc_base, nseg, bd,
+        qt_ut.data_ptr(), qt_idf.data_ptr(), u_begin.data_ptr(),
+        u_end.data_ptr(), bounds.data_ptr(), scores.data_ptr(),
+        B, U, N, doc_base, nseg, bd,
         norm_a, norm_b, k1 + 1.0, _ext.stream_ptr())
     return scores
 

@@ -27,6 +27,8 @@ def main() -> None:
     ap.add_argument("--k", type=int, default=100)
     ap.add_argument("--segments", type=int, default=1,
                     help="split the corpus into this many segments")
+    ap.add_argument("--bd", type=int, default=0,
+                    help="override the doc-block size")
     args = ap.parse_args()
 
     from infomesh_amd.ops import _build
@@ -74,30 +76,36 @@ def main() -> None:
     # kernel-only (no topk): time score writes via events
     from infomesh_amd.ops import kernels as K
     qrows, tset = shard.dedupe_terms(qterms)
+    uterms, qt_ut = np.unique(tset, return_inverse=True)
     idf = shard._idf_table()[tset]
     qt_off = np.zeros(B + 1, dtype=np.int64)
     np.cumsum(np.bincount(qrows, minlength=B), out=qt_off[1:])
     qt_off_d = torch.from_numpy(qt_off).to(torch.int32).to(dev)
+    qt_ut_d = torch.from_numpy(qt_ut.astype(np.int32)).to(dev)
     qt_idf_d = torch.from_numpy(idf).to(dev)
+    bd = args.bd or shard._pick_bd(B)
     segs = []
     for seg in shard.segments:
-        qb = torch.from_numpy(seg.h_offs[tset]).to(dev)
-        qe = torch.from_numpy(seg.h_offs[tset + 1]).to(dev)
-        segs.append((seg, qb, qe))
-    bd = shard._pick_bd(B)
+        qb = torch.from_numpy(seg.h_offs[uterms]).to(dev)
+        qe = torch.from_numpy(seg.h_offs[uterms + 1]).to(dev)
+        nblocks = (seg.n_docs + bd - 1) // bd
+        bw = torch.empty(len(uterms) * nblocks * 2, dtype=torch.int32,
+                         device=dev)
+        segs.append((seg, qb, qe, bw))
     ev0, ev1 = torch.cuda.Event(True), torch.cuda.Event(True)
     ev0.record()
     for _ in range(args.iters):
-        for seg, qb, qe in segs:
-            K.bm25_block(seg.doc_ids, seg.tfdl, qt_off_d, qb, qe, qt_idf_d,
-                         scores, seg.doc_base, seg.n_docs, bd, shard.avgdl)
+        for seg, qb, qe, bw in segs:
+            K.bm25_block(seg.doc_ids, seg.tfdl, qt_off_d, qt_ut_d,
+                         qt_idf_d, qb, qe, bw, scores,
+                         seg.doc_base, seg.n_docs, bd, shard.avgdl)
     ev1.record()
     torch.cuda.synchronize()
     kern_ms = ev0.elapsed_time(ev1) / args.iters
 
     npost = int(qt_idf_d.numel() and sum(
         int((s.h_offs[tset + 1] - s.h_offs[tset]).sum())
-        for s, _, _ in segs))
+        for s, _, _, _ in segs))
     traffic_gb = (npost * 8 + B * N * 4) / 1e9
     print(f"B={B} N={N} bd={bd} postings/batch={npost}")
     print(f"bm25 kernel: {kern_ms * 1e3:.0f} us/batch  "

@@ -285,25 +285,38 @@ def test_bm25_parity():
     avgdl = float(doc_lens.float().mean())
     dev = "cuda"
     B = len(queries)
-    # per-(q, term) tuple table (deduped, per-query CSR)
+    # per-(q, term) tuple table (deduped per query, unique across
+    # queries for the bounds pre-pass)
+    import numpy as np2
+    qrows, tlist = [], []
+    for qi, terms in enumerate(queries):
+        for t in sorted(set(terms)):
+            qrows.append(qi); tlist.append(t)
+    uterms, qt_ut = np2.unique(np2.array(tlist), return_inverse=True)
     qt_off = [0]
-    qb, qe, qi_ = [], [], []
-    for terms in queries:
+    qi_ = []
+    for qi, terms in enumerate(queries):
         for t in sorted(set(terms)):
             df = offsets[t + 1] - offsets[t]
-            idf = math.log(1.0 + (n_docs - df + 0.5) / (df + 0.5))
-            qb.append(offsets[t]); qe.append(offsets[t + 1]); qi_.append(idf)
-        qt_off.append(len(qb))
+            qi_.append(math.log(1.0 + (n_docs - df + 0.5) / (df + 0.5)))
+        qt_off.append(len(qi_))
+    bd = 4096
+    nblocks = (n_docs + bd - 1) // bd
+    bounds = torch.empty(len(uterms) * nblocks * 2, dtype=torch.int32,
+                         device=dev)
     # garbage-filled output: the kernel must write every column
     scores = torch.full((B, n_docs), float("nan"), device=dev)
     K.bm25_block(
         torch.tensor(doc_ids, dtype=torch.int32, device=dev),
         torch.tensor(tfdl, dtype=torch.int32, device=dev),
         torch.tensor(qt_off, dtype=torch.int32, device=dev),
-        torch.tensor(qb, dtype=torch.int64, device=dev),
-        torch.tensor(qe, dtype=torch.int64, device=dev),
+        torch.from_numpy(qt_ut.astype(np2.int32)).to(dev),
         torch.tensor(qi_, dtype=torch.float32, device=dev),
-        scores, doc_base=0, nseg=n_docs, bd=4096, avgdl=avgdl)
+        torch.from_numpy(np2.array([offsets[t] for t in uterms],
+                                   dtype=np2.int64)).to(dev),
+        torch.from_numpy(np2.array([offsets[t + 1] for t in uterms],
+                                   dtype=np2.int64)).to(dev),
+        bounds, scores, doc_base=0, nseg=n_docs, bd=bd, avgdl=avgdl)
     assert not torch.isnan(scores).any(), "kernel left columns unwritten"
     _assert_close(scores, ref, rtol=1e-3, atol=1e-3, what="bm25")
 
