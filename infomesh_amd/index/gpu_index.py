@@ -433,9 +433,10 @@ class GpuShard:
         return qrows.astype(np.int64), srt[qrows, cols]
 
     def _pick_bd(self, B: int) -> int:
-        """Doc-block size: 16K docs (64 KB LDS -> 2 workgroups/CU)
-        unless that underfills the chip, then shrink."""
-        for bd in (16384, 8192, 4096):
+        """Doc-block size: 8K docs (32 KB LDS -> 5 workgroups/CU;
+        measured 405 us vs 471/713 us for 16K/32K at 1.25M docs B=128,
+        scripts/bm25_probe.py) unless that underfills the chip."""
+        for bd in (8192, 4096):
             blocks = sum((s.n_docs + bd - 1) // bd for s in self.segments)
             if blocks * max(B, 1) >= 2048:
                 return bd

@@ -270,7 +270,11 @@ def test_bm25_parity():
             doc_lens[d] += tf
         postings[t] = plist
     queries = [[1, 2, 3], [10, 250], [0], [299, 5, 5]]
-    ref = R.bm25_scores(postings, doc_lens, queries, n_docs)
+    # engine semantics: query terms are DEDUPED per query (qtf dropped,
+    # matching round-1 gpu_index dedupe + FTS5 MATCH behavior), so the
+    # oracle gets the deduped lists
+    ref = R.bm25_scores(postings, doc_lens,
+                        [sorted(set(q)) for q in queries], n_docs)
 
     # Build segment CSR (doc ids asc per term, tf|dl packed) like
     # index/gpu_index.py _install_segment does.
