@@ -71,6 +71,9 @@ class SearchConfig:
     rerank_top_n: int = 100
     rerank_keep: int = 10
     hybrid: bool = True
+    # dynamic micro-batching between entry points and the GPU plane
+    batch_max: int = 128               # matches the benched batch size
+    batch_wait_ms: float = 1.5         # max collect latency per request
 
 
 @dataclass(frozen=True)

@@ -121,28 +121,51 @@ class HybridEngine:
     def search(self, query: str, limit: int = 10,
                use_dense: bool | None = None) -> list[SearchHit]:
         """Single-query search against the engine (shard-fused)."""
-        if self.shard.n_docs == 0:
-            return []
-        terms = [bm25_term_ids(query)]
+        return self.search_many([query], limit=limit,
+                                use_dense=use_dense)[0]
+
+    def search_many(self, queries: list[str], limit: int = 10,
+                    use_dense: bool | None = None) -> list[list[SearchHit]]:
+        """Batched search: ONE collective plane.search_batch for the
+        whole list (the batcher's execute path). Fusion happens exactly
+        once, on the plane (RRF of the per-shard BM25 + dense top-k);
+        callers hydrate url/title from the LocalStore."""
+        if self.shard.n_docs == 0 or not queries:
+            return [[] for _ in queries]
+        B = len(queries)
+        # pad the batch to a power-of-two bucket: the plane's hipGraph
+        # cache, scores buffers and the encoder graphs are per-(B, k)
+        # shape — unbounded serving batch sizes would thrash them
+        Bp = 1
+        while Bp < B:
+            Bp *= 2
+        terms = [bm25_term_ids(q) for q in queries]
+        terms += [np.zeros(0, dtype=np.int64)] * (Bp - B)
         emb = None
         if use_dense is None:
             use_dense = self.encoder is not None
         if use_dense and self.encoder is not None:
-            emb = self.encoder.encode_texts([query])
+            texts = queries + [""] * (Bp - B)
+            emb = self.encoder.encode_texts(texts)
         fused = self.plane.search_batch(
-            terms, emb, B=1, dim=emb.shape[1] if emb is not None else 384,
+            terms, emb, B=Bp, dim=emb.shape[1] if emb is not None else 384,
             n_results=limit, use_dense=use_dense and emb is not None)
         if fused is None:
-            return []
-        hits: list[SearchHit] = []
-        for gid, score in zip(fused.ids[0].tolist(),
-                              fused.scores[0].tolist()):
-            if gid < 0:
-                continue
-            hits.append(SearchHit(doc_id=int(gid), url="", title="",
-                                  snippet="", bm25=0.0, score=float(score),
-                                  source="gpu-hybrid"))
-        return hits
+            return [[] for _ in queries]
+        ids = fused.ids.tolist()
+        scores = fused.scores.tolist()
+        out: list[list[SearchHit]] = []
+        for qi in range(B):
+            hits: list[SearchHit] = []
+            for gid, score in zip(ids[qi], scores[qi]):
+                if gid < 0:
+                    continue
+                hits.append(SearchHit(doc_id=int(gid), url="", title="",
+                                      snippet="", bm25=0.0,
+                                      score=float(score),
+                                      source="gpu-hybrid"))
+            out.append(hits)
+        return out
 
     def stats(self) -> dict:
         return {

@@ -197,7 +197,10 @@ class McpServer:
                 except json.JSONDecodeError:
                     await _respond(send, 400, {"error": "bad json"})
                     return
-                resp = server.handle_message(msg)
+                # thread-dispatch so concurrent requests overlap and
+                # the query batcher can group them into one GPU batch
+                import asyncio
+                resp = await asyncio.to_thread(server.handle_message, msg)
                 await _respond(send, 200, resp or {})
             elif scope["method"] == "GET" and scope["path"] == "/health":
                 await _respond(send, 200, {"ok": True})

@@ -25,10 +25,12 @@ from .engine import HybridEngine
 from .errors import InfoMeshError
 from .index.link_graph import LinkGraph
 from .index.local_store import Document, LocalStore
+from .search.batcher import QueryBatcher
 from .search.cache import QueryCache
-from .search.nlp import RelatedSearchTracker
-from .search.query import (SearchResponse, search_hybrid,
-                           search_local)
+from .search.nlp import RelatedSearchTracker, parse_query_filters
+from .search.passage import highlight, select_best_passage
+from .search.query import (SearchResponse, preprocess_query,
+                           search_hybrid, search_local)
 from .trust.attestation import create_attestation
 from .trust.dmca import TakedownManager
 from .trust.gdpr import DeletionManager
@@ -59,6 +61,7 @@ class AppContext:
     related: RelatedSearchTracker
     worker: CrawlWorker | None = None
     engine: HybridEngine | None = None
+    batcher: QueryBatcher | None = None
     attestations: list = field(default_factory=list)
     farming: FarmingDetector = field(default_factory=FarmingDetector)
     started_at: float = field(default_factory=time.time)
@@ -122,6 +125,12 @@ class AppContext:
                 ctx.engine = HybridEngine(k_per_shard=cfg.search.max_results_per_shard)
             except Exception as e:
                 log.warning("engine unavailable: %s", e)
+        if ctx.engine is not None:
+            # all entry points share one micro-batching queue so
+            # concurrent requests reach the GPU plane as one batch
+            ctx.batcher = QueryBatcher(
+                ctx.engine, max_batch=cfg.search.batch_max,
+                max_wait_ms=cfg.search.batch_wait_ms)
         return ctx
 
     # ------------------------------------------------------------ ingest
@@ -216,16 +225,27 @@ class AppContext:
         trust_fn = self.trust.trust_fn()
         engine_ready = (self.engine is not None
                         and self.engine.shard.n_docs > 0)
+        # Metadata filters (site:/language/date) live in the SQLite
+        # plane only — filtered queries take the FTS path even when the
+        # engine is up (reference local_store.py:253-352 filter SQL).
+        pq = parse_query_filters(query)
+        has_filters = bool(pq.site or pq.after or pq.before or pq.language)
         if mode == "local" or (mode == "auto" and not engine_ready):
             resp = search_local(self.store, query, limit=limit,
                                 authority_fn=authority, trust_fn=trust_fn)
         elif mode in ("auto", "hybrid", "distributed"):
-            resp = search_hybrid(
-                self.store, _EngineDense(self, limit), query, limit=limit,
-                authority_fn=authority, trust_fn=trust_fn,
-                rrf_k=self.config.search.rrf_k)
-            if mode == "distributed":
-                resp.mode = "distributed"
+            if engine_ready and not has_filters:
+                # single-fusion path: the GPU plane RRF-fuses BM25 +
+                # dense exactly once; results are hydrated from the
+                # LocalStore (round-1 double-fusion fix)
+                resp = self._engine_search(query, limit, mode)
+            else:
+                resp = search_hybrid(
+                    self.store, None, query, limit=limit,
+                    authority_fn=authority, trust_fn=trust_fn,
+                    rrf_k=self.config.search.rrf_k)
+                if mode == "distributed":
+                    resp.mode = "distributed"
         else:
             raise InfoMeshError("SRCH001", f"unknown mode {mode!r}")
         resp = GLOBAL_PLUGINS.run("post_search", resp, query=query)
@@ -233,6 +253,36 @@ class AppContext:
             self.cache.put(key, resp)
         self.ledger.record_action(Action.QUERY_SERVED, 1.0)
         return resp
+
+    def _engine_search(self, query: str, limit: int,
+                       mode: str) -> SearchResponse:
+        """GPU-plane search via the micro-batcher, hydrated from the
+        LocalStore (url/title/domain/snippet — engine hits carry only
+        global doc ids + fused scores)."""
+        t0 = time.time()
+        eff = preprocess_query(query)
+        if self.batcher is not None:
+            if self.batcher.engine is not self.engine:
+                self.batcher.engine = self.engine  # engine was rebound
+            hits = self.batcher.submit(eff, limit)
+        else:
+            hits = self.engine.search(eff, limit=limit)
+        hydrated = []
+        for h in hits:
+            doc = self.store.get_document(h.doc_id)
+            if doc is None:
+                continue
+            h.url, h.title = doc.url, doc.title
+            h.domain, h.crawled_at = doc.domain, doc.crawled_at
+            if doc.text:
+                h.snippet = highlight(
+                    select_best_passage(eff, doc.text), eff)
+            hydrated.append(h)
+        return SearchResponse(
+            query=query, effective_query=eff, results=hydrated,
+            elapsed_ms=(time.time() - t0) * 1e3,
+            mode="distributed" if mode == "distributed" else "hybrid",
+            total_candidates=len(hits))
 
     def fetch_page(self, url: str) -> Document | None:
         """Cache-first page fetch (reference: services.py:220-335);
@@ -254,6 +304,7 @@ class AppContext:
             "engine": self.engine.stats() if self.engine else None,
             "credits": self.ledger.stats(),
             "cache": self.cache.stats(),
+            "batcher": self.batcher.stats() if self.batcher else None,
             "crawler": self.worker.stats if self.worker else None,
             "link_edges": self.link_graph.edge_count(),
         }
@@ -263,6 +314,8 @@ class AppContext:
             self.feeds.save()
         except Exception:
             pass
+        if self.batcher is not None:
+            self.batcher.close()
         for c in (self.store, self.link_graph, self.dedup, self.ledger,
                   self.trust, self.takedowns, self.deletions):
             try:
@@ -271,27 +324,3 @@ class AppContext:
                 pass
 
 
-class _EngineDense:
-    """Adapter: HybridEngine as the DenseSearcher for search_hybrid,
-    resolving engine global ids back to LocalStore documents."""
-
-    def __init__(self, ctx: AppContext, limit: int):
-        self.ctx = ctx
-        self.limit = limit
-
-    def search(self, query: str, limit: int = 10):
-        engine = self.ctx.engine
-        if engine is None or engine.shard.n_docs == 0:
-            return []
-        hits = engine.search(query, limit=limit)
-        out = []
-        for h in hits:
-            doc = self.ctx.store.get_document(h.doc_id)
-            if doc is None:
-                continue
-            h.url = doc.url
-            h.title = doc.title
-            h.domain = doc.domain
-            h.crawled_at = doc.crawled_at
-            out.append(h)
-        return out
