@@ -289,6 +289,22 @@ class LocalStore:
             "SELECT * FROM documents WHERE id=?", (doc_id,)).fetchone()
         return self._row_to_doc(r) if r else None
 
+    def get_documents(self, ids: list[int]) -> dict[int, "Document"]:
+        """Batched lookup (one IN query) — the serving batcher hydrates
+        a whole GPU batch's results in one round trip."""
+        if not ids:
+            return {}
+        out: dict[int, Document] = {}
+        uniq = list({int(i) for i in ids})
+        for i in range(0, len(uniq), 500):   # SQLite var limit safety
+            chunk = uniq[i:i + 500]
+            q = ("SELECT * FROM documents WHERE id IN (%s)"
+                 % ",".join("?" * len(chunk)))
+            for r in self.conn.execute(q, chunk).fetchall():
+                d = self._row_to_doc(r)
+                out[d.doc_id] = d
+        return out
+
     def get_document_by_url(self, url: str) -> Document | None:
         r = self.conn.execute(
             "SELECT * FROM documents WHERE url=?", (url,)).fetchone()

@@ -199,14 +199,30 @@ class McpServer:
                     return
                 # thread-dispatch so concurrent requests overlap and
                 # the query batcher can group them into one GPU batch
+                # (dedicated wide pool: asyncio's default executor is
+                # cpu_count+4 workers, which caps in-flight requests
+                # far below the GPU batch size)
                 import asyncio
-                resp = await asyncio.to_thread(server.handle_message, msg)
+                resp = await asyncio.get_running_loop().run_in_executor(
+                    _handler_pool(), server.handle_message, msg)
                 await _respond(send, 200, resp or {})
             elif scope["method"] == "GET" and scope["path"] == "/health":
                 await _respond(send, 200, {"ok": True})
             else:
                 await _respond(send, 404, {"error": "not found"})
         return app
+
+
+_POOL = None
+
+
+def _handler_pool():
+    global _POOL
+    if _POOL is None:
+        from concurrent.futures import ThreadPoolExecutor
+        _POOL = ThreadPoolExecutor(max_workers=256,
+                                   thread_name_prefix="mcp-handler")
+    return _POOL
 
 
 async def _respond(send, status: int, payload: dict) -> None:

@@ -88,6 +88,28 @@ def select_best_passage(query: str, text: str,
     return out
 
 
+def fast_snippet(query: str, text: str, width: int = 200) -> str:
+    """Cheap snippet for the batched serving path: a window around the
+    first query-term occurrence, terms highlighted. The FTS5 snippet()
+    analogue (reference local_store.py:253-352 uses SQL snippet());
+    full passage scoring (select_best_passage) stays on the unbatched
+    local path where per-query CPU time is not the bottleneck."""
+    terms = sorted({t for t in _WORD_RE.findall(query.lower())
+                    if len(t) > 1}, key=len, reverse=True)
+    if not terms or not text:
+        return highlight(text[:width], query)
+    pattern = re.compile(
+        r"\b(" + "|".join(re.escape(t) for t in terms) + r")\b", re.I)
+    m = pattern.search(text)
+    if m is None:
+        return text[:width]
+    lo = max(0, m.start() - width // 3)
+    window = text[lo:lo + width]
+    if lo > 0:
+        window = "…" + window.lstrip()
+    return pattern.sub(lambda mm: f"<b>{mm.group(0)}</b>", window)
+
+
 def highlight(text: str, query: str, tag: str = "b") -> str:
     """Wrap query terms in <b>…</b> (reference: passage.py:233)."""
     terms = sorted({t for t in _WORD_RE.findall(query.lower()) if len(t) > 1},

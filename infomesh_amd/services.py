@@ -28,7 +28,7 @@ from .index.local_store import Document, LocalStore
 from .search.batcher import QueryBatcher
 from .search.cache import QueryCache
 from .search.nlp import RelatedSearchTracker, parse_query_filters
-from .search.passage import highlight, select_best_passage
+from .search.passage import fast_snippet
 from .search.query import (SearchResponse, preprocess_query,
                            search_hybrid, search_local)
 from .trust.attestation import create_attestation
@@ -127,10 +127,13 @@ class AppContext:
                 log.warning("engine unavailable: %s", e)
         if ctx.engine is not None:
             # all entry points share one micro-batching queue so
-            # concurrent requests reach the GPU plane as one batch
+            # concurrent requests reach the GPU plane as one batch;
+            # hydration happens batch-wide inside the executor (one
+            # SQLite IN query per batch, fast snippets)
             ctx.batcher = QueryBatcher(
                 ctx.engine, max_batch=cfg.search.batch_max,
-                max_wait_ms=cfg.search.batch_wait_ms)
+                max_wait_ms=cfg.search.batch_wait_ms,
+                execute=ctx._batch_execute)
         return ctx
 
     # ------------------------------------------------------------ ingest
@@ -254,32 +257,55 @@ class AppContext:
         self.ledger.record_action(Action.QUERY_SERVED, 1.0)
         return resp
 
+    def ensure_batcher(self) -> QueryBatcher | None:
+        """Create the micro-batcher if an engine was attached after
+        create() (tests/benches assign ctx.engine directly)."""
+        if self.batcher is None and self.engine is not None:
+            self.batcher = QueryBatcher(
+                self.engine, max_batch=self.config.search.batch_max,
+                max_wait_ms=self.config.search.batch_wait_ms,
+                execute=self._batch_execute)
+        return self.batcher
+
+    def _batch_execute(self, queries: list[str],
+                       limit: int) -> list[list]:
+        """Batcher executor: ONE plane call + ONE hydration round trip
+        for the whole batch (engine hits carry only global doc ids +
+        fused scores; url/title/domain/snippet come from LocalStore)."""
+        if self.batcher is not None \
+                and self.batcher.engine is not self.engine:
+            self.batcher.engine = self.engine   # engine was rebound
+        per_q = self.engine.search_many(queries, limit=limit)
+        gids = [h.doc_id for hits in per_q for h in hits]
+        docs = self.store.get_documents(gids)
+        out: list[list] = []
+        for q, hits in zip(queries, per_q):
+            hydrated = []
+            for h in hits:
+                doc = docs.get(h.doc_id)
+                if doc is None:
+                    continue
+                h.url, h.title = doc.url, doc.title
+                h.domain, h.crawled_at = doc.domain, doc.crawled_at
+                if doc.text:
+                    h.snippet = fast_snippet(q, doc.text)
+                hydrated.append(h)
+            out.append(hydrated)
+        return out
+
     def _engine_search(self, query: str, limit: int,
                        mode: str) -> SearchResponse:
-        """GPU-plane search via the micro-batcher, hydrated from the
-        LocalStore (url/title/domain/snippet — engine hits carry only
-        global doc ids + fused scores)."""
+        """GPU-plane search via the micro-batcher (fused once on the
+        plane, hydrated batch-wide in _batch_execute)."""
         t0 = time.time()
         eff = preprocess_query(query)
-        if self.batcher is not None:
-            if self.batcher.engine is not self.engine:
-                self.batcher.engine = self.engine  # engine was rebound
-            hits = self.batcher.submit(eff, limit)
+        batcher = self.ensure_batcher()
+        if batcher is not None:
+            hits = batcher.submit(eff, limit)
         else:
-            hits = self.engine.search(eff, limit=limit)
-        hydrated = []
-        for h in hits:
-            doc = self.store.get_document(h.doc_id)
-            if doc is None:
-                continue
-            h.url, h.title = doc.url, doc.title
-            h.domain, h.crawled_at = doc.domain, doc.crawled_at
-            if doc.text:
-                h.snippet = highlight(
-                    select_best_passage(eff, doc.text), eff)
-            hydrated.append(h)
+            hits = self._batch_execute([eff], limit)[0]
         return SearchResponse(
-            query=query, effective_query=eff, results=hydrated,
+            query=query, effective_query=eff, results=hits,
             elapsed_ms=(time.time() - t0) * 1e3,
             mode="distributed" if mode == "distributed" else "hybrid",
             total_candidates=len(hits))
