@@ -334,7 +334,7 @@ class GpuShard:
         self._h_idf = None
         # dense hipGraphs captured the old embeddings view/shape
         self._dense_graphs = {}
-        self._dense_scores_buf = None
+        self._dense_bufs = {}
 
     def optimize(self) -> None:
         """Merge all posting segments into one, entirely on-device:
@@ -641,14 +641,21 @@ class GpuShard:
 
     def _get_dense_buf(self, B: int) -> torch.Tensor:
         """Persistent dense scores buffer, separate from the BM25 one
-        (which may still be feeding its top-k on another stream). Reuse
-        avoids a multi-GB alloc/free per batch at large N."""
+        (which may still be feeding its top-k on another stream). Keyed
+        per batch size: captured hipGraphs keep writing the buffer they
+        recorded, so a shared buffer reallocated for a different B
+        would leave older graphs writing freed memory (the serving
+        batcher sends pow2-bucketed batch sizes, so this cache is
+        bounded). Reuse avoids a multi-GB alloc/free per batch."""
         N = self.n_docs
-        if (getattr(self, "_dense_scores_buf", None) is None
-                or self._dense_scores_buf.shape != (1, B, N)):
-            self._dense_scores_buf = torch.empty(
+        bufs = getattr(self, "_dense_bufs", None)
+        if bufs is None:
+            bufs = self._dense_bufs = {}
+        buf = bufs.get(B)
+        if buf is None or buf.shape != (1, B, N):
+            buf = bufs[B] = torch.empty(
                 1, B, N, device=self.device, dtype=torch.float32)
-        return self._dense_scores_buf
+        return buf
 
     def to_global(self, idx: torch.Tensor) -> torch.Tensor:
         safe = idx.clamp(min=0).long()

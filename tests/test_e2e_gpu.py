@@ -98,3 +98,59 @@ def test_gpu_incremental_ingest(gpu_ctx):
                           use_cache=False, deduct=False)
     assert any("new.example" in getattr(r, "url", "")
                for r in resp.results)
+
+
+def test_gpu_variable_batch_serving():
+    """Interleaved batch sizes through the dense hipGraph path — the
+    serving batcher sends pow2-bucketed B values; per-B graph/buffer
+    caches must never let an older graph write a freed buffer (round-2
+    regression: GPU memory fault in serve_probe)."""
+    import numpy as np
+    from infomesh_amd.index.synth import build_synth_shard, synth_queries
+    from infomesh_amd.parallel.query_plane import DistributedQueryPlane
+
+    shard = build_synth_shard(60_000, avg_len=60, device="cuda", seed=3)
+    plane = DistributedQueryPlane(shard, k_per_shard=50)
+    terms_all, emb_all = synth_queries(128, n_terms=4, seed=9,
+                                       device="cuda")
+    # oracle at B=128
+    fused128 = plane.search_batch(terms_all, emb_all, B=128,
+                                  n_results=10)
+    for B in (1, 64, 128, 16, 2, 128, 32, 1, 128):
+        fused = plane.search_batch(terms_all[:B], emb_all[:B], B=B,
+                                   n_results=10)
+        assert fused.ids.shape[0] == B
+        assert torch.equal(fused.ids[:B], fused128.ids[:B]), \
+            f"batch-size {B} results diverge"
+    torch.cuda.synchronize()
+
+
+def test_gpu_batched_services_path():
+    """AppContext.search through the real batcher + hydration on GPU."""
+    import threading
+    ctx = AppContext.create(config=Config(), with_engine=False,
+                            with_worker=False, in_memory=True)
+    ctx.engine = HybridEngine(device="cuda", use_encoder=True)
+    for url, title, text in CORPUS:
+        ctx.index_document(Document(url=url, title=title, text=text),
+                           attest=False, credit=False)
+    ctx.flush_engine()
+    assert ctx.ensure_batcher() is not None
+    out = {}
+
+    def client(i, q):
+        out[i] = ctx.search(q, use_cache=False, deduct=False)
+
+    qs = [q for q, _ in QUERIES] * 3
+    threads = [threading.Thread(target=client, args=(i, q))
+               for i, q in enumerate(qs)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(30)
+    assert len(out) == len(qs)
+    for i, q in enumerate(qs):
+        assert out[i].results, f"no results for {q!r}"
+        assert out[i].results[0].url and out[i].results[0].title
+    assert ctx.batcher.stats()["queries"] >= len(qs)
+    ctx.close()
