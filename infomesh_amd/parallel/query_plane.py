@@ -34,6 +34,7 @@ class FusedHits:
     bm25_scores: torch.Tensor
     dense_ids: torch.Tensor
     dense_scores: torch.Tensor
+    degraded: bool = False   # served by a shrunk rank group
 
 
 def rrf_fuse(ids_lists: list[torch.Tensor], score_lists: list[torch.Tensor],
@@ -94,6 +95,7 @@ class DistributedQueryPlane:
         self.shard = shard
         self.fabric = fabric or Fabric()
         self.k = k_per_shard
+        self.fault_stale_s = 5.0
         self._scores_buf: torch.Tensor | None = None
         self._bm25_stream = None
         # RRF fusion is ~30 tiny fixed-shape kernels (~0.5 ms of launch
@@ -104,35 +106,97 @@ class DistributedQueryPlane:
     def world_size(self) -> int:
         return self.fabric.world
 
-    def _pack_queries(self, queries_terms: list[np.ndarray] | None,
-                      query_emb: torch.Tensor | None, B: int, dim: int):
-        """Rank 0 packs -> broadcast -> every rank unpacks."""
-        dev = self.fabric.device
-        terms_t = torch.full((B, MAX_QUERY_TERMS), -1, dtype=torch.int64)
-        if queries_terms is not None:
-            for i, t in enumerate(queries_terms):
-                t = t[:MAX_QUERY_TERMS]
-                terms_t[i, :len(t)] = torch.from_numpy(t.astype(np.int64))
-        terms_t = terms_t.to(dev)
-        emb_t = (query_emb.to(dev).float() if query_emb is not None
-                 else torch.zeros(B, dim, device=dev))
-        self.fabric.broadcast(terms_t)
-        self.fabric.broadcast(emb_t)
-        out_terms = []
-        tt = terms_t.cpu().numpy()
-        for i in range(B):
-            row = tt[i]
-            out_terms.append(row[row >= 0])
-        return out_terms, emb_t
-
     def search_batch(self, queries_terms: list[np.ndarray] | None,
                      query_emb: torch.Tensor | None,
                      B: int, dim: int = 384, n_results: int = 10,
                      use_dense: bool = True,
                      phase_t: dict | None = None,
-                     encode_fn=None) -> FusedHits | None:
-        """Collective search. Rank 0 passes real queries and gets the
-        FusedHits; other ranks pass None and get None.
+                     encode_fn=None,
+                     encode_shard=None) -> FusedHits | None:
+        """Collective search with rank-fault tolerance: on a collective
+        failure (dead rank → timeout) the fabric shrinks to the
+        pre-created exclusion subgroup and the batch retries once,
+        flagged degraded (reference parity: local-only serving when
+        peers are absent, infomesh/search/query.py:471-490)."""
+        try:
+            return self._search_batch(queries_terms, query_emb, B, dim,
+                                      n_results, use_dense, phase_t,
+                                      encode_fn, encode_shard)
+        except RuntimeError as e:
+            if not self.fabric.initialized or not self.fabric.collective_ok:
+                raise
+            import logging
+            import time as _t
+            log = logging.getLogger("infomesh.plane")
+            # heartbeat staleness disambiguates a dead rank from a
+            # plain kernel error; survivors may hit their collective
+            # failures at different times, so give the heartbeat a
+            # staleness window before deciding
+            dead = self.fabric.dead_ranks(stale_s=self.fault_stale_s)
+            if not dead:
+                _t.sleep(self.fault_stale_s)
+                dead = self.fabric.dead_ranks(stale_s=self.fault_stale_s)
+            if not dead:
+                raise   # not a rank failure (e.g. a kernel error)
+            log.warning("collective failed (%s); dead ranks %s — "
+                        "shrinking group", type(e).__name__, dead)
+            self.fabric.degrade(dead)
+            # resync through the store (collective-free: a timed-out op
+            # closes its gloo pairs / poisons its NCCL comm, so the
+            # fresh subgroup must stay untouched until every survivor
+            # has drained its own failure)
+            if self.fabric.collective_ok and not self.fabric.resync():
+                more = self.fabric.dead_ranks(stale_s=self.fault_stale_s)
+                log.warning("resync failed (dead %s) — serving "
+                            "local-only", more)
+                self.fabric.degrade(more or list(
+                    r for r in self.fabric.active_ranks
+                    if r != self.fabric.rank))
+            return self._search_batch(queries_terms, query_emb, B, dim,
+                                      n_results, use_dense, phase_t,
+                                      encode_fn, encode_shard)
+
+    def _sharded_encode(self, encode_shard: dict, B: int,
+                        dim: int) -> torch.Tensor:
+        """Per-rank query-encode sharding (BACKLOG 8-GPU win): rank 0
+        broadcasts the token-id batch; every ACTIVE rank encodes an
+        equal slice with its own (identically-seeded) encoder; slices
+        all-gather back into the full [B, dim] embedding matrix on
+        every rank — which also replaces the embedding broadcast.
+        Encoder wall time scales 1/W at the cost of one [B,S] i32
+        broadcast + one [B/W, dim] all-gather (tens of KB over xGMI)."""
+        fa = self.fabric
+        dev = fa.device
+        S = int(encode_shard["S"])
+        fn = encode_shard["fn"]
+        W = fa.effective_world
+        Bs = (B + W - 1) // W
+        Bp = Bs * W
+        ids_t = torch.zeros(Bp, S, dtype=torch.int32, device=dev)
+        lens_t = torch.ones(Bp, dtype=torch.int32, device=dev)
+        if fa.rank == 0:
+            qids = encode_shard["qids"]
+            qlens = encode_shard["qlens"]
+            ids_t[:B] = qids.to(dev)
+            lens_t[:B] = qlens.to(dev)
+        fa.broadcast(ids_t)
+        fa.broadcast(lens_t)
+        my = fa.active_ranks.index(fa.rank)
+        sl = slice(my * Bs, (my + 1) * Bs)
+        emb_slice = fn(ids_t[sl].contiguous(),
+                       lens_t[sl].contiguous())
+        gath = fa.all_gather(emb_slice.contiguous())
+        return gath.reshape(Bp, dim)[:B]
+
+    def _search_batch(self, queries_terms: list[np.ndarray] | None,
+                      query_emb: torch.Tensor | None,
+                      B: int, dim: int = 384, n_results: int = 10,
+                      use_dense: bool = True,
+                      phase_t: dict | None = None,
+                      encode_fn=None, encode_shard=None
+                      ) -> FusedHits | None:
+        """One collective search attempt. Rank 0 passes real queries and
+        gets the FusedHits; other ranks pass None and get None.
 
         With encode_fn (rank 0), the BM25 plane is launched on a side
         stream BEFORE the query encoding runs, overlapping the encoder's
@@ -154,13 +218,21 @@ class DistributedQueryPlane:
         if torch.cuda.is_available() and self.shard.device.type == "cuda":
             hits = self._search_overlapped(queries_terms, query_emb,
                                            encode_fn, B, dim, use_dense,
-                                           phase_t=phase_t)
+                                           phase_t=phase_t,
+                                           encode_shard=encode_shard)
             tp = mark("plane.shard", tp)
         else:
-            if encode_fn is not None:
-                query_emb = encode_fn()
-            terms, emb = self._pack_queries(queries_terms, query_emb, B,
-                                            dim)
+            terms = self._broadcast_terms(queries_terms, B)
+            if encode_shard is not None:
+                emb = self._sharded_encode(encode_shard, B, dim).float()
+            else:
+                if encode_fn is not None and self.fabric.rank == 0:
+                    query_emb = encode_fn()
+                dev = self.fabric.device
+                emb = (query_emb.to(dev).float()
+                       if query_emb is not None
+                       else torch.zeros(B, dim, device=dev))
+                self.fabric.broadcast(emb)
             tp = mark("plane.pack", tp)
             hits = self.shard.search(
                 terms, emb if use_dense else None, k=self.k,
@@ -207,10 +279,12 @@ class DistributedQueryPlane:
         mark("plane.fuse", tp)
         return FusedHits(ids=ids, scores=scores, bm25_ids=bm_i,
                          bm25_scores=bm_s, dense_ids=dn_i,
-                         dense_scores=dn_s)
+                         dense_scores=dn_s,
+                         degraded=self.fabric.degraded)
 
     def _search_overlapped(self, queries_terms, query_emb, encode_fn,
-                           B, dim, use_dense, phase_t=None) -> ShardHits:
+                           B, dim, use_dense, phase_t=None,
+                           encode_shard=None) -> ShardHits:
         """BM25 on a side stream || query encoding on the main stream.
 
         Broadcast order (terms, then embeddings) is identical on all
@@ -243,13 +317,16 @@ class DistributedQueryPlane:
                 terms, self.k, scores_buf=self._get_scores_buf(B))
             bm_ids = self.shard.to_global(bm_idx)
         tp = mark("ov.bm25", tp)
-        if self.fabric.rank == 0:
-            emb = encode_fn() if encode_fn is not None else query_emb
+        if encode_shard is not None:
+            emb_t = self._sharded_encode(encode_shard, B, dim).float()
         else:
-            emb = None
-        emb_t = (emb.to(dev).float() if emb is not None
-                 else torch.zeros(B, dim, device=dev))
-        self.fabric.broadcast(emb_t)
+            if self.fabric.rank == 0:
+                emb = encode_fn() if encode_fn is not None else query_emb
+            else:
+                emb = None
+            emb_t = (emb.to(dev).float() if emb is not None
+                     else torch.zeros(B, dim, device=dev))
+            self.fabric.broadcast(emb_t)
         tp = mark("ov.encode", tp)
         if use_dense and self.shard.embeddings is not None:
             dn_vals, dn_idx = self.shard.search_dense(emb_t, self.k)

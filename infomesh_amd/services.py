@@ -308,7 +308,9 @@ class AppContext:
             query=query, effective_query=eff, results=hits,
             elapsed_ms=(time.time() - t0) * 1e3,
             mode="distributed" if mode == "distributed" else "hybrid",
-            total_candidates=len(hits))
+            total_candidates=len(hits),
+            degraded=bool(getattr(self.engine.plane.fabric,
+                                  "degraded", False)))
 
     def fetch_page(self, url: str) -> Document | None:
         """Cache-first page fetch (reference: services.py:220-335);

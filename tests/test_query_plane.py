@@ -7,6 +7,7 @@ backed MockInfoMeshDHT) with CpuShard + a world-1 Fabric, plus a real
 from __future__ import annotations
 
 import os
+import time
 
 import numpy as np
 import torch
@@ -287,3 +288,136 @@ def test_plane_gloo_world4_matches_single_shard(tmp_path):
     valid = ids[ids >= 0]
     assert len({int(i) % _W4 for i in valid}) >= 3, \
         "fused results should draw from (nearly) all shards"
+
+
+# ----------------------------------- rank-fault degraded serving (gloo)
+
+_W3 = 3
+
+
+def _fault_worker(rank: int, port: int, out_file: str):
+    """3 ranks; rank 2 crashes after batch 1. Survivors detect via
+    heartbeat staleness after the collective timeout, shrink to the
+    pre-created exclusion subgroup, and batch 2 serves degraded from
+    shards 0+1 (VERDICT #4: reference query.py:471-490 behavior)."""
+    os.environ.update({
+        "RANK": str(rank), "WORLD_SIZE": str(_W3),
+        "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+    })
+    fabric = Fabric(backend="gloo", timeout_s=6, heartbeat_s=0.3)
+    docs, emb = _corpus_arrays(_N_DOCS4, seed=13)
+    mine = list(range(rank, _N_DOCS4, _W3))
+    shard = _build_from_docs([docs[i] for i in mine], emb[mine], mine)
+    plane = DistributedQueryPlane(shard, fabric, k_per_shard=20)
+    plane.fault_stale_s = 2.0
+    qterms = [np.array([7, 13, 40]), np.array([99, 100])]
+    g = torch.Generator().manual_seed(23)
+    qemb = torch.nn.functional.normalize(torch.randn(2, 16, generator=g),
+                                         dim=-1)
+
+    def run_batch():
+        return plane.search_batch(qterms if rank == 0 else None,
+                                  qemb if rank == 0 else None,
+                                  B=2, dim=16, n_results=12)
+
+    f1 = run_batch()
+    if rank == 0:
+        assert f1 is not None and not f1.degraded
+    if rank == 2:
+        os._exit(0)   # crash-stop: no destroy, heartbeats cease
+    time.sleep(3.0)   # let rank 2's heartbeat go stale
+    f2 = run_batch()  # collective times out -> degrade -> retry
+    assert fabric.degraded and fabric.active_ranks == [0, 1]
+    f3 = run_batch()  # subsequent batches stay on the shrunk group
+    if rank == 0:
+        assert f2 is not None and f2.degraded and f3.degraded
+        valid = f2.ids[f2.ids >= 0]
+        # results now come only from shards 0 and 1 (gid mod 3)
+        assert ((valid % _W3) != 2).all()
+        assert (valid % _W3 == 0).any() and (valid % _W3 == 1).any()
+        torch.save({"ok": True}, out_file)
+    fabric.destroy()
+
+
+def test_plane_rank_drop_degraded(tmp_path):
+    import torch.multiprocessing as mp
+    out_file = str(tmp_path / "ok")
+    port = 29531
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_fault_worker, args=(r, port, out_file))
+             for r in range(_W3)]
+    for p in procs:
+        p.start()
+    deadline = time.time() + 150
+    for i, p in enumerate(procs):
+        p.join(timeout=max(1.0, deadline - time.time()))
+        if i != 2:
+            assert p.exitcode == 0, f"rank {i} exit {p.exitcode}"
+    assert os.path.exists(out_file)
+
+
+# ------------------------------------- sharded encode (gloo, world 2)
+
+def _fake_encode(ids: torch.Tensor, lens: torch.Tensor) -> torch.Tensor:
+    """Deterministic stand-in encoder: every rank computes the same
+    embedding for the same token rows (like the seed-identical BERT)."""
+    freqs = torch.arange(1, 9, dtype=torch.float32) * 0.13
+    x = torch.sin(ids.float().unsqueeze(2) * freqs).sum(dim=1)
+    return torch.nn.functional.normalize(x, dim=-1)
+
+
+def _enc_shard_worker(rank: int, port: int, out_file: str):
+    os.environ.update({
+        "RANK": str(rank), "WORLD_SIZE": "2",
+        "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+    })
+    fabric = Fabric(backend="gloo")
+    docs, emb = _corpus_arrays(200, seed=31)
+    # embeddings must live in the fake encoder's 8-dim space: re-embed
+    # docs with the same fake encoder over their first terms
+    doc_ids = torch.zeros(200, 6, dtype=torch.int32)
+    for i, d in enumerate(docs):
+        t = torch.from_numpy(d[:6].astype(np.int64))
+        doc_ids[i, :len(t)] = t.to(torch.int32)
+    demb = _fake_encode(doc_ids, None).bfloat16()
+    mine = list(range(rank, 200, 2))
+    shard = _build_from_docs([docs[i] for i in mine], demb[mine], mine)
+    plane = DistributedQueryPlane(shard, fabric, k_per_shard=15)
+    B, S = 3, 6
+    g = torch.Generator().manual_seed(41)
+    qids = torch.randint(0, 800, (B, S), generator=g, dtype=torch.int32)
+    qlens = torch.full((B,), S, dtype=torch.int32)
+    qterms = [qids[i].numpy().astype(np.int64) for i in range(B)]
+    shard_arg = {"S": S, "fn": _fake_encode,
+                 "qids": qids if rank == 0 else None,
+                 "qlens": qlens if rank == 0 else None}
+    fused = plane.search_batch(qterms if rank == 0 else None, None,
+                               B=B, dim=8, n_results=10,
+                               encode_shard=shard_arg)
+    # oracle: same plane, embeddings computed on rank 0 + broadcast
+    qemb = _fake_encode(qids, qlens)
+    fused2 = plane.search_batch(qterms if rank == 0 else None,
+                                qemb if rank == 0 else None,
+                                B=B, dim=8, n_results=10)
+    if rank == 0:
+        assert torch.equal(fused.ids, fused2.ids), \
+            "sharded encode diverges from rank-0 encode + broadcast"
+        assert torch.allclose(fused.scores, fused2.scores, atol=1e-5)
+        with open(out_file, "w") as f:
+            f.write("ok")
+    fabric.destroy()
+
+
+def test_plane_sharded_encode_matches_broadcast(tmp_path):
+    import torch.multiprocessing as mp
+    out_file = str(tmp_path / "ok")
+    port = 29541
+    ctx = mp.get_context("spawn")
+    procs = [ctx.Process(target=_enc_shard_worker,
+                         args=(r, port, out_file)) for r in range(2)]
+    for p in procs:
+        p.start()
+    for p in procs:
+        p.join(timeout=120)
+        assert p.exitcode == 0
+    assert os.path.exists(out_file)
