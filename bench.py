@@ -97,9 +97,13 @@ def main():
                                   k_per_shard=args.k_per_shard)
 
     encoder = reranker = summarizer = None
-    if rank == 0 and not args.bm25_only and use_gpu:
+    if not args.bm25_only and use_gpu:
+        # EVERY rank builds the (seed-identical) encoder: multi-GPU
+        # runs shard the query encode across ranks (rank 0 broadcasts
+        # token ids, each rank encodes B/W rows, slices all-gather)
         from infomesh_amd.models.encoder import EmbeddingEncoder
-        encoder = EmbeddingEncoder(device=device)
+        if rank == 0 or world > 1:
+            encoder = EmbeddingEncoder(device=device)
     if rank == 0 and args.rerank and use_gpu:
         from infomesh_amd.models.reranker import CrossEncoderReranker
         reranker = CrossEncoderReranker(device=device)
@@ -176,8 +180,15 @@ def main():
         j = i % n_batches
         emb = None
         encode_fn = None
+        encode_shard = None
         tp = time.perf_counter()
-        if rank == 0 and encoder is not None:
+        if encoder is not None and world > 1:
+            # shard the query encode across ranks (plane broadcasts the
+            # token ids and all-gathers the embedding slices)
+            encode_shard = {"S": args.query_len, "fn": encoder.encode_ids,
+                            "qids": qids_all[j] if rank == 0 else None,
+                            "qlens": qlens_all[j] if rank == 0 else None}
+        elif rank == 0 and encoder is not None:
             # passed as a callback: the query plane overlaps the BM25
             # side-stream work with the encoder forward
             def encode_fn():
@@ -191,7 +202,8 @@ def main():
                 args.n_results,
                 args.rerank_candidates if args.rerank else 0),
             use_dense=not args.bm25_only, phase_t=phase_t
-            if args.phase_timers else None, encode_fn=encode_fn)
+            if args.phase_timers else None, encode_fn=encode_fn,
+            encode_shard=encode_shard)
         tp = _mark("search+fuse", tp)
         if rank == 0 and reranker is not None:
             run_rerank(fused)
