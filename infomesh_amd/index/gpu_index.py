@@ -618,7 +618,7 @@ class GpuShard:
                 # the deferred overflow check by hand (same static
                 # workspace/layout as at capture; need=0 -> exact).
                 cnt_off = B * 512 + 3 * B
-                flag_off = B * 512 + 4 * B
+                flag_off = B * 512 + 5 * B
                 tk._pending.append((ws, cnt_off, flag_off, B, 0))
             # detach from the graph's static outputs (next replay
             # overwrites them)

@@ -9,10 +9,11 @@
 //   2. select1: find byte bin containing the Kth value.
 //   3. hist2: 256-bin histogram of byte 2 within that bin.
 //   4. select2: 16-bit threshold prefix.
-//   5. compact: gather (value, idx) with prefix >= threshold (cap 8192).
+//   5. compact: gather (value, idx) — strictly-above candidates into a
+//      reserved region (provably < K of them), threshold-prefix-equal
+//      candidates until the cap (excess ties dropped: sound, they are
+//      interchangeable at rank K).
 //   6. sort: one block per row bitonic-sorts candidates, emits top-K.
-// Ties sharing a 16-bit prefix can overflow the candidate cap on
-// pathological distributions: the overflow flag is checked by the wrapper.
 #include "common.h"
 
 #define TOPK_CAP 8192
@@ -119,15 +120,34 @@ __global__ void select2_kernel(const unsigned* __restrict__ hist2,
   }
 }
 
+// Candidate layout per row (TOPK_CAP slots):
+//   [0, HI_RES)        strictly-above-threshold candidates (exact
+//                      select guarantees < K <= HI_RES of them)
+//   [HI_RES, TOPK_CAP) threshold-PREFIX-EQUAL candidates, kept until
+//                      the region fills. Prefix-equal candidates are
+//                      interchangeable at rank K (any K of them is a
+//                      valid top-k), so DROPPING the excess is sound:
+//                      massively tied planes (BM25 over tiny-vocab
+//                      corpora) no longer overflow. The final bitonic
+//                      sorts by the full 32-bit ordered value, so
+//                      prefix-equal candidates still order exactly.
+//                      The overflow flag now only fires when the
+//                      strictly-above region overflows — impossible
+//                      for the exact path, possible for the sampled
+//                      threshold (checked loudly by the wrapper).
+#define HI_RES 1024
+
 __global__ __launch_bounds__(256) void compact_kernel(
     const float* __restrict__ scores, const unsigned* __restrict__ thresh16,
     unsigned long long* __restrict__ cand, unsigned* __restrict__ cnt,
-    unsigned* __restrict__ overflow, long N) {
+    unsigned* __restrict__ cnt_eq, unsigned* __restrict__ overflow,
+    long N) {
   const int b = blockIdx.y;
   const unsigned t16 = thresh16[b];
   const long start = (long)blockIdx.x * blockDim.x + threadIdx.x;
   const long step = (long)gridDim.x * blockDim.x;
   const float* row = scores + (long)b * N;
+  unsigned long long* crow = cand + (long)b * TOPK_CAP;
   const long n4 = N / 4;
   for (long i = start; i < n4; i += step) {
     const float4 v = reinterpret_cast<const float4*>(row)[i];
@@ -135,28 +155,33 @@ __global__ __launch_bounds__(256) void compact_kernel(
 #pragma unroll
     for (int j = 0; j < 4; ++j) {
       const unsigned o = float_to_ordered(f[j]);
-      if ((o >> 16) >= t16) {
+      const unsigned p16 = o >> 16;
+      // Ascending sort key: ~ordered in high bits (desc value),
+      // raw idx in low bits (ties -> smaller idx first).
+      const unsigned long long key =
+          ((unsigned long long)(~o) << 32) | (unsigned)(i * 4 + j);
+      if (p16 > t16) {
         const unsigned pos = atomicAdd(&cnt[b], 1u);
-        if (pos < TOPK_CAP)
-          cand[(long)b * TOPK_CAP + pos] =
-              ((unsigned long long)(~o) << 32) | (unsigned)(i * 4 + j);
-        else
-          *overflow = 1u;
+        if (pos < HI_RES) crow[pos] = key;
+        else *overflow = 1u;
+      } else if (p16 == t16) {
+        const unsigned pos = atomicAdd(&cnt_eq[b], 1u);
+        if (pos < TOPK_CAP - HI_RES) crow[HI_RES + pos] = key;
       }
     }
   }
   for (long i = n4 * 4 + start; i < N; i += step) {
     const unsigned o = float_to_ordered(row[i]);
-    if ((o >> 16) >= t16) {
+    const unsigned p16 = o >> 16;
+    const unsigned long long key =
+        ((unsigned long long)(~o) << 32) | (unsigned)i;
+    if (p16 > t16) {
       const unsigned pos = atomicAdd(&cnt[b], 1u);
-      if (pos < TOPK_CAP) {
-        // Ascending sort key: ~ordered in high bits (desc value),
-        // raw idx in low bits (ties -> smaller idx first).
-        cand[(long)b * TOPK_CAP + pos] =
-            ((unsigned long long)(~o) << 32) | (unsigned)i;
-      } else {
-        *overflow = 1u;
-      }
+      if (pos < HI_RES) crow[pos] = key;
+      else *overflow = 1u;
+    } else if (p16 == t16) {
+      const unsigned pos = atomicAdd(&cnt_eq[b], 1u);
+      if (pos < TOPK_CAP - HI_RES) crow[HI_RES + pos] = key;
     }
   }
 }
@@ -183,17 +208,23 @@ __global__ void sample_gather_kernel(const float* __restrict__ scores,
 
 __global__ __launch_bounds__(256) void sort_emit_kernel(
     unsigned long long* __restrict__ cand, const unsigned* __restrict__ cnt,
+    const unsigned* __restrict__ cnt_eq,
     float* __restrict__ out_vals, int* __restrict__ out_idx, int K) {
   __shared__ unsigned long long d[TOPK_CAP];
   const int b = blockIdx.x;
-  const int n = min(cnt[b], (unsigned)TOPK_CAP);
+  const int nhi = min(cnt[b], (unsigned)HI_RES);
+  const int neq = min(cnt_eq[b], (unsigned)(TOPK_CAP - HI_RES));
+  const int n = nhi + neq;
   // Bitonic network size: next pow2 of the actual candidate count —
   // typical rows carry ~K+epsilon candidates, so this cuts the sort
   // from CAP=8192 to 128/256 most of the time.
   int n2 = 64;
   while (n2 < n) n2 <<= 1;
+  const unsigned long long* crow = cand + (long)b * TOPK_CAP;
   for (int i = threadIdx.x; i < n2; i += blockDim.x)
-    d[i] = (i < n) ? cand[(long)b * TOPK_CAP + i] : ~0ULL;  // pad = worst
+    d[i] = (i < nhi) ? crow[i]
+         : (i < n) ? crow[HI_RES + (i - nhi)]
+         : ~0ULL;  // pad = worst
   __syncthreads();
   for (int k = 2; k <= n2; k <<= 1) {
     for (int j = k >> 1; j > 0; j >>= 1) {
@@ -224,9 +255,9 @@ __global__ __launch_bounds__(256) void sort_emit_kernel(
 
 // Workspace layout (u32 units), provided zero-initialized by the wrapper:
 //   hist1 [B*256] | hist2 [B*256] | bin1 [B] | chi1 [B] | thresh16 [B]
-//   | cnt [B] | overflow [1] | cand (u64) [B*TOPK_CAP] (8-byte aligned)
+//   | cnt [B] | cnt_eq [B] | overflow [1] | cand (u64) [B*TOPK_CAP]
 extern "C" long infomesh_topk_workspace_u32(int B) {
-  long u = (long)B * 256 * 2 + (long)B * 4 + 1;
+  long u = (long)B * 256 * 2 + (long)B * 5 + 1;
   u = (u + 1) & ~1L;  // align cand to 8 bytes
   return u + (long)B * TOPK_CAP * 2;
 }
@@ -243,8 +274,9 @@ extern "C" void infomesh_topk(const void* scores, void* workspace,
   unsigned* chi1 = bin1 + B;
   unsigned* thresh16 = chi1 + B;
   unsigned* cnt = thresh16 + B;
-  unsigned* overflow = cnt + B;
-  long off = (long)B * 256 * 2 + (long)B * 4 + 1;
+  unsigned* cnt_eq = cnt + B;
+  unsigned* overflow = cnt_eq + B;
+  long off = (long)B * 256 * 2 + (long)B * 5 + 1;
   off = (off + 1) & ~1L;
   auto* cand = reinterpret_cast<unsigned long long*>(ws + off);
 
@@ -275,7 +307,7 @@ extern "C" void infomesh_topk(const void* scores, void* workspace,
                        hist2, bin1, chi1, thresh16, Kp);
     hipLaunchKernelGGL(compact_kernel, g1, blk, 0, s,
                        (const float*)scores, thresh16, cand, cnt,
-                       overflow, N);
+                       cnt_eq, overflow, N);
   } else {
     hipLaunchKernelGGL(hist1_kernel, g1, blk, 0, s,
                        (const float*)scores, hist1, N);
@@ -287,8 +319,9 @@ extern "C" void infomesh_topk(const void* scores, void* workspace,
                        hist2, bin1, chi1, thresh16, K);
     hipLaunchKernelGGL(compact_kernel, g1, blk, 0, s,
                        (const float*)scores, thresh16, cand, cnt,
-                       overflow, N);
+                       cnt_eq, overflow, N);
   }
   hipLaunchKernelGGL(sort_emit_kernel, dim3(B), blk, 0, s,
-                     cand, cnt, (float*)out_vals, (int*)out_idx, K);
+                     cand, cnt, cnt_eq, (float*)out_vals, (int*)out_idx,
+                     K);
 }

@@ -275,15 +275,15 @@ class TopK:
             self._ws = torch.empty(nu32, device=scores.device,
                                    dtype=torch.int32)
             self._ws_b = nu32
-        self._ws[: B * 512 + 4 * B + 2].zero_()  # hists + counters region
+        self._ws[: B * 512 + 5 * B + 2].zero_()  # hists + counters region
         vals = torch.empty((B, k), device=scores.device, dtype=torch.float32)
         idx = torch.empty((B, k), device=scores.device, dtype=torch.int32)
         lib.infomesh_topk(scores.data_ptr(), self._ws.data_ptr(),
                           vals.data_ptr(), idx.data_ptr(), B, N, k,
                           int(sampled), _ext.stream_ptr())
-        # cnt[B] then the overflow flag live after the histograms.
+        # cnt[B], cnt_eq[B], then the overflow flag after the histograms.
         cnt_off = B * 512 + 3 * B
-        flag_off = B * 512 + 4 * B
+        flag_off = B * 512 + 5 * B
         need = min(k, N) if sampled else 0
         if self.defer_check:
             self._pending.append((self._ws, cnt_off, flag_off, B, need))
@@ -295,11 +295,13 @@ class TopK:
     def _verify(ws, cnt_off, flag_off, B, need) -> None:
         if int(ws[flag_off].item()) != 0:
             raise RuntimeError(
-                "topk candidate overflow (cap exceeded — pathologically "
-                "tied scores or an unlucky sampled threshold)")
+                "topk strictly-above-threshold candidates exceeded the "
+                "reserve (an unlucky sampled threshold); rerun with "
+                "sampled=False")
         if need:
-            cnts = ws[cnt_off:cnt_off + B]
-            if int(cnts.min().item()) < need:
+            # available candidates = strictly-above + threshold-equal
+            avail = ws[cnt_off:cnt_off + B] + ws[cnt_off + B:cnt_off + 2 * B]
+            if int(avail.min().item()) < need:
                 raise RuntimeError(
                     "topk sampled threshold undershot (cnt < k); "
                     "rerun with sampled=False")

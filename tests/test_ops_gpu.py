@@ -625,3 +625,28 @@ def test_dense_scores_matches_generic_tile():
     generic = K.gemm_nt(a, b, out_f32=True)
     torch.cuda.synchronize()
     assert torch.equal(stream, generic)
+
+
+def test_topk_massively_tied_scores():
+    """Degenerate planes (tiny-vocab corpora) produce tens of thousands
+    of EXACTLY tied scores; the selector must return a valid top-k
+    (correct values; tie membership arbitrary) instead of overflowing
+    (round-2 regression: serving 500s on tied corpora)."""
+    g = torch.Generator(device="cuda").manual_seed(0)
+    B, N, k = 8, 200_000, 100
+    # three tied levels: 5.0 (50k), 3.0 (100k), 1.0 (rest)
+    scores = torch.ones(B, N, device="cuda")
+    scores[:, :50_000] = 5.0
+    scores[:, 50_000:150_000] = 3.0
+    perm = torch.randperm(N, generator=g, device="cuda")
+    scores = scores[:, perm].contiguous()
+    vals, idx = K.topk(scores, k)
+    assert (vals == 5.0).all(), "all top-100 must come from the 5.0 tier"
+    assert (scores[0, idx[0].long()] == 5.0).all()
+    assert len(set(idx[0].tolist())) == k, "indices must be distinct"
+    # threshold tier bigger than the whole candidate region, k spanning
+    scores2 = torch.full((2, N), 2.0, device="cuda")
+    scores2[:, :10] = 7.0
+    vals2, idx2 = K.topk(scores2, 50)
+    assert (vals2[:, :10] == 7.0).all() and (vals2[:, 10:] == 2.0).all()
+    assert len(set(idx2[0].tolist())) == 50
