@@ -82,6 +82,23 @@ class CreditLedger(SQLiteStore):
         self.kp = kp
         self._off_peak_fn = off_peak_fn
         self._debt_since: float | None = None
+        # batched accounting (record_action_async)
+        import threading
+        self._accum: dict[Action, float] = {}
+        self._accum_searches = 0
+        self._accum_lock = threading.Lock()
+        self._flusher = None
+        self._flush_stop = None
+        self.flush_interval_s = 0.5
+
+    def close(self) -> None:
+        if self._flush_stop is not None:
+            self._flush_stop.set()
+        try:
+            self.flush_pending()
+        except Exception:
+            pass
+        super().close()
 
     # ------------------------------------------------------------ record
     def _last_hash(self) -> str:
@@ -156,6 +173,57 @@ class CreditLedger(SQLiteStore):
         cost = self.search_cost()
         self.record_action(Action.SEARCH_SPEND, cost)
         return cost
+
+    # --------------------------------------------- batched accounting
+    def record_action_async(self, action: Action,
+                            quantity: float = 1.0) -> None:
+        """Coalesce high-rate accounting into periodic signed entries:
+        ONE hash-chained + Ed25519-signed entry per action type per
+        flush interval, with the quantities summed. Pure-python signing
+        costs ~3 ms under the GIL — per-request record_action() would
+        cap the whole serving node at ~130 QPS. Credit totals are
+        identical (C = W·Q·M is linear in Q)."""
+        with self._accum_lock:
+            self._accum[action] = self._accum.get(action, 0.0) + quantity
+            self._ensure_flusher()
+
+    def deduct_search_cost_async(self) -> None:
+        """Batched deduct: the per-search cost is evaluated at flush
+        time (tier drift within one interval is negligible)."""
+        with self._accum_lock:
+            self._accum_searches += 1
+            self._ensure_flusher()
+
+    def _ensure_flusher(self) -> None:
+        import threading
+        if self._flusher is None or not self._flusher.is_alive():
+            self._flush_stop = threading.Event()
+            self._flusher = threading.Thread(
+                target=self._flush_loop, name="infomesh-ledger-flush",
+                daemon=True)
+            self._flusher.start()
+
+    def _flush_loop(self) -> None:
+        while not self._flush_stop.wait(self.flush_interval_s):
+            try:
+                self.flush_pending()
+            except Exception:
+                return
+
+    def flush_pending(self) -> int:
+        """Write one signed entry per accumulated action type."""
+        with self._accum_lock:
+            accum, self._accum = self._accum, {}
+            n_search, self._accum_searches = self._accum_searches, 0
+        n = 0
+        for action, qty in accum.items():
+            self.record_action(action, qty)
+            n += 1
+        if n_search:
+            self.record_action(Action.SEARCH_SPEND,
+                               self.search_cost() * n_search)
+            n += 1
+        return n
 
     # ------------------------------------------------------ verification
     def verify_chain(self) -> bool:

@@ -121,8 +121,10 @@ __global__ void select2_kernel(const unsigned* __restrict__ hist2,
 }
 
 // Candidate layout per row (TOPK_CAP slots):
-//   [0, HI_RES)        strictly-above-threshold candidates (exact
-//                      select guarantees < K <= HI_RES of them)
+//   [0, HI_RES)        strictly-above-threshold candidates (the exact
+//                      select guarantees < K <= 1024 of them; the
+//                      sampled threshold targets ~3K with tail slack,
+//                      hence the 4096 reserve)
 //   [HI_RES, TOPK_CAP) threshold-PREFIX-EQUAL candidates, kept until
 //                      the region fills. Prefix-equal candidates are
 //                      interchangeable at rank K (any K of them is a
@@ -135,7 +137,7 @@ __global__ void select2_kernel(const unsigned* __restrict__ hist2,
 //                      strictly-above region overflows — impossible
 //                      for the exact path, possible for the sampled
 //                      threshold (checked loudly by the wrapper).
-#define HI_RES 1024
+#define HI_RES 4096
 
 __global__ __launch_bounds__(256) void compact_kernel(
     const float* __restrict__ scores, const unsigned* __restrict__ thresh16,

@@ -222,7 +222,7 @@ class AppContext:
             if cached is not None:
                 return cached
         if deduct and self.config.credits.enabled:
-            self.ledger.deduct_search_cost()
+            self.ledger.deduct_search_cost_async()
         self.related.record(query)
         authority = self.link_graph.url_authority
         trust_fn = self.trust.trust_fn()
@@ -254,7 +254,7 @@ class AppContext:
         resp = GLOBAL_PLUGINS.run("post_search", resp, query=query)
         if use_cache:
             self.cache.put(key, resp)
-        self.ledger.record_action(Action.QUERY_SERVED, 1.0)
+        self.ledger.record_action_async(Action.QUERY_SERVED, 1.0)
         return resp
 
     def ensure_batcher(self) -> QueryBatcher | None:

@@ -113,3 +113,24 @@ def test_farming_detector():
     assert d.on_probation(now=t0 + 1)
     assert d.multiplier(now=t0 + 1) == 0.0
     assert d.multiplier(now=t0 + 25 * 3600) == 1.0
+
+
+def test_ledger_batched_accounting():
+    """record_action_async coalesces events into one signed entry per
+    action per flush; totals and the hash chain stay exact."""
+    from infomesh_amd.credits.ledger import (ACTION_WEIGHTS, Action,
+                                             CreditLedger)
+    from infomesh_amd.trust.keys import KeyPair
+    led = CreditLedger(":memory:", kp=KeyPair.generate())
+    for _ in range(50):
+        led.record_action_async(Action.QUERY_SERVED, 1.0)
+    for _ in range(7):
+        led.deduct_search_cost_async()
+    n = led.flush_pending()
+    assert n == 2   # one QUERY_SERVED entry + one SEARCH_SPEND entry
+    st = led.stats()
+    assert abs(st["by_action"]["query_served"]
+               - 50 * ACTION_WEIGHTS[Action.QUERY_SERVED]) < 1e-9
+    assert st["by_action"]["search_spend"] < 0
+    assert led.verify_chain()
+    led.close()

@@ -120,7 +120,10 @@ def test_gpu_variable_batch_serving():
         fused = plane.search_batch(terms_all[:B], emb_all[:B], B=B,
                                    n_results=10)
         assert fused.ids.shape[0] == B
-        assert torch.equal(fused.ids[:B], fused128.ids[:B]), \
+        # compare SCORES (tie membership in top-k is arbitrary and the
+        # tie-tolerant selector picks by atomics order)
+        assert torch.allclose(fused.scores, fused128.scores[:B],
+                              atol=1e-5), \
             f"batch-size {B} results diverge"
     torch.cuda.synchronize()
 
