@@ -7,6 +7,7 @@ classification). The GPU batch passage scorer is optional
 """
 from __future__ import annotations
 
+import functools as _functools
 import re
 from dataclasses import dataclass
 
@@ -88,18 +89,27 @@ def select_best_passage(query: str, text: str,
     return out
 
 
+@_functools.lru_cache(maxsize=512)
+def _snippet_pattern(query: str) -> "re.Pattern | None":
+    terms = sorted({t for t in _WORD_RE.findall(query.lower())
+                    if len(t) > 1}, key=len, reverse=True)
+    if not terms:
+        return None
+    return re.compile(
+        r"\b(" + "|".join(re.escape(t) for t in terms) + r")\b", re.I)
+
+
 def fast_snippet(query: str, text: str, width: int = 200) -> str:
     """Cheap snippet for the batched serving path: a window around the
     first query-term occurrence, terms highlighted. The FTS5 snippet()
     analogue (reference local_store.py:253-352 uses SQL snippet());
     full passage scoring (select_best_passage) stays on the unbatched
-    local path where per-query CPU time is not the bottleneck."""
-    terms = sorted({t for t in _WORD_RE.findall(query.lower())
-                    if len(t) > 1}, key=len, reverse=True)
-    if not terms or not text:
+    local path where per-query CPU time is not the bottleneck. The
+    compiled pattern is cached per query — a serving batch snippets
+    ~10 docs per query."""
+    pattern = _snippet_pattern(query)
+    if pattern is None or not text:
         return highlight(text[:width], query)
-    pattern = re.compile(
-        r"\b(" + "|".join(re.escape(t) for t in terms) + r")\b", re.I)
     m = pattern.search(text)
     if m is None:
         return text[:width]

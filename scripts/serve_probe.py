@@ -15,6 +15,7 @@ from __future__ import annotations
 
 import argparse
 import json
+import os
 import multiprocessing as mp
 import sys
 import time
@@ -38,6 +39,8 @@ def _client_proc(port: int, conc: int, secs: int, seed: int, out_q):
     rng = random.Random(seed)
     queries = [" ".join(rng.sample(WORDS, 3)) for _ in range(512)]
 
+    batch = int(os.environ.get("PROBE_BATCH", "0"))
+
     async def run():
         lat = []
         n = 0
@@ -50,13 +53,19 @@ def _client_proc(port: int, conc: int, secs: int, seed: int, out_q):
                 nonlocal n, errors
                 i = wid
                 while time.perf_counter() < deadline:
-                    q = queries[i % len(queries)]
                     i += conc
+                    if batch > 1:
+                        qs = [queries[(i + j) % len(queries)]
+                              for j in range(batch)]
+                        args = {"queries": qs, "limit": 10}
+                        name = "batch_search"
+                    else:
+                        args = {"query": queries[i % len(queries)],
+                                "limit": 10}
+                        name = "web_search"
                     body = {"jsonrpc": "2.0", "id": i,
                             "method": "tools/call",
-                            "params": {"name": "web_search",
-                                       "arguments": {"query": q,
-                                                     "limit": 10}}}
+                            "params": {"name": name, "arguments": args}}
                     t0 = time.perf_counter()
                     try:
                         r = await client.post("/mcp", json=body)
@@ -83,7 +92,12 @@ def main() -> None:
     ap.add_argument("--secs", type=int, default=10)
     ap.add_argument("--port", type=int, default=8931)
     ap.add_argument("--no-dense", action="store_true")
+    ap.add_argument("--batch", type=int, default=0,
+                    help="use the batch_search tool with N queries per "
+                         "HTTP request (bulk-caller mode)")
     args = ap.parse_args()
+    if args.batch:
+        os.environ["PROBE_BATCH"] = str(args.batch)
 
     import random
     import threading
@@ -152,9 +166,11 @@ def main() -> None:
     for p in procs:
         p.join()
     lats.sort()
-    qps = total / args.secs
+    per_req = max(args.batch, 1)
+    qps = total * per_req / args.secs
     res = {
         "http_qps": round(qps, 1),
+        "queries_per_request": per_req,
         "requests": total,
         "errors": errs,
         "p50_ms": round(lats[len(lats) // 2], 2) if lats else None,
