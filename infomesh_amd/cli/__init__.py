@@ -216,8 +216,8 @@ def dashboard(text, refresh):
         finally:
             ctx.close()
         return
-    from ..dashboard.app import run_dashboard
-    run_dashboard(refresh_s=refresh)
+    from ..dashboard.tabs import run_tabbed_dashboard
+    run_tabbed_dashboard(refresh_s=refresh)
 
 
 # ----------------------------------------------------------------- index

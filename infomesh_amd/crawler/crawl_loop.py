@@ -35,6 +35,13 @@ async def seed_and_crawl_loop(ctx: AppContext,
     governor = governor or getattr(ctx, "governor", None) \
         or ResourceGovernor()
     sched = ctx.worker.scheduler
+    # trigger-driven priority queue (reference freshness.py:67-212):
+    # drained before the BFS scheduler so feed/user triggers beat
+    # breadth-first discovery
+    from .freshness import PriorityRecrawlQueue, RecrawlTrigger
+    pq = getattr(ctx, "recrawl_queue", None)
+    if pq is None:
+        pq = ctx.recrawl_queue = PriorityRecrawlQueue()
     for url in load_seeds(seed_category):
         sched.add_url(url, depth=0)
     last_feed = last_recrawl = last_flush = 0.0
@@ -76,15 +83,22 @@ async def seed_and_crawl_loop(ctx: AppContext,
                 except Exception:
                     feed = None
                 for item in ctx.feeds.record_poll(mf.url, feed, now):
-                    sched.add_url(item.url, depth=0, priority=2)
+                    pq.enqueue(item.url, RecrawlTrigger.RSS_UPDATE,
+                               source=mf.url)
 
-        # priority recrawl batch (crawl_loop.py:110-180)
+        # adaptive recrawl feeds the SCHEDULED tier of the queue
         if now - last_recrawl > RECRAWL_EVERY_S:
             last_recrawl = now
             for doc in ctx.store.due_for_recrawl(limit=10):
-                sched.add_url(doc.url, depth=0, priority=3)
+                pq.enqueue(doc.url, RecrawlTrigger.SCHEDULED)
 
-        item = await sched.get_url(timeout=2.0)
+        # drain the priority queue first (reference crawl_loop.py:110-180)
+        pitem = pq.dequeue()
+        if pitem is not None:
+            item = (pitem.url, -1 if pitem.trigger
+                    == RecrawlTrigger.USER_REQUEST else 0)
+        else:
+            item = await sched.get_url(timeout=2.0)
         if item is None:
             if now - last_activity > IDLE_RESEED_AFTER_S:
                 for url in load_seeds(seed_category):

@@ -49,17 +49,74 @@ def load_starter(store: LocalStore, search_dirs: list[Path],
     return res
 
 
-def download_starter(url: str, dest: Path, timeout: float = 120.0) -> Path:
+MAX_STARTER_BYTES = 1 << 30   # snapshot import guard (snapshot.py)
+
+
+def download_starter(url: str, dest: Path, timeout: float = 120.0,
+                     progress: Callable[[int, int], None] | None = None,
+                     client=None) -> Path:
     """Fetch a snapshot over HTTP when connectivity exists (offline
-    deployments use find_local_starters instead)."""
+    deployments use find_local_starters instead). Streams with a size
+    guard and an optional progress(done, total) callback."""
     import httpx
     dest.parent.mkdir(parents=True, exist_ok=True)
     tmp = dest.with_suffix(".part")
-    with httpx.stream("GET", url, timeout=timeout,
-                      follow_redirects=True) as resp:
-        resp.raise_for_status()
-        with open(tmp, "wb") as f:
-            for chunk in resp.iter_bytes():
-                f.write(chunk)
+    own = client is None
+    client = client or httpx.Client(follow_redirects=True, timeout=timeout)
+    try:
+        with client.stream("GET", url) as resp:
+            resp.raise_for_status()
+            total = int(resp.headers.get("content-length", 0))
+            if total > MAX_STARTER_BYTES:
+                raise ValueError(f"starter too large ({total} bytes)")
+            done = 0
+            with open(tmp, "wb") as f:
+                for chunk in resp.iter_bytes():
+                    done += len(chunk)
+                    if done > MAX_STARTER_BYTES:
+                        raise ValueError("starter exceeded size guard")
+                    f.write(chunk)
+                    if progress:
+                        progress(done, total)
+    finally:
+        if own:
+            client.close()
     tmp.replace(dest)
     return dest
+
+
+def fetch_release_starter(store: LocalStore, dest_dir: Path,
+                          repo: str = "dotnetpower/infomesh",
+                          progress: Callable[[int, int], None] | None = None,
+                          client=None) -> dict | None:
+    """Community-release discovery + download + import (reference flow:
+    infomesh/index/starter.py:76-192): query the GitHub releases API,
+    pick the newest `.infomesh-snapshot` asset, stream it down with the
+    size guard, then import into the LocalStore. `client` is injectable
+    for offline tests."""
+    import httpx
+    own = client is None
+    client = client or httpx.Client(follow_redirects=True, timeout=30.0)
+    try:
+        r = client.get(
+            f"https://api.github.com/repos/{repo}/releases/latest",
+            headers={"Accept": "application/vnd.github+json"})
+        if r.status_code != 200:
+            log.info("no starter release available (%s)", r.status_code)
+            return None
+        assets = r.json().get("assets", [])
+        asset = next((a for a in assets
+                      if a.get("name", "").endswith(SUFFIX)
+                      and a.get("size", 0) <= MAX_STARTER_BYTES), None)
+        if asset is None:
+            return None
+        dest = dest_dir / asset["name"]
+        download_starter(asset["browser_download_url"], dest,
+                         progress=progress, client=client)
+        res = import_snapshot(store, dest)
+        res["path"] = str(dest)
+        res["release_asset"] = asset["name"]
+        return res
+    finally:
+        if own:
+            client.close()

@@ -38,6 +38,7 @@ class Handlers:
                    domain_allowlist: list[str] | None = None,
                    domain_blocklist: list[str] | None = None,
                    fetch_full_content: bool = False,
+                   validate: bool = False,
                    **_) -> dict[str, Any]:
         if not query or not query.strip():
             raise InfoMeshError("SRCH001", "empty query")
@@ -107,6 +108,8 @@ class Handlers:
             "degraded": resp.degraded,
             "results": results,
         }
+        if validate:
+            out["validation"] = self._cross_validate(query, results)
         if chunk_size:
             rag = format_rag_output(query, results, chunk_size=chunk_size,
                                     answer_mode=answer_mode,
@@ -229,6 +232,37 @@ class Handlers:
                               "hbm_bytes": eng.shard.hbm_bytes(),
                               "device": str(eng.shard.device)}]
         return out
+
+    def _cross_validate(self, query: str,
+                        engine_results: list[dict]) -> dict[str, Any]:
+        """Cross-validate the engine plane against the SQLite FTS plane
+        (two independent scorers of the same corpus — the intra-node
+        analogue of the reference's multi-peer check,
+        infomesh/mcp/handlers.py:432-452 + search/cross_validate.py).
+        A corrupted GPU shard shows up as wild score deviation or
+        snippet disagreement vs the FTS ground truth."""
+        from ..search.cross_validate import (SourceResult,
+                                             cross_validate_results)
+        from ..search.query import search_local
+        fts = search_local(self.ctx.store, query, limit=10,
+                           enhance_snippets=False)
+        sources = {
+            "engine": [SourceResult(url=r.get("url", ""),
+                                    title=r.get("title", ""),
+                                    snippet=r.get("snippet", ""),
+                                    score=float(r.get("score", 0.0)))
+                       for r in engine_results],
+            "fts": [SourceResult(url=h.url, title=h.title,
+                                 snippet=h.snippet, score=h.score)
+                    for h in fts.results],
+        }
+        report = cross_validate_results(sources)
+        return {
+            "n_sources": report.n_sources,
+            "n_suspicious": report.n_suspicious,
+            "suspicious_urls": report.suspicious_urls,
+            "verdicts": {r.url: r.verdict for r in report.results},
+        }
 
     def batch_search(self, queries: list[str] | None = None,
                      limit: int = 10, **_) -> dict[str, Any]:

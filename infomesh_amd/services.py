@@ -62,6 +62,7 @@ class AppContext:
     worker: CrawlWorker | None = None
     engine: HybridEngine | None = None
     batcher: QueryBatcher | None = None
+    recrawl_queue: Any = None   # crawler.freshness.PriorityRecrawlQueue
     attestations: list = field(default_factory=list)
     farming: FarmingDetector = field(default_factory=FarmingDetector)
     started_at: float = field(default_factory=time.time)

@@ -153,3 +153,87 @@ def alert_rules_yaml() -> str:
     expr: engine_hbm_free_fraction < 0.05
     for: 5m
 """
+
+
+# ---------------------------------------------------------- OTLP export
+
+def trace_to_otlp(trace: QueryTrace, service_name: str = "infomesh-amd",
+                  trace_id: str | None = None) -> dict:
+    """Encode a QueryTrace as an OTLP/HTTP JSON ExportTraceServiceRequest
+    (reference parity: infomesh/observability/metrics.py:24-70 uses the
+    OTel SDK + OTLP exporter; here the OTLP JSON wire format is emitted
+    directly — no SDK dependency, same collectors accept it)."""
+    import os as _os
+    tid = trace_id or _os.urandom(16).hex()
+    root_sid = _os.urandom(8).hex()
+    t0 = trace.started
+    t_end = max((s.end or time.time()) for s in trace.spans) \
+        if trace.spans else time.time()
+
+    def ns(t: float) -> str:
+        return str(int(t * 1e9))
+
+    spans = [{
+        "traceId": tid, "spanId": root_sid,
+        "name": "search",
+        "kind": 2,  # SERVER
+        "startTimeUnixNano": ns(t0), "endTimeUnixNano": ns(t_end),
+        "attributes": [{"key": "query",
+                        "value": {"stringValue": trace.query}}],
+    }]
+    for s in trace.spans:
+        spans.append({
+            "traceId": tid, "spanId": _os.urandom(8).hex(),
+            "parentSpanId": root_sid,
+            "name": s.name, "kind": 1,  # INTERNAL
+            "startTimeUnixNano": ns(s.start),
+            "endTimeUnixNano": ns(s.end or time.time()),
+        })
+    return {"resourceSpans": [{
+        "resource": {"attributes": [{
+            "key": "service.name",
+            "value": {"stringValue": service_name}}]},
+        "scopeSpans": [{"scope": {"name": "infomesh_amd"},
+                        "spans": spans}],
+    }]}
+
+
+class OtlpExporter:
+    """Batched OTLP/HTTP JSON trace exporter. Disabled unless an
+    endpoint is configured (reference behavior: OTel optional,
+    import-guarded)."""
+
+    def __init__(self, endpoint: str = "", service_name: str = "infomesh-amd",
+                 batch_size: int = 32, post_fn=None):
+        self.endpoint = endpoint.rstrip("/")
+        self.service_name = service_name
+        self.batch_size = batch_size
+        self._buf: list[dict] = []
+        self._post = post_fn or self._default_post
+        self.exported = 0
+        self.errors = 0
+
+    @property
+    def enabled(self) -> bool:
+        return bool(self.endpoint)
+
+    def export(self, trace: QueryTrace) -> None:
+        if not self.enabled:
+            return
+        self._buf.append(trace_to_otlp(trace, self.service_name))
+        if len(self._buf) >= self.batch_size:
+            self.flush()
+
+    def flush(self) -> None:
+        buf, self._buf = self._buf, []
+        for payload in buf:
+            try:
+                self._post(f"{self.endpoint}/v1/traces", payload)
+                self.exported += 1
+            except Exception:
+                self.errors += 1
+
+    @staticmethod
+    def _default_post(url: str, payload: dict) -> None:
+        import httpx
+        httpx.post(url, json=payload, timeout=5.0).raise_for_status()

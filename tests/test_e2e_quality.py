@@ -129,3 +129,26 @@ def test_engine_reindex_after_deletion(quality_ctx):
     ctx.deletions.request_deletion("https://gone.com/x")
     assert ctx.store.count() == 0
     ctx.close()
+
+
+# ------------------------------------------------- NDCG / MRR regression
+
+def test_ndcg_mrr_regression_gate(quality_ctx):
+    """Relevance regression gate (VERDICT weak #10): NDCG@5 and MRR
+    over the labeled query set, computed with search/quality.py's own
+    metrics, must clear fixed thresholds. A ranking regression (fusion
+    bug, tokenizer change, scoring sign flip) fails this before any
+    human notices."""
+    from infomesh_amd.search.quality import mrr, ndcg
+    ndcgs, hits_ranked = [], []
+    for query, expected in QUERIES:
+        resp = quality_ctx.search(query, limit=5, use_cache=False,
+                                  deduct=False)
+        urls = [getattr(r, "url", "") for r in resp.results][:5]
+        rels = [1.0 if u == expected else 0.0 for u in urls]
+        ndcgs.append(ndcg(rels, k=5))
+        hits_ranked.append([u == expected for u in urls])
+    mean_ndcg = sum(ndcgs) / len(ndcgs)
+    mean_mrr = sum(mrr(h) for h in hits_ranked) / len(hits_ranked)
+    assert mean_ndcg >= 0.9, f"NDCG@5 regressed: {mean_ndcg:.3f} {ndcgs}"
+    assert mean_mrr >= 0.9, f"MRR regressed: {mean_mrr:.3f}"
