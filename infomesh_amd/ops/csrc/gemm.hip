@@ -154,9 +154,55 @@ __global__ __launch_bounds__(256) void gemm_bf16_nt_kernel(
     cur ^= 1;
   }
 
-  // ---- epilogue: C/D map col=lane&15, row=(lane>>4)*4+reg ----------------
+  // ---- epilogue ---------------------------------------------------------
+  // Huge-N f32 output (the dense scoring plane): stage the wave's
+  // 64-col row segments through LDS and emit 16-B dwordx4 stores,
+  // NONTEMPORAL — C is written once and next read by the streaming
+  // top-k, so allocating it in L2 only evicts the B rows being
+  // streamed. Smaller/bf16 outputs keep the direct scalar path.
   const int crow_base = (lane >> 4) * 4;
   const int ccol = lane & 15;
+  if (OUT_F32 && (BN == 128 || BN == 64) && n0 + BN <= N
+      && m0 + BM <= M) {
+    // reuse the (drained) double buffer as an f32 staging tile:
+    // per wave a [BM/WM][BN/WN] block = 64x64 (or 64x32/32x32) f32
+    float* stage_f32 = reinterpret_cast<float*>(&smem[0][0])
+        + wid * (BM / WM) * (BN / WN);
+    __syncthreads();   // all MFMA reads of smem are done
+#pragma unroll
+    for (int i = 0; i < FM; ++i)
+#pragma unroll
+      for (int j = 0; j < FN; ++j)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int lm = i * 16 + crow_base + r;      // wave-local row
+          const int ln = j * 16 + ccol;               // wave-local col
+          const int n = n0 + wn * (BN / WN) + ln;
+          const float bv = bias ? bias[n] : 0.0f;
+          stage_f32[lm * (BN / WN) + ln] =
+              apply_act(alpha * acc[i][j][r] + bv, act);
+        }
+    __builtin_amdgcn_s_waitcnt(0);  // lgkmcnt: LDS writes visible
+    __builtin_amdgcn_wave_barrier();
+    // each lane streams 16 B of a row segment: WAVEN cols * 4 B / 16
+    constexpr int SEGN = BN / WN;                  // cols per wave block
+    constexpr int LPR = SEGN * 4 / 16;             // lanes per row
+    constexpr int ROWS_PER_IT = 64 / LPR;
+    const int srow = lane / LPR, scol4 = (lane % LPR) * 4;
+#pragma unroll
+    for (int base = 0; base < BM / WM; base += ROWS_PER_IT) {
+      const int lm = base + srow;
+      const int m = m0 + wm * (BM / WM) + lm;
+      const int n = n0 + wn * SEGN + scol4;
+      f32x4 v = *reinterpret_cast<const f32x4*>(
+          &stage_f32[lm * SEGN + scol4]);
+      __builtin_nontemporal_store(
+          v, reinterpret_cast<f32x4*>(
+              reinterpret_cast<float*>(C)
+              + (long)g * strideC + (long)m * N + n));
+    }
+    return;
+  }
 #pragma unroll
   for (int i = 0; i < FM; ++i) {
 #pragma unroll
