@@ -224,6 +224,168 @@ __global__ __launch_bounds__(256) void gemm_bf16_nt_kernel(
   }
 }
 
+// ---- 3-buffer glds-span pipelined 64x64 tile ---------------------------
+// For the latency-bound regime (encoder projections: M~4096, small N/K,
+// few K-steps): the serial stage->vmcnt(0)->barrier->compute structure
+// of the generic tile pays a full LDS-fill latency every K-step. Here
+// stages t+1 AND t+2 stay in flight across RAW s_barriers with counted
+// vmcnt waits (guide "pipelining across barriers": 3-buf span +83% over
+// serial at 1-block/CU), and 16 KiB/buffer keeps 3 workgroups resident
+// per CU on top. Round-1 BACKLOG item "counted-vmcnt small-tile
+// pipeline (gemm8-style scheduling at 64^2)".
+template <bool OUT_F32>
+__global__ __launch_bounds__(256) void gemm_pipe64_kernel(
+    const bf16* __restrict__ A, const bf16* __restrict__ B,
+    void* __restrict__ C, const float* __restrict__ bias,
+    int M, int N, int K,
+    long strideA, long strideB, long strideC,
+    int act, float alpha) {
+  constexpr int BM = 64, BN = 64, BK = 64;
+  __shared__ bf16 smem[3][(BM + BN) * BK];   // 3 x 16 KiB
+
+  const int tiles_n = (N + BN - 1) / BN;
+  const int tiles_m = (M + BM - 1) / BM;
+  const int tile = xcd_swizzle(blockIdx.x, tiles_m * tiles_n);
+  const int m0 = (tile / tiles_n) * BM, n0 = (tile % tiles_n) * BN;
+  const int g = blockIdx.y;
+  const bf16* Ag = A + (long)g * strideA;
+  const bf16* Bg = B + (long)g * strideB;
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wm = wid >> 1, wn = wid & 1;    // 2x2 waves, 32x32 each
+  const int n_ksteps = K / BK;
+
+  // tile = 16 KiB = 16 chunks of 1 KiB (8 A chunks then 8 B chunks);
+  // 4 glds per wave per stage
+  auto stage = [&](int buf, int t) {
+    const int ks = t < n_ksteps ? t : n_ksteps - 1;  // clamp (static cnt)
+    const long k0 = (long)ks * BK;
+#pragma unroll
+    for (int c = 0; c < 4; ++c) {
+      const int chunk = wid * 4 + c;
+      const int lin = chunk * 1024 + lane * 16;
+      const int src = swz64(lin);
+      const int row = src / (BK * 2);
+      const int colb = src % (BK * 2);
+      const bool is_b = chunk >= 8;
+      int grow = is_b ? (n0 + row - BM) : (m0 + row);
+      const int lim = is_b ? N : M;
+      grow = grow < lim ? grow : lim - 1;
+      const bf16* gsrc = (is_b ? Bg : Ag) + (long)grow * K + k0 + colb / 2;
+      auto* dst = (__attribute__((address_space(3))) unsigned int*)
+          ((char*)&smem[buf][0] + (long)chunk * 1024);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)gsrc,
+          dst, 16, 0, 0);
+    }
+  };
+
+  const int fr = lane & 15;
+  const int fk = (lane >> 4) * 8;
+  f32x4 acc[2][2];
+#pragma unroll
+  for (int i = 0; i < 2; ++i)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  stage(0, 0);
+  stage(1, 1);
+  asm volatile("s_waitcnt vmcnt(4)" ::: "memory");   // t0 landed
+  __builtin_amdgcn_s_barrier();
+
+  for (int t = 0; t < n_ksteps; ++t) {
+    const char* abase = (const char*)&smem[t % 3][0];
+    const char* bbase = abase + BM * BK * 2;
+    bf16x8 a[2][2], b[2][2];
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        const int row = wm * 32 + i * 16 + fr;
+        a[i][ks] = *reinterpret_cast<const bf16x8*>(
+            abase + swz64((row * BK + ks * 32 + fk) * 2));
+        const int brow = wn * 32 + i * 16 + fr;
+        b[i][ks] = *reinterpret_cast<const bf16x8*>(
+            bbase + swz64((brow * BK + ks * 32 + fk) * 2));
+      }
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+#pragma unroll
+      for (int j = 0; j < 2; ++j)
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              a[i][ks], b[j][ks], acc[i][j], 0, 0, 0);
+    stage((t + 2) % 3, t + 2);
+    // own t+1 chunks landed (t+2's 4 stay in flight); cross-wave
+    // visibility via the raw barrier — __syncthreads() would emit
+    // vmcnt(0) and drain the span (guide pitfall)
+    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+  }
+
+  // drain the clamped tail prefetches before reusing smem
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  const int crow_base = (lane >> 4) * 4;
+  const int ccol = lane & 15;
+  if (OUT_F32 && n0 + BN <= N && m0 + BM <= M) {
+    // staged dwordx4 nontemporal epilogue (see the generic tile)
+    float* stage_f32 = reinterpret_cast<float*>(&smem[0][0])
+        + wid * 32 * 32;
+#pragma unroll
+    for (int i = 0; i < 2; ++i)
+#pragma unroll
+      for (int j = 0; j < 2; ++j)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int lm = i * 16 + crow_base + r;
+          const int ln = j * 16 + ccol;
+          const float bv = bias ? bias[n0 + wn * 32 + ln] : 0.0f;
+          stage_f32[lm * 32 + ln] =
+              apply_act(alpha * acc[i][j][r] + bv, act);
+        }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_wave_barrier();
+    const int srow = lane / 8, scol4 = (lane % 8) * 4;
+#pragma unroll
+    for (int base = 0; base < 32; base += 8) {
+      const int lm = base + srow;
+      const int m = m0 + wm * 32 + lm;
+      const int n = n0 + wn * 32 + scol4;
+      f32x4 v = *reinterpret_cast<const f32x4*>(
+          &stage_f32[lm * 32 + scol4]);
+      __builtin_nontemporal_store(
+          v, reinterpret_cast<f32x4*>(
+              reinterpret_cast<float*>(C)
+              + (long)g * strideC + (long)m * N + n));
+    }
+    return;
+  }
+#pragma unroll
+  for (int i = 0; i < 2; ++i) {
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      const int n = n0 + wn * 32 + j * 16 + ccol;
+      if (n >= N) continue;
+      const float bv = bias ? bias[n] : 0.0f;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int m = m0 + wm * 32 + i * 16 + crow_base + r;
+        if (m >= M) continue;
+        float v = apply_act(alpha * acc[i][j][r] + bv, act);
+        if (OUT_F32)
+          reinterpret_cast<float*>(C)[(long)g * strideC + (long)m * N + n] = v;
+        else
+          reinterpret_cast<bf16*>(C)[(long)g * strideC + (long)m * N + n] =
+              f2bf(v);
+      }
+    }
+  }
+}
+
 }  // namespace
 
 extern "C" void infomesh_gemm_bf16_nt(
@@ -265,7 +427,24 @@ extern "C" void infomesh_gemm_bf16_nt(
   // (BK=128 for the 64^2 tile was probe-tested and is ~60% SLOWER:
   // doubling LDS to 64 KB halves resident blocks per CU, which costs
   // more latency hiding than the halved K-step drains save.)
-  if (bm == 64 && bn == 64) PICK(64, 64);
+  static const int pipe_ov = [] {
+    const char* e = getenv("INFOMESH_GEMM_PIPE");
+    return e ? atoi(e) : 1;
+  }();
+  if (bm == 64 && bn == 64 && bk64 && pipe_ov) {
+    // latency-bound regime: 3-buffer glds-span pipeline
+    if (out_f32)
+      hipLaunchKernelGGL(gemm_pipe64_kernel<true>, grid, block, 0, s,
+                         (const bf16*)A, (const bf16*)B, C,
+                         (const float*)bias, M, N, K, strideA, strideB,
+                         strideC, act, alpha);
+    else
+      hipLaunchKernelGGL(gemm_pipe64_kernel<false>, grid, block, 0, s,
+                         (const bf16*)A, (const bf16*)B, C,
+                         (const float*)bias, M, N, K, strideA, strideB,
+                         strideC, act, alpha);
+  }
+  else if (bm == 64 && bn == 64) PICK(64, 64);
   else if (bm == 64) PICK(64, 128);
   else PICK(128, 128);
 #undef PICK
