@@ -233,14 +233,14 @@ __global__ __launch_bounds__(256) void gemm_bf16_nt_kernel(
 // serial at 1-block/CU), and 16 KiB/buffer keeps 3 workgroups resident
 // per CU on top. Round-1 BACKLOG item "counted-vmcnt small-tile
 // pipeline (gemm8-style scheduling at 64^2)".
-template <bool OUT_F32>
+template <int BNT, bool OUT_F32>
 __global__ __launch_bounds__(256) void gemm_pipe64_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ B,
     void* __restrict__ C, const float* __restrict__ bias,
     int M, int N, int K,
     long strideA, long strideB, long strideC,
     int act, float alpha) {
-  constexpr int BM = 64, BN = 64, BK = 64;
+  constexpr int BM = 64, BN = BNT, BK = 64;
   __shared__ bf16 smem[3][(BM + BN) * BK];   // 3 x 16 KiB
 
   const int tiles_n = (N + BN - 1) / BN;
@@ -253,22 +253,24 @@ __global__ __launch_bounds__(256) void gemm_pipe64_kernel(
 
   const int lane = threadIdx.x & 63;
   const int wid = threadIdx.x >> 6;
-  const int wm = wid >> 1, wn = wid & 1;    // 2x2 waves, 32x32 each
+  const int wm = wid >> 1, wn = wid & 1;    // 2x2 waves, 32 x BN/2 each
+  constexpr int FN = BN / 2 / 16;           // n-frags per wave
   const int n_ksteps = K / BK;
 
-  // tile = 16 KiB = 16 chunks of 1 KiB (8 A chunks then 8 B chunks);
-  // 4 glds per wave per stage
+  // tile = (BM+BN)*BK*2 bytes in 1 KiB chunks (A chunks then B);
+  // CPW glds per wave per stage
+  constexpr int CPW = (BM + BN) * BK * 2 / 1024 / 4;
   auto stage = [&](int buf, int t) {
     const int ks = t < n_ksteps ? t : n_ksteps - 1;  // clamp (static cnt)
     const long k0 = (long)ks * BK;
 #pragma unroll
-    for (int c = 0; c < 4; ++c) {
-      const int chunk = wid * 4 + c;
+    for (int c = 0; c < CPW; ++c) {
+      const int chunk = wid * CPW + c;
       const int lin = chunk * 1024 + lane * 16;
       const int src = swz64(lin);
       const int row = src / (BK * 2);
       const int colb = src % (BK * 2);
-      const bool is_b = chunk >= 8;
+      const bool is_b = chunk >= (BM * BK * 2 / 1024);
       int grow = is_b ? (n0 + row - BM) : (m0 + row);
       const int lim = is_b ? N : M;
       grow = grow < lim ? grow : lim - 1;
@@ -283,21 +285,21 @@ __global__ __launch_bounds__(256) void gemm_pipe64_kernel(
 
   const int fr = lane & 15;
   const int fk = (lane >> 4) * 8;
-  f32x4 acc[2][2];
+  f32x4 acc[2][FN];
 #pragma unroll
   for (int i = 0; i < 2; ++i)
 #pragma unroll
-    for (int j = 0; j < 2; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+    for (int j = 0; j < FN; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   stage(0, 0);
   stage(1, 1);
-  asm volatile("s_waitcnt vmcnt(4)" ::: "memory");   // t0 landed
+  asm volatile("s_waitcnt vmcnt(%0)" :: "n"(CPW) : "memory");  // t0 in
   __builtin_amdgcn_s_barrier();
 
   for (int t = 0; t < n_ksteps; ++t) {
     const char* abase = (const char*)&smem[t % 3][0];
     const char* bbase = abase + BM * BK * 2;
-    bf16x8 a[2][2], b[2][2];
+    bf16x8 a[2][2], b[FN][2];
 #pragma unroll
     for (int i = 0; i < 2; ++i)
 #pragma unroll
@@ -305,23 +307,28 @@ __global__ __launch_bounds__(256) void gemm_pipe64_kernel(
         const int row = wm * 32 + i * 16 + fr;
         a[i][ks] = *reinterpret_cast<const bf16x8*>(
             abase + swz64((row * BK + ks * 32 + fk) * 2));
-        const int brow = wn * 32 + i * 16 + fr;
-        b[i][ks] = *reinterpret_cast<const bf16x8*>(
+      }
+#pragma unroll
+    for (int j = 0; j < FN; ++j)
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        const int brow = wn * (BN / 2) + j * 16 + fr;
+        b[j][ks] = *reinterpret_cast<const bf16x8*>(
             bbase + swz64((brow * BK + ks * 32 + fk) * 2));
       }
 #pragma unroll
     for (int i = 0; i < 2; ++i)
 #pragma unroll
-      for (int j = 0; j < 2; ++j)
+      for (int j = 0; j < FN; ++j)
 #pragma unroll
         for (int ks = 0; ks < 2; ++ks)
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               a[i][ks], b[j][ks], acc[i][j], 0, 0, 0);
     stage((t + 2) % 3, t + 2);
-    // own t+1 chunks landed (t+2's 4 stay in flight); cross-wave
+    // own t+1 chunks landed (t+2's CPW stay in flight); cross-wave
     // visibility via the raw barrier — __syncthreads() would emit
     // vmcnt(0) and drain the span (guide pitfall)
-    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    asm volatile("s_waitcnt vmcnt(%0)" :: "n"(CPW) : "memory");
     __builtin_amdgcn_s_barrier();
   }
 
@@ -331,32 +338,35 @@ __global__ __launch_bounds__(256) void gemm_pipe64_kernel(
 
   const int crow_base = (lane >> 4) * 4;
   const int ccol = lane & 15;
+  constexpr int SEGN = BN / 2;
   if (OUT_F32 && n0 + BN <= N && m0 + BM <= M) {
     // staged dwordx4 nontemporal epilogue (see the generic tile)
     float* stage_f32 = reinterpret_cast<float*>(&smem[0][0])
-        + wid * 32 * 32;
+        + wid * 32 * SEGN;
 #pragma unroll
     for (int i = 0; i < 2; ++i)
 #pragma unroll
-      for (int j = 0; j < 2; ++j)
+      for (int j = 0; j < FN; ++j)
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const int lm = i * 16 + crow_base + r;
           const int ln = j * 16 + ccol;
-          const float bv = bias ? bias[n0 + wn * 32 + ln] : 0.0f;
-          stage_f32[lm * 32 + ln] =
+          const float bv = bias ? bias[n0 + wn * SEGN + ln] : 0.0f;
+          stage_f32[lm * SEGN + ln] =
               apply_act(alpha * acc[i][j][r] + bv, act);
         }
     asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     __builtin_amdgcn_wave_barrier();
-    const int srow = lane / 8, scol4 = (lane % 8) * 4;
+    constexpr int LPR = SEGN / 4;
+    constexpr int RPI = 64 / LPR;
+    const int srow = lane / LPR, scol4 = (lane % LPR) * 4;
 #pragma unroll
-    for (int base = 0; base < 32; base += 8) {
+    for (int base = 0; base < 32; base += RPI) {
       const int lm = base + srow;
       const int m = m0 + wm * 32 + lm;
-      const int n = n0 + wn * 32 + scol4;
+      const int n = n0 + wn * SEGN + scol4;
       f32x4 v = *reinterpret_cast<const f32x4*>(
-          &stage_f32[lm * 32 + scol4]);
+          &stage_f32[lm * SEGN + scol4]);
       __builtin_nontemporal_store(
           v, reinterpret_cast<f32x4*>(
               reinterpret_cast<float*>(C)
@@ -367,8 +377,8 @@ __global__ __launch_bounds__(256) void gemm_pipe64_kernel(
 #pragma unroll
   for (int i = 0; i < 2; ++i) {
 #pragma unroll
-    for (int j = 0; j < 2; ++j) {
-      const int n = n0 + wn * 32 + j * 16 + ccol;
+    for (int j = 0; j < FN; ++j) {
+      const int n = n0 + wn * SEGN + j * 16 + ccol;
       if (n >= N) continue;
       const float bv = bias ? bias[n] : 0.0f;
 #pragma unroll
@@ -432,14 +442,30 @@ extern "C" void infomesh_gemm_bf16_nt(
     return e ? atoi(e) : 1;
   }();
   if (bm == 64 && bn == 64 && bk64 && pipe_ov) {
-    // latency-bound regime: 3-buffer glds-span pipeline
-    if (out_f32)
-      hipLaunchKernelGGL(gemm_pipe64_kernel<true>, grid, block, 0, s,
+    // latency-bound regime: 3-buffer glds-span pipeline; BN=128 when N
+    // has whole 128-tiles (pipe_ov=2 forces BN=64, =3 forces BN=128)
+    const bool wide = (pipe_ov == 3) ||
+        (pipe_ov != 2 && N >= 512);
+    if (wide) {
+      const int tiles_w = ((M + 63) / 64) * ((N + 127) / 128);
+      dim3 gw(tiles_w, batch);
+      if (out_f32)
+        hipLaunchKernelGGL((gemm_pipe64_kernel<128, true>), gw, block,
+                           0, s, (const bf16*)A, (const bf16*)B, C,
+                           (const float*)bias, M, N, K, strideA,
+                           strideB, strideC, act, alpha);
+      else
+        hipLaunchKernelGGL((gemm_pipe64_kernel<128, false>), gw, block,
+                           0, s, (const bf16*)A, (const bf16*)B, C,
+                           (const float*)bias, M, N, K, strideA,
+                           strideB, strideC, act, alpha);
+    } else if (out_f32)
+      hipLaunchKernelGGL((gemm_pipe64_kernel<64, true>), grid, block, 0, s,
                          (const bf16*)A, (const bf16*)B, C,
                          (const float*)bias, M, N, K, strideA, strideB,
                          strideC, act, alpha);
     else
-      hipLaunchKernelGGL(gemm_pipe64_kernel<false>, grid, block, 0, s,
+      hipLaunchKernelGGL((gemm_pipe64_kernel<64, false>), grid, block, 0, s,
                          (const bf16*)A, (const bf16*)B, C,
                          (const float*)bias, M, N, K, strideA, strideB,
                          strideC, act, alpha);
