@@ -444,8 +444,10 @@ extern "C" void infomesh_gemm_bf16_nt(
   if (bm == 64 && bn == 64 && bk64 && pipe_ov) {
     // latency-bound regime: 3-buffer glds-span pipeline; BN=128 when N
     // has whole 128-tiles (pipe_ov=2 forces BN=64, =3 forces BN=128)
-    const bool wide = (pipe_ov == 3) ||
-        (pipe_ov != 2 && N >= 512);
+    // BN=128 measured SLOWER on the encoder shapes (1278 vs 990 us
+    // full-encoder: 2 WGs/CU vs 3 loses more than wider MFMA gains);
+    // kept behind =3 for future shapes
+    const bool wide = (pipe_ov == 3);
     if (wide) {
       const int tiles_w = ((M + 63) / 64) * ((N + 127) / 128);
       dim3 gw(tiles_w, batch);
