@@ -79,10 +79,14 @@ __global__ __launch_bounds__(256) void bm25_block_kernel(
     long doc_base, long nseg, int BD, int nblocks,
     int blk_base, int total_blocks, int k_sel,
     float norm_a, float norm_b, float k1p1) {
-  // dynamic LDS: scores [BD] f32 | hist1 [256] | hist2 [256] | ctl [8]
+  // dynamic LDS: scores [BD] f32 | hist1 [8 x 264 padded copies] |
+  // hist2 [256] | ctl [8]. 8 bank-padded histogram copies (copy c at
+  // c*264 u32 — 264%64=8, so a bin's copies land in distinct banks)
+  // break the atomic serialization of concentrated score
+  // distributions (same trick as topk.hip hist1).
   extern __shared__ float lds_scores[];
   unsigned* hist1 = reinterpret_cast<unsigned*>(lds_scores + BD);
-  unsigned* hist2 = hist1 + 256;
+  unsigned* hist2 = hist1 + 8 * 264;
   unsigned* ctl = hist2 + 256;   // [0]=b1 [1]=chi [2]=t16 [3]=strict [4]=eq
   // grid: x = query (fast), y = doc-block — adjacent workgroups are
   // the SAME posting sub-range for different queries, so the XCD's L2
@@ -92,9 +96,8 @@ __global__ __launch_bounds__(256) void bm25_block_kernel(
   const long d0 = (long)blk * BD;
   const int nd = (int)min((long)BD, nseg - d0);
   for (int i = threadIdx.x; i < nd; i += blockDim.x) lds_scores[i] = 0.0f;
-  for (int i = threadIdx.x; i < 256; i += blockDim.x) {
-    hist1[i] = 0; hist2[i] = 0;
-  }
+  for (int i = threadIdx.x; i < 8 * 264; i += blockDim.x) hist1[i] = 0;
+  for (int i = threadIdx.x; i < 256; i += blockDim.x) hist2[i] = 0;
   if (threadIdx.x < 8) ctl[threadIdx.x] = 0;
   __syncthreads();
   const int t0 = qt_off[q], t1 = qt_off[q + 1];
@@ -119,66 +122,105 @@ __global__ __launch_bounds__(256) void bm25_block_kernel(
   __syncthreads();
 
   // ---- in-LDS 2-level radix select of the block's top-k ------------
+  // BM25 scores are >= 0; ZEROS (unmatched docs, the vast majority of
+  // a block) are skipped everywhere — histogramming them would
+  // serialize thousands of atomics on one bin. The zero prefix is
+  // 0x8000, so "score > 0" == "prefix > 0x8000".
   const int kq = nd < k_sel ? nd : k_sel;
-  for (int i = threadIdx.x; i < nd; i += blockDim.x)
-    atomicAdd(&hist1[float_to_ordered(lds_scores[i]) >> 24], 1u);
+  unsigned* my_h1 = hist1 + (threadIdx.x & 7) * 264;
+  for (int i = threadIdx.x; i < nd; i += blockDim.x) {
+    const float v = lds_scores[i];
+    if (v != 0.0f)
+      atomicAdd(&my_h1[float_to_ordered(v) >> 24], 1u);
+  }
+  __syncthreads();
+  if (threadIdx.x < 64) {   // merge the 8 copies (wave 0)
+#pragma unroll
+    for (int b = threadIdx.x; b < 256; b += 64) {
+      unsigned s = hist1[b];
+#pragma unroll
+      for (int c = 1; c < 8; ++c) s += hist1[c * 264 + b];
+      hist1[b] = s;
+    }
+  }
   __syncthreads();
   if (threadIdx.x == 0) {
     unsigned cum = 0;
-    for (int i = 255; i >= 0; --i) {
+    int b1 = -1;
+    for (int i = 255; i > 128; --i) {   // positive floats only
       const unsigned c = hist1[i];
-      if (cum + c >= (unsigned)kq || i == 0) {
-        ctl[0] = (unsigned)i; ctl[1] = cum; break;
-      }
+      if (cum + c >= (unsigned)kq) { b1 = i; break; }
       cum += c;
+    }
+    if (b1 < 0) {
+      // fewer than kq nonzero scores: every nonzero is strict, zeros
+      // fill the quota (t16 = the zero prefix)
+      ctl[0] = 0; ctl[1] = cum; ctl[2] = 0x8000u;
+    } else {
+      ctl[0] = (unsigned)b1; ctl[1] = cum; ctl[2] = 0;
     }
   }
   __syncthreads();
   const unsigned b1 = ctl[0];
-  for (int i = threadIdx.x; i < nd; i += blockDim.x) {
-    const unsigned o = float_to_ordered(lds_scores[i]);
-    if ((o >> 24) == b1) atomicAdd(&hist2[(o >> 16) & 255], 1u);
-  }
-  __syncthreads();
-  if (threadIdx.x == 0) {
-    unsigned cum = ctl[1];
-    for (int i = 255; i >= 0; --i) {
-      const unsigned c = hist2[i];
-      if (cum + c >= (unsigned)kq || i == 0) {
-        ctl[2] = (b1 << 8) | (unsigned)i; break;
-      }
-      cum += c;
+  if (b1) {
+    for (int i = threadIdx.x; i < nd; i += blockDim.x) {
+      const float v = lds_scores[i];
+      if (v == 0.0f) continue;
+      const unsigned o = float_to_ordered(v);
+      if ((o >> 24) == b1) atomicAdd(&hist2[(o >> 16) & 255], 1u);
     }
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      unsigned cum = ctl[1];
+      for (int i = 255; i >= 0; --i) {
+        const unsigned c = hist2[i];
+        if (cum + c >= (unsigned)kq || i == 0) {
+          ctl[2] = (b1 << 8) | (unsigned)i; break;
+        }
+        cum += c;
+      }
+    }
+    __syncthreads();
   }
-  __syncthreads();
   const unsigned t16 = ctl[2];
   const long obase = ((long)q * total_blocks + blk_base + blk)
       * (long)k_sel;
   // strictly-above candidates first (provably < kq of them) ...
   for (int i = threadIdx.x; i < nd; i += blockDim.x) {
-    const unsigned o = float_to_ordered(lds_scores[i]);
+    const float v = lds_scores[i];
+    if (v == 0.0f) continue;
+    const unsigned o = float_to_ordered(v);
     if ((o >> 16) > t16) {
       const unsigned pos = atomicAdd(&ctl[3], 1u);
-      out_vals[obase + pos] = lds_scores[i];
-      out_idx[obase + pos] = (int)(doc_base + d0) + i;
-    }
-  }
-  __syncthreads();
-  const unsigned strict = ctl[3];
-  // ... then threshold-prefix-equal until the quota (interchangeable
-  // at rank k — matching the global selector's tie semantics)
-  for (int i = threadIdx.x; i < nd; i += blockDim.x) {
-    const unsigned o = float_to_ordered(lds_scores[i]);
-    if ((o >> 16) == t16) {
-      const unsigned pos = strict + atomicAdd(&ctl[4], 1u);
-      if (pos < (unsigned)kq) {
-        out_vals[obase + pos] = lds_scores[i];
+      if (pos < (unsigned)k_sel) {   // bound guard (denormal edge)
+        out_vals[obase + pos] = v;
         out_idx[obase + pos] = (int)(doc_base + d0) + i;
       }
     }
   }
   __syncthreads();
-  const int filled = min((int)(strict + ctl[4]), kq);
+  const unsigned strict = min(ctl[3], (unsigned)kq);
+  const unsigned need_eq = (unsigned)kq - strict;
+  // ... then threshold-prefix-equal until the quota (interchangeable
+  // at rank k — matching the global selector's tie semantics). The
+  // ctl[4] pre-check gates the atomic once the quota is filled: a
+  // massively tied bin (or the zero-fill case) would otherwise
+  // serialize thousands of increments on one LDS counter.
+  for (int i = threadIdx.x; i < nd; i += blockDim.x) {
+    if (ctl[4] >= need_eq) break;    // racy fast-exit; exact via pos
+    const float v = lds_scores[i];
+    const unsigned o = float_to_ordered(v);
+    if ((o >> 16) == t16) {
+      const unsigned pos = atomicAdd(&ctl[4], 1u);
+      if (pos < need_eq) {
+        out_vals[obase + strict + pos] = v;
+        out_idx[obase + strict + pos] = (int)(doc_base + d0) + i;
+      }
+    }
+  }
+  __syncthreads();
+  const unsigned eq = ctl[4] < need_eq ? ctl[4] : need_eq;
+  const int filled = (int)(strict + eq);
   for (int j = filled + threadIdx.x; j < k_sel; j += blockDim.x) {
     out_vals[obase + j] = -INFINITY;
     out_idx[obase + j] = -1;
@@ -216,7 +258,8 @@ extern "C" void infomesh_bm25_block(
                        (const long*)u_end, (int*)bounds, U, nblocks, BD);
   }
   dim3 grid((unsigned)B, (unsigned)nblocks);
-  const size_t lds = (size_t)BD * sizeof(float) + (256 + 256 + 8) * 4;
+  const size_t lds = (size_t)BD * sizeof(float)
+      + (8 * 264 + 256 + 8) * 4;
   hipLaunchKernelGGL(bm25_block_kernel, grid, dim3(256), lds, s,
                      (const int*)doc_ids, (const unsigned int*)tfdl,
                      (const int*)qt_off, (const int*)qt_ut,

@@ -348,7 +348,7 @@ def bm25_block(doc_ids: torch.Tensor, tfdl: torch.Tensor,
     assert nblocks <= 65535 and blk_base + nblocks <= total_blocks
     assert doc_ids.dtype == torch.int32 and tfdl.dtype == torch.int32
     assert bounds.numel() >= U * nblocks * 2
-    assert bd * 4 + 2080 <= 160 * 1024 and k_sel <= bd
+    assert bd * 4 + 9504 <= 160 * 1024 and k_sel <= bd
     norm_a = k1 * (1.0 - b)
     norm_b = k1 * b / max(avgdl, 1e-9)
     _ext.lib().infomesh_bm25_block(
