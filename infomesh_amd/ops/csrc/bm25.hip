@@ -120,6 +120,11 @@ __global__ __launch_bounds__(256) void bm25_block_kernel(
     }
   }
   __syncthreads();
+  if (k_sel == 0) {   // profiling mode: accumulate only, emit nothing
+    if (lds_scores[threadIdx.x] > 1e30f)   // never true; defeat DCE
+      out_vals[0] = lds_scores[threadIdx.x];
+    return;
+  }
 
   // ---- in-LDS 2-level radix select of the block's top-k ------------
   // BM25 scores are >= 0; ZEROS (unmatched docs, the vast majority of

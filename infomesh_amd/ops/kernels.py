@@ -349,6 +349,8 @@ def bm25_block(doc_ids: torch.Tensor, tfdl: torch.Tensor,
     assert doc_ids.dtype == torch.int32 and tfdl.dtype == torch.int32
     assert bounds.numel() >= U * nblocks * 2
     assert bd * 4 + 9504 <= 160 * 1024 and k_sel <= bd
+    if k_sel == 0:   # accumulate-only profiling mode
+        pass
     norm_a = k1 * (1.0 - b)
     norm_b = k1 * b / max(avgdl, 1e-9)
     _ext.lib().infomesh_bm25_block(
