@@ -332,7 +332,6 @@ class GpuShard:
 
     def _invalidate_query_caches(self) -> None:
         self._h_idf = None
-        self._cand_cache = {}
         # dense hipGraphs captured the old embeddings view/shape
         self._dense_graphs = {}
         self._dense_bufs = {}
@@ -476,11 +475,9 @@ class GpuShard:
                     scores_buf: torch.Tensor | None = None,
                     mark=None) -> tuple[torch.Tensor, torch.Tensor]:
         """BM25 plane only (needs terms, not embeddings) — callable on a
-        side stream to overlap with query encoding. The fused kernel
-        emits each (query, doc-block)'s top-k candidates straight from
-        LDS; a small global top-k over [B, total_blocks*k] finishes —
-        no [B, N] score matrix exists (scores_buf is accepted for API
-        compatibility and ignored)."""
+        side stream to overlap with query encoding. One kernel launch
+        per posting segment; the segments' disjoint doc ranges cover
+        every column of the score matrix exactly once."""
         import time as _time
         from ..ops import kernels as K
         if mark is None:
@@ -492,6 +489,10 @@ class GpuShard:
         k = min(k, N)
         topk = self._get_topk()
         tp = _time.perf_counter()
+        if scores_buf is not None and scores_buf.shape == (B, N):
+            scores = scores_buf
+        else:
+            scores = torch.empty(B, N, device=dev, dtype=torch.float32)
         qrows, terms = self.dedupe_terms(queries_terms)
         # queries share Zipf-common terms: dedupe ACROSS queries so the
         # bounds pre-pass searches each term's postings once per block
@@ -501,9 +502,6 @@ class GpuShard:
         np.cumsum(np.bincount(qrows, minlength=B), out=qt_off[1:])
         bd = self._pick_bd(B)
         U = len(uterms)
-        seg_blocks = [(s.n_docs + bd - 1) // bd for s in self.segments]
-        total_blocks = sum(seg_blocks)
-        cand_v, cand_i = self._cand_bufs(B, total_blocks * k)
         tp = mark("shard.chunks", tp)
         # don't overwrite the pinned staging buffers while a prior
         # step's async H2D copy could still be in flight
@@ -513,44 +511,24 @@ class GpuShard:
         qt_off_d = self._h2d("qt_off", qt_off, torch.int32)
         qt_ut_d = self._h2d("qt_ut", qt_ut, torch.int32)
         qt_idf_d = self._h2d("qt_idf", idf, torch.float32)
-        blk_base = 0
         for si, seg in enumerate(self.segments):
-            bounds = self._bounds_ws(U * seg_blocks[si] * 2)
+            nblocks = (seg.n_docs + bd - 1) // bd
+            bounds = self._bounds_ws(U * nblocks * 2)
             K.bm25_block(
                 seg.doc_ids, seg.tfdl, qt_off_d, qt_ut_d, qt_idf_d,
                 self._h2d(f"qb{si}", seg.h_offs[uterms], torch.int64),
                 self._h2d(f"qe{si}", seg.h_offs[uterms + 1], torch.int64),
-                bounds, cand_v, cand_i, seg.doc_base, seg.n_docs, bd,
-                blk_base, total_blocks, k,
+                bounds, scores, seg.doc_base, seg.n_docs, bd,
                 self.avgdl, k1=BM25_K1, b=BM25_B)
-            blk_base += seg_blocks[si]
         if dev.type == "cuda":
             if evt is None:
                 evt = self._h2d_evt = torch.cuda.Event()
             evt.record()
         tp = mark("shard.bm25", tp)
-        vals, pos = topk(cand_v, k)
-        idx = torch.gather(cand_i, 1, pos.long()).to(torch.int32)
-        idx = torch.where(vals > -float("inf"), idx,
-                          torch.full_like(idx, -1))
+        out = topk(scores, k)
         mark("shard.bm25topk", tp)
-        return vals, idx
-
-    def _cand_bufs(self, B: int, cols: int):
-        """Per-(B, cols) candidate buffers for the fused BM25 kernel
-        (persistent — [B, nblocks*k] is ~O(MB), not O(N))."""
-        bufs = getattr(self, "_cand_cache", None)
-        if bufs is None:
-            bufs = self._cand_cache = {}
-        key = (B, cols)
-        entry = bufs.get(key)
-        if entry is None:
-            entry = bufs[key] = (
-                torch.empty(B, cols, device=self.device,
-                            dtype=torch.float32),
-                torch.empty(B, cols, device=self.device,
-                            dtype=torch.int32))
-        return entry
+        self._bm25_scores_buf = scores
+        return out
 
     def search(self, queries_terms: list[np.ndarray],
                query_emb: torch.Tensor | None, k: int = 100,

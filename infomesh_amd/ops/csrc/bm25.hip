@@ -1,5 +1,5 @@
-// GPU BM25 scorer over per-segment CSR posting lists in HBM — v3:
-// doc-block LDS accumulation FUSED with per-block top-k selection.
+// GPU BM25 scorer over per-segment CSR posting lists in HBM — v2:
+// doc-block LDS accumulation instead of global atomic scatter-add.
 //
 // Replaces: SQLite FTS5 `MATCH ... ORDER BY bm25()` (reference
 // infomesh/index/local_store.py:316-332) for the GPU shards; the CPU
@@ -7,23 +7,18 @@
 //
 // score(q, d) = Σ_t idf(t) · tf·(k1+1) / (tf + k1·(1−b+b·dl/avgdl))
 //
-// Evolution (profiles/):
-//   v1  global atomic scatter-add into a [B, N] score matrix: ~982 us
-//       kernel + ~0.6 ms zero-fill at 1.25M docs / B=128.
-//   v2  doc-block LDS accumulation + bounds pre-pass + query-major
-//       grid (L2 posting reuse): 405 us kernel — but the [B, N] score
-//       write (5.1 GB at 10M docs) plus the 3-pass global top-k
-//       (15 GB reads) still bounded the leg (6.4 ms at 10M).
-//   v3  the block's scores never leave LDS: a 2-level in-LDS radix
-//       select emits each (query, doc-block)'s top-k candidates
-//       (~100 x 8 B instead of 32 KB of scores), and one small global
-//       top-k over [B, nblocks*k] candidates finishes. Global traffic
-//       drops to postings + O(B * nblocks * k) — the [B, N] matrix no
-//       longer exists.
-//
-// Tie semantics match ops/csrc/topk.hip: candidates sharing the
-// 16-bit ordered-float threshold prefix are interchangeable at rank k;
-// any k of them is a valid top-k.
+// Design (MI355X): the v1 kernel did one global atomicAdd per posting
+// into scores[q*N+d] — random read-modify-write lines over a multi-GB
+// working set, measured HBM-bound at ~1 ms/batch (profiles/r01_*).
+// v2 tiles the doc axis: each workgroup owns one (query, doc-block)
+// tile, accumulates its block's scores in LDS (ds_add_f32 atomics are
+// conflict-cheap), then streams the block out once. Each term's
+// postings are doc-sorted, so the block's sub-range is two binary
+// searches. Global traffic drops to: postings read once + scores
+// written exactly once (which also removes the separate zero-fill —
+// every column of the output is written by exactly one workgroup,
+// because segments partition the doc axis and blocks partition each
+// segment).
 //
 // Per-posting doc length travels packed with tf (tf | dl<<16) so the
 // norm is computed in-kernel from the CURRENT global avgdl — this is
@@ -46,9 +41,9 @@ DEVINL long lower_bound_i32(const int* __restrict__ a, long lo, long hi,
 // posting sub-range once into a bounds table (i32 offsets relative to
 // the term's posting begin). The main kernel then has ZERO serial
 // binary searches — per (q,term) they were ~34 dependent HBM loads,
-// poorly hidden at a few workgroups/CU; here 1 thread per pair with
-// tens of thousands in flight hides them completely, and queries
-// sharing a term (Zipf-common) reuse the same entry.
+// poorly hidden at 2 workgroups/CU; here 1 thread per pair with tens
+// of thousands in flight hides them completely, and queries sharing a
+// term (Zipf-common) reuse the same entry.
 __global__ __launch_bounds__(256) void bm25_bounds_kernel(
     const int* __restrict__ doc_ids,
     const long* __restrict__ u_begin,   // [U] per unique term
@@ -74,20 +69,10 @@ __global__ __launch_bounds__(256) void bm25_block_kernel(
     const float* __restrict__ qt_idf,       // [T]
     const long* __restrict__ u_begin,       // [U]
     const int* __restrict__ bounds,         // [U * nblocks * 2]
-    float* __restrict__ out_vals,           // [B, total_blocks * k]
-    int* __restrict__ out_idx,              // [B, total_blocks * k]
-    long doc_base, long nseg, int BD, int nblocks,
-    int blk_base, int total_blocks, int k_sel,
+    float* __restrict__ scores,             // [B, rowN]
+    long rowN, long doc_base, long nseg, int BD, int nblocks,
     float norm_a, float norm_b, float k1p1) {
-  // dynamic LDS: scores [BD] f32 | hist1 [8 x 264 padded copies] |
-  // hist2 [256] | ctl [8]. 8 bank-padded histogram copies (copy c at
-  // c*264 u32 — 264%64=8, so a bin's copies land in distinct banks)
-  // break the atomic serialization of concentrated score
-  // distributions (same trick as topk.hip hist1).
-  extern __shared__ float lds_scores[];
-  unsigned* hist1 = reinterpret_cast<unsigned*>(lds_scores + BD);
-  unsigned* hist2 = hist1 + 8 * 264;
-  unsigned* ctl = hist2 + 256;   // [0]=b1 [1]=chi [2]=t16 [3]=strict [4]=eq
+  extern __shared__ float lds_scores[];     // [BD]
   // grid: x = query (fast), y = doc-block — adjacent workgroups are
   // the SAME posting sub-range for different queries, so the XCD's L2
   // serves the repeat reads instead of HBM
@@ -96,9 +81,6 @@ __global__ __launch_bounds__(256) void bm25_block_kernel(
   const long d0 = (long)blk * BD;
   const int nd = (int)min((long)BD, nseg - d0);
   for (int i = threadIdx.x; i < nd; i += blockDim.x) lds_scores[i] = 0.0f;
-  for (int i = threadIdx.x; i < 8 * 264; i += blockDim.x) hist1[i] = 0;
-  for (int i = threadIdx.x; i < 256; i += blockDim.x) hist2[i] = 0;
-  if (threadIdx.x < 8) ctl[threadIdx.x] = 0;
   __syncthreads();
   const int t0 = qt_off[q], t1 = qt_off[q + 1];
   const int wave = threadIdx.x / WAVE, lane = threadIdx.x % WAVE;
@@ -120,116 +102,12 @@ __global__ __launch_bounds__(256) void bm25_block_kernel(
     }
   }
   __syncthreads();
-  if (k_sel == 0) {   // profiling mode: accumulate only, emit nothing
-    if (lds_scores[threadIdx.x] > 1e30f)   // never true; defeat DCE
-      out_vals[0] = lds_scores[threadIdx.x];
-    return;
-  }
-
-  // ---- in-LDS 2-level radix select of the block's top-k ------------
-  // BM25 scores are >= 0; ZEROS (unmatched docs, the vast majority of
-  // a block) are skipped everywhere — histogramming them would
-  // serialize thousands of atomics on one bin. The zero prefix is
-  // 0x8000, so "score > 0" == "prefix > 0x8000".
-  const int kq = nd < k_sel ? nd : k_sel;
-  unsigned* my_h1 = hist1 + (threadIdx.x & 7) * 264;
-  for (int i = threadIdx.x; i < nd; i += blockDim.x) {
-    const float v = lds_scores[i];
-    if (v != 0.0f)
-      atomicAdd(&my_h1[float_to_ordered(v) >> 24], 1u);
-  }
-  __syncthreads();
-  if (threadIdx.x < 64) {   // merge the 8 copies (wave 0)
-#pragma unroll
-    for (int b = threadIdx.x; b < 256; b += 64) {
-      unsigned s = hist1[b];
-#pragma unroll
-      for (int c = 1; c < 8; ++c) s += hist1[c * 264 + b];
-      hist1[b] = s;
-    }
-  }
-  __syncthreads();
-  if (threadIdx.x == 0) {
-    unsigned cum = 0;
-    int b1 = -1;
-    for (int i = 255; i > 128; --i) {   // positive floats only
-      const unsigned c = hist1[i];
-      if (cum + c >= (unsigned)kq) { b1 = i; break; }
-      cum += c;
-    }
-    if (b1 < 0) {
-      // fewer than kq nonzero scores: every nonzero is strict, zeros
-      // fill the quota (t16 = the zero prefix)
-      ctl[0] = 0; ctl[1] = cum; ctl[2] = 0x8000u;
-    } else {
-      ctl[0] = (unsigned)b1; ctl[1] = cum; ctl[2] = 0;
-    }
-  }
-  __syncthreads();
-  const unsigned b1 = ctl[0];
-  if (b1) {
-    for (int i = threadIdx.x; i < nd; i += blockDim.x) {
-      const float v = lds_scores[i];
-      if (v == 0.0f) continue;
-      const unsigned o = float_to_ordered(v);
-      if ((o >> 24) == b1) atomicAdd(&hist2[(o >> 16) & 255], 1u);
-    }
-    __syncthreads();
-    if (threadIdx.x == 0) {
-      unsigned cum = ctl[1];
-      for (int i = 255; i >= 0; --i) {
-        const unsigned c = hist2[i];
-        if (cum + c >= (unsigned)kq || i == 0) {
-          ctl[2] = (b1 << 8) | (unsigned)i; break;
-        }
-        cum += c;
-      }
-    }
-    __syncthreads();
-  }
-  const unsigned t16 = ctl[2];
-  const long obase = ((long)q * total_blocks + blk_base + blk)
-      * (long)k_sel;
-  // strictly-above candidates first (provably < kq of them) ...
-  for (int i = threadIdx.x; i < nd; i += blockDim.x) {
-    const float v = lds_scores[i];
-    if (v == 0.0f) continue;
-    const unsigned o = float_to_ordered(v);
-    if ((o >> 16) > t16) {
-      const unsigned pos = atomicAdd(&ctl[3], 1u);
-      if (pos < (unsigned)k_sel) {   // bound guard (denormal edge)
-        out_vals[obase + pos] = v;
-        out_idx[obase + pos] = (int)(doc_base + d0) + i;
-      }
-    }
-  }
-  __syncthreads();
-  const unsigned strict = min(ctl[3], (unsigned)kq);
-  const unsigned need_eq = (unsigned)kq - strict;
-  // ... then threshold-prefix-equal until the quota (interchangeable
-  // at rank k — matching the global selector's tie semantics). The
-  // ctl[4] pre-check gates the atomic once the quota is filled: a
-  // massively tied bin (or the zero-fill case) would otherwise
-  // serialize thousands of increments on one LDS counter.
-  for (int i = threadIdx.x; i < nd; i += blockDim.x) {
-    if (ctl[4] >= need_eq) break;    // racy fast-exit; exact via pos
-    const float v = lds_scores[i];
-    const unsigned o = float_to_ordered(v);
-    if ((o >> 16) == t16) {
-      const unsigned pos = atomicAdd(&ctl[4], 1u);
-      if (pos < need_eq) {
-        out_vals[obase + strict + pos] = v;
-        out_idx[obase + strict + pos] = (int)(doc_base + d0) + i;
-      }
-    }
-  }
-  __syncthreads();
-  const unsigned eq = ctl[4] < need_eq ? ctl[4] : need_eq;
-  const int filled = (int)(strict + eq);
-  for (int j = filled + threadIdx.x; j < k_sel; j += blockDim.x) {
-    out_vals[obase + j] = -INFINITY;
-    out_idx[obase + j] = -1;
-  }
+  // scores are written once and only re-read by the top-k streaming
+  // pass — nontemporal keeps them out of L2, which the posting reads
+  // (shared across adjacent same-block workgroups) actually want
+  float* __restrict__ srow = scores + (long)q * rowN + doc_base + d0;
+  for (int i = threadIdx.x; i < nd; i += blockDim.x)
+    __builtin_nontemporal_store(lds_scores[i], srow + i);
 }
 
 // Fused score-combine: out = wa * a + wb * b (optional linear hybrid;
@@ -248,10 +126,9 @@ __global__ void combine_kernel(const float* __restrict__ a,
 extern "C" void infomesh_bm25_block(
     const void* doc_ids, const void* tfdl, const void* qt_off,
     const void* qt_ut, const void* qt_idf, const void* u_begin,
-    const void* u_end, void* bounds, void* out_vals, void* out_idx,
-    int B, int U, long doc_base, long nseg,
-    int BD, int blk_base, int total_blocks, int k_sel,
-    float norm_a, float norm_b, float k1p1, void* stream) {
+    const void* u_end, void* bounds, void* scores,
+    int B, int U, long rowN, long doc_base, long nseg,
+    int BD, float norm_a, float norm_b, float k1p1, void* stream) {
   if (nseg <= 0 || B <= 0) return;
   auto s = reinterpret_cast<hipStream_t>(stream);
   const int nblocks = (int)((nseg + BD - 1) / BD);
@@ -263,15 +140,14 @@ extern "C" void infomesh_bm25_block(
                        (const long*)u_end, (int*)bounds, U, nblocks, BD);
   }
   dim3 grid((unsigned)B, (unsigned)nblocks);
-  const size_t lds = (size_t)BD * sizeof(float)
-      + (8 * 264 + 256 + 8) * 4;
-  hipLaunchKernelGGL(bm25_block_kernel, grid, dim3(256), lds, s,
+  hipLaunchKernelGGL(bm25_block_kernel, grid, dim3(256),
+                     (size_t)BD * sizeof(float), s,
                      (const int*)doc_ids, (const unsigned int*)tfdl,
                      (const int*)qt_off, (const int*)qt_ut,
                      (const float*)qt_idf, (const long*)u_begin,
-                     (const int*)bounds, (float*)out_vals, (int*)out_idx,
-                     doc_base, nseg, BD, nblocks, blk_base, total_blocks,
-                     k_sel, norm_a, norm_b, k1p1);
+                     (const int*)bounds, (float*)scores,
+                     rowN, doc_base, nseg, BD, nblocks,
+                     norm_a, norm_b, k1p1);
 }
 
 extern "C" void infomesh_score_combine(const void* a, const void* b,

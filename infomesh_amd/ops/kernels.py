@@ -320,46 +320,39 @@ def bm25_block(doc_ids: torch.Tensor, tfdl: torch.Tensor,
                qt_off: torch.Tensor, qt_ut: torch.Tensor,
                qt_idf: torch.Tensor, u_begin: torch.Tensor,
                u_end: torch.Tensor, bounds: torch.Tensor,
-               out_vals: torch.Tensor, out_idx: torch.Tensor,
-               doc_base: int, nseg: int, bd: int,
-               blk_base: int, total_blocks: int, k_sel: int,
-               avgdl: float, k1: float = 1.2,
-               b: float = 0.75) -> None:
-    """Fused BM25 score + per-block top-k for ONE posting segment.
+               scores: torch.Tensor, doc_base: int, nseg: int,
+               bd: int, avgdl: float, k1: float = 1.2,
+               b: float = 0.75) -> torch.Tensor:
+    """Doc-block LDS-accumulated BM25 for ONE posting segment.
 
     Two launches: a bounds pre-pass binary-searching each (unique term,
     doc-block) posting sub-range once, then the block kernel (one
     workgroup per (query, doc-block), query-major so adjacent
-    workgroups reuse posting reads through L2). Each block accumulates
-    its scores in LDS, radix-selects its top-k_sel in LDS, and emits
-    ONLY those candidates to out_vals/out_idx[B, total_blocks*k_sel]
-    at column blk_base*k_sel onward (-inf/-1 padded) — the [B, N]
-    score matrix never exists. Norm is computed in-kernel from the
-    packed per-posting doc length and the current global avgdl.
+    workgroups reuse posting reads through L2). Writes every element of
+    scores[:, doc_base:doc_base+nseg] exactly once (zero where no
+    posting hits) — no pre-zeroing needed when segments partition the
+    doc axis. Norm is computed in-kernel from the packed per-posting
+    doc length and the current global avgdl.
 
     bounds: caller-provided i32 workspace of at least
     U * ceil(nseg/bd) * 2 elements."""
-    B = qt_off.numel() - 1
+    B, N = scores.shape
     U = u_begin.numel()
     nblocks = (nseg + bd - 1) // bd
-    _check(out_vals, torch.float32, "out_vals")
-    assert out_idx.dtype == torch.int32
-    assert out_vals.shape == (B, total_blocks * k_sel)
-    assert nblocks <= 65535 and blk_base + nblocks <= total_blocks
+    _check(scores, torch.float32, "scores")
+    assert nblocks <= 65535 and qt_off.numel() == B + 1
     assert doc_ids.dtype == torch.int32 and tfdl.dtype == torch.int32
     assert bounds.numel() >= U * nblocks * 2
-    assert bd * 4 + 9504 <= 160 * 1024 and k_sel <= bd
-    if k_sel == 0:   # accumulate-only profiling mode
-        pass
+    assert bd * 4 <= 160 * 1024
     norm_a = k1 * (1.0 - b)
     norm_b = k1 * b / max(avgdl, 1e-9)
     _ext.lib().infomesh_bm25_block(
         doc_ids.data_ptr(), tfdl.data_ptr(), qt_off.data_ptr(),
         qt_ut.data_ptr(), qt_idf.data_ptr(), u_begin.data_ptr(),
-        u_end.data_ptr(), bounds.data_ptr(), out_vals.data_ptr(),
-        out_idx.data_ptr(), B, U, doc_base, nseg, bd,
-        blk_base, total_blocks, k_sel,
+        u_end.data_ptr(), bounds.data_ptr(), scores.data_ptr(),
+        B, U, N, doc_base, nseg, bd,
         norm_a, norm_b, k1 + 1.0, _ext.stream_ptr())
+    return scores
 
 
 def score_combine(a: torch.Tensor, b: torch.Tensor, wa: float,

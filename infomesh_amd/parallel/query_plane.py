@@ -96,6 +96,7 @@ class DistributedQueryPlane:
         self.fabric = fabric or Fabric()
         self.k = k_per_shard
         self.fault_stale_s = 5.0
+        self._scores_buf: torch.Tensor | None = None
         self._bm25_stream = None
         # RRF fusion is ~30 tiny fixed-shape kernels (~0.5 ms of launch
         # overhead per batch) -> hipGraph-captured per n_results.
@@ -235,7 +236,7 @@ class DistributedQueryPlane:
             tp = mark("plane.pack", tp)
             hits = self.shard.search(
                 terms, emb if use_dense else None, k=self.k,
-                phase_t=phase_t)
+                scores_buf=self._get_scores_buf(B), phase_t=phase_t)
             tp = mark("plane.shard", tp)
         # all-gather fixed [B,k] blocks -> [W, B, k]
         bm_s = self.fabric.all_gather(hits.bm25_scores)
@@ -312,7 +313,8 @@ class DistributedQueryPlane:
         main = torch.cuda.current_stream(dev)
         self._bm25_stream.wait_stream(main)
         with torch.cuda.stream(self._bm25_stream):
-            bm_vals, bm_idx = self.shard.search_bm25(terms, self.k)
+            bm_vals, bm_idx = self.shard.search_bm25(
+                terms, self.k, scores_buf=self._get_scores_buf(B))
             bm_ids = self.shard.to_global(bm_idx)
         tp = mark("ov.bm25", tp)
         if encode_shard is not None:
@@ -354,3 +356,12 @@ class DistributedQueryPlane:
         tt = terms_t.cpu().numpy()
         return [tt[i][tt[i] >= 0] for i in range(B)]
 
+    def _get_scores_buf(self, B: int) -> torch.Tensor | None:
+        N = self.shard.n_docs
+        if N == 0:
+            return None
+        if (self._scores_buf is None or
+                self._scores_buf.shape != (B, N)):
+            self._scores_buf = torch.zeros(
+                B, N, device=self.shard.device, dtype=torch.float32)
+        return self._scores_buf

@@ -308,10 +308,8 @@ def test_bm25_parity():
     nblocks = (n_docs + bd - 1) // bd
     bounds = torch.empty(len(uterms) * nblocks * 2, dtype=torch.int32,
                          device=dev)
-    k_sel = 64
-    cand_v = torch.full((B, nblocks * k_sel), float("nan"), device=dev)
-    cand_i = torch.full((B, nblocks * k_sel), -7, dtype=torch.int32,
-                        device=dev)
+    # garbage-filled output: the kernel must write every column
+    scores = torch.full((B, n_docs), float("nan"), device=dev)
     K.bm25_block(
         torch.tensor(doc_ids, dtype=torch.int32, device=dev),
         torch.tensor(tfdl, dtype=torch.int32, device=dev),
@@ -322,22 +320,9 @@ def test_bm25_parity():
                                    dtype=np2.int64)).to(dev),
         torch.from_numpy(np2.array([offsets[t + 1] for t in uterms],
                                    dtype=np2.int64)).to(dev),
-        bounds, cand_v, cand_i, doc_base=0, nseg=n_docs, bd=bd,
-        blk_base=0, total_blocks=nblocks, k_sel=k_sel, avgdl=avgdl)
-    assert not torch.isnan(cand_v).any(), "kernel left slots unwritten"
-    # fused per-block top-k: the global top-k over candidates must
-    # match torch.topk over the reference score matrix (values), and
-    # each candidate's value must equal the reference at its index
-    vals, pos = torch.topk(cand_v, k_sel, dim=1)
-    idx = torch.gather(cand_i, 1, pos)
-    rv, _ = torch.topk(ref.to(dev), k_sel, dim=1)
-    _assert_close(vals, rv, rtol=1e-3, atol=1e-3, what="bm25 topk vals")
-    for qi in range(B):
-        v = vals[qi].cpu()
-        i = idx[qi].cpu()
-        valid = i >= 0
-        _assert_close(v[valid], ref[qi][i[valid].long()],
-                      rtol=1e-3, atol=1e-3, what="bm25 cand self")
+        bounds, scores, doc_base=0, nseg=n_docs, bd=bd, avgdl=avgdl)
+    assert not torch.isnan(scores).any(), "kernel left columns unwritten"
+    _assert_close(scores, ref, rtol=1e-3, atol=1e-3, what="bm25")
 
 
 def test_bm25_segmented_matches_merged():
@@ -366,9 +351,8 @@ def test_bm25_segmented_matches_merged():
                for _ in range(8)]
     hi = inc.search(queries, None, k=20)     # BM25-only branch
     hb = bulk.search(queries, None, k=20)
-    # tie membership may differ across block partitions (the fused
-    # selector's accepted semantics) — compare top-k VALUES and check
-    # the ids returned actually carry those scores
+    # tie membership at rank k is arbitrary (the selector's accepted
+    # semantics) — compare top-k VALUES, and require equal result counts
     _assert_close(hi.bm25_scores, hb.bm25_scores, rtol=1e-4, atol=1e-4,
                   what="seg-bm25")
     assert (hi.bm25_ids >= 0).sum() == (hb.bm25_ids >= 0).sum()
