@@ -17,11 +17,21 @@ pmc = tb("rocpd_pmc_event")
 disp = tb("rocpd_kernel_dispatch")
 sym = tb("rocpd_info_kernel_symbol")
 info = tb("rocpd_info_pmc")
+pcols = [c[1] for c in conn.execute(f"PRAGMA table_info({pmc})")]
+dcols = [c[1] for c in conn.execute(f"PRAGMA table_info({disp})")]
+print("# pmc cols:", pcols, file=sys.stderr)
+print("# disp cols:", dcols, file=sys.stderr)
+join_col = next((c for c in ("dispatch_id", "event_id", "id")
+                 if c in pcols), pcols[0])
+d_col = "dispatch_id" if "dispatch_id" in dcols else "id"
+val_col = "value" if "value" in pcols else pcols[-1]
+key_col = next((c for c in ("pmc_id", "counter_id", "info_id")
+                if c in pcols), None)
 rows = conn.execute(f"""
-  SELECT s.kernel_name, i.name, SUM(p.value)
-  FROM {pmc} p JOIN {disp} d ON p.dispatch_id = d.dispatch_id
+  SELECT s.kernel_name, i.name, SUM(p.{val_col})
+  FROM {pmc} p JOIN {disp} d ON p.{join_col} = d.{d_col}
   JOIN {sym} s ON d.kernel_id = s.id
-  JOIN {info} i ON p.pmc_id = i.id
+  JOIN {info} i ON p.{key_col} = i.id
   GROUP BY s.kernel_name, i.name""").fetchall()
 agg = collections.defaultdict(dict)
 for name, counter, val in rows:
