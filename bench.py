@@ -226,10 +226,12 @@ def main():
     sync()
     phase_t.clear()  # drop one-time warmup costs from phase stats
     lat_ms = []
+    submit_ms = []
     t0 = time.perf_counter()
     for i in range(args.steps):
         ts = time.perf_counter()
         step(args.warmup + i)
+        submit_ms.append((time.perf_counter() - ts) * 1e3)
         if use_gpu:
             torch.cuda.synchronize()
         lat_ms.append((time.perf_counter() - ts) * 1e3)
@@ -278,6 +280,9 @@ def main():
                     sorted(lat_ms)[int(len(lat_ms) * 0.95) - 1], 3),
                 "p99_batch_ms": round(
                     sorted(lat_ms)[max(int(len(lat_ms) * 0.99) - 1, 0)], 3),
+                # host submit time of a step (work enqueue until the
+                # final sync): the gap between this and p50 is GPU tail
+                "p50_submit_ms": round(statistics.median(submit_ms), 3),
                 "setup_s": round(setup_s, 1),
             },
         }
