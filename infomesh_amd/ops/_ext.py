@@ -108,10 +108,43 @@ def _try_load() -> ctypes.CDLL | None:
     ws = lib.infomesh_topk_workspace_u32
     ws.argtypes = [c_int]
     ws.restype = c_long
+    import os
+    if os.environ.get("INFOMESH_SYNC_DEBUG"):
+        return _SyncDebugLib(lib)
     return lib
 
 
-def lib() -> ctypes.CDLL:
+class _SyncDebugLib:
+    """Sanitizer-style launch checking (INFOMESH_SYNC_DEBUG=1): after
+    every kernel launch, synchronize the device so an async fault
+    (bad pointer, OOB write surfacing later) raises AT THE LAUNCH SITE
+    with the kernel's name, instead of poisoning a later op. The GPU
+    analogue of the reference's structural concurrency checks
+     — debugging only, serializes every launch."""
+
+    def __init__(self, real: ctypes.CDLL):
+        self._real = real
+
+    def __getattr__(self, name: str):
+        fn = getattr(self._real, name)
+        if not name.startswith("infomesh_"):
+            return fn
+
+        def checked(*args, _fn=fn, _name=name):
+            r = _fn(*args)
+            import torch
+            if torch.cuda.is_available():
+                try:
+                    torch.cuda.synchronize()
+                except RuntimeError as e:
+                    raise RuntimeError(
+                        f"async GPU fault surfaced at {_name}: {e}"
+                    ) from e
+            return r
+        return checked
+
+
+def lib():
     """The loaded extension; raises GPU001 when missing."""
     global _LIB
     if _LIB is None:

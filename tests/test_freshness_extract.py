@@ -256,3 +256,25 @@ def test_dashboard_tab_switching():
     assert next_tab("overview", "h") == TABS[-1]   # wraps
     assert next_tab("overview", "q") is None
     assert next_tab("overview", "x") == "overview"
+
+
+# --------------------------------------------------- sync-debug proxy
+
+def test_sync_debug_lib_proxy(monkeypatch):
+    """INFOMESH_SYNC_DEBUG=1 wraps every kernel entry with a
+    synchronize-and-raise check (sanitizer-style launch validation)."""
+    import importlib
+    from infomesh_amd.ops import _ext as ext
+    monkeypatch.setenv("INFOMESH_SYNC_DEBUG", "1")
+    mod = importlib.reload(ext)
+    try:
+        if not mod.available():
+            pytest.skip("extension not built in this environment")
+        wrapped = mod.lib()
+        assert isinstance(wrapped, mod._SyncDebugLib)
+        fn = wrapped.infomesh_topk_workspace_u32
+        assert callable(fn)
+        assert fn(4) > 0   # passthrough result survives wrapping
+    finally:
+        monkeypatch.delenv("INFOMESH_SYNC_DEBUG")
+        importlib.reload(mod)
