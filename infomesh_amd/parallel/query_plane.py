@@ -238,11 +238,21 @@ class DistributedQueryPlane:
                 terms, emb if use_dense else None, k=self.k,
                 scores_buf=self._get_scores_buf(B), phase_t=phase_t)
             tp = mark("plane.shard", tp)
-        # all-gather fixed [B,k] blocks -> [W, B, k]
-        bm_s = self.fabric.all_gather(hits.bm25_scores)
-        bm_i = self.fabric.all_gather(hits.bm25_ids)
-        dn_s = self.fabric.all_gather(hits.dense_scores)
-        dn_i = self.fabric.all_gather(hits.dense_ids)
+        # ONE packed all-gather instead of four: xGMI collectives at
+        # this payload size (a few hundred KB) are latency-dominated,
+        # so the [B,k] score blocks and the bit-reinterpreted i64 id
+        # blocks ride together -> [W, B, 6k], unpacked after
+        B_, k_ = hits.bm25_scores.shape
+        packed = torch.cat([
+            hits.bm25_scores, hits.dense_scores,
+            hits.bm25_ids.view(torch.float32).reshape(B_, 2 * k_),
+            hits.dense_ids.view(torch.float32).reshape(B_, 2 * k_),
+        ], dim=1)
+        g = self.fabric.all_gather(packed)        # [W, B, 6k]
+        bm_s = g[:, :, :k_].contiguous()
+        dn_s = g[:, :, k_:2 * k_].contiguous()
+        bm_i = g[:, :, 2 * k_:4 * k_].contiguous().view(torch.int64)
+        dn_i = g[:, :, 4 * k_:6 * k_].contiguous().view(torch.int64)
         for tk in (self.shard._get_topk(),
                    getattr(self.shard, "_topk_dense", None)):
             if tk is not None and getattr(tk, "defer_check", False):
