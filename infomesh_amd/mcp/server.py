@@ -65,6 +65,13 @@ class McpServer:
                 result = self._call_tool(params)
             elif method == "resources/list":
                 result = {"resources": RESOURCES}
+            elif method == "resources/templates/list":
+                result = {"resourceTemplates": [
+                    {"uriTemplate": "infomesh://doc/{url}",
+                     "name": "document",
+                     "description": "An indexed document by URL",
+                     "mimeType": "application/json"},
+                ]}
             elif method == "resources/read":
                 result = self._read_resource(params)
             elif method == "prompts/list":

@@ -310,6 +310,11 @@ def test_mcp_resources_and_prompts(tmp_path, monkeypatch):
         import json as _json
         body = _json.loads(rd["result"]["contents"][0]["text"])
         assert "documents" in body or body  # stats dict present
+        rt = srv.handle_message({"jsonrpc": "2.0", "id": 4,
+                                 "method": "resources/templates/list"})
+        tmpl = rt["result"]["resourceTemplates"]
+        assert any(t["uriTemplate"] == "infomesh://doc/{url}"
+                   for t in tmpl)
         pl = srv.handle_message({"jsonrpc": "2.0", "id": 4,
                                  "method": "prompts/list"})
         names = [p["name"] for p in pl["result"]["prompts"]]
