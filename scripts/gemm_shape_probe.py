@@ -13,12 +13,19 @@ import torch
 
 sys.path.insert(0, str(Path(__file__).resolve().parent.parent))
 
+import os
 SHAPES = [
     (4096, 1152, 384),   # encoder QKV
     (4096, 384, 384),    # attn out
     (4096, 1536, 384),   # FFN up
     (4096, 384, 1536),   # FFN down
 ]
+if os.environ.get("PROBE_RERANK"):
+    # bge-reranker-base shapes at 12800 pairs x 160 tokens (M=2.05M);
+    # probe at M/4 to keep memory/time sane (same regime)
+    M = 512_000
+    SHAPES = [(M, 2304, 768), (M, 768, 768), (M, 3072, 768),
+              (M, 768, 3072)]
 
 
 def bench(fn, iters=200):
