@@ -223,8 +223,50 @@ __global__ __launch_bounds__(512, 1) void gemm8_kernel(
   }
 
   // ---- epilogue
+  // bf16 full tiles: stage each wave's 128x64 slab through LDS and
+  // emit 16-B dwordx4 NONTEMPORAL stores. At short K (encoder /
+  // reranker projections, K<=3072) the C write is a large share of
+  // the kernel's traffic, and the scalar 2-B stores ran it at ~1 TB/s
+  // — measured 2.2x behind the vendor library at M=512k,K=768 shapes
+  // before this (profiles/r02_summary.md). The in-flight tail
+  // prefetches must drain first (they write the smem we re-use).
   const int crow0 = (lane >> 4) * 4;
   const int ccol = lane & 15;
+  if (!OUT_F32 && m0 + G8_BM <= M && n0 + G8_BN <= N) {
+    asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    bf16* stage_bf = reinterpret_cast<bf16*>(&smem[0][0])
+        + (long)wid * 128 * 64;
+#pragma unroll
+    for (int i = 0; i < 8; ++i)
+#pragma unroll
+      for (int j = 0; j < 4; ++j) {
+        const int ln = j * 16 + ccol;
+        const float bv = bias ? bias[n0 + wn * 64 + ln] : 0.0f;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int lm = i * 16 + crow0 + r;
+          stage_bf[lm * 64 + ln] =
+              f2bf(apply_act(alpha * acc[i][j][r] + bv, act));
+        }
+      }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_wave_barrier();
+    const int srow = lane >> 3, scol8 = (lane & 7) * 8;  // 8 lanes/row
+#pragma unroll
+    for (int base = 0; base < 128; base += 8) {
+      const int lm = base + srow;
+      const int m = m0 + wm * 128 + lm;
+      const int n = n0 + wn * 64 + scol8;
+      bf16x8 v = *reinterpret_cast<const bf16x8*>(
+          &stage_bf[lm * 64 + scol8]);
+      __builtin_nontemporal_store(
+          v, reinterpret_cast<bf16x8*>(
+              reinterpret_cast<bf16*>(C)
+              + (long)g * strideC + (long)m * N + n));
+    }
+    return;
+  }
 #pragma unroll
   for (int i = 0; i < 8; ++i) {
 #pragma unroll
