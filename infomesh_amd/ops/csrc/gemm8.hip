@@ -23,6 +23,7 @@
 // vector_store.py:120-157) at the shapes where the 128^2 tile is
 // pipeline-bound.
 #include "common.h"
+#include <cstdlib>
 
 #define G8_BM 256
 #define G8_BN 256
@@ -289,6 +290,221 @@ __global__ __launch_bounds__(512, 1) void gemm8_kernel(
   }
 }
 
+// ---- persistent-tile variant -------------------------------------------
+// At short K (encoder/reranker projections: 6-48 K-tiles) the 8-phase
+// schedule's prologue (6 half-stages) + drain serialize per tile — and
+// at 1 block/CU there is no second block to overlap them, costing ~2x
+// vs the vendor library at K=768 (profiles/r02_summary.md). Here ONE
+// resident block per CU walks a contiguous range of tiles and the
+// (tile, kstep) sequence is flattened into one absolute step stream:
+// the double-buffer rotation, counted vmcnt waits and raw barriers
+// carry straight across tile boundaries, so the pipeline ramps once
+// per RANGE instead of once per tile. The epilogue stays scalar (the
+// LDS staging would force a vmcnt(0) drain mid-stream); its stores
+// ride the same vm counter and retire during the next tile's MFMAs.
+template <bool OUT_F32>
+__global__ __launch_bounds__(512, 1) void gemm8p_kernel(
+    const bf16* __restrict__ A, const bf16* __restrict__ B,
+    void* __restrict__ C, const float* __restrict__ bias,
+    int M, int N, int K,
+    long strideA, long strideB, long strideC,
+    int act, float alpha) {
+  __shared__ bf16 smem[2][(G8_BM + G8_BN) * G8_BK];
+
+  const int tiles_n = (N + G8_BN - 1) / G8_BN;
+  const int tiles_m = (M + G8_BM - 1) / G8_BM;
+  const int tiles_total = tiles_m * tiles_n;
+  const int per = (tiles_total + gridDim.x - 1) / gridDim.x;
+  const int t_begin = blockIdx.x * per;
+  const int t_end = min(t_begin + per, tiles_total);
+  if (t_begin >= tiles_total) return;
+  const int g = blockIdx.y;
+  const bf16* Ag = A + (long)g * strideA;
+  const bf16* Bg = B + (long)g * strideB;
+
+  const int lane = threadIdx.x & 63;
+  const int wid = threadIdx.x >> 6;
+  const int wm = wid >> 2, wn = wid & 3;
+  const int n_ksteps = K / G8_BK;
+  const long S_total = (long)(t_end - t_begin) * n_ksteps;
+
+  // absolute step s -> (m0, n0, kstep); prefetch overruns clamp to the
+  // final step so the vmcnt counts stay static
+  auto stage_half = [&](int buf, long s, int half) {
+    if (s >= S_total) s = S_total - 1;
+    const int t = t_begin + (int)(s / n_ksteps);
+    const long k0 = (long)(s % n_ksteps) * G8_BK;
+    const int m0 = (t / tiles_n) * G8_BM;
+    const int n0 = (t % tiles_n) * G8_BN;
+    const bool is_b = half >= 2;
+    const int row_base = (half & 1) * 128;
+    const long region = (is_b ? (long)G8_BM * G8_BK * 2 : 0) +
+                        (long)row_base * 128;
+#pragma unroll
+    for (int c2 = 0; c2 < 2; ++c2) {
+      const int chunk = wid * 2 + c2;
+      const int lin_byte = chunk * 1024 + lane * 16;
+      const int s_byte = swz(lin_byte);
+      const int lrow = s_byte / 128;
+      const int colb = s_byte % 128;
+      int grow = (is_b ? n0 : m0) + row_base + lrow;
+      const int lim = is_b ? N : M;
+      grow = grow < lim ? grow : lim - 1;
+      const bf16* gsrc = (is_b ? Bg : Ag) + (long)grow * K + k0 + colb / 2;
+      auto* dst = (__attribute__((address_space(3))) unsigned int*)
+          ((char*)&smem[buf][0] + region + (long)chunk * 1024);
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) unsigned int*)gsrc,
+          dst, 16, 0, 0);
+    }
+  };
+
+  const int fr = lane & 15;
+  const int fkb = (lane >> 4) * 16;
+  const int crow0 = (lane >> 4) * 4;
+  const int ccol = lane & 15;
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  stage_half(0, 0, 0); stage_half(0, 0, 1);
+  stage_half(0, 0, 2); stage_half(0, 0, 3);
+  stage_half(1, 1, 0); stage_half(1, 1, 1);
+  asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  int cur = 0;
+  for (long s = 0; s < S_total; ++s) {
+    const char* abase = (const char*)&smem[cur][0];
+    const char* bbase = abase + (long)G8_BM * G8_BK * 2;
+    bf16x8 afr[4][2], bfr[2][2];
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        const int row = wm * 128 + i * 16 + fr;
+        afr[i][ks] = *reinterpret_cast<const bf16x8*>(
+            abase + swz(row * 128 + ks * 64 + fkb));
+      }
+#pragma unroll
+    for (int j = 0; j < 2; ++j)
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        const int row = wn * 64 + j * 16 + fr;
+        bfr[j][ks] = *reinterpret_cast<const bf16x8*>(
+            bbase + swz(row * 128 + ks * 64 + fkb));
+      }
+    stage_half(cur ^ 1, s + 1, 2);
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_sched_barrier(0);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < 2; ++j)
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afr[i][ks], bfr[j][ks], acc[i][j], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+
+    bf16x8 bfr2[2][2];
+#pragma unroll
+    for (int j = 0; j < 2; ++j)
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        const int row = wn * 64 + (j + 2) * 16 + fr;
+        bfr2[j][ks] = *reinterpret_cast<const bf16x8*>(
+            bbase + swz(row * 128 + ks * 64 + fkb));
+      }
+    stage_half(cur ^ 1, s + 1, 3);
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_sched_barrier(0);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < 2; ++j)
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks)
+          acc[i][j + 2] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afr[i][ks], bfr2[j][ks], acc[i][j + 2], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        const int row = wm * 128 + (i + 4) * 16 + fr;
+        afr[i][ks] = *reinterpret_cast<const bf16x8*>(
+            abase + swz(row * 128 + ks * 64 + fkb));
+      }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_sched_barrier(0);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < 2; ++j)
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks)
+          acc[i + 4][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afr[i][ks], bfr[j][ks], acc[i + 4][j], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+
+    __builtin_amdgcn_s_barrier();
+    stage_half(cur, s + 2, 0);
+    stage_half(cur, s + 2, 1);
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int j = 0; j < 2; ++j)
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks)
+          acc[i + 4][j + 2] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afr[i][ks], bfr2[j][ks], acc[i + 4][j + 2], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+
+    asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    cur ^= 1;
+
+    if ((s + 1) % n_ksteps == 0) {
+      // tile finished: scalar epilogue (no smem use — the pipeline's
+      // prefetches for the NEXT tile stay in flight), reset acc
+      const int t = t_begin + (int)(s / n_ksteps);
+      const int m0 = (t / tiles_n) * G8_BM;
+      const int n0 = (t % tiles_n) * G8_BN;
+#pragma unroll
+      for (int i = 0; i < 8; ++i) {
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          const int n = n0 + wn * 64 + j * 16 + ccol;
+          if (n >= N) continue;
+          const float bv = bias ? bias[n] : 0.0f;
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const int m = m0 + wm * 128 + i * 16 + crow0 + r;
+            if (m >= M) continue;
+            float v = apply_act(alpha * acc[i][j][r] + bv, act);
+            if (OUT_F32)
+              reinterpret_cast<float*>(C)[
+                  (long)g * strideC + (long)m * N + n] = v;
+            else
+              reinterpret_cast<bf16*>(C)[
+                  (long)g * strideC + (long)m * N + n] = f2bf(v);
+          }
+          acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+        }
+      }
+    }
+  }
+}
+
 }  // namespace
 
 extern "C" void infomesh_gemm8_bf16_nt(
@@ -297,8 +513,32 @@ extern "C" void infomesh_gemm8_bf16_nt(
     long strideA, long strideB, long strideC,
     int act, float alpha, int out_f32, void* stream) {
   const int tiles = ((M + G8_BM - 1) / G8_BM) * ((N + G8_BN - 1) / G8_BN);
-  dim3 grid(tiles, batch), block(512);
   auto s = reinterpret_cast<hipStream_t>(stream);
+  // Short-K regime: per-tile prologue/drain dominates at 1 block/CU —
+  // the persistent variant carries the pipeline across tiles.
+  // INFOMESH_GEMM8P=0 disables, =1 forces for all K.
+  static const int p_ov = [] {
+    const char* e = getenv("INFOMESH_GEMM8P");
+    return e ? atoi(e) : -1;
+  }();
+  const int n_ksteps = K / G8_BK;
+  const bool persistent =
+      (p_ov == 1) || (p_ov != 0 && n_ksteps <= 48 && tiles > 256);
+  if (persistent) {
+    dim3 grid((unsigned)(tiles < 256 ? tiles : 256), batch);
+    if (out_f32)
+      hipLaunchKernelGGL(gemm8p_kernel<true>, grid, dim3(512), 0, s,
+                         (const bf16*)A, (const bf16*)B, C,
+                         (const float*)bias, M, N, K, strideA, strideB,
+                         strideC, act, alpha);
+    else
+      hipLaunchKernelGGL(gemm8p_kernel<false>, grid, dim3(512), 0, s,
+                         (const bf16*)A, (const bf16*)B, C,
+                         (const float*)bias, M, N, K, strideA, strideB,
+                         strideC, act, alpha);
+    return;
+  }
+  dim3 grid(tiles, batch), block(512);
   if (out_f32)
     hipLaunchKernelGGL(gemm8_kernel<true>, grid, block, 0, s,
                        (const bf16*)A, (const bf16*)B, C,
