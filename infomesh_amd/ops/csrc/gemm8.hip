@@ -522,8 +522,13 @@ extern "C" void infomesh_gemm8_bf16_nt(
     return e ? atoi(e) : -1;
   }();
   const int n_ksteps = K / G8_BK;
-  const bool persistent =
-      (p_ov == 1) || (p_ov != 0 && n_ksteps <= 48 && tiles > 256);
+  // Measured WORSE than per-tile blocks at M=512k,K=768 (495 vs
+  // 564 GF/s): the scalar epilogue's stores ride the same vm counter
+  // as the glds span, so every post-boundary vmcnt(4) drains them —
+  // stalling the pipeline it was meant to keep alive — and the
+  // contiguous chunk assignment loses xcd_swizzle's L2 locality.
+  // Kept behind INFOMESH_GEMM8P=1 for future reruns; default OFF.
+  const bool persistent = (p_ov == 1) && n_ksteps >= 1 && tiles >= 1;
   if (persistent) {
     dim3 grid((unsigned)(tiles < 256 ? tiles : 256), batch);
     if (out_f32)
