@@ -14,6 +14,8 @@ from dataclasses import dataclass
 _PRE_CODE_RE = re.compile(
     r"<pre[^>]*>\s*<code([^>]*)>(.*?)</code>\s*</pre>",
     re.IGNORECASE | re.DOTALL)
+_PRE_BARE_RE = re.compile(r"<pre([^>]*)>(.*?)</pre>",
+                          re.IGNORECASE | re.DOTALL)
 _FENCE_RE = re.compile(r"```([A-Za-z0-9_+-]*)\n(.*?)```", re.DOTALL)
 _LANG_ATTR_RE = re.compile(
     r"(?:class|data-lang(?:uage)?)=[\"'][^\"']*?"
@@ -65,8 +67,10 @@ def _clean(fragment: str) -> str:
 
 
 def extract_code_blocks(text: str, max_blocks: int = 50) -> list[CodeBlock]:
-    """<pre><code class="language-x"> blocks plus markdown fences."""
+    """<pre><code class="language-x"> blocks, bare <pre> blocks, plus
+    markdown fences."""
     out: list[CodeBlock] = []
+    spans: list[tuple[int, int]] = []
     for m in _PRE_CODE_RE.finditer(text):
         attrs, body = m.group(1), m.group(2)
         lang = ""
@@ -76,6 +80,15 @@ def extract_code_blocks(text: str, max_blocks: int = 50) -> list[CodeBlock]:
         code = html.unescape(_TAG_RE.sub("", body)).strip("\n")
         if code.strip():
             out.append(CodeBlock(code=code, language=lang))
+            spans.append(m.span())
+        if len(out) >= max_blocks:
+            return out
+    for m in _PRE_BARE_RE.finditer(text):
+        if any(a <= m.start() < b for a, b in spans):
+            continue   # already captured as <pre><code>
+        code = html.unescape(_TAG_RE.sub("", m.group(2))).strip("\n")
+        if code.strip():
+            out.append(CodeBlock(code=code, language=""))
         if len(out) >= max_blocks:
             return out
     for m in _FENCE_RE.finditer(text):

@@ -55,31 +55,28 @@ def extract_open_graph(html: str) -> dict[str, str]:
 
 
 def extract_code_blocks(html: str, max_blocks: int = 20) -> list[str]:
+    """String view over content_extract's typed extraction (one
+    implementation; content_extract.py carries language detection,
+    markdown fences and the reference's richer surface)."""
+    from .content_extract import extract_code_blocks as _rich
     out = []
-    for m in _CODE_RE.finditer(html):
-        code = _TAG_RE.sub("", m.group(1) or m.group(2) or "").strip()
-        if len(code) >= 20:
-            out.append(code[:5000])
-        if len(out) >= max_blocks:
-            break
-    return out
+    for cb in _rich(html, max_blocks=max_blocks):
+        if len(cb.code) >= 20:
+            out.append(cb.code[:5000])
+    return out[:max_blocks]
 
 
 def extract_tables(html: str, max_tables: int = 5) -> list[list[list[str]]]:
-    tables = []
-    for m in _TABLE_RE.finditer(html):
-        rows = []
-        for row_html in re.findall(r"<tr[^>]*>(.*?)</tr>", m.group(1),
-                                   re.S | re.I):
-            cells = [_TAG_RE.sub("", c).strip() for c in re.findall(
-                r"<t[hd][^>]*>(.*?)</t[hd]>", row_html, re.S | re.I)]
-            if cells:
-                rows.append(cells)
+    """Rows-of-cells view over content_extract's typed tables (headers
+    included as the first row when present)."""
+    from .content_extract import extract_tables as _rich
+    out: list[list[list[str]]] = []
+    for t in _rich(html, max_tables=max_tables):
+        rows = ([list(t.headers)] if t.headers else []) \
+            + [list(r) for r in t.rows]
         if rows:
-            tables.append(rows)
-        if len(tables) >= max_tables:
-            break
-    return tables
+            out.append(rows)
+    return out[:max_tables]
 
 
 def extract_structured(html: str) -> StructuredData:
