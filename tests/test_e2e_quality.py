@@ -47,6 +47,45 @@ CORPUS = [
      "via reciprocal rank fusion improves recall."),
 ]
 
+CORPUS += [
+    ("https://rocm.docs/rocprof", "Profiling with rocprofv3",
+     "rocprofv3 collects kernel traces and performance counters such as "
+     "MFMA utilization, LDS bank conflicts and wave cycles on AMD GPUs."),
+    ("https://rocm.docs/hbm", "HBM3E memory subsystem",
+     "HBM3E stacks deliver eight terabytes per second of bandwidth; "
+     "kernels should stream coalesced reads and avoid random writes."),
+    ("https://python.docs/threading", "threading — thread-based parallelism",
+     "Threads share memory under the global interpreter lock; use locks "
+     "queues and events to coordinate concurrent workers in Python."),
+    ("https://python.docs/json", "json — JSON encoder and decoder",
+     "The json module serializes Python objects to strings and parses "
+     "JSON documents with loads and dumps functions."),
+    ("https://web.example/robots", "Robots exclusion protocol",
+     "Crawlers fetch robots.txt to learn disallowed paths, crawl delay "
+     "and sitemap locations before requesting pages from a site."),
+    ("https://web.example/rss", "RSS and Atom feeds",
+     "Feeds syndicate new articles; aggregators poll feed XML, parse "
+     "items with titles links and dates, and schedule fresh fetches."),
+    ("https://ir.example/rrf", "Reciprocal rank fusion",
+     "RRF merges ranked lists by summing one over k plus rank, a robust "
+     "fusion baseline that needs no score calibration between systems."),
+    ("https://ir.example/ndcg", "Evaluating search with NDCG",
+     "Normalized discounted cumulative gain scores graded relevance "
+     "with a logarithmic position discount; MRR tracks the first hit."),
+    ("https://ml.example/transformer", "Transformer encoders",
+     "Self attention mixes token representations with query key value "
+     "projections, layer norm and feed forward networks per layer."),
+    ("https://ml.example/quantize", "Quantization for inference",
+     "Low precision formats like bf16 and fp8 shrink models and raise "
+     "matrix throughput on tensor core hardware with minimal loss."),
+    ("https://cooking.example/curry", "Weeknight lentil curry",
+     "Simmer red lentils with onion garlic ginger turmeric and coconut "
+     "milk, finish with lime and cilantro for a quick dinner."),
+    ("https://gardening.example/tomato", "Growing tomatoes",
+     "Tomato seedlings need warmth, staking, consistent watering and "
+     "pruning of suckers to set heavy trusses of fruit."),
+]
+
 QUERIES = [
     ("writing hip kernels mfma lds", "https://rocm.docs/hip-kernels"),
     ("rccl all-gather xgmi collectives", "https://rocm.docs/rccl-guide"),
@@ -55,6 +94,15 @@ QUERIES = [
     ("carbonara recipe eggs pecorino", "https://cooking.example/pasta"),
     ("bm25 term frequency ranking", "https://ir.example/bm25"),
     ("dense retrieval cosine embeddings", "https://ir.example/dense"),
+    ("rocprofv3 performance counters", "https://rocm.docs/rocprof"),
+    ("hbm3e bandwidth coalesced", "https://rocm.docs/hbm"),
+    ("python threading locks queues", "https://python.docs/threading"),
+    ("robots.txt crawl delay sitemap", "https://web.example/robots"),
+    ("rss atom feed aggregator poll", "https://web.example/rss"),
+    ("reciprocal rank fusion merge lists", "https://ir.example/rrf"),
+    ("ndcg graded relevance discount", "https://ir.example/ndcg"),
+    ("transformer self attention layer norm", "https://ml.example/transformer"),
+    ("lentil curry coconut turmeric", "https://cooking.example/curry"),
 ]
 
 
