@@ -16,15 +16,40 @@ from .gpu_index import BM25_VOCAB, GpuShard
 def synth_corpus_arrays(n_docs: int, avg_len: int = 120,
                         vocab: int = BM25_VOCAB, seed: int = 0,
                         zipf_a: float = 1.3):
-    """Returns (flat_terms, flat_docs, doc_lens) numpy arrays."""
+    """Returns (flat_terms, flat_docs, doc_lens) numpy arrays.
+
+    Zipf term sampling dominates generation at 10M docs (~110 s
+    single-threaded for 567M draws); large corpora fan the draw out
+    over worker processes (deterministic per (seed, chunk), same
+    distribution — numpy's exact rejection sampler throughout)."""
     rng = np.random.default_rng(seed)
     doc_lens = np.clip(
         rng.lognormal(mean=np.log(avg_len), sigma=0.4, size=n_docs),
         8, avg_len * 6).astype(np.int64)
     total = int(doc_lens.sum())
-    flat_terms = (rng.zipf(zipf_a, size=total) - 1) % vocab
+    if total >= 50_000_000:
+        import multiprocessing as mp
+        import os
+        workers = min(16, os.cpu_count() or 4)
+        per = (total + workers - 1) // workers
+        sizes = [min(per, total - i * per) for i in range(workers)]
+        sizes = [s for s in sizes if s > 0]
+        jobs = [(seed * 1_000_003 + 17 * i + 1, s, zipf_a, vocab)
+                for i, s in enumerate(sizes)]
+        # spawn, not fork: the caller may hold a live HIP/RCCL context
+        # (multi-rank bench generates after device binding) and forked
+        # children of a CUDA process are a known hazard even when they
+        # never touch the GPU themselves
+        from ._synthworker import zipf_chunk
+        ctx = mp.get_context("spawn")
+        with ctx.Pool(len(jobs)) as pool:
+            parts = pool.map(zipf_chunk, jobs)
+        flat_terms = np.concatenate(parts)
+    else:
+        flat_terms = ((rng.zipf(zipf_a, size=total) - 1)
+                      % vocab).astype(np.int64)
     flat_docs = np.repeat(np.arange(n_docs, dtype=np.int64), doc_lens)
-    return flat_terms.astype(np.int64), flat_docs, doc_lens
+    return flat_terms, flat_docs, doc_lens
 
 
 def synth_embeddings(n_docs: int, dim: int = 384, seed: int = 0,
