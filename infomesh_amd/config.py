@@ -111,6 +111,8 @@ class ApiConfig:
     port: int = 8080
     api_key: str = ""
     rate_limit_per_min: int = 120
+    # OTLP/HTTP collector base url ("http://host:4318"); empty = off
+    otlp_endpoint: str = ""
 
 
 @dataclass(frozen=True)

@@ -70,6 +70,7 @@ class AppContext:
     # index_document refuses writes at READ_ONLY (reference
     # governor.py:49-74 semantics)
     governor: Any = None
+    otlp: Any = None   # utils.observability.OtlpExporter (off by default)
 
     # ------------------------------------------------------------ build
     @classmethod
@@ -111,6 +112,9 @@ class AppContext:
         )
         from .utils.governor import ResourceGovernor
         ctx.governor = ResourceGovernor()
+        if cfg.api.otlp_endpoint:
+            from .utils.observability import OtlpExporter
+            ctx.otlp = OtlpExporter(endpoint=cfg.api.otlp_endpoint)
         if with_worker if with_worker is not None else role in ("full", "crawler"):
             ctx.worker = CrawlWorker(
                 cfg.crawl,
@@ -299,9 +303,18 @@ class AppContext:
         """GPU-plane search via the micro-batcher (fused once on the
         plane, hydrated batch-wide in _batch_execute)."""
         t0 = time.time()
+        trace = None
+        if self.otlp is not None and self.otlp.enabled:
+            from .utils.observability import QueryTrace
+            trace = QueryTrace(query)
         eff = preprocess_query(query)
         batcher = self.ensure_batcher()
-        if batcher is not None:
+        if trace is not None:
+            with trace.span("engine"):
+                hits = (batcher.submit(eff, limit) if batcher is not None
+                        else self._batch_execute([eff], limit)[0])
+            self.otlp.export(trace)
+        elif batcher is not None:
             hits = batcher.submit(eff, limit)
         else:
             hits = self._batch_execute([eff], limit)[0]

@@ -278,3 +278,27 @@ def test_sync_debug_lib_proxy(monkeypatch):
     finally:
         monkeypatch.delenv("INFOMESH_SYNC_DEBUG")
         importlib.reload(mod)
+
+
+def test_otlp_wired_into_search(monkeypatch):
+    """With api.otlp_endpoint configured, every engine search exports a
+    trace to the collector (batched in the exporter)."""
+    import dataclasses
+    from infomesh_amd.config import Config
+    from infomesh_amd.index.local_store import Document
+    from infomesh_amd.services import AppContext
+    cfg = Config(api=dataclasses.replace(Config().api,
+                                         otlp_endpoint="http://c:4318"))
+    ctx = AppContext.create(config=cfg, with_engine=True,
+                            with_worker=False, in_memory=True)
+    assert ctx.otlp is not None and ctx.otlp.enabled
+    posted = []
+    ctx.otlp._post = lambda url, payload: posted.append(url)
+    ctx.otlp.batch_size = 1
+    ctx.index_document(Document(url="https://t/1", title="T",
+                                text="otlp traced search body"),
+                       attest=False, credit=False)
+    ctx.flush_engine()
+    ctx.search("traced search", use_cache=False, deduct=False)
+    assert posted and posted[0].endswith("/v1/traces")
+    ctx.close()
