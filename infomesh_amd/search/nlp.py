@@ -172,24 +172,31 @@ def parse_query_filters(query: str) -> ParsedQuery:
 @dataclass
 class RelatedSearchTracker:
     """Session-local co-occurrence tracker for `related searches`
-    (reference: nlp.py:972+)."""
+    (reference: nlp.py:972+). Locked: the MCP HTTP transport records
+    from many handler threads, and an unguarded OrderedDict iteration
+    concurrent with a record() raises mid-request."""
     max_entries: int = 1000
     _recent: OrderedDict = field(default_factory=OrderedDict)
+    _lock: "threading.Lock" = field(
+        default_factory=lambda: __import__("threading").Lock())
 
     def record(self, query: str) -> None:
         q = query.strip().lower()
         if not q:
             return
-        self._recent[q] = self._recent.pop(q, 0) + 1
-        while len(self._recent) > self.max_entries:
-            self._recent.popitem(last=False)
+        with self._lock:
+            self._recent[q] = self._recent.pop(q, 0) + 1
+            while len(self._recent) > self.max_entries:
+                self._recent.popitem(last=False)
 
     def related(self, query: str, limit: int = 5) -> list[str]:
         terms = set(query.lower().split())
         if not terms:
             return []
+        with self._lock:
+            snapshot = list(self._recent.items())
         scored = []
-        for past, cnt in self._recent.items():
+        for past, cnt in snapshot:
             if past == query.lower():
                 continue
             overlap = len(terms & set(past.split()))
