@@ -342,3 +342,45 @@ def test_tokenizer_and_scorer_plugin_slots():
         GLOBAL_PLUGINS._hooks.clear()
     assert ScoreBreakdown(bm25=1.0, freshness=0, trust=0, authority=0,
                           title_match=0, url_path=0).total > 0
+
+
+def test_crawl_loop_drains_priority_queue_first(tmp_data_dir):
+    """URLs enqueued with triggers (RSS/user) are crawled ahead of the
+    BFS scheduler (reference freshness.py:67-212 behavior)."""
+    import dataclasses
+    from infomesh_amd.config import Config, CrawlConfig
+    from infomesh_amd.crawler.crawl_loop import seed_and_crawl_loop
+    from infomesh_amd.crawler.freshness import (PriorityRecrawlQueue,
+                                                RecrawlTrigger)
+    from infomesh_amd.services import AppContext
+
+    crawled: list[str] = []
+
+    def handler(request):
+        crawled.append(str(request.url))
+        return httpx.Response(
+            200, text="<html><head><title>P</title></head><body><p>" +
+                      "priority queue crawl body text. " * 6 +
+                      "</p></body></html>",
+            headers={"content-type": "text/html"})
+
+    cfg = dataclasses.replace(Config(), crawl=CrawlConfig(
+        politeness_delay_s=0, respect_robots=False,
+        max_urls_per_hour=1000))
+    ctx = AppContext.create(config=cfg, with_engine=False,
+                            with_worker=True, in_memory=True)
+    ctx.worker._client = httpx.AsyncClient(
+        transport=httpx.MockTransport(handler))
+    ctx.worker.resolve_dns = False
+    pq = ctx.recrawl_queue = PriorityRecrawlQueue()
+    pq.enqueue("https://feed.example/fresh-item",
+               RecrawlTrigger.RSS_UPDATE, source="https://feed.example/rss")
+    pq.enqueue("https://user.example/forced", RecrawlTrigger.USER_REQUEST)
+
+    stats = asyncio.run(seed_and_crawl_loop(ctx, max_iterations=3))
+    assert stats["crawled"] >= 2
+    # priority items beat the seed list, USER_REQUEST first
+    assert crawled[0].startswith("https://user.example/")
+    assert crawled[1].startswith("https://feed.example/")
+    assert len(pq) == 0 and pq.total_dequeued == 2
+    ctx.close()
