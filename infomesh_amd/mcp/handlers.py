@@ -323,10 +323,15 @@ class Handlers:
                 "deletion_recorded": True}
 
     def call(self, tool: str, args: dict[str, Any]) -> dict[str, Any]:
-        from .tools import resolve_tool
+        from .tools import resolve_tool, validate_args
         resolved = resolve_tool(tool)
         if resolved is None:
             raise InfoMeshError("SRCH001", f"unknown tool {tool!r}")
+        violations = validate_args(tool, args)
+        if violations:
+            raise InfoMeshError("SRCH001",
+                                "invalid arguments: " + "; ".join(
+                                    violations[:5]))
         # legacy arg adaptation
         if tool == "explain":
             args = {**args, "explain": True}

@@ -541,3 +541,36 @@ def test_mcp_suggest_tool(tmp_path, monkeypatch):
         assert payload["suggestions"] == ["Quantum computing primer"]
     finally:
         ctx.close()
+
+
+def test_tool_argument_validation():
+    """Tool calls are validated against their declared inputSchema
+    (types, bounds, string/list caps) before dispatch."""
+    from infomesh_amd.mcp.tools import validate_args
+    assert validate_args("web_search", {"query": "ok", "limit": 5}) == []
+    assert validate_args("web_search", {"query": ""})      # missing/empty
+    assert validate_args("web_search", {"query": 42})      # wrong type
+    assert validate_args("web_search", {"query": "x" * 20_000})
+    assert validate_args("crawl_url", {"url": "https://a", "depth": 99})
+    assert validate_args("crawl_url", {"url": "https://a", "depth": 1}) == []
+    assert validate_args("batch_search",
+                         {"queries": ["a"] * 2000})        # list cap
+
+
+def test_handlers_reject_invalid_args(mcp_ctx=None):
+    import pytest as _pt
+    from infomesh_amd.config import Config
+    from infomesh_amd.mcp.handlers import Handlers
+    from infomesh_amd.services import AppContext
+    ctx = AppContext.create(config=Config(), with_engine=False,
+                            with_worker=False, in_memory=True)
+    h = Handlers(ctx)
+    out = None
+    try:
+        out = h.call("web_search", {"query": 42})
+    except Exception as e:
+        out = {"error": str(e)}
+    assert out and "error" in out or "invalid" in str(out).lower()
+    ok = h.call("status", {})
+    assert "error" not in ok
+    ctx.close()
