@@ -43,6 +43,9 @@ def parse_args():
     p.add_argument("--avg-doc-len", type=int, default=120)
     p.add_argument("--query-len", type=int, default=32)
     p.add_argument("--bm25-only", action="store_true")
+    p.add_argument("--emb-dtype", choices=("bf16", "fp8"), default="bf16",
+                   help="dense-plane embedding storage (fp8 = OCP e4m3, "
+                        "half HBM + half read traffic, opt-in)")
     p.add_argument("--rerank", action="store_true",
                    help="add cross-encoder top-100->10 (BASELINE config 3)")
     p.add_argument("--rag", action="store_true",
@@ -81,7 +84,7 @@ def main():
         shard = build_synth_shard(
             docs_per_gpu, shard_rank=rank, world=world,
             avg_len=args.avg_doc_len, device=device, seed=args.seed,
-            with_dense=not args.bm25_only)
+            with_dense=not args.bm25_only, emb_dtype=args.emb_dtype)
     else:
         from infomesh_amd.index.gpu_index import CpuShard
         from infomesh_amd.index.synth import synth_corpus_arrays
@@ -260,7 +263,8 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "bf16" if use_gpu else "fp32-cpu-sanity",
+            "dtype": (("bf16+fp8emb" if args.emb_dtype == "fp8"
+                       else "bf16") if use_gpu else "fp32-cpu-sanity"),
             "data": "synthetic",
             "config": {
                 "model": "bge-small-en(random-init) encoder + "

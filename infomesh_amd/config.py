@@ -22,7 +22,7 @@ log = logging.getLogger("infomesh.config")
 
 NODE_ROLES = ("full", "crawler", "search")
 FTS_TOKENIZERS = ("unicode61", "ascii", "porter", "trigram")
-COMPUTE_DTYPES = ("bf16", "fp16", "fp32")
+COMPUTE_DTYPES = ("bf16", "fp16", "fp32", "fp8")
 
 
 def default_data_dir() -> Path:

@@ -27,13 +27,15 @@ log = logging.getLogger("infomesh.engine")
 class HybridEngine:
     def __init__(self, device: str | None = None, k_per_shard: int = 100,
                  use_encoder: bool = True, encoder_max_len: int = 128,
-                 fabric: Fabric | None = None):
+                 fabric: Fabric | None = None, emb_dtype: str = "bf16"):
         self.gpu = torch.cuda.is_available() if device is None \
             else device.startswith("cuda")
         self.device = device or ("cuda" if self.gpu else "cpu")
         self.fabric = fabric or Fabric()
-        self.shard: GpuShard = (GpuShard(self.device) if self.gpu
-                                else CpuShard())
+        self.emb_dtype = emb_dtype
+        self.shard: GpuShard = (
+            GpuShard(self.device, emb_dtype=emb_dtype) if self.gpu
+            else CpuShard(emb_dtype=emb_dtype))
         self.plane = DistributedQueryPlane(self.shard, self.fabric,
                                            k_per_shard=k_per_shard)
         self.encoder = None
@@ -102,7 +104,8 @@ class HybridEngine:
                           else np.zeros(0, np.int64))
             flat_docs = np.repeat(np.arange(len(tokens), dtype=np.int64),
                                   [len(t) for t in tokens])
-            shard = GpuShard(self.device) if self.gpu else CpuShard()
+            shard = (GpuShard(self.device, emb_dtype=self.emb_dtype)
+                     if self.gpu else CpuShard(emb_dtype=self.emb_dtype))
             shard.build_from_arrays(
                 flat_terms, flat_docs, lens,
                 np.asarray(ids, dtype=np.int64), new_emb)

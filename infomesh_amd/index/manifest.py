@@ -46,6 +46,7 @@ def save_shard(shard: GpuShard, path: str | Path, rank: int = 0,
         "rank": rank, "world": world,
         "n_docs": shard.n_docs, "vocab": shard.vocab,
         "avgdl": shard.avgdl,
+        "emb_dtype": shard.emb_dtype,
         "n_segments": len(shard.segments),
         "created_at": time.time(),
     }
@@ -75,6 +76,7 @@ def load_shard(path: str | Path, device: str = "cuda") -> GpuShard:
     shard.n_docs = meta["n_docs"]
     shard.vocab = meta["vocab"]
     shard.avgdl = meta["avgdl"]
+    shard.emb_dtype = meta.get("emb_dtype", "bf16")
     shard.df = np.asarray(blob["df"], dtype=np.int64)
     shard._doc_lens = np.asarray(blob["doc_lens"], dtype=np.int64)
     dev = shard.device

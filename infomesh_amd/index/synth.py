@@ -68,7 +68,8 @@ def synth_embeddings(n_docs: int, dim: int = 384, seed: int = 0,
 def build_synth_shard(n_docs: int, shard_rank: int = 0, world: int = 1,
                       avg_len: int = 120, dim: int = 384,
                       device: str = "cuda", seed: int = 0,
-                      with_dense: bool = True) -> GpuShard:
+                      with_dense: bool = True,
+                      emb_dtype: str = "bf16") -> GpuShard:
     """Build one GPU shard of a world-sharded synthetic corpus.
 
     Global ids interleave round-robin (gid = local * world + rank) so
@@ -78,7 +79,7 @@ def build_synth_shard(n_docs: int, shard_rank: int = 0, world: int = 1,
     emb = synth_embeddings(n_docs, dim, seed=seed * 1000 + shard_rank,
                            device=device) if with_dense else None
     gids = np.arange(n_docs, dtype=np.int64) * world + shard_rank
-    shard = GpuShard(device=device)
+    shard = GpuShard(device=device, emb_dtype=emb_dtype)
     shard.build_from_arrays(terms, docs, lens, gids, emb)
     return shard
 

@@ -68,6 +68,9 @@ _SIGNATURES: dict[str, list] = {
                            c_void_p, c_int, c_int, c_int, c_int, c_void_p],
     "infomesh_dense_scores": [c_void_p, c_void_p, c_void_p,
                               c_int, c_long, c_int, c_float, c_void_p],
+    "infomesh_dense_scores_fp8": [c_void_p, c_void_p, c_void_p,
+                                  c_int, c_long, c_int, c_float,
+                                  c_void_p],
     "infomesh_gemm8_bf16_nt": [c_void_p, c_void_p, c_void_p, c_void_p,
                                c_int, c_int, c_int, c_int,
                                c_long, c_long, c_long,
@@ -104,7 +107,9 @@ def _try_load() -> ctypes.CDLL | None:
         fn = getattr(lib, name)
         fn.argtypes = argtypes
         # dense_scores reports dispatch eligibility (0 = launched)
-        fn.restype = c_int if name == "infomesh_dense_scores" else None
+        fn.restype = c_int if name in ("infomesh_dense_scores",
+                                       "infomesh_dense_scores_fp8") \
+            else None
     ws = lib.infomesh_topk_workspace_u32
     ws.argtypes = [c_int]
     ws.restype = c_long

@@ -101,6 +101,28 @@ def dense_scores(a: torch.Tensor, b: torch.Tensor, alpha: float = 1.0,
     return out if rc == 0 else None
 
 
+def dense_scores_fp8(a: torch.Tensor, b: torch.Tensor,
+                     alpha: float = 1.0,
+                     out: torch.Tensor | None = None
+                     ) -> torch.Tensor | None:
+    """FP8 (OCP e4m3) streaming dense-score GEMM (densescore8.hip):
+    C[M,N] f32 = A[M,K] @ B[N,K]^T with fp8 operands — half the read
+    traffic and half the HBM footprint of the bf16 plane (the plane is
+    bandwidth-bound; non-scaled fp8 MFMA matches the bf16 rate). Both
+    operands must be torch.float8_e4m3fn. Returns None when the shape
+    is ineligible."""
+    M, K = a.shape
+    N = b.shape[0]
+    assert a.dtype == torch.float8_e4m3fn and a.is_cuda and a.is_contiguous()
+    assert b.dtype == torch.float8_e4m3fn and b.is_contiguous()
+    if out is None:
+        out = torch.empty(M, N, device=a.device, dtype=torch.float32)
+    rc = _ext.lib().infomesh_dense_scores_fp8(
+        a.data_ptr(), b.data_ptr(), out.data_ptr(), M, N, K, alpha,
+        _ext.stream_ptr())
+    return out if rc == 0 else None
+
+
 def layernorm(x: torch.Tensor, gamma: torch.Tensor, beta: torch.Tensor,
               residual: torch.Tensor | None = None, eps: float = 1e-12,
               return_residual: bool = False):

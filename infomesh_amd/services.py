@@ -127,7 +127,9 @@ class AppContext:
             else role in ("full", "search")
         if want_engine:
             try:
-                ctx.engine = HybridEngine(k_per_shard=cfg.search.max_results_per_shard)
+                ctx.engine = HybridEngine(
+                    k_per_shard=cfg.search.max_results_per_shard,
+                    emb_dtype="fp8" if cfg.gpu.dtype == "fp8" else "bf16")
             except Exception as e:
                 log.warning("engine unavailable: %s", e)
         if ctx.engine is not None:
