@@ -337,3 +337,25 @@ def test_bm25_term_ids_unicode_folding():
     shard.build()
     hits = shard.search([bm25_term_ids("квантовый")], None, k=2)
     assert int(hits.bm25_ids[0, 0]) == 1
+
+
+def test_cpu_shard_fp8_mode():
+    """fp8 embedding storage works on the CPU oracle too (upcast
+    scoring), so the mode is testable without a GPU."""
+    import numpy as np
+    import torch
+    from infomesh_amd.index.gpu_index import CpuShard
+    rng = np.random.default_rng(3)
+    docs = [rng.integers(0, 200, size=10).astype(np.int64)
+            for _ in range(50)]
+    g = torch.Generator().manual_seed(3)
+    emb = torch.nn.functional.normalize(
+        torch.randn(50, 32, generator=g), dim=-1)
+    shard = CpuShard(emb_dtype="fp8")
+    for i, d in enumerate(docs):
+        shard.add_document(i, d, emb[i])
+    shard.build()
+    assert shard.embeddings.dtype == torch.float8_e4m3fn
+    hits = shard.search([docs[7][:3]], emb[7:8], k=5)
+    assert int(hits.dense_ids[0, 0]) == 7   # self-retrieval
+    assert float(hits.dense_scores[0, 0]) > 0.9
