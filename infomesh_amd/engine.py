@@ -27,12 +27,14 @@ log = logging.getLogger("infomesh.engine")
 class HybridEngine:
     def __init__(self, device: str | None = None, k_per_shard: int = 100,
                  use_encoder: bool = True, encoder_max_len: int = 128,
-                 fabric: Fabric | None = None, emb_dtype: str = "bf16"):
+                 fabric: Fabric | None = None, emb_dtype: str = "bf16",
+                 hbm_budget_gb: float = 260.0):
         self.gpu = torch.cuda.is_available() if device is None \
             else device.startswith("cuda")
         self.device = device or ("cuda" if self.gpu else "cpu")
         self.fabric = fabric or Fabric()
         self.emb_dtype = emb_dtype
+        self.hbm_budget_bytes = int(hbm_budget_gb * 1e9)
         self.shard: GpuShard = (
             GpuShard(self.device, emb_dtype=emb_dtype) if self.gpu
             else CpuShard(emb_dtype=emb_dtype))
@@ -74,6 +76,22 @@ class HybridEngine:
         if not self._pending_ids:
             return 0
         n_new = len(self._pending_ids)
+        # HBM budget guard (config gpu.hbm_budget_gb — the 288 GB/GPU
+        # sizing knob): estimate the new segment before committing and
+        # refuse the flush instead of tripping the allocator mid-build.
+        # Pendings stay pending; the crawl loop logs and retries later
+        # (the reference governor's degrade-before-OOM behavior).
+        esize = 1 if self.emb_dtype == "fp8" else 2
+        est_new = sum(len(t) for t in self._pending_tokens) * 8 \
+            + (n_new * 384 * esize if self.encoder is not None else 0) \
+            + n_new * 8
+        if self.shard.hbm_bytes() + est_new > self.hbm_budget_bytes:
+            from .errors import InfoMeshError
+            raise InfoMeshError(
+                "GPU002",
+                f"HBM budget exceeded: shard {self.shard.hbm_bytes()/1e9:.1f}"
+                f" GB + ~{est_new/1e9:.2f} GB new > "
+                f"{self.hbm_budget_bytes/1e9:.0f} GB budget")
         # Encode BEFORE committing any state: a failed embed (OOM, ...)
         # leaves the engine exactly as it was — the old epoch keeps
         # serving and the pending docs stay pending for a retry.

@@ -129,7 +129,8 @@ class AppContext:
             try:
                 ctx.engine = HybridEngine(
                     k_per_shard=cfg.search.max_results_per_shard,
-                    emb_dtype="fp8" if cfg.gpu.dtype == "fp8" else "bf16")
+                    emb_dtype="fp8" if cfg.gpu.dtype == "fp8" else "bf16",
+                    hbm_budget_gb=cfg.gpu.hbm_budget_gb)
             except Exception as e:
                 log.warning("engine unavailable: %s", e)
         if ctx.engine is not None:

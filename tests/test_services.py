@@ -273,3 +273,25 @@ def test_plugin_hooks_fire_in_real_paths(tmp_path, monkeypatch):
             ctx.close()
     finally:
         GLOBAL_PLUGINS._hooks.clear()
+
+
+def test_engine_hbm_budget_guard():
+    """gpu.hbm_budget_gb refuses a flush that would overflow HBM,
+    leaving pendings intact (degrade-before-OOM, reference governor
+    RSS-limit analogue)."""
+    import numpy as np
+    import pytest as _pt
+    from infomesh_amd.engine import HybridEngine
+    from infomesh_amd.errors import InfoMeshError
+    from infomesh_amd.index.local_store import Document
+    eng = HybridEngine(device="cpu", use_encoder=False,
+                       hbm_budget_gb=1e-6)   # ~1 KB budget
+    d = Document(url="https://x/1", title="T",
+                 text="budget guard test body " * 50)
+    d.doc_id = 1
+    eng.add_document(d)
+    with _pt.raises(InfoMeshError):
+        eng.flush()
+    assert eng.pending_count == 1          # nothing lost
+    eng.hbm_budget_bytes = 10**9
+    assert eng.flush() == 1                # retry succeeds
