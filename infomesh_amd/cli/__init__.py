@@ -235,8 +235,10 @@ def export(path, max_docs):
     from ..index.snapshot import export_snapshot
     ctx = _ctx(with_worker=False, with_engine=False)
     try:
-        header = export_snapshot(ctx.store, path, max_docs=max_docs,
-                                 node_name=ctx.keys.node_id[:12])
+        header = export_snapshot(
+            ctx.store, path, max_docs=max_docs,
+            node_name=ctx.keys.node_id[:12],
+            level=ctx.config.index.snapshot_compression_level)
         click.echo(json.dumps(header, indent=2))
     finally:
         ctx.close()

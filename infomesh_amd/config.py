@@ -84,7 +84,6 @@ class GpuConfig:
     hbm_budget_gb: float = 260.0       # of 288 GB HBM3E per GPU
     dtype: str = "bf16"
     topk_per_shard: int = 100
-    query_batch: int = 64
     require_extension: bool = True     # GPU present but ext missing → hard error
 
 
@@ -155,7 +154,6 @@ _CLAMPS: dict[tuple[str, str], tuple[float, float]] = {
     ("gpu", "n_shards"): (0, 8),
     ("gpu", "hbm_budget_gb"): (1.0, 288.0),
     ("gpu", "topk_per_shard"): (1, 4096),
-    ("gpu", "query_batch"): (1, 4096),
     ("api", "port"): (1, 65535),
 }
 

@@ -77,11 +77,24 @@ class CreditLedger(SQLiteStore):
 
     def __init__(self, path: str | Path = ":memory:",
                  kp: KeyPair | None = None,
-                 off_peak_fn=None):
+                 off_peak_fn=None,
+                 crawl_reward: float | None = None,
+                 query_reward: float | None = None,
+                 search_cost: float | None = None,
+                 grace_hours: float | None = None):
         super().__init__(path)
         self.kp = kp
         self._off_peak_fn = off_peak_fn
         self._debt_since: float | None = None
+        # config overrides (credits.* section); None = module defaults
+        self._weights = dict(ACTION_WEIGHTS)
+        if crawl_reward is not None:
+            self._weights[Action.CRAWL] = float(crawl_reward)
+        if query_reward is not None:
+            self._weights[Action.QUERY_SERVED] = float(query_reward)
+        self._base_search_cost = search_cost
+        self._grace_hours = float(grace_hours) if grace_hours is not None \
+            else GRACE_HOURS
         # batched accounting (record_action_async)
         import threading
         self._accum: dict[Action, float] = {}
@@ -114,7 +127,7 @@ class CreditLedger(SQLiteStore):
         mult = 1.0
         if action in LLM_ACTIONS and self._is_off_peak(ts):
             mult = OFF_PEAK_MULTIPLIER
-        credits = ACTION_WEIGHTS[action] * quantity * mult
+        credits = self._weights[action] * quantity * mult
         prev = self._last_hash()
         payload = json.dumps(
             {"action": action.value, "q": quantity, "m": mult,
@@ -153,7 +166,14 @@ class CreditLedger(SQLiteStore):
         return t
 
     def search_cost(self) -> float:
-        cost = TIER_SEARCH_COST[self.tier() - 1]
+        if self._base_search_cost is not None:
+            # config override scales the tier-1 rate; tier discounts
+            # keep their published ratios (ledger.py:12-15)
+            cost = (self._base_search_cost
+                    * TIER_SEARCH_COST[self.tier() - 1]
+                    / TIER_SEARCH_COST[0])
+        else:
+            cost = TIER_SEARCH_COST[self.tier() - 1]
         if self.in_debt_mode():
             cost *= DEBT_MULTIPLIER
         return cost
@@ -167,7 +187,7 @@ class CreditLedger(SQLiteStore):
         if self._debt_since is None:
             self._debt_since = now
             return False
-        return (now - self._debt_since) > GRACE_HOURS * 3600.0
+        return (now - self._debt_since) > self._grace_hours * 3600.0
 
     def deduct_search_cost(self) -> float:
         cost = self.search_cost()

@@ -28,13 +28,24 @@ class HybridEngine:
     def __init__(self, device: str | None = None, k_per_shard: int = 100,
                  use_encoder: bool = True, encoder_max_len: int = 128,
                  fabric: Fabric | None = None, emb_dtype: str = "bf16",
-                 hbm_budget_gb: float = 260.0):
+                 hbm_budget_gb: float = 260.0,
+                 embed_max_chars: int = 2000,
+                 require_extension: bool = True):
         self.gpu = torch.cuda.is_available() if device is None \
             else device.startswith("cuda")
         self.device = device or ("cuda" if self.gpu else "cpu")
         self.fabric = fabric or Fabric()
         self.emb_dtype = emb_dtype
         self.hbm_budget_bytes = int(hbm_budget_gb * 1e9)
+        self.embed_max_chars = int(embed_max_chars)
+        if self.gpu and require_extension:
+            # fail at startup, not at the first query (gpu.require_extension)
+            from .ops import _ext
+            if not _ext.available():
+                from .errors import GpuExtensionMissing
+                raise GpuExtensionMissing(
+                    "GPU present but the HIP extension failed to load "
+                    "(gpu.require_extension=true)")
         self.shard: GpuShard = (
             GpuShard(self.device, emb_dtype=emb_dtype) if self.gpu
             else CpuShard(emb_dtype=emb_dtype))
@@ -58,7 +69,9 @@ class HybridEngine:
         assert doc.doc_id is not None
         text = f"{doc.title}\n{doc.text}"[:4000]
         self._pending_tokens.append(bm25_term_ids(text))
-        self._pending_texts.append(text[:2000])   # embed truncation
+        # embed truncation (config index.embed_max_chars; reference
+        # vector_store.py:144-157 truncates at 2000 chars)
+        self._pending_texts.append(text[:self.embed_max_chars])
         self._pending_ids.append(doc.doc_id)
 
     @property

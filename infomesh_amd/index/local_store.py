@@ -126,11 +126,13 @@ def sanitize_fts_query(query: str, match_any: bool = False) -> str:
 class LocalStore:
     """The canonical FTS5-backed document store."""
 
-    def __init__(self, path: str | Path = ":memory:", tokenizer: str = "unicode61"):
+    def __init__(self, path: str | Path = ":memory:", tokenizer: str = "unicode61",
+                 max_text_chars: int = 500_000):
         if tokenizer not in FTS_TOKENIZERS:
             raise InfoMeshError("IDX003", tokenizer)
         self.path = str(path)
         self.tokenizer = tokenizer
+        self.max_text_chars = int(max_text_chars)
         if self.path != ":memory:":
             Path(self.path).parent.mkdir(parents=True, exist_ok=True)
         self.conn = sqlite3.connect(self.path, check_same_thread=False)
@@ -166,6 +168,8 @@ class LocalStore:
         Returns the rowid, or None when skipped as a duplicate
         (reference: local_store.py:198-251)."""
         now = time.time()
+        if len(doc.text) > self.max_text_chars:   # config index.max_text_chars
+            doc.text = doc.text[:self.max_text_chars]
         text_hash = doc.text_hash or content_hash(doc.text)
         domain = doc.domain or extract_domain(doc.url)
         existing = self.conn.execute(

@@ -32,7 +32,8 @@ SUFFIX = ".infomesh-snapshot"
 
 def export_snapshot(store: LocalStore, path: str | Path,
                     max_docs: int = MAX_DOCS,
-                    node_name: str = "") -> dict[str, Any]:
+                    node_name: str = "",
+                    level: int = compression.LEVEL_SNAPSHOT) -> dict[str, Any]:
     """Export the store to a snapshot file; returns the metadata header."""
     path = Path(path)
     docs: list[dict[str, Any]] = []
@@ -56,7 +57,7 @@ def export_snapshot(store: LocalStore, path: str | Path,
         "node": node_name,
         "generator": "infomesh-amd",
     }
-    comp = compression.Compressor(compression.LEVEL_SNAPSHOT)
+    comp = compression.Compressor(level)   # config index.snapshot_compression_level
     header_z = comp.compress(json.dumps(header).encode("utf-8"))
     docs_z = comp.compress(msgpack.packb(docs, use_bin_type=True))
     path.parent.mkdir(parents=True, exist_ok=True)

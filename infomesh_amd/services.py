@@ -92,10 +92,16 @@ class AppContext:
             load_plugins_from_config(
                 [x.strip() for x in cfg.node.plugins.split(",")
                  if x.strip()])
-        store = LocalStore(p("index.db"), tokenizer=cfg.index.fts_tokenizer)
+        store = LocalStore(p("index.db"), tokenizer=cfg.index.fts_tokenizer,
+                           max_text_chars=cfg.index.max_text_chars)
         keys = KeyPair.generate() if in_memory else ensure_keys(data)
-        ledger = CreditLedger(p("ledger.db"), kp=keys)
-        trust = TrustStore(p("trust.db"))
+        ledger = CreditLedger(p("ledger.db"), kp=keys,
+                              crawl_reward=cfg.credits.crawl_reward,
+                              query_reward=cfg.credits.query_reward,
+                              search_cost=cfg.credits.search_cost,
+                              grace_hours=cfg.credits.grace_hours)
+        trust = TrustStore(p("trust.db"),
+                           isolation_failures=cfg.trust.isolation_failures)
         ctx = cls(
             config=cfg,
             store=store,
@@ -128,9 +134,11 @@ class AppContext:
         if want_engine:
             try:
                 ctx.engine = HybridEngine(
-                    k_per_shard=cfg.search.max_results_per_shard,
+                    k_per_shard=cfg.gpu.topk_per_shard,
                     emb_dtype="fp8" if cfg.gpu.dtype == "fp8" else "bf16",
-                    hbm_budget_gb=cfg.gpu.hbm_budget_gb)
+                    hbm_budget_gb=cfg.gpu.hbm_budget_gb,
+                    embed_max_chars=cfg.index.embed_max_chars,
+                    require_extension=cfg.gpu.require_extension)
             except Exception as e:
                 log.warning("engine unavailable: %s", e)
         if ctx.engine is not None:
@@ -274,6 +282,18 @@ class AppContext:
                 max_wait_ms=self.config.search.batch_wait_ms,
                 execute=self._batch_execute)
         return self.batcher
+
+    def make_auditor(self, fetch_fn):
+        """Build the random-audit scheduler with the configured rate
+        (trust.audits_per_hour) and quorum size (trust.auditors).
+        fetch_fn: async url -> text | None (the node's own fetcher)."""
+        from .trust.audit import AuditScheduler
+        from .trust.detector import MaliciousNodeDetector
+        return AuditScheduler(
+            self.store, self.trust, fetch_fn,
+            rate_per_hour=self.config.trust.audits_per_hour,
+            auditors=self.config.trust.auditors,
+            detector=MaliciousNodeDetector())
 
     def _batch_execute(self, queries: list[str],
                        limit: int) -> list[list]:

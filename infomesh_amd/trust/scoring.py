@@ -51,8 +51,12 @@ class TrustStore(SQLiteStore):
     );
     """
 
-    def __init__(self, path: str | Path = ":memory:"):
+    def __init__(self, path: str | Path = ":memory:",
+                 isolation_failures: int = ISOLATION_FAILURES):
         super().__init__(path)
+        # config trust.isolation_failures (reference trust_system.py
+        # isolates after 3 consecutive failed audits)
+        self.isolation_failures = int(isolation_failures)
 
     def _get(self, subject: str):
         return self.execute(
@@ -80,7 +84,7 @@ class TrustStore(SQLiteStore):
         self._ensure(subject)
         row = self._get(subject)
         fails = 0 if passed else int(row["consecutive_failures"]) + 1
-        isolated = 1 if fails >= ISOLATION_FAILURES else int(row["isolated"])
+        isolated = 1 if fails >= self.isolation_failures else int(row["isolated"])
         if passed:
             isolated = 0
         self.execute(

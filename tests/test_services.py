@@ -295,3 +295,58 @@ def test_engine_hbm_budget_guard():
     assert eng.pending_count == 1          # nothing lost
     eng.hbm_budget_bytes = 10**9
     assert eng.flush() == 1                # retry succeeds
+
+
+def test_config_knobs_reach_behavior(tmp_path):
+    """Every documented config knob must change runtime behavior:
+    credits.* → ledger math, trust.isolation_failures → TrustStore,
+    index.max_text_chars → LocalStore truncation, trust.audits_per_hour
+    → auditor rate (round-2 dead-knob audit)."""
+    import dataclasses as dc
+    cfg = Config()
+    cfg = dc.replace(
+        cfg,
+        credits=dc.replace(cfg.credits, crawl_reward=2.5, search_cost=0.2,
+                           grace_hours=1.0),
+        trust=dc.replace(cfg.trust, isolation_failures=2,
+                         audits_per_hour=7.0, auditors=5),
+        index=dc.replace(cfg.index, max_text_chars=50))
+    ctx = AppContext.create(config=cfg, with_worker=False, with_engine=False,
+                            in_memory=True)
+    try:
+        from infomesh_amd.credits.ledger import Action
+        e = ctx.ledger.record_action(Action.CRAWL, 4)
+        assert e.credits == 2.5 * 4
+        assert abs(ctx.ledger.search_cost() - 0.2) < 1e-9  # tier 1 override
+        assert ctx.ledger._grace_hours == 1.0
+        assert ctx.trust.isolation_failures == 2
+        # 2 failed audits isolate under the tightened threshold
+        ctx.trust.record_audit("nodeX", False)
+        ctx.trust.record_audit("nodeX", False)
+        assert ctx.trust.tier("nodeX") == "isolated"
+        # oversized text is truncated at ingest
+        from infomesh_amd.index.local_store import Document
+        did = ctx.store.add_document(Document(url="http://x/1", title="t",
+                                              text="a" * 500))
+        assert len(ctx.store.get_document(did).text) == 50
+
+        async def fake_fetch(url):
+            return None
+        aud = ctx.make_auditor(fake_fetch)
+        assert aud.rate == 7.0 and aud.auditors == 5
+        assert aud.due(now=aud.last_audit + 3600.0 / 7.0 + 1)
+    finally:
+        ctx.close()
+
+
+def test_snapshot_compression_level_param(tmp_path):
+    from infomesh_amd.index.local_store import Document, LocalStore
+    from infomesh_amd.index.snapshot import export_snapshot, import_snapshot
+
+    store = LocalStore(":memory:")
+    store.add_document(Document(url="http://x/1", title="t", text="hello " * 200))
+    p = tmp_path / "s.infomesh-snapshot"
+    export_snapshot(store, p, level=1)
+    dst = LocalStore(":memory:")
+    info = import_snapshot(dst, p)
+    assert info["imported"] == 1 or dst.count() == 1
