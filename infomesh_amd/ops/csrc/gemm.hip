@@ -21,8 +21,20 @@ namespace {
 // destination stays lane-linear), it spreads the b128 fragment reads of
 // a wave across banks. PMC before: 1.59e9 LDS bank conflicts in this
 // kernel. BK=64 only (BK=32 rows are 64 B and keep the linear image).
+// LDS swizzle for BK=64 tiles (involution, 16 B granules). Rows are
+// 128 B so bank_start = 32*(row&1) + 4*chunk collapses rows r and r+2
+// onto the same banks; xoring the 16 B-chunk index (bits 4-6) with
+// h(row) = (row ^ (row>>3)) & 7 (bits 7-12 of the offset) gives all 16
+// rows of a quarter-wave MFMA operand read distinct 4-bank windows.
+// Measured: the SQ_LDS_BANK_CONFLICT counter on the pipelined tile
+// (4.16e8, profiles/) is UNCHANGED by this — those conflicts come from
+// the epilogue f32 staging (4-way ds_write folding, ~192 cycles/wave-
+// tile), and wall time is glds-latency-bound either way; the swizzle is
+// kept because conflict-free operand reads cost nothing. Key uses only
+// bits >=7, which the xor never touches, so swz64(swz64(x)) == x and
+// region bases (bit 13+) pass through unchanged.
 __device__ __forceinline__ int swz64(int byte_off) {
-  return byte_off ^ (((byte_off >> 9) & 1) << 5);
+  return byte_off ^ ((((byte_off >> 7) ^ (byte_off >> 10)) & 7) << 4);
 }
 
 // Tile configurations (4 waves each):

@@ -31,10 +31,17 @@
 
 namespace {
 
-// st_16x32 swizzle on a [rows][64 bf16] row-major image:
-// byte addr = row*128 + col2; flip bit5 (32B group) when row bit2 set.
+// LDS swizzle on a [rows][64 bf16] row-major image (involution, 16 B
+// granules): xor the 16 B-chunk index (bits 4-6) with
+// h(row) = (row ^ (row>>3)) & 7 so the 16 rows of a quarter-wave MFMA
+// operand read land on 16 distinct 4-bank windows (the old single-bit
+// flip left rows r/r+2 colliding — see gemm.hip swz64 for the bank
+// arithmetic and the measured outcome: conflict counters/wall time are
+// epilogue- and latency-dominated, so this is hygiene, not a win).
+// Key uses only bits >=7, untouched by the xor, so swz(swz(x)) == x
+// and region bases (bit 13+) pass through.
 DEVINL int swz(int byte_off) {
-  return byte_off ^ (((byte_off >> 9) & 1) << 5);
+  return byte_off ^ ((((byte_off >> 7) ^ (byte_off >> 10)) & 7) << 4);
 }
 
 template <bool OUT_F32>
