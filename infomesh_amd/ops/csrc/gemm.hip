@@ -15,13 +15,10 @@
 
 namespace {
 
-// st_16x32 swizzle on a [rows][64 bf16] row-major LDS image (byte addr
-// = row*128 + col2): flip bit5 (32B group) when row bit2 is set. Applied
-// to the glds SOURCE address and to the ds_read address (rule 21:
-// destination stays lane-linear), it spreads the b128 fragment reads of
-// a wave across banks. PMC before: 1.59e9 LDS bank conflicts in this
-// kernel. BK=64 only (BK=32 rows are 64 B and keep the linear image).
-// LDS swizzle for BK=64 tiles (involution, 16 B granules). Rows are
+// LDS swizzle for BK=64 tiles (involution, 16 B granules), applied to
+// the glds SOURCE address and the ds_read address (rule 21: the LDS
+// destination stays lane-linear). BK=32 rows are 64 B and keep the
+// linear image. Rows are
 // 128 B so bank_start = 32*(row&1) + 4*chunk collapses rows r and r+2
 // onto the same banks; xoring the 16 B-chunk index (bits 4-6) with
 // h(row) = (row ^ (row>>3)) & 7 (bits 7-12 of the offset) gives all 16
