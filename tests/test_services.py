@@ -350,3 +350,15 @@ def test_snapshot_compression_level_param(tmp_path):
     dst = LocalStore(":memory:")
     info = import_snapshot(dst, p)
     assert info["imported"] == 1 or dst.count() == 1
+
+
+def test_require_extension_fails_loudly(monkeypatch):
+    """gpu.require_extension: a GPU engine must refuse to start when the
+    HIP extension is missing — never a silent eager fallback (the
+    round-end 'native code not loaded' check)."""
+    from infomesh_amd.errors import GpuExtensionMissing
+    from infomesh_amd.ops import _ext
+
+    monkeypatch.setattr(_ext, "available", lambda: False)
+    with pytest.raises(GpuExtensionMissing):
+        HybridEngine(device="cuda", require_extension=True)
