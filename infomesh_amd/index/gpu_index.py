@@ -388,6 +388,7 @@ class GpuShard:
         # dense hipGraphs captured the old embeddings view/shape
         self._dense_graphs = {}
         self._dense_bufs = {}
+        self._bm25_hists = {}
 
     def optimize(self) -> None:
         """Merge all posting segments into one, entirely on-device:
@@ -555,6 +556,18 @@ class GpuShard:
         np.cumsum(np.bincount(qrows, minlength=B), out=qt_off[1:])
         bd = self._pick_bd(B)
         U = len(uterms)
+        # fused topk pass 1: the block kernel histograms every score it
+        # writes, so the top-k select skips one full read of the [B, N]
+        # array (~0.9 ms/batch at 10M docs)
+        hists = getattr(self, "_bm25_hists", None)
+        if hists is None:
+            hists = self._bm25_hists = {}
+        hist = hists.get(B)
+        if hist is None:
+            hist = hists[B] = torch.zeros(
+                B * 256, device=dev, dtype=torch.int32)
+        else:
+            hist.zero_()
         tp = mark("shard.chunks", tp)
         # don't overwrite the pinned staging buffers while a prior
         # step's async H2D copy could still be in flight
@@ -572,13 +585,13 @@ class GpuShard:
                 self._h2d(f"qb{si}", seg.h_offs[uterms], torch.int64),
                 self._h2d(f"qe{si}", seg.h_offs[uterms + 1], torch.int64),
                 bounds, scores, seg.doc_base, seg.n_docs, bd,
-                self.avgdl, k1=BM25_K1, b=BM25_B)
+                self.avgdl, k1=BM25_K1, b=BM25_B, hist1=hist)
         if dev.type == "cuda":
             if evt is None:
                 evt = self._h2d_evt = torch.cuda.Event()
             evt.record()
         tp = mark("shard.bm25", tp)
-        out = topk(scores, k)
+        out = topk(scores, k, ext_hist1=hist)
         mark("shard.bm25topk", tp)
         self._bm25_scores_buf = scores
         return out

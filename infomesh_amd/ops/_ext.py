@@ -40,12 +40,12 @@ _SIGNATURES: dict[str, list] = {
                       c_int, c_int, c_void_p],
     "infomesh_argmax": [c_void_p, c_void_p, c_long, c_int, c_void_p],
     "infomesh_topk": [c_void_p, c_void_p, c_void_p, c_void_p, c_int,
-                      c_long, c_int, c_int, c_void_p],
+                      c_long, c_int, c_int, c_void_p, c_void_p],
     "infomesh_bm25_block": [c_void_p, c_void_p, c_void_p, c_void_p,
                             c_void_p, c_void_p, c_void_p, c_void_p,
-                            c_void_p, c_int, c_int, c_long, c_long,
-                            c_long, c_int, c_float, c_float, c_float,
-                            c_void_p],
+                            c_void_p, c_void_p, c_int, c_int, c_long,
+                            c_long, c_long, c_int, c_float, c_float,
+                            c_float, c_void_p],
     "infomesh_score_combine": [c_void_p, c_void_p, c_void_p, c_float,
                                c_float, c_long, c_void_p],
     "infomesh_simhash_fingerprint": [c_void_p, c_void_p, c_void_p, c_long,

@@ -61,6 +61,15 @@ __global__ __launch_bounds__(256) void bm25_bounds_kernel(
   bounds[2 * i + 1] = (int)(hi - b);
 }
 
+// Fused hist1: every written score is also counted into a per-query
+// 256-bin ordered-top-byte histogram (8 padded LDS copies, zeros
+// register-aggregated per wave since empty blocks dominate), merged by
+// global atomics. infomesh_topk(ext_hist1=...) then SKIPS its first
+// full read of the score array — at 10M docs that pass alone is
+// ~0.9 ms/batch of HBM streaming. Counts are exact (each column is
+// written and histogrammed exactly once across segments x blocks), so
+// downstream select/compact semantics are bit-identical.
+template <bool HIST>
 __global__ __launch_bounds__(256) void bm25_block_kernel(
     const int* __restrict__ doc_ids,        // [P] segment-local, asc per term
     const unsigned int* __restrict__ tfdl,  // [P] tf | dl<<16
@@ -70,9 +79,10 @@ __global__ __launch_bounds__(256) void bm25_block_kernel(
     const long* __restrict__ u_begin,       // [U]
     const int* __restrict__ bounds,         // [U * nblocks * 2]
     float* __restrict__ scores,             // [B, rowN]
+    unsigned* __restrict__ hist1,           // [B * 256] or null
     long rowN, long doc_base, long nseg, int BD, int nblocks,
     float norm_a, float norm_b, float k1p1) {
-  extern __shared__ float lds_scores[];     // [BD]
+  extern __shared__ float lds_scores[];     // [BD] (+ hist copies)
   // grid: x = query (fast), y = doc-block — adjacent workgroups are
   // the SAME posting sub-range for different queries, so the XCD's L2
   // serves the repeat reads instead of HBM
@@ -80,7 +90,12 @@ __global__ __launch_bounds__(256) void bm25_block_kernel(
   const int blk = blockIdx.y;
   const long d0 = (long)blk * BD;
   const int nd = (int)min((long)BD, nseg - d0);
+  // 8 padded histogram copies after the score tile (same bank spread
+  // as topk.hip hist1_kernel: 264-int stride)
+  unsigned* lh = reinterpret_cast<unsigned*>(lds_scores + BD);
   for (int i = threadIdx.x; i < nd; i += blockDim.x) lds_scores[i] = 0.0f;
+  if (HIST)
+    for (int i = threadIdx.x; i < 8 * 264; i += blockDim.x) lh[i] = 0;
   __syncthreads();
   const int t0 = qt_off[q], t1 = qt_off[q + 1];
   const int wave = threadIdx.x / WAVE, lane = threadIdx.x % WAVE;
@@ -106,8 +121,33 @@ __global__ __launch_bounds__(256) void bm25_block_kernel(
   // pass — nontemporal keeps them out of L2, which the posting reads
   // (shared across adjacent same-block workgroups) actually want
   float* __restrict__ srow = scores + (long)q * rowN + doc_base + d0;
-  for (int i = threadIdx.x; i < nd; i += blockDim.x)
-    __builtin_nontemporal_store(lds_scores[i], srow + i);
+  if (!HIST) {
+    for (int i = threadIdx.x; i < nd; i += blockDim.x)
+      __builtin_nontemporal_store(lds_scores[i], srow + i);
+    return;
+  }
+  unsigned* my = lh + (threadIdx.x & 7) * 264;
+  unsigned zc = 0;  // zero scores dominate: aggregate in a register
+  for (int i = threadIdx.x; i < nd; i += blockDim.x) {
+    const float v = lds_scores[i];
+    __builtin_nontemporal_store(v, srow + i);
+    if (v == 0.0f) ++zc;
+    else atomicAdd(&my[float_to_ordered(v) >> 24], 1u);
+  }
+  // wave-reduce the zero count; one LDS atomic per wave
+  for (int off = WAVE / 2; off; off >>= 1)
+    zc += __shfl_xor(zc, off, WAVE);
+  constexpr unsigned ZBIN = 0x80000000u >> 24;  // ordered(+0.0f) byte
+  if ((threadIdx.x % WAVE) == 0 && zc)
+    atomicAdd(&my[ZBIN], zc);
+  __syncthreads();
+  unsigned* hrow = hist1 + (long)q * 256;
+  for (int i = threadIdx.x; i < 256; i += blockDim.x) {
+    unsigned sum = 0;
+#pragma unroll
+    for (int c = 0; c < 8; ++c) sum += lh[c * 264 + i];
+    if (sum) atomicAdd(&hrow[i], sum);
+  }
 }
 
 // Fused score-combine: out = wa * a + wb * b (optional linear hybrid;
@@ -126,7 +166,7 @@ __global__ void combine_kernel(const float* __restrict__ a,
 extern "C" void infomesh_bm25_block(
     const void* doc_ids, const void* tfdl, const void* qt_off,
     const void* qt_ut, const void* qt_idf, const void* u_begin,
-    const void* u_end, void* bounds, void* scores,
+    const void* u_end, void* bounds, void* scores, void* hist1,
     int B, int U, long rowN, long doc_base, long nseg,
     int BD, float norm_a, float norm_b, float k1p1, void* stream) {
   if (nseg <= 0 || B <= 0) return;
@@ -140,14 +180,25 @@ extern "C" void infomesh_bm25_block(
                        (const long*)u_end, (int*)bounds, U, nblocks, BD);
   }
   dim3 grid((unsigned)B, (unsigned)nblocks);
-  hipLaunchKernelGGL(bm25_block_kernel, grid, dim3(256),
-                     (size_t)BD * sizeof(float), s,
-                     (const int*)doc_ids, (const unsigned int*)tfdl,
-                     (const int*)qt_off, (const int*)qt_ut,
-                     (const float*)qt_idf, (const long*)u_begin,
-                     (const int*)bounds, (float*)scores,
-                     rowN, doc_base, nseg, BD, nblocks,
-                     norm_a, norm_b, k1p1);
+  const size_t shm = (size_t)BD * sizeof(float)
+      + (hist1 ? 8 * 264 * sizeof(unsigned) : 0);
+  if (hist1)
+    hipLaunchKernelGGL(bm25_block_kernel<true>, grid, dim3(256), shm, s,
+                       (const int*)doc_ids, (const unsigned int*)tfdl,
+                       (const int*)qt_off, (const int*)qt_ut,
+                       (const float*)qt_idf, (const long*)u_begin,
+                       (const int*)bounds, (float*)scores,
+                       (unsigned*)hist1,
+                       rowN, doc_base, nseg, BD, nblocks,
+                       norm_a, norm_b, k1p1);
+  else
+    hipLaunchKernelGGL(bm25_block_kernel<false>, grid, dim3(256), shm, s,
+                       (const int*)doc_ids, (const unsigned int*)tfdl,
+                       (const int*)qt_off, (const int*)qt_ut,
+                       (const float*)qt_idf, (const long*)u_begin,
+                       (const int*)bounds, (float*)scores, nullptr,
+                       rowN, doc_base, nseg, BD, nblocks,
+                       norm_a, norm_b, k1p1);
 }
 
 extern "C" void infomesh_score_combine(const void* a, const void* b,

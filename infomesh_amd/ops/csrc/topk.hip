@@ -264,10 +264,14 @@ extern "C" long infomesh_topk_workspace_u32(int B) {
   return u + (long)B * TOPK_CAP * 2;
 }
 
+// ext_hist1: non-null = a producer-fused [B*256] top-byte histogram of
+// the FULL score rows (exact counts, zeros included); pass 1 is skipped
+// and select1 reads it directly. Incompatible with sampled (the sampled
+// path histograms its own gathered sample).
 extern "C" void infomesh_topk(const void* scores, void* workspace,
                               void* out_vals, void* out_idx,
                               int B, long N, int K, int sampled,
-                              void* stream) {
+                              const void* ext_hist1, void* stream) {
   auto s = reinterpret_cast<hipStream_t>(stream);
   unsigned* ws = reinterpret_cast<unsigned*>(workspace);
   unsigned* hist1 = ws;
@@ -311,10 +315,15 @@ extern "C" void infomesh_topk(const void* scores, void* workspace,
                        (const float*)scores, thresh16, cand, cnt,
                        cnt_eq, overflow, N);
   } else {
-    hipLaunchKernelGGL(hist1_kernel, g1, blk, 0, s,
-                       (const float*)scores, hist1, N);
+    const unsigned* h1 = hist1;
+    if (ext_hist1) {
+      h1 = (const unsigned*)ext_hist1;   // producer-fused pass 1
+    } else {
+      hipLaunchKernelGGL(hist1_kernel, g1, blk, 0, s,
+                         (const float*)scores, hist1, N);
+    }
     hipLaunchKernelGGL(select1_kernel, dim3(B), dim3(64), 0, s,
-                       hist1, bin1, chi1, K);
+                       h1, bin1, chi1, K);
     hipLaunchKernelGGL(hist2_kernel, g1, blk, 0, s,
                        (const float*)scores, bin1, hist2, N);
     hipLaunchKernelGGL(select2_kernel, dim3(B), dim3(64), 0, s,

@@ -275,7 +275,8 @@ class TopK:
         self._pending: list[tuple[torch.Tensor, int]] = []
 
     def __call__(self, scores: torch.Tensor, k: int,
-                 sampled: bool | None = None
+                 sampled: bool | None = None,
+                 ext_hist1: torch.Tensor | None = None
                  ) -> tuple[torch.Tensor, torch.Tensor]:
         assert scores.dim() == 2
         _check(scores, torch.float32, "scores")
@@ -300,9 +301,17 @@ class TopK:
         self._ws[: B * 512 + 5 * B + 2].zero_()  # hists + counters region
         vals = torch.empty((B, k), device=scores.device, dtype=torch.float32)
         idx = torch.empty((B, k), device=scores.device, dtype=torch.int32)
+        if ext_hist1 is not None:
+            # producer-fused pass 1 (bm25_block hist1=): exact counts,
+            # mutually exclusive with the sampled threshold
+            assert not sampled
+            assert ext_hist1.dtype == torch.int32 \
+                and ext_hist1.numel() >= B * 256
         lib.infomesh_topk(scores.data_ptr(), self._ws.data_ptr(),
                           vals.data_ptr(), idx.data_ptr(), B, N, k,
-                          int(sampled), _ext.stream_ptr())
+                          int(sampled),
+                          0 if ext_hist1 is None else ext_hist1.data_ptr(),
+                          _ext.stream_ptr())
         # cnt[B], cnt_eq[B], then the overflow flag after the histograms.
         cnt_off = B * 512 + 3 * B
         flag_off = B * 512 + 5 * B
@@ -344,7 +353,8 @@ def bm25_block(doc_ids: torch.Tensor, tfdl: torch.Tensor,
                u_end: torch.Tensor, bounds: torch.Tensor,
                scores: torch.Tensor, doc_base: int, nseg: int,
                bd: int, avgdl: float, k1: float = 1.2,
-               b: float = 0.75) -> torch.Tensor:
+               b: float = 0.75,
+               hist1: torch.Tensor | None = None) -> torch.Tensor:
     """Doc-block LDS-accumulated BM25 for ONE posting segment.
 
     Two launches: a bounds pre-pass binary-searching each (unique term,
@@ -365,13 +375,19 @@ def bm25_block(doc_ids: torch.Tensor, tfdl: torch.Tensor,
     assert nblocks <= 65535 and qt_off.numel() == B + 1
     assert doc_ids.dtype == torch.int32 and tfdl.dtype == torch.int32
     assert bounds.numel() >= U * nblocks * 2
-    assert bd * 4 <= 160 * 1024
+    assert bd * 4 + (8 * 264 * 4 if hist1 is not None else 0) <= 160 * 1024
+    if hist1 is not None:
+        # fused topk pass 1: exact [B,256] ordered-top-byte counts of the
+        # written score columns (caller zeroes it before the first
+        # segment and hands it to topk(ext_hist1=...))
+        assert hist1.dtype == torch.int32 and hist1.numel() >= B * 256
     norm_a = k1 * (1.0 - b)
     norm_b = k1 * b / max(avgdl, 1e-9)
     _ext.lib().infomesh_bm25_block(
         doc_ids.data_ptr(), tfdl.data_ptr(), qt_off.data_ptr(),
         qt_ut.data_ptr(), qt_idf.data_ptr(), u_begin.data_ptr(),
         u_end.data_ptr(), bounds.data_ptr(), scores.data_ptr(),
+        0 if hist1 is None else hist1.data_ptr(),
         B, U, N, doc_base, nseg, bd,
         norm_a, norm_b, k1 + 1.0, _ext.stream_ptr())
     return scores
