@@ -680,9 +680,8 @@ class GpuShard:
                 # Graph replay skips the wrapper's python, so re-arm
                 # the deferred overflow check by hand (same static
                 # workspace/layout as at capture; need=0 -> exact).
-                cnt_off = B * 512 + 3 * B
-                flag_off = B * 512 + 5 * B
-                tk._pending.append((ws, cnt_off, flag_off, B, 0))
+                tk._pending.append((ws, K.topk_cnt_off(B),
+                                    K.topk_flag_off(B), B, 0))
             # detach from the graph's static outputs (next replay
             # overwrites them)
             out = (vals.clone(), idx.clone())

@@ -260,6 +260,20 @@ def argmax(logits: torch.Tensor) -> torch.Tensor:
     return out
 
 
+# Workspace layout offsets (u32 units) — keep in sync with topk.hip:
+# hist1[B*256] | hist2[B*256] | bin1 | chi1 | thresh16 | cnt | cnt_eq
+# | overflow. Exposed as functions because the dense plane's
+# graph-replay path re-arms the deferred overflow check by hand (a
+# hardcoded copy of these offsets once went stale and read cnt_eq as
+# the overflow flag — round-2 fix).
+def topk_cnt_off(B: int) -> int:
+    return B * 512 + 3 * B
+
+
+def topk_flag_off(B: int) -> int:
+    return B * 512 + 5 * B
+
+
 class TopK:
     """Reusable top-k selector (keeps its workspace allocated).
 
@@ -313,8 +327,8 @@ class TopK:
                           0 if ext_hist1 is None else ext_hist1.data_ptr(),
                           _ext.stream_ptr())
         # cnt[B], cnt_eq[B], then the overflow flag after the histograms.
-        cnt_off = B * 512 + 3 * B
-        flag_off = B * 512 + 5 * B
+        cnt_off = topk_cnt_off(B)
+        flag_off = topk_flag_off(B)
         need = min(k, N) if sampled else 0
         if self.defer_check:
             self._pending.append((self._ws, cnt_off, flag_off, B, need))
