@@ -286,6 +286,10 @@ __global__ __launch_bounds__(256) void gemm_pipe64_kernel(
       const bf16* gsrc = (is_b ? Bg : Ag) + (long)grow * K + k0 + colb / 2;
       auto* dst = (__attribute__((address_space(3))) unsigned int*)
           ((char*)&smem[buf][0] + (long)chunk * 1024);
+      // NT on the A loads was tried here (gemm8-style): encoder
+      // 993 -> 1135 us. At these shapes BOTH operands fit the XCD L2
+      // together (A 3.1 MB x 18-way n-reuse + B 0.9 MB), so the NT
+      // hint only destroyed A's own reuse. Keep default caching.
       __builtin_amdgcn_global_load_lds(
           (const __attribute__((address_space(1))) unsigned int*)gsrc,
           dst, 16, 0, 0);

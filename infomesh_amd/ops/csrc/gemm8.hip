@@ -98,9 +98,21 @@ __global__ __launch_bounds__(512, 1) void gemm8_kernel(
       const bf16* gsrc = (is_b ? Bg : Ag) + (long)grow * K + k0 + colb / 2;
       auto* dst = (__attribute__((address_space(3))) unsigned int*)
           ((char*)&smem[buf][0] + region + (long)chunk * 1024);
-      __builtin_amdgcn_global_load_lds(
-          (const __attribute__((address_space(1))) unsigned int*)gsrc,
-          dst, 16, 0, 0);
+      // A (M up to ~512k rows) streams through exactly once; B
+      // (N*K*2 <= a few MB at the shapes this kernel serves) wants to
+      // stay in the XCD's 4 MB L2 across m-tiles. Marking the A loads
+      // non-temporal (CPol nt, aux=2) stops the A stream from evicting
+      // B — without it B thrashes and the kernel runs at the
+      // read-everything-from-HBM roofline (~700 GF/s at the reranker
+      // shapes; torch/rocBLAS was 2x faster purely on this reuse).
+      if (is_b)
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) unsigned int*)gsrc,
+            dst, 16, 0, 0);
+      else
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) unsigned int*)gsrc,
+            dst, 16, 0, /*nt*/ 2);
     }
   };
 
@@ -360,9 +372,21 @@ __global__ __launch_bounds__(512, 1) void gemm8p_kernel(
       const bf16* gsrc = (is_b ? Bg : Ag) + (long)grow * K + k0 + colb / 2;
       auto* dst = (__attribute__((address_space(3))) unsigned int*)
           ((char*)&smem[buf][0] + region + (long)chunk * 1024);
-      __builtin_amdgcn_global_load_lds(
-          (const __attribute__((address_space(1))) unsigned int*)gsrc,
-          dst, 16, 0, 0);
+      // A (M up to ~512k rows) streams through exactly once; B
+      // (N*K*2 <= a few MB at the shapes this kernel serves) wants to
+      // stay in the XCD's 4 MB L2 across m-tiles. Marking the A loads
+      // non-temporal (CPol nt, aux=2) stops the A stream from evicting
+      // B — without it B thrashes and the kernel runs at the
+      // read-everything-from-HBM roofline (~700 GF/s at the reranker
+      // shapes; torch/rocBLAS was 2x faster purely on this reuse).
+      if (is_b)
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) unsigned int*)gsrc,
+            dst, 16, 0, 0);
+      else
+        __builtin_amdgcn_global_load_lds(
+            (const __attribute__((address_space(1))) unsigned int*)gsrc,
+            dst, 16, 0, /*nt*/ 2);
     }
   };
 
