@@ -135,7 +135,6 @@ def validate_shard_hits(fused_ids, bm25_ids, bm25_scores,
     wildly off its siblings' (relative deviation of per-shard max
     scores) is flagged. Cheap (host, top-k only) and runs on the
     already-gathered tensors."""
-    import torch
     B, WK = bm25_scores.shape
     k = WK // max(world, 1)
     per_shard_max = bm25_scores.view(B, world, k).amax(dim=2)  # [B, W]
