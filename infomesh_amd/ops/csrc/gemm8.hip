@@ -48,7 +48,7 @@ template <bool OUT_F32>
 __global__ __launch_bounds__(512, 1) void gemm8_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ B,
     void* __restrict__ C, const float* __restrict__ bias,
-    int M, int N, int K,
+    int M, int N, int K, int ldc,
     long strideA, long strideB, long strideC,
     int act, float alpha) {
   // one buffer = A tile (256x64) + B tile (256x64) = 64 KiB
@@ -283,7 +283,7 @@ __global__ __launch_bounds__(512, 1) void gemm8_kernel(
       __builtin_nontemporal_store(
           v, reinterpret_cast<bf16x8*>(
               reinterpret_cast<bf16*>(C)
-              + (long)g * strideC + (long)m * N + n));
+              + (long)g * strideC + (long)m * ldc + n));
     }
     return;
   }
@@ -300,9 +300,9 @@ __global__ __launch_bounds__(512, 1) void gemm8_kernel(
         if (m >= M) continue;
         float v = apply_act(alpha * acc[i][j][r] + bv, act);
         if (OUT_F32)
-          reinterpret_cast<float*>(C)[(long)g * strideC + (long)m * N + n] = v;
+          reinterpret_cast<float*>(C)[(long)g * strideC + (long)m * ldc + n] = v;
         else
-          reinterpret_cast<bf16*>(C)[(long)g * strideC + (long)m * N + n] =
+          reinterpret_cast<bf16*>(C)[(long)g * strideC + (long)m * ldc + n] =
               f2bf(v);
       }
     }
@@ -325,7 +325,7 @@ template <bool OUT_F32>
 __global__ __launch_bounds__(512, 1) void gemm8p_kernel(
     const bf16* __restrict__ A, const bf16* __restrict__ B,
     void* __restrict__ C, const float* __restrict__ bias,
-    int M, int N, int K,
+    int M, int N, int K, int ldc,
     long strideA, long strideB, long strideC,
     int act, float alpha) {
   __shared__ bf16 smem[2][(G8_BM + G8_BN) * G8_BK];
@@ -524,10 +524,10 @@ __global__ __launch_bounds__(512, 1) void gemm8p_kernel(
             float v = apply_act(alpha * acc[i][j][r] + bv, act);
             if (OUT_F32)
               reinterpret_cast<float*>(C)[
-                  (long)g * strideC + (long)m * N + n] = v;
+                  (long)g * strideC + (long)m * ldc + n] = v;
             else
               reinterpret_cast<bf16*>(C)[
-                  (long)g * strideC + (long)m * N + n] = f2bf(v);
+                  (long)g * strideC + (long)m * ldc + n] = f2bf(v);
           }
           acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
         }
@@ -565,24 +565,66 @@ extern "C" void infomesh_gemm8_bf16_nt(
     if (out_f32)
       hipLaunchKernelGGL(gemm8p_kernel<true>, grid, dim3(512), 0, s,
                          (const bf16*)A, (const bf16*)B, C,
-                         (const float*)bias, M, N, K, strideA, strideB,
+                         (const float*)bias, M, N, K, N, strideA, strideB,
                          strideC, act, alpha);
     else
       hipLaunchKernelGGL(gemm8p_kernel<false>, grid, dim3(512), 0, s,
                          (const bf16*)A, (const bf16*)B, C,
-                         (const float*)bias, M, N, K, strideA, strideB,
+                         (const float*)bias, M, N, K, N, strideA, strideB,
                          strideC, act, alpha);
     return;
+  }
+  // N-chunking for over-L2 B panels (default OFF — INFOMESH_GEMM8_CHUNK=1;
+  // designed/unvalidated this round, see BACKLOG): when the B panel
+  // (N*K*2 bytes) exceeds the 4 MB per-XCD L2 even the NT-A stream
+  // cannot keep it resident; splitting N into chunks whose slice fits
+  // ~2.5 MB trades (n_chunks-1) extra reads of A for eliminating the
+  // B thrash. Only applied when that trade is favorable.
+  static const int chunk_ov = [] {
+    const char* e = getenv("INFOMESH_GEMM8_CHUNK");
+    return e ? atoi(e) : 0;
+  }();
+  const long b_bytes = (long)N * K * 2;
+  if (chunk_ov && b_bytes > 3 * 1024 * 1024 && batch == 1) {
+    int cols = (int)(2.5 * 1024 * 1024 / (K * 2)) / G8_BN * G8_BN;
+    if (cols >= G8_BN) {
+      const int nch = (N + cols - 1) / cols;
+      const long a_bytes = (long)M * K * 2;
+      // extra A reads must undercut the avoided B re-reads (one per
+      // 512 rows of m-tiles is a conservative thrash estimate)
+      if ((long)(nch - 1) * a_bytes < b_bytes * (M / 512)) {
+        for (int c0 = 0; c0 < N; c0 += cols) {
+          const int nc = (N - c0) < cols ? (N - c0) : cols;
+          const int t = ((M + G8_BM - 1) / G8_BM) * ((nc + G8_BN - 1) / G8_BN);
+          dim3 gc(t, 1);
+          const bf16* Bc = (const bf16*)B + (long)c0 * K;
+          const float* bc = bias ? (const float*)bias + c0 : nullptr;
+          if (out_f32)
+            hipLaunchKernelGGL(gemm8_kernel<true>, gc, dim3(512), 0, s,
+                               (const bf16*)A, Bc,
+                               (char*)C + (long)c0 * 4, bc,
+                               M, nc, K, N, strideA, strideB, strideC,
+                               act, alpha);
+          else
+            hipLaunchKernelGGL(gemm8_kernel<false>, gc, dim3(512), 0, s,
+                               (const bf16*)A, Bc,
+                               (char*)C + (long)c0 * 2, bc,
+                               M, nc, K, N, strideA, strideB, strideC,
+                               act, alpha);
+        }
+        return;
+      }
+    }
   }
   dim3 grid(tiles, batch), block(512);
   if (out_f32)
     hipLaunchKernelGGL(gemm8_kernel<true>, grid, block, 0, s,
                        (const bf16*)A, (const bf16*)B, C,
-                       (const float*)bias, M, N, K, strideA, strideB,
+                       (const float*)bias, M, N, K, N, strideA, strideB,
                        strideC, act, alpha);
   else
     hipLaunchKernelGGL(gemm8_kernel<false>, grid, block, 0, s,
                        (const bf16*)A, (const bf16*)B, C,
-                       (const float*)bias, M, N, K, strideA, strideB,
+                       (const float*)bias, M, N, K, N, strideA, strideB,
                        strideC, act, alpha);
 }
