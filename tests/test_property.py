@@ -401,3 +401,88 @@ def test_segmented_build_any_split_matches_bulk(batch_sizes, do_opt):
     hi = inc.search(qs, None, k=min(10, total))
     hb = bulk.search(qs, None, k=min(10, total))
     assert torch.allclose(hi.bm25_scores, hb.bm25_scores, atol=1e-5)
+
+
+@given(st.lists(st.tuples(st.sampled_from(["crawl", "query_served",
+                                           "hosting", "uptime"]),
+                          st.floats(0.01, 50.0)), min_size=1, max_size=30),
+       st.floats(0.1, 5.0), st.floats(0.1, 5.0))
+@settings(max_examples=25, deadline=None)
+def test_ledger_invariants_under_arbitrary_actions(actions, cw, qw):
+    """Balance == Σ entry credits, the hash chain verifies after any
+    action sequence, and config weight overrides scale linearly."""
+    from infomesh_amd.credits.ledger import Action, CreditLedger
+    from infomesh_amd.trust.keys import KeyPair
+
+    led = CreditLedger(":memory:", kp=KeyPair.generate(),
+                       crawl_reward=cw, query_reward=qw,
+                       off_peak_fn=lambda ts: False)
+    total = 0.0
+    for name, qty in actions:
+        act = Action(name)
+        e = led.record_action(act, qty)
+        w = {Action.CRAWL: cw, Action.QUERY_SERVED: qw,
+             Action.HOSTING: 0.1, Action.UPTIME: 0.5}[act]
+        assert abs(e.credits - w * qty) < 1e-9
+        total += e.credits
+    assert abs(led.balance() - total) < 1e-6
+    assert led.verify_chain()
+    led.close()
+
+
+@given(st.lists(st.tuples(st.sampled_from(["crawl", "query_served"]),
+                          st.floats(0.5, 3.0)), min_size=1, max_size=20))
+@settings(max_examples=15, deadline=None)
+def test_ledger_batched_equals_eager_totals(actions):
+    """record_action_async + flush yields the same balance as eager
+    record_action (C = W·Q·M is linear in Q)."""
+    from infomesh_amd.credits.ledger import Action, CreditLedger
+
+    eager = CreditLedger(":memory:", off_peak_fn=lambda ts: False)
+    batched = CreditLedger(":memory:", off_peak_fn=lambda ts: False)
+    for name, qty in actions:
+        eager.record_action(Action(name), qty)
+        with batched._accum_lock:   # accumulate without the timer thread
+            batched._accum[Action(name)] = \
+                batched._accum.get(Action(name), 0.0) + qty
+    batched.flush_pending()
+    assert abs(eager.balance() - batched.balance()) < 1e-6
+    assert batched.verify_chain()
+    eager.close(); batched.close()
+
+
+@given(st.lists(st.text(alphabet=st.characters(min_codepoint=32,
+                                               max_codepoint=1000),
+                        min_size=1, max_size=40),
+                min_size=1, max_size=12, unique=True))
+@settings(max_examples=20, deadline=None)
+def test_manifest_roundtrip_cpu(texts):
+    """save_shard/load_shard preserves postings, doc lengths and search
+    behavior for any document set (CPU shard)."""
+    import tempfile
+    from pathlib import Path as P
+
+    import torch
+
+    from infomesh_amd.index.gpu_index import CpuShard, bm25_term_ids
+    from infomesh_amd.index.manifest import load_shard, save_shard
+
+    shard = CpuShard()
+    for i, t in enumerate(texts):
+        shard.add_document(i, bm25_term_ids(t), None)
+    shard.build()
+    with tempfile.TemporaryDirectory() as d:
+        meta = save_shard(shard, P(d) / "s.pt")
+        back = load_shard(P(d) / "s.pt", device="cpu")
+    assert meta["n_docs"] == len(texts)
+    assert back.n_docs == shard.n_docs
+    assert np.array_equal(back.df, shard.df)
+    assert np.array_equal(back._doc_lens, shard._doc_lens)
+    q = [bm25_term_ids(texts[0])]
+    a = shard.search(q, None, k=min(5, shard.n_docs))
+    b = back.search(q, None, k=min(5, shard.n_docs))
+    assert torch.allclose(a.bm25_scores, b.bm25_scores, atol=1e-6)
+    # warm-started shard keeps accepting appends (round-1 ADVICE fix)
+    back.add_document(len(texts), bm25_term_ids("appended doc"), None)
+    back.build()
+    assert back.n_docs == len(texts) + 1
