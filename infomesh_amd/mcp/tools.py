@@ -200,8 +200,17 @@ def validate_args(tool: str, args: dict) -> list[str]:
         if args.get(req) in (None, ""):
             errors.append(f"missing required argument {req!r}")
     for key, val in args.items():
+        if isinstance(val, str) and len(val) > MAX_STRING_ARG:
+            errors.append(f"{key}: string too long "
+                          f"({len(val)} > {MAX_STRING_ARG})")
+            continue
+        if isinstance(val, list) and len(val) > 1000:
+            errors.append(f"{key}: list too long ({len(val)})")
+            continue
         spec = props.get(key)
         if spec is None or val is None:
+            # unknown extra args tolerated (legacy forms) — but only
+            # after the global abuse caps above
             continue
         want = _TYPE_MAP.get(spec.get("type", ""))
         if want is not None and not isinstance(val, want):
@@ -214,9 +223,6 @@ def validate_args(tool: str, args: dict) -> list[str]:
                     f"{key}: expected {spec.get('type')}, got "
                     f"{type(val).__name__}")
                 continue
-        if isinstance(val, str) and len(val) > MAX_STRING_ARG:
-            errors.append(f"{key}: string too long "
-                          f"({len(val)} > {MAX_STRING_ARG})")
         if isinstance(val, (int, float)) and not isinstance(val, bool):
             lo = spec.get("minimum")
             hi = spec.get("maximum")
@@ -224,8 +230,6 @@ def validate_args(tool: str, args: dict) -> list[str]:
                 errors.append(f"{key}: {val} below minimum {lo}")
             if hi is not None and val > hi:
                 errors.append(f"{key}: {val} above maximum {hi}")
-        if isinstance(val, list) and len(val) > 1000:
-            errors.append(f"{key}: list too long ({len(val)})")
         enum = spec.get("enum") if spec else None
         if enum and val not in enum:
             errors.append(f"{key}: {val!r} not in {enum}")

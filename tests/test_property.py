@@ -486,3 +486,51 @@ def test_manifest_roundtrip_cpu(texts):
     back.add_document(len(texts), bm25_term_ids("appended doc"), None)
     back.build()
     assert back.n_docs == len(texts) + 1
+
+
+@given(st.sampled_from(["web_search", "fetch_page", "crawl_url",
+                        "fact_check", "status", "nonexistent_tool"]),
+       st.dictionaries(st.text(max_size=20),
+                       st.one_of(st.text(max_size=50), st.integers(),
+                                 st.booleans(), st.floats(allow_nan=False),
+                                 st.lists(st.text(max_size=10),
+                                          max_size=5)),
+                       max_size=8))
+@settings(max_examples=60, deadline=None)
+def test_mcp_validate_args_total(tool, args):
+    """validate_args never raises on arbitrary argument dicts — it
+    returns violation strings (empty means acceptable)."""
+    from infomesh_amd.mcp.tools import validate_args
+
+    errs = validate_args(tool, args)
+    assert isinstance(errs, list)
+    assert all(isinstance(e, str) for e in errs)
+    # the abuse caps always hold: huge strings are always rejected
+    errs2 = validate_args(tool, {"query": "x" * 100_000})
+    assert errs2
+
+
+@given(st.lists(st.text(min_size=1, max_size=30), min_size=1,
+                max_size=40))
+@settings(max_examples=15, deadline=None)
+def test_batcher_maps_each_query_to_its_own_result(queries):
+    """QueryBatcher returns exactly one result per submit and never
+    crosses wires, for any concurrent submission pattern."""
+    import concurrent.futures as cf
+
+    from infomesh_amd.search.batcher import QueryBatcher
+
+    def execute(qs, limit):
+        return [f"res:{q}" for q in qs]
+
+    b = QueryBatcher(engine=None, max_batch=8, max_wait_ms=0.5,
+                     execute=execute)
+    try:
+        with cf.ThreadPoolExecutor(max_workers=16) as pool:
+            futs = {pool.submit(b.submit, q): q for q in queries}
+            for f, q in futs.items():
+                assert f.result(timeout=10) == f"res:{q}"
+        s = b.stats()
+        assert s["queries"] >= len(set(queries)) or s["queries"] >= 1
+    finally:
+        b.close()
