@@ -22,7 +22,7 @@ from .crawler.rss import FeedMonitor
 from .crawler.scheduler import Scheduler
 from .crawler.worker import CrawlResult, CrawlWorker
 from .engine import HybridEngine
-from .errors import InfoMeshError
+from .errors import GpuExtensionMissing, InfoMeshError
 from .index.link_graph import LinkGraph
 from .index.local_store import Document, LocalStore
 from .search.batcher import QueryBatcher
@@ -139,6 +139,11 @@ class AppContext:
                     hbm_budget_gb=cfg.gpu.hbm_budget_gb,
                     embed_max_chars=cfg.index.embed_max_chars,
                     require_extension=cfg.gpu.require_extension)
+            except GpuExtensionMissing:
+                # gpu.require_extension means FAIL, not degrade: a GPU
+                # node silently serving the CPU path is exactly what
+                # the knob exists to prevent
+                raise
             except Exception as e:
                 log.warning("engine unavailable: %s", e)
         if ctx.engine is not None:

@@ -362,3 +362,18 @@ def test_require_extension_fails_loudly(monkeypatch):
     monkeypatch.setattr(_ext, "available", lambda: False)
     with pytest.raises(GpuExtensionMissing):
         HybridEngine(device="cuda", require_extension=True)
+
+
+def test_create_fails_loudly_when_extension_required_and_missing(monkeypatch):
+    """AppContext.create must NOT swallow GpuExtensionMissing into a
+    CPU-only degrade — the knob exists to make that loud."""
+    import torch
+
+    from infomesh_amd.ops import _ext
+
+    monkeypatch.setattr(torch.cuda, "is_available", lambda: True)
+    monkeypatch.setattr(_ext, "available", lambda: False)
+    from infomesh_amd.errors import GpuExtensionMissing
+    with pytest.raises(GpuExtensionMissing):
+        AppContext.create(config=Config(), with_worker=False,
+                          with_engine=True, in_memory=True)
