@@ -30,7 +30,11 @@ def synth_corpus_arrays(n_docs: int, avg_len: int = 120,
     if total >= 50_000_000:
         import multiprocessing as mp
         import os
-        workers = min(16, os.cpu_count() or 4)
+        # under torchrun every rank generates concurrently — share the
+        # host's cores instead of spawning world*16 workers (the 8-GPU
+        # SCALE run would oversubscribe and slow ALL ranks' setup)
+        world = int(os.environ.get("WORLD_SIZE", "1") or 1)
+        workers = max(2, min(16, (os.cpu_count() or 4) // max(world, 1)))
         per = (total + workers - 1) // workers
         sizes = [min(per, total - i * per) for i in range(workers)]
         sizes = [s for s in sizes if s > 0]
