@@ -574,3 +574,30 @@ def test_handlers_reject_invalid_args(mcp_ctx=None):
     ok = h.call("status", {})
     assert "error" not in ok
     ctx.close()
+
+
+def test_bench_multi_rank_cpu_contract(tmp_path):
+    """The driver's N>1 launch shape (torch.distributed.run, gloo on
+    CPU) must produce exactly one valid JSON contract line from rank 0
+    — de-risks the round-end 8-GPU SCALE run."""
+    import json as _json
+    import subprocess
+    import sys
+    from pathlib import Path
+
+    root = Path(__file__).resolve().parent.parent
+    proc = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29515", str(root / "bench.py"),
+         "--gpus", "2", "--steps", "2", "--warmup", "1",
+         "--docs-per-gpu", "5000"],
+        capture_output=True, text=True, timeout=240, cwd=root)
+    assert proc.returncode == 0, proc.stderr[-2000:]
+    lines = [ln for ln in proc.stdout.splitlines()
+             if ln.startswith("{") and '"metric"' in ln]
+    assert len(lines) == 1, proc.stdout[-2000:]
+    rec = _json.loads(lines[0])
+    assert rec["n_gpus"] == 2 and rec["scaling"] == "weak"
+    assert rec["config"]["parallelism"] == "shard2"
+    assert rec["value"] > 0 and rec["ms_per_step"] > 0
