@@ -580,10 +580,11 @@ extern "C" void infomesh_gemm8_bf16_nt(
   // cannot keep it resident; splitting N into chunks whose slice fits
   // ~2.5 MB trades (n_chunks-1) extra reads of A for eliminating the
   // B thrash. Only applied when that trade is favorable.
-  static const int chunk_ov = [] {
-    const char* e = getenv("INFOMESH_GEMM8_CHUNK");
-    return e ? atoi(e) : 0;
-  }();
+  // read per call (a getenv is ~100 ns vs the 100s-of-us GEMMs it
+  // gates): a first-call-latched static made the flag unusable from
+  // code that builds the extension before configuring the env
+  const char* chunk_env = getenv("INFOMESH_GEMM8_CHUNK");
+  const int chunk_ov = chunk_env ? atoi(chunk_env) : 0;
   const long b_bytes = (long)N * K * 2;
   if (chunk_ov && b_bytes > 3 * 1024 * 1024 && batch == 1) {
     int cols = (int)(2.5 * 1024 * 1024 / (K * 2)) / G8_BN * G8_BN;
