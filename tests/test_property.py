@@ -534,3 +534,46 @@ def test_batcher_maps_each_query_to_its_own_result(queries):
         assert s["queries"] >= len(set(queries)) or s["queries"] >= 1
     finally:
         b.close()
+
+
+@given(st.lists(st.floats(0.1, 10.0), min_size=1, max_size=15),
+       st.integers(0, 2 ** 31 - 1),
+       st.sampled_from(["root", "signature", "entry_hash", "proof_path"]))
+@settings(max_examples=20, deadline=None)
+def test_credit_proof_verifies_and_rejects_any_tamper(qtys, seed, field):
+    """A freshly built Merkle credit proof always verifies; tampering
+    with ANY of root/signature/sampled-entry/proof-path is detected."""
+    import json
+    import random as _random
+
+    from infomesh_amd.credits.ledger import Action, CreditLedger
+    from infomesh_amd.credits.verification import CreditProofBuilder
+    from infomesh_amd.trust.keys import KeyPair
+
+    kp = KeyPair.generate()
+    led = CreditLedger(":memory:", kp=kp, off_peak_fn=lambda ts: False)
+    for q in qtys:
+        led.record_action(Action.CRAWL, q)
+    proof = CreditProofBuilder(led, kp).build_proof(
+        n_samples=2, rng=_random.Random(seed))
+    assert CreditProofBuilder.verify_proof(proof)
+
+    bad = json.loads(json.dumps(proof))   # deep copy
+    if field == "root":
+        bad["root"] = "00" * 32
+    elif field == "signature":
+        bad["signature"] = "11" * 64
+    elif field == "entry_hash" and bad["samples"]:
+        bad["samples"][0]["entry_hash"] = "ff" * 32
+    elif field == "proof_path" and bad["samples"] \
+            and bad["samples"][0]["proof"].get("path"):
+        p = bad["samples"][0]["proof"]["path"][0]
+        if isinstance(p, (list, tuple)) and len(p) == 2:
+            bad["samples"][0]["proof"]["path"][0] = [p[0], "ee" * 32]
+        else:
+            bad["samples"][0]["proof"]["path"][0] = "ee" * 32
+    else:
+        led.close()
+        return   # nothing to tamper (no samples)
+    assert not CreditProofBuilder.verify_proof(bad)
+    led.close()
