@@ -577,3 +577,33 @@ def test_credit_proof_verifies_and_rejects_any_tamper(qtys, seed, field):
         return   # nothing to tamper (no samples)
     assert not CreditProofBuilder.verify_proof(bad)
     led.close()
+
+
+@given(st.integers(1, 4096), st.floats(1.0, 288.0),
+       st.sampled_from(["unicode61", "ascii", "porter", "trigram"]),
+       st.integers(1, 19))
+@settings(max_examples=20, deadline=None)
+def test_config_save_load_roundtrip(topk, hbm, tok, lvl):
+    """save_config -> load_config preserves every non-default value
+    (TOML round-trip; values chosen inside the clamp ranges —
+    hbm_budget_gb clamps at the physical 288 GB)."""
+    import dataclasses as dc
+    import tempfile
+    from pathlib import Path as P
+
+    from infomesh_amd.config import Config, load_config, save_config
+
+    cfg = Config()
+    cfg = dc.replace(
+        cfg,
+        gpu=dc.replace(cfg.gpu, topk_per_shard=topk, hbm_budget_gb=hbm),
+        index=dc.replace(cfg.index, fts_tokenizer=tok,
+                         snapshot_compression_level=lvl))
+    with tempfile.TemporaryDirectory() as d:
+        p = P(d) / "config.toml"
+        save_config(cfg, p)
+        back = load_config(p, env={})
+    assert back.gpu.topk_per_shard == topk
+    assert abs(back.gpu.hbm_budget_gb - hbm) < 1e-6
+    assert back.index.fts_tokenizer == tok
+    assert back.index.snapshot_compression_level == lvl
