@@ -21,33 +21,43 @@ class Session:
 
 class SessionStore:
     def __init__(self, ttl_s: float = 3600.0, max_sessions: int = 1000):
+        import threading
+        self._lock = threading.Lock()
         self.ttl_s = ttl_s
         self.max_sessions = max_sessions
         self._sessions: dict[str, Session] = {}
 
     def create(self) -> Session:
-        self._gc()
-        s = Session(session_id=secrets.token_hex(16))
-        self._sessions[s.session_id] = s
-        return s
+        # handlers run on a 256-thread pool: every read-modify of the
+        # session dict must hold the lock (iteration during _gc raced
+        # with concurrent inserts)
+        with self._lock:
+            self._gc()
+            s = Session(session_id=secrets.token_hex(16))
+            self._sessions[s.session_id] = s
+            return s
 
     def get(self, session_id: str) -> Session | None:
-        s = self._sessions.get(session_id)
-        if s is None:
-            return None
-        if time.time() - s.last_seen > self.ttl_s:
-            del self._sessions[session_id]
-            return None
-        s.last_seen = time.time()
-        return s
+        with self._lock:
+            s = self._sessions.get(session_id)
+            if s is None:
+                return None
+            if time.time() - s.last_seen > self.ttl_s:
+                del self._sessions[session_id]
+                return None
+            s.last_seen = time.time()
+            return s
 
     def touch(self, session_id: str) -> Session:
-        s = self.get(session_id)
-        if s is None:
-            s = Session(session_id=session_id)
-            self._sessions[session_id] = s
-        s.calls += 1
-        return s
+        with self._lock:
+            s = self._sessions.get(session_id)
+            now = time.time()
+            if s is None or now - s.last_seen > self.ttl_s:
+                s = Session(session_id=session_id)
+                self._sessions[session_id] = s
+            s.last_seen = now
+            s.calls += 1
+            return s
 
     def _gc(self) -> None:
         now = time.time()
@@ -67,21 +77,26 @@ class AnalyticsTracker:
     """Per-tool call counts + latency percentiles."""
 
     def __init__(self, max_samples: int = 1000):
+        import threading
         self.max_samples = max_samples
+        self._lock = threading.Lock()   # recorded from the handler pool
         self._latency: dict[str, list[float]] = defaultdict(list)
         self._errors: dict[str, int] = defaultdict(int)
 
     def record(self, tool: str, elapsed_ms: float, error: bool = False) -> None:
-        lst = self._latency[tool]
-        lst.append(elapsed_ms)
-        if len(lst) > self.max_samples:
-            del lst[: len(lst) // 2]
-        if error:
-            self._errors[tool] += 1
+        with self._lock:
+            lst = self._latency[tool]
+            lst.append(elapsed_ms)
+            if len(lst) > self.max_samples:
+                del lst[: len(lst) // 2]
+            if error:
+                self._errors[tool] += 1
 
     def report(self) -> dict:
         out = {}
-        for tool, lst in self._latency.items():
+        with self._lock:
+            snap = {t: list(v) for t, v in self._latency.items()}
+        for tool, lst in snap.items():
             s = sorted(lst)
             n = len(s)
             out[tool] = {

@@ -601,3 +601,40 @@ def test_bench_multi_rank_cpu_contract(tmp_path):
     assert rec["n_gpus"] == 2 and rec["scaling"] == "weak"
     assert rec["config"]["parallelism"] == "shard2"
     assert rec["value"] > 0 and rec["ms_per_step"] > 0
+
+
+def test_session_and_analytics_thread_safety():
+    """SessionStore/AnalyticsTracker are hammered by the 256-thread
+    handler pool: concurrent create/touch/gc/record must never corrupt
+    state or raise (dict-changed-during-iteration class of bug)."""
+    import concurrent.futures as cf
+    import random
+
+    from infomesh_amd.mcp.session import AnalyticsTracker, SessionStore
+
+    store = SessionStore(ttl_s=0.05, max_sessions=50)
+    tracker = AnalyticsTracker(max_samples=64)
+    rng = random.Random(3)
+    errs = []
+
+    def worker(i):
+        try:
+            for j in range(200):
+                r = rng.random()
+                if r < 0.4:
+                    store.create()
+                elif r < 0.8:
+                    store.touch(f"sess-{j % 20}")
+                else:
+                    store.get(f"sess-{j % 20}")
+                tracker.record(f"tool{j % 5}", r * 10.0,
+                               error=(j % 17 == 0))
+        except Exception as e:   # pragma: no cover - the assertion
+            errs.append(repr(e))
+
+    with cf.ThreadPoolExecutor(max_workers=32) as pool:
+        list(pool.map(worker, range(32)))
+    assert not errs, errs[:3]
+    assert store.count() <= 50 + 32   # bounded (gc races are benign)
+    rep = tracker.report()
+    assert rep and all(v["calls"] > 0 for v in rep.values())
