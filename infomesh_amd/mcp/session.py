@@ -114,14 +114,19 @@ class WebhookRegistry:
     """Registered webhooks notified on index events (in-process)."""
 
     def __init__(self):
+        import threading
+        self._lock = threading.Lock()
         self._hooks: dict[str, list] = defaultdict(list)
 
     def register(self, event: str, callback) -> None:
-        self._hooks[event].append(callback)
+        with self._lock:
+            self._hooks[event].append(callback)
 
     def fire(self, event: str, payload: dict) -> int:
         n = 0
-        for cb in self._hooks.get(event, []):
+        with self._lock:   # snapshot: register() may run concurrently
+            hooks = list(self._hooks.get(event, []))
+        for cb in hooks:
             try:
                 cb(payload)
                 n += 1
