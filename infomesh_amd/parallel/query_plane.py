@@ -87,6 +87,18 @@ def rrf_fuse(ids_lists: list[torch.Tensor], score_lists: list[torch.Tensor],
     return out_ids, out_scores
 
 
+def _as_i64(sl: torch.Tensor) -> torch.Tensor:
+    """Bit-cast a float32 id slice back to int64 [..., 2k] -> [..., k],
+    robust to storage offset/stride parity (a contiguous slice keeps
+    its parent's offset, and float->int64 view demands even offset,
+    stride-1 last dim and even length)."""
+    shp = sl.shape[:-1] + (sl.shape[-1] // 2,)
+    flat = sl.contiguous().reshape(-1)
+    if flat.storage_offset() % 2:
+        flat = flat.clone()
+    return flat.view(torch.int64).reshape(shp)
+
+
 class DistributedQueryPlane:
     """SPMD query plane: every rank calls search_batch collectively."""
 
@@ -251,8 +263,11 @@ class DistributedQueryPlane:
         g = self.fabric.all_gather(packed)        # [W, B, 6k]
         bm_s = g[:, :, :k_].contiguous()
         dn_s = g[:, :, k_:2 * k_].contiguous()
-        bm_i = g[:, :, 2 * k_:4 * k_].contiguous().view(torch.int64)
-        dn_i = g[:, :, 4 * k_:6 * k_].contiguous().view(torch.int64)
+        # .contiguous() is a no-op for an already-contiguous slice, which
+        # at world=1 leaves an ODD float storage offset when k is odd —
+        # view(int64) then raises. _as_i64 guarantees offset-0 storage.
+        bm_i = _as_i64(g[:, :, 2 * k_:4 * k_])
+        dn_i = _as_i64(g[:, :, 4 * k_:6 * k_])
         for tk in (self.shard._get_topk(),
                    getattr(self.shard, "_topk_dense", None)):
             if tk is not None and getattr(tk, "defer_check", False):

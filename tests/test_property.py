@@ -607,3 +607,30 @@ def test_config_save_load_roundtrip(topk, hbm, tok, lvl):
     assert abs(back.gpu.hbm_budget_gb - hbm) < 1e-6
     assert back.index.fts_tokenizer == tok
     assert back.index.snapshot_compression_level == lvl
+
+
+@given(st.integers(1, 16), st.integers(1, 64), st.integers(0, 2**31))
+@settings(max_examples=25, deadline=None)
+def test_packed_gather_idbitcast_roundtrip(B, k, seed):
+    """The query plane packs int64 ids into the float32 all-gather
+    payload via bit-casting (one collective instead of four). The
+    pack/unpack must be bit-exact for ANY id values including
+    negatives (-1 sentinels), NaN-pattern-aliasing bit patterns and
+    ids > 2^24 (float-precision traps if anyone ever 'simplified'
+    the cast to a value conversion)."""
+    import torch as t
+
+    g = t.Generator().manual_seed(seed)
+    ids = t.randint(-1, 2**62, (B, k), dtype=t.int64, generator=g)
+    ids[0, 0] = -1
+    ids[-1, -1] = (1 << 62) + 12345
+    scores = t.randn(B, k, generator=g)
+    packed = t.cat([scores, ids.view(t.float32).reshape(B, 2 * k)], dim=1)
+    from infomesh_amd.parallel.query_plane import _as_i64
+
+    # simulate the wire: contiguous copy as float32 (what gloo/RCCL do)
+    wire = packed.contiguous().clone()
+    back_scores = wire[:, :k]
+    back_ids = _as_i64(wire[:, k:])
+    assert t.equal(back_ids, ids)
+    assert t.equal(back_scores, scores)
