@@ -634,3 +634,33 @@ def test_packed_gather_idbitcast_roundtrip(B, k, seed):
     back_ids = _as_i64(wire[:, k:])
     assert t.equal(back_ids, ids)
     assert t.equal(back_scores, scores)
+
+
+@given(st.integers(1, 9), st.integers(0, 10_000))
+@settings(max_examples=10, deadline=None)
+def test_engine_search_many_padding_never_leaks(B, seed):
+    """search_many pads the batch to a pow2 bucket — the pad queries
+    (empty term lists) must never leak hits into or displace the real
+    B results."""
+    import random as _r
+
+    from infomesh_amd.engine import HybridEngine
+    from infomesh_amd.index.local_store import Document
+
+    rng = _r.Random(seed)
+    eng = HybridEngine(device="cpu", use_encoder=False)
+    words = ["rocm", "hip", "kernel", "wave", "lds", "mfma", "xgmi",
+             "shard", "index", "query"]
+    for i in range(30):
+        txt = " ".join(rng.choices(words, k=12))
+        d = Document(url=f"http://x/{i}", title=f"t{i}", text=txt)
+        d.doc_id = i + 1
+        eng.add_document(d)
+    eng.flush()
+    queries = [" ".join(rng.choices(words, k=2)) for _ in range(B)]
+    out = eng.search_many(queries, limit=5)
+    assert len(out) == B
+    for hits in out:
+        assert len(hits) <= 5
+        for h in hits:
+            assert 1 <= h.doc_id <= 30
