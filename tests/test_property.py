@@ -664,3 +664,26 @@ def test_engine_search_many_padding_never_leaks(B, seed):
         assert len(hits) <= 5
         for h in hits:
             assert 1 <= h.doc_id <= 30
+
+
+@given(st.lists(st.text(min_size=1, max_size=25), min_size=1, max_size=8),
+       st.integers(1, 5))
+@settings(max_examples=15, deadline=None)
+def test_rag_chunking_covers_and_bounds(passages, max_chunks):
+    """RAG output chunking: every produced chunk's text comes from the
+    inputs, counts are bounded, and the formatter is total."""
+    from infomesh_amd.search.rag import format_rag_output
+
+    results = [{"url": f"http://x/{i}", "title": f"t{i}",
+                "snippet": p, "score": 1.0 / (i + 1)}
+               for i, p in enumerate(passages)]
+    out = format_rag_output("some query", results,
+                            chunk_size=64 * max_chunks)
+    assert out.chunks is not None
+    for ch in out.chunks:
+        # chunk text derives from the snippets (allowing truncation)
+        assert ch.text == "" or any(ch.text[:20] in p or p in ch.text
+                                    for p in passages)
+    # answer mode is total too
+    out2 = format_rag_output("some query", results, answer_mode=True)
+    assert out2.confidence is None or 0.0 <= out2.confidence <= 1.0
