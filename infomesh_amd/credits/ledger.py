@@ -152,10 +152,28 @@ class CreditLedger(SQLiteStore):
 
     # ----------------------------------------------------------- balance
     def balance(self) -> float:
+        """Durable entries plus accrued-but-unflushed async credits —
+        callers of record_action_async see their contribution
+        immediately even though the signed entry lands at the next
+        flush interval."""
         row = self.execute(
             "SELECT COALESCE(SUM(credits), 0) AS b FROM credit_entries"
         ).fetchone()
-        return float(row["b"])
+        pending = 0.0
+        with self._accum_lock:
+            for action, qty in self._accum.items():
+                pending += self._weights[action] * qty
+            n_search = self._accum_searches
+        if n_search:
+            pending += self._weights[Action.SEARCH_SPEND] \
+                * self._base_cost_now() * n_search
+        return float(row["b"]) + pending
+
+    def _base_cost_now(self) -> float:
+        """Current per-search cost WITHOUT re-entering balance()."""
+        if self._base_search_cost is not None:
+            return self._base_search_cost
+        return TIER_SEARCH_COST[0]
 
     def tier(self) -> int:
         b = self.balance()

@@ -42,11 +42,24 @@ class ContentAttestation:
 
 
 def create_attestation(kp: KeyPair, url: str, raw_hash: str,
-                       text_hash: str) -> ContentAttestation:
+                       text_hash: str,
+                       sign: bool = True) -> ContentAttestation:
+    """sign=False defers the ~3 ms pure-python Ed25519 signature —
+    ingest stores the claim unsigned and sign_attestation() completes
+    it on first serve (attestations are write-mostly: the reference
+    publishes them to the DHT on demand, not per crawl)."""
     att = ContentAttestation(url=url, raw_hash=raw_hash,
                              text_hash=text_hash, node_id=kp.node_id,
                              ts=time.time(), public_key=kp.public.hex())
-    att.signature = kp.sign(att.payload()).hex()
+    if sign:
+        att.signature = kp.sign(att.payload()).hex()
+    return att
+
+
+def sign_attestation(kp: KeyPair, att: ContentAttestation) -> ContentAttestation:
+    """Complete a deferred signature (idempotent)."""
+    if not att.signature:
+        att.signature = kp.sign(att.payload()).hex()
     return att
 
 

@@ -377,3 +377,28 @@ def test_create_fails_loudly_when_extension_required_and_missing(monkeypatch):
     with pytest.raises(GpuExtensionMissing):
         AppContext.create(config=Config(), with_worker=False,
                           with_engine=True, in_memory=True)
+
+
+def test_ingest_defers_signatures_but_serves_them_signed():
+    """index_document must not pay per-page Ed25519 on the hot path:
+    attestations are stored unsigned and signed on first serve; crawl
+    credits accrue via the batched ledger but are visible in balance()
+    immediately."""
+    from infomesh_amd.trust.attestation import verify_attestation
+
+    ctx = AppContext.create(config=Config(), with_worker=False,
+                            with_engine=False, in_memory=True)
+    try:
+        for i in range(5):
+            ctx.index_document(Document(url=f"http://d/{i}", title="t",
+                                        text=f"body {i} unique"))
+        assert len(ctx.attestations) == 5
+        assert all(not a.signature for a in ctx.attestations)  # deferred
+        assert ctx.ledger.balance() > 0                        # pending-aware
+        served = ctx.signed_attestations(3)
+        assert len(served) == 3
+        assert all(a.signature and verify_attestation(a) for a in served)
+        n = ctx.ledger.flush_pending()
+        assert n >= 1 and ctx.ledger.verify_chain()
+    finally:
+        ctx.close()
