@@ -73,6 +73,7 @@ class TakedownManager(SQLiteStore):
             (url_pattern, reason, claimant, notice.ts,
              json.dumps(notice.to_dict())))
         self.commit()
+        self._pat_cache = None
         return notice
 
     @staticmethod
@@ -112,11 +113,28 @@ class TakedownManager(SQLiteStore):
             (now - COMPLIANCE_WINDOW_S,)).fetchall()]
 
     def is_blocked(self, url: str) -> bool:
-        """Would this URL be rejected at (re)index time?"""
-        from ..index.local_store import extract_domain
-        domain = extract_domain(url)
-        for row in self.execute("SELECT url_pattern FROM takedowns"):
-            p = row["url_pattern"]
-            if p == url or (p.startswith("domain:") and p[7:] == domain):
-                return True
+        """Would this URL be rejected at (re)index time? Runs on every
+        index_document call — patterns are cached in memory (takedowns
+        change rarely) instead of a table scan per page."""
+        exact, domains = self._patterns()
+        if not exact and not domains:
+            return False
+        if url in exact:
+            return True
+        if domains:
+            from ..index.local_store import extract_domain
+            return extract_domain(url) in domains
         return False
+
+    def _patterns(self):
+        cached = getattr(self, "_pat_cache", None)
+        if cached is None:
+            exact, domains = set(), set()
+            for row in self.execute("SELECT url_pattern FROM takedowns"):
+                p = row["url_pattern"]
+                if p.startswith("domain:"):
+                    domains.add(p[7:])
+                else:
+                    exact.add(p)
+            cached = self._pat_cache = (exact, domains)
+        return cached

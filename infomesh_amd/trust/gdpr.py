@@ -54,6 +54,7 @@ class DeletionManager(SQLiteStore):
             (subject, reason, content_hash(requester)[:16], ts,
              json.dumps(record)))
         self.commit()
+        self._subj_cache = None
         self.enforce()
         return record
 
@@ -80,13 +81,28 @@ class DeletionManager(SQLiteStore):
         return removed
 
     def is_forgotten(self, url: str) -> bool:
-        """Index-time guard: refuses re-indexing of deleted subjects."""
-        domain = extract_domain(url)
-        for row in self.execute("SELECT subject FROM deletions"):
-            s = row["subject"]
-            if s == url or (s.startswith("domain:") and s[7:] == domain):
-                return True
-        return False
+        """Index-time guard: refuses re-indexing of deleted subjects.
+        Runs on every index_document call — subjects cached in memory
+        (deletion records change rarely)."""
+        exact, domains = self._subjects()
+        if not exact and not domains:
+            return False
+        if url in exact:
+            return True
+        return bool(domains) and extract_domain(url) in domains
+
+    def _subjects(self):
+        cached = getattr(self, "_subj_cache", None)
+        if cached is None:
+            exact, domains = set(), set()
+            for row in self.execute("SELECT subject FROM deletions"):
+                s = row["subject"]
+                if s.startswith("domain:"):
+                    domains.add(s[7:])
+                else:
+                    exact.add(s)
+            cached = self._subj_cache = (exact, domains)
+        return cached
 
     def export_records(self) -> list[dict]:
         return [json.loads(r["record"]) for r in
