@@ -687,3 +687,40 @@ def test_rag_chunking_covers_and_bounds(passages, max_chunks):
     # answer mode is total too
     out2 = format_rag_output("some query", results, answer_mode=True)
     assert out2.confidence is None or 0.0 <= out2.confidence <= 1.0
+
+
+@given(st.integers(0, 23), st.integers(0, 23), st.integers(0, 23),
+       st.integers(0, 59))
+@settings(max_examples=60, deadline=None)
+def test_off_peak_window_wraps_correctly(start_h, end_h, hour, minute):
+    """is_off_peak handles wrap-around windows (23:00-07:00) for every
+    (start, end, sample-time) combination: inside iff the hour lies in
+    the half-open cyclic interval [start, end)."""
+    import datetime as dt
+
+    from infomesh_amd.credits.scheduling import is_off_peak
+
+    ts = dt.datetime(2026, 3, 10, hour, minute).timestamp()
+    got = is_off_peak(ts, start_h=start_h, end_h=end_h)
+    if start_h == end_h:
+        expected = False   # empty window
+    elif start_h < end_h:
+        expected = start_h <= hour < end_h
+    else:
+        expected = hour >= start_h or hour < end_h
+    assert got == expected, (start_h, end_h, hour, got, expected)
+
+
+@given(st.integers(1, 400), st.floats(0.01, 5.0))
+@settings(max_examples=20, deadline=None)
+def test_farming_detector_monotone_suspicion(n, spacing_s):
+    """Perfectly regular high-rate action streams must never RAISE the
+    reward multiplier; the multiplier stays within [0, 1]."""
+    from infomesh_amd.credits.farming import FarmingDetector
+
+    det = FarmingDetector()
+    t0 = 1_700_000_000.0
+    for i in range(n):
+        det.record("crawl", ts=t0 + i * spacing_s)
+    m = det.multiplier(now=t0 + n * spacing_s)
+    assert 0.0 <= m <= 1.0
