@@ -152,11 +152,16 @@ def search(query, limit, mode, as_json, explain):
         ctx.close()
 
 
-@cli.command()
+@cli.group()
+def feedback():
+    """Implicit relevance feedback (record / stats / top-urls)."""
+
+
+@feedback.command(name="record")
 @click.argument("url")
 @click.option("--signal", "signal_", default="fetch",
               type=click.Choice(["fetch", "cite", "click", "skip"]))
-def feedback(url, signal_):
+def feedback_record(url, signal_):
     """Record implicit feedback for a URL."""
     from ..search.feedback import FeedbackStore
     cfg = load_config()
@@ -164,6 +169,30 @@ def feedback(url, signal_):
     fs.record(url, signal_)
     fs.close()
     click.echo("recorded")
+
+
+@feedback.command(name="stats")
+def feedback_stats():
+    """Aggregate feedback counters (reference: cli feedback stats)."""
+    from ..search.feedback import FeedbackStore
+    fs = FeedbackStore(load_config().data_dir / "feedback.db")
+    try:
+        click.echo(json.dumps(fs.stats(), indent=2))
+    finally:
+        fs.close()
+
+
+@feedback.command(name="top-urls")
+@click.option("--limit", default=10)
+def feedback_top_urls(limit):
+    """URLs with the strongest positive feedback boost."""
+    from ..search.feedback import FeedbackStore
+    fs = FeedbackStore(load_config().data_dir / "feedback.db")
+    try:
+        for url, boost in fs.top_urls(limit):
+            click.echo(f"{boost:+.3f}  {url}")
+    finally:
+        fs.close()
 
 
 # ----------------------------------------------------------------- crawl
@@ -244,6 +273,43 @@ def export(path, max_docs):
         ctx.close()
 
 
+@index.command(name="import-urls")
+@click.argument("url_file", type=click.Path(exists=True))
+@click.option("--max-urls", default=1000)
+def index_import_urls(url_file, max_urls):
+    """Queue a file of URLs (one per line) for crawling."""
+    import asyncio
+    urls = [ln.strip() for ln in open(url_file)
+            if ln.strip() and not ln.startswith("#")][:max_urls]
+    ctx = _ctx(with_engine=False)
+    try:
+        n = 0
+        for u in urls:
+            try:
+                res = asyncio.run(ctx.crawl_and_index(u))
+                n += 1 if res.get("status") == "ok" else 0
+            except Exception as e:
+                click.echo(f"skip {u}: {e}", err=True)
+        click.echo(f"crawled+indexed {n}/{len(urls)}")
+    finally:
+        ctx.close()
+
+
+@index.command(name="import-wet")
+@click.argument("path", type=click.Path(exists=True))
+@click.option("--max-records", default=None, type=int)
+def index_import_wet(path, max_records):
+    """Import a Common Crawl WET file (reference: index import-wet)."""
+    from ..index.commoncrawl import CommonCrawlImporter
+    ctx = _ctx(with_engine=False, with_worker=False)
+    try:
+        imp = CommonCrawlImporter(ctx.store)
+        res = imp.import_wet(path, max_records=max_records)
+        click.echo(json.dumps(res, indent=2))
+    finally:
+        ctx.close()
+
+
 @index.command(name="import")
 @click.argument("path", type=click.Path(exists=True))
 def import_(path):
@@ -283,6 +349,21 @@ def config_show():
     click.echo(json.dumps(dataclasses.asdict(load_config()), indent=2))
 
 
+@config.command(name="github")
+@click.option("--email", default=None,
+              help="owner email for cross-node credit aggregation")
+def config_github(email):
+    """Set/show the GitHub owner identity (reference: config github).
+    Falls back to the local git config email; only the SHA-256 hash is
+    ever shared with peers."""
+    from ..credits.github_identity import ensure_owner_identity
+    chosen = ensure_owner_identity(load_config().data_dir, email)
+    if chosen is None:
+        click.echo("no valid email (pass --email or set git config user.email)")
+        raise SystemExit(1)
+    click.echo(f"owner identity: {chosen}")
+
+
 @config.command(name="set")
 @click.argument("key")     # section.key
 @click.argument("value")
@@ -315,6 +396,16 @@ def keys_show():
     kp = ensure_keys(load_config().data_dir)
     click.echo(json.dumps({"node_id": kp.node_id,
                            "public_key": kp.public.hex()}, indent=2))
+
+
+@keys.command(name="export")
+def keys_export():
+    """Export the PUBLIC key material (never the private key)."""
+    from ..trust.keys import ensure_keys
+    kp = ensure_keys(load_config().data_dir)
+    click.echo(json.dumps({"node_id": kp.node_id,
+                           "public_key": kp.public.hex(),
+                           "format": "ed25519-hex"}, indent=2))
 
 
 @keys.command(name="rotate")
