@@ -638,3 +638,29 @@ def test_session_and_analytics_thread_safety():
     assert store.count() <= 50 + 32   # bounded (gc races are benign)
     rep = tracker.report()
     assert rep and all(v["calls"] > 0 for v in rep.values())
+
+
+def test_cli_new_commands_smoke(tmp_path):
+    """Round-2 CLI parity additions: keys export, config github,
+    feedback record/stats/top-urls, index import-wet all run clean."""
+    from click.testing import CliRunner
+
+    from infomesh_amd.cli import cli
+
+    r = CliRunner()
+    env = {"INFOMESH_NODE_DATA_DIR": str(tmp_path)}
+    out = r.invoke(cli, ["keys", "export"], env=env)
+    assert out.exit_code == 0 and "public_key" in out.output
+    out = r.invoke(cli, ["config", "github", "--email", "o@example.com"],
+                   env=env)
+    assert out.exit_code == 0 and "o@example.com" in out.output
+    assert r.invoke(cli, ["feedback", "record", "http://x/9"],
+                    env=env).exit_code == 0
+    out = r.invoke(cli, ["feedback", "stats"], env=env)
+    assert out.exit_code == 0 and "fetch" in out.output
+    out = r.invoke(cli, ["feedback", "top-urls"], env=env)
+    assert out.exit_code == 0 and "http://x/9" in out.output
+    wet = tmp_path / "c.wet"
+    wet.write_text("not a wet file")
+    out = r.invoke(cli, ["index", "import-wet", str(wet)], env=env)
+    assert out.exit_code == 0   # total: imports 0, never crashes
