@@ -106,8 +106,11 @@ def rank_local_results(query: str, hits: Sequence[SearchHit],
                        trust_fn: TrustFn | None = None,
                        now: float | None = None,
                        explain: bool = False,
+                       boost_fn=None,
                        ) -> list[SearchHit] | list[tuple[SearchHit, ScoreBreakdown]]:
-    """Composite-rank FTS hits in place (reference: ranking.py:241-285)."""
+    """Composite-rank FTS hits in place (reference: ranking.py:241-285).
+    boost_fn: implicit-feedback additive boost per url (capped by
+    FeedbackStore; reference applies it in ranking, feedback.py)."""
     if not hits:
         return []
     batch_max = max(h.bm25 for h in hits)
@@ -121,7 +124,7 @@ def rank_local_results(query: str, hits: Sequence[SearchHit],
             title_match=title_match_score(query, h.title),
             url_path=url_path_score(query, h.url),
         )
-        h.score = bd.total
+        h.score = bd.total + (boost_fn(h.url) if boost_fn else 0.0)
         out.append((h, bd))
     out.sort(key=lambda p: p[0].score, reverse=True)
     if explain:

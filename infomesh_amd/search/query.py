@@ -58,6 +58,7 @@ def preprocess_query(query: str, language: str = "en") -> str:
 
 def search_local(store: LocalStore, query: str, limit: int = 10,
                  authority_fn: AuthorityFn | None = None,
+                 boost_fn=None,
                  trust_fn: TrustFn | None = None,
                  enhance_snippets: bool = True) -> SearchResponse:
     t0 = time.perf_counter()
@@ -77,7 +78,7 @@ def search_local(store: LocalStore, query: str, limit: int = 10,
             if len(hits) >= limit:
                 break
     ranked = rank_local_results(eff, hits, authority_fn=authority_fn,
-                                trust_fn=trust_fn)
+                                trust_fn=trust_fn, boost_fn=boost_fn)
     ranked = ranked[:limit]
     if enhance_snippets:
         for h in ranked:

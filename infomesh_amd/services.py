@@ -59,6 +59,7 @@ class AppContext:
     cache: QueryCache
     feeds: FeedMonitor
     related: RelatedSearchTracker
+    feedback: object | None = None
     worker: CrawlWorker | None = None
     engine: HybridEngine | None = None
     batcher: QueryBatcher | None = None
@@ -116,6 +117,8 @@ class AppContext:
             feeds=FeedMonitor(path=None if in_memory else p("feeds.json")),
             related=RelatedSearchTracker(),
         )
+        from .search.feedback import FeedbackStore
+        ctx.feedback = FeedbackStore(p("feedback.db"))
         from .utils.governor import ResourceGovernor
         ctx.governor = ResourceGovernor()
         if cfg.api.otlp_endpoint:
@@ -272,7 +275,9 @@ class AppContext:
         has_filters = bool(pq.site or pq.after or pq.before or pq.language)
         if mode == "local" or (mode == "auto" and not engine_ready):
             resp = search_local(self.store, query, limit=limit,
-                                authority_fn=authority, trust_fn=trust_fn)
+                                authority_fn=authority, trust_fn=trust_fn,
+                                boost_fn=(self.feedback.url_boost
+                                          if self.feedback else None))
         elif mode in ("auto", "hybrid", "distributed"):
             if engine_ready and not has_filters:
                 # single-fusion path: the GPU plane RRF-fuses BM25 +
@@ -403,7 +408,8 @@ class AppContext:
         if self.batcher is not None:
             self.batcher.close()
         for c in (self.store, self.link_graph, self.dedup, self.ledger,
-                  self.trust, self.takedowns, self.deletions):
+                  self.trust, self.takedowns, self.deletions,
+                  self.feedback):
             try:
                 c.close()
             except Exception:

@@ -402,3 +402,28 @@ def test_ingest_defers_signatures_but_serves_them_signed():
         assert n >= 1 and ctx.ledger.verify_chain()
     finally:
         ctx.close()
+
+
+def test_feedback_boost_influences_local_ranking():
+    """Recorded implicit feedback must actually move local ranking
+    (reference applies FeedbackStore boosts in ranking)."""
+    ctx = AppContext.create(config=Config(), with_worker=False,
+                            with_engine=False, in_memory=True)
+    try:
+        for i in range(2):
+            ctx.index_document(Document(
+                url=f"http://b/{i}", title="same terms here",
+                text=f"identical body words rocm hip {'pad' * i}"))
+        base = ctx.search("identical body words", limit=2, mode="local",
+                          use_cache=False, deduct=False)
+        assert len(base.results) == 2
+        loser = base.results[-1].url
+        for _ in range(40):            # strong positive signal
+            ctx.feedback.record(loser, "cite")
+        ctx.feedback._boost_cache.clear()
+        ctx.cache.invalidate()
+        boosted = ctx.search("identical body words", limit=2,
+                             mode="local", use_cache=False, deduct=False)
+        assert boosted.results[0].url == loser
+    finally:
+        ctx.close()
