@@ -144,7 +144,9 @@ def search(query, limit, mode, as_json, explain):
     ctx = _ctx(with_worker=False)
     try:
         if explain:
-            click.echo(render_explanation(explain_search(ctx.store, q, limit)))
+            click.echo(render_explanation(explain_search(
+                ctx.store, q, limit,
+                boost_fn=ctx.feedback.url_boost if ctx.feedback else None)))
             return
         resp = ctx.search(q, limit=limit, mode=mode)
         click.echo(format_json(resp) if as_json else format_text(resp))

@@ -62,7 +62,9 @@ class Handlers:
                     "results": explain_search(
                         self.ctx.store, query, limit,
                         authority_fn=self.ctx.link_graph.url_authority,
-                        trust_fn=self.ctx.trust.trust_fn())}
+                        trust_fn=self.ctx.trust.trust_fn(),
+                        boost_fn=(self.ctx.feedback.url_boost
+                                  if self.ctx.feedback else None))}
         over = limit * 2 if (domain_allowlist or domain_blocklist
                              or recency_days) else limit
         resp = self.ctx.search(query, limit=over, mode=mode)

@@ -14,16 +14,20 @@ from ..index.ranking import (W_AUTHORITY, W_BM25,
 
 
 def explain_search(store: LocalStore, query: str, limit: int = 5,
-                   authority_fn=None, trust_fn=None) -> list[dict[str, Any]]:
+                   authority_fn=None, trust_fn=None,
+                   boost_fn=None) -> list[dict[str, Any]]:
     hits = store.search(query, limit=limit * 2)
     explained = rank_local_results(query, hits, authority_fn=authority_fn,
-                                   trust_fn=trust_fn, explain=True)
+                                   trust_fn=trust_fn, explain=True,
+                                   boost_fn=boost_fn)
     out = []
     for hit, bd in explained[:limit]:
         out.append({
             "url": hit.url,
             "title": hit.title,
             "total": round(hit.score, 4),
+            "feedback_boost": round(boost_fn(hit.url), 4) if boost_fn
+            else 0.0,
             "components": {
                 "bm25": {"value": round(bd.bm25, 4), "weight": W_BM25},
                 "freshness": {"value": round(bd.freshness, 4),
