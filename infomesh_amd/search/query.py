@@ -96,11 +96,11 @@ def search_hybrid(store: LocalStore, dense: DenseSearcher | None,
                   query: str, limit: int = 10,
                   authority_fn: AuthorityFn | None = None,
                   trust_fn: TrustFn | None = None,
-                  rrf_k: int = 60) -> SearchResponse:
+                  rrf_k: int = 60, boost_fn=None) -> SearchResponse:
     t0 = time.perf_counter()
     local = search_local(store, query, limit=limit * 2,
                          authority_fn=authority_fn, trust_fn=trust_fn,
-                         enhance_snippets=False)
+                         enhance_snippets=False, boost_fn=boost_fn)
     lists: list[Sequence[SearchHit]] = [local.results]
     sources = ["fts"]
     weights = [1.0]

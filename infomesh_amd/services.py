@@ -288,7 +288,9 @@ class AppContext:
                 resp = search_hybrid(
                     self.store, None, query, limit=limit,
                     authority_fn=authority, trust_fn=trust_fn,
-                    rrf_k=self.config.search.rrf_k)
+                    rrf_k=self.config.search.rrf_k,
+                    boost_fn=(self.feedback.url_boost
+                              if self.feedback else None))
                 if mode == "distributed":
                     resp.mode = "distributed"
         else:
