@@ -346,6 +346,13 @@ class AppContext:
                 if doc.text:
                     h.snippet = fast_snippet(q, doc.text)
                 hydrated.append(h)
+            if self.feedback is not None and hydrated:
+                # implicit-feedback boost on the GPU plane too: additive
+                # on the fused RRF score, then reorder (cheap: <=limit
+                # hits; url_boost is 60 s-cached per url)
+                for h in hydrated:
+                    h.score += self.feedback.url_boost(h.url)
+                hydrated.sort(key=lambda x: x.score, reverse=True)
             out.append(hydrated)
         return out
 
