@@ -39,6 +39,17 @@ class FeedbackStore(SQLiteStore):
             (url, query, signal, time.time()))
         self.commit()
         self._boost_cache.pop(url, None)
+        self._has_cache = True
+
+    def has_signals(self) -> bool:
+        """Cheap guard for hot paths: False until anything is recorded
+        (cached; invalidated by record())."""
+        cached = getattr(self, "_has_cache", None)
+        if cached is None:
+            row = self.execute(
+                "SELECT EXISTS(SELECT 1 FROM feedback) AS e").fetchone()
+            cached = self._has_cache = bool(row["e"])
+        return cached
 
     def url_boost(self, url: str) -> float:
         """Saturating additive boost in [-CAP, +CAP]."""

@@ -346,7 +346,8 @@ class AppContext:
                 if doc.text:
                     h.snippet = fast_snippet(q, doc.text)
                 hydrated.append(h)
-            if self.feedback is not None and hydrated:
+            if self.feedback is not None and hydrated \
+                    and self.feedback.has_signals():
                 # implicit-feedback boost on the GPU plane too: additive
                 # on the fused RRF score, then reorder (cheap: <=limit
                 # hits; url_boost is 60 s-cached per url)
