@@ -80,6 +80,19 @@ class SummaryReputation:
         """The [0,1] term consumed by trust/scoring.py (weight 0.20)."""
         return self.get(node_id).ewma
 
+    def sync_to_trust(self, trust) -> int:
+        """Push each node's summary-quality EWMA into the TrustStore's
+        0.20-weighted component (the reputation->trust bridge the
+        reference runs in its maintenance loop). Returns nodes synced."""
+        n = 0
+        for node_id in list(self._state):
+            rep = self.get(node_id)
+            if rep.samples >= 1:
+                trust.update_component(node_id, "summary_quality",
+                                       rep.ewma)
+                n += 1
+        return n
+
     def leaderboard(self) -> list[Reputation]:
         return sorted((self.get(n) for n in self._state),
                       key=lambda r: -r.ewma)

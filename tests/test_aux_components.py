@@ -175,3 +175,21 @@ def test_dashboard_sparkline_and_history(tmp_path, monkeypatch):
     assert layout is not None
     # two renders with history grow sparkline series without error
     render_dashboard(data, h)
+
+
+def test_reputation_syncs_into_trust_score():
+    """Summary-quality grades must reach the composite trust score
+    (the 0.20-weighted component, reference scoring)."""
+    from infomesh_amd.trust.reputation import SummaryReputation
+    from infomesh_amd.trust.scoring import TrustStore
+
+    rep = SummaryReputation()
+    trust = TrustStore(":memory:")
+    for _ in range(6):
+        rep.record("good-node", 0.95)
+        rep.record("bad-node", 0.05)
+    assert rep.sync_to_trust(trust) == 2
+    g = trust.score("good-node")
+    b = trust.score("bad-node")
+    assert g > b
+    trust.close()
