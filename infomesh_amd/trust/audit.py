@@ -92,7 +92,11 @@ class AuditScheduler:
             domain = extract_domain(url)
             self.trust.record_audit(domain, passed)
             if self.detector is not None and not passed:
-                self.detector.record(domain, "audit_fail")
+                threat = self.detector.record(domain, "audit_fail")
+                if threat.isolate:
+                    # detector verdict overrides the consecutive-
+                    # failures ladder (CRITICAL always isolates)
+                    self.trust.isolate(domain)
         self.history.append(result)
         return result
 

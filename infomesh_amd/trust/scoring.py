@@ -94,6 +94,16 @@ class TrustStore(SQLiteStore):
         self.commit()
         self.update_component(subject, "audit_pass", 1.0 if passed else 0.0)
 
+    def isolate(self, subject: str) -> None:
+        """Direct isolation (MaliciousNodeDetector verdicts: CRITICAL
+        always isolates; HIGH isolates on repeat forgery) — bypasses
+        the consecutive-failures ladder."""
+        self._ensure(subject)
+        self.execute(
+            "UPDATE trust SET isolated=1, updated_at=? WHERE subject=?",
+            (time.time(), subject))
+        self.commit()
+
     def score(self, subject: str) -> float:
         row = self._get(subject)
         if row is None:

@@ -193,3 +193,32 @@ def test_reputation_syncs_into_trust_score():
     b = trust.score("bad-node")
     assert g > b
     trust.close()
+
+
+def test_detector_critical_verdict_isolates_immediately():
+    """A CRITICAL detector verdict isolates the domain even before the
+    consecutive-failure ladder would."""
+    import asyncio
+
+    from infomesh_amd.trust.audit import AuditScheduler
+    from infomesh_amd.trust.detector import MaliciousNodeDetector
+    from infomesh_amd.trust.scoring import TrustStore
+    from infomesh_amd.index.local_store import Document, LocalStore
+
+    store = LocalStore(":memory:")
+    store.add_document(Document(url="http://evil.example/p", title="t",
+                                text="page body content here"))
+    trust = TrustStore(":memory:", isolation_failures=99)  # ladder off
+    det = MaliciousNodeDetector()
+    det.record("evil.example", "invalid_signature", 3)  # pre-escalated
+    det.record("evil.example", "invalid_proof", 3)
+
+    async def fetch(url):
+        return "DIFFERENT content — audit mismatch"
+
+    sched = AuditScheduler(store, trust, fetch, auditors=3,
+                           detector=det)
+    res = asyncio.run(sched.run_audit())
+    assert res is not None and not res.passed
+    assert trust.tier("evil.example") == "isolated"
+    store.close(); trust.close()
