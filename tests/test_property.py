@@ -728,3 +728,36 @@ def test_farming_detector_monotone_suspicion(n, spacing_s):
         det.record("crawl", ts=t0 + i * spacing_s)
     m = det.multiplier(now=t0 + n * spacing_s)
     assert 0.0 <= m <= 1.0
+
+
+@given(st.text(max_size=120))
+@settings(max_examples=150, deadline=None)
+def test_ssrf_guard_total_and_never_wrong_type(raw):
+    """validate_url either returns the url or raises InfoMeshError —
+    never any other exception, for arbitrary junk input."""
+    from infomesh_amd.errors import InfoMeshError
+    from infomesh_amd.security import is_url_safe, validate_url
+
+    try:
+        out = validate_url(raw)
+        assert out == raw
+        assert is_url_safe(raw)
+    except InfoMeshError:
+        assert not is_url_safe(raw)
+
+
+@given(st.sampled_from([
+    "http://127.0.0.1/x", "http://localhost/a", "https://[::1]/",
+    "http://10.0.0.8/", "http://192.168.1.1/p", "http://172.16.5.5/",
+    "http://169.254.169.254/latest/meta-data", "file:///etc/passwd",
+    "ftp://host/x", "gopher://host", "http://0.0.0.0/",
+    "http://user:pass@evil.com@10.0.0.1/", "http://[fe80::1]/",
+    "http://224.0.0.1/", "javascript:alert(1)",
+]))
+@settings(max_examples=30, deadline=None)
+def test_ssrf_guard_blocks_known_bad(url):
+    """The classic SSRF corpus is always rejected (metadata endpoint,
+    loopback, RFC1918, link-local, schemes, userinfo tricks)."""
+    from infomesh_amd.security import is_url_safe
+
+    assert not is_url_safe(url), url
