@@ -781,3 +781,35 @@ def test_scheduler_depth_and_domain_bounds(items):
             assert not ok          # depth cap
         added += 1 if ok else 0
     assert s.qsize() == added <= len(items)
+
+
+@given(st.lists(st.floats(0.1, 500.0), min_size=1, max_size=50),
+       st.floats(1.0, 1000.0))
+@settings(max_examples=20, deadline=None)
+def test_slo_tracker_p95_and_breach_consistency(lats, target):
+    """SLOTracker: reported p95 is a real order statistic of the
+    samples, and breach status agrees with target comparison."""
+    from infomesh_amd.utils.slo import SLOTracker
+
+    t = SLOTracker()
+    t.define("search", target_p95_ms=target)
+    for v in lats:
+        t.record("search", v, ok=True)
+    rep = t.report()["search"]
+    s = sorted(lats)
+    p95 = s[min(len(s) - 1, int(len(s) * 0.95))]
+    assert abs(rep["p95_ms"] - p95) < 0.006   # report rounds to 2 dp
+    assert rep["met"] == (p95 <= target) or abs(p95 - target) < 0.01
+
+
+@given(st.floats(0, 2e9), st.floats(0, 1), st.integers(0, 100_000),
+       st.booleans())
+@settings(max_examples=40, deadline=None)
+def test_data_quality_grade_total(crawled_at, trust, text_len, has_title):
+    """Quality grading is total with a known grade and bounded score
+    for any (age, trust, length, title) combination."""
+    from infomesh_amd.utils.data_quality import grade_document
+
+    g = grade_document(crawled_at, trust, text_len, has_title,
+                       now=2e9)
+    assert g.grade in ("A", "B", "C", "D", "F")
