@@ -664,3 +664,17 @@ def test_cli_new_commands_smoke(tmp_path):
     wet.write_text("not a wet file")
     out = r.invoke(cli, ["index", "import-wet", str(wet)], env=env)
     assert out.exit_code == 0   # total: imports 0, never crashes
+
+
+def test_examples_quickstart_runs():
+    """The shipped quickstart example must keep working end to end."""
+    import subprocess
+    import sys
+    from pathlib import Path
+
+    root = Path(__file__).resolve().parent.parent
+    out = subprocess.run([sys.executable, str(root / "examples" /
+                                              "quickstart.py")],
+                         capture_output=True, text=True, timeout=120)
+    assert out.returncode == 0, out.stderr[-800:]
+    assert "restored docs: 3" in out.stdout
