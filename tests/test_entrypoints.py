@@ -678,3 +678,16 @@ def test_examples_quickstart_runs():
                          capture_output=True, text=True, timeout=120)
     assert out.returncode == 0, out.stderr[-800:]
     assert "restored docs: 3" in out.stdout
+
+
+def test_examples_serve_and_query_runs():
+    import subprocess
+    import sys
+    from pathlib import Path
+
+    root = Path(__file__).resolve().parent.parent
+    out = subprocess.run([sys.executable, str(root / "examples" /
+                                              "serve_and_query.py")],
+                         capture_output=True, text=True, timeout=120)
+    assert out.returncode == 0, out.stderr[-800:]
+    assert "health: True" in out.stdout
