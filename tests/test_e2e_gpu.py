@@ -157,3 +157,18 @@ def test_gpu_batched_services_path():
         assert out[i].results[0].url and out[i].results[0].title
     assert ctx.batcher.stats()["queries"] >= len(qs)
     ctx.close()
+
+
+@pytest.mark.gpu
+def test_examples_gpu_engine_runs():
+    """The shipped GPU example must keep working on real hardware."""
+    import subprocess
+    import sys
+    from pathlib import Path
+
+    root = Path(__file__).resolve().parent.parent
+    out = subprocess.run([sys.executable, str(root / "examples" /
+                                              "gpu_engine.py")],
+                         capture_output=True, text=True, timeout=300)
+    assert out.returncode == 0, out.stderr[-800:]
+    assert "warm start: 10001 docs restored" in out.stdout
