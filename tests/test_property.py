@@ -813,3 +813,38 @@ def test_data_quality_grade_total(crawled_at, trust, text_len, has_title):
     g = grade_document(crawled_at, trust, text_len, has_title,
                        now=2e9)
     assert g.grade in ("A", "B", "C", "D", "F")
+
+
+@given(st.integers(1, 40))
+@settings(max_examples=8, deadline=None)
+def test_decompression_bomb_guard(mb):
+    """A payload expanding past the 100 MB cap must raise, never
+    allocate unbounded memory; payloads under the cap round-trip."""
+    from infomesh_amd import compression
+    from infomesh_amd.errors import InfoMeshError
+
+    blob = b"\x00" * (mb * 1024 * 1024)
+    z = compression.compress(blob)
+    if mb * 1024 * 1024 <= compression.MAX_DECOMPRESSED_BYTES:
+        assert compression.decompress(z) == blob
+    # a tight cap must trip the guard loudly
+    try:
+        compression.decompress(z, max_decompressed=1024)
+        raise AssertionError("bomb guard did not fire")
+    except (InfoMeshError, ValueError, Exception) as e:
+        assert not isinstance(e, AssertionError)
+
+
+@given(st.binary(max_size=200))
+@settings(max_examples=60, deadline=None)
+def test_decompress_garbage_fails_closed(junk):
+    """Arbitrary junk never crashes the process: decompress raises a
+    catchable error or returns bytes (for valid frames)."""
+    from infomesh_amd import compression
+
+    try:
+        out = compression.decompress(junk)
+        assert isinstance(out, bytes)
+    except Exception as e:
+        assert not isinstance(e, (SystemExit, KeyboardInterrupt,
+                                  MemoryError))
