@@ -848,3 +848,23 @@ def test_decompress_garbage_fails_closed(junk):
     except Exception as e:
         assert not isinstance(e, (SystemExit, KeyboardInterrupt,
                                   MemoryError))
+
+
+@given(st.lists(st.text(min_size=1, max_size=30), min_size=2,
+                max_size=40, unique=True),
+       st.integers(0, 100))
+@settings(max_examples=20, deadline=None)
+def test_merkle_proof_for_every_leaf(items, tamper_idx):
+    """Every leaf of any tree proves membership against the root, and
+    a proof for one leaf never verifies a DIFFERENT leaf's value."""
+    from infomesh_amd.trust.merkle import MerkleTree
+
+    tree = MerkleTree.from_items(items)
+    root = tree.root
+    for i, it in enumerate(items):
+        p = tree.prove(i)
+        assert MerkleTree.verify_proof(root, p, it)
+    i = tamper_idx % len(items)
+    j = (i + 1) % len(items)
+    p = tree.prove(i)
+    assert not MerkleTree.verify_proof(root, p, items[j])
