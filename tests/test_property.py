@@ -868,3 +868,37 @@ def test_merkle_proof_for_every_leaf(items, tamper_idx):
     j = (i + 1) % len(items)
     p = tree.prove(i)
     assert not MerkleTree.verify_proof(root, p, items[j])
+
+
+@given(st.text(min_size=1, max_size=60), st.text(max_size=32),
+       st.text(max_size=32),
+       st.sampled_from(["url", "raw_hash", "text_hash", "signature",
+                        "ts"]))
+@settings(max_examples=20, deadline=None)
+def test_attestation_roundtrip_and_tamper(url, rh, th, field):
+    """Attestations serde-roundtrip and verification rejects any field
+    tamper (including the deferred-signing completion path)."""
+    from infomesh_amd.trust.attestation import (create_attestation,
+                                                sign_attestation,
+                                                verify_attestation)
+    from infomesh_amd.trust.keys import KeyPair
+
+    kp = KeyPair.generate()
+    att = create_attestation(kp, url, rh, th, sign=False)
+    assert not att.signature
+    sign_attestation(kp, att)
+    assert verify_attestation(att)
+    # serde roundtrip
+    back = type(att)(**att.to_dict()) if hasattr(att, "to_dict") else att
+    if hasattr(att, "to_dict"):
+        assert verify_attestation(back)
+    # tamper
+    import dataclasses as dc
+    bad = dc.replace(att)
+    if field == "ts":
+        bad.ts = att.ts + 1.0
+    elif field == "signature":
+        bad.signature = "ab" * 64
+    else:
+        setattr(bad, field, getattr(att, field) + "x")
+    assert not verify_attestation(bad)
