@@ -902,3 +902,28 @@ def test_attestation_roundtrip_and_tamper(url, rh, th, field):
     else:
         setattr(bad, field, getattr(att, field) + "x")
     assert not verify_attestation(bad)
+
+
+@given(st.lists(st.tuples(st.text(min_size=1, max_size=20),
+                          st.floats(0.0, 1.0)),
+                min_size=1, max_size=25))
+@settings(max_examples=20, deadline=None)
+def test_trust_score_bounds_and_tier_consistency(updates):
+    """Trust scores stay in [0,1] under any update sequence and the
+    tier label always matches the score thresholds."""
+    from infomesh_amd.trust.scoring import (TIER_TRUSTED, TIER_SUSPECT,
+                                            TIER_UNTRUSTED, TrustStore)
+
+    t = TrustStore(":memory:")
+    comps = ("uptime", "contribution", "audit_pass", "summary_quality")
+    for i, (subj, val) in enumerate(updates):
+        t.update_component(subj, comps[i % 4], val)
+    for subj, _ in updates:
+        s = t.score(subj)
+        assert 0.0 <= s <= 1.0
+        tier = t.tier(subj)
+        if tier == "trusted":
+            assert s >= TIER_TRUSTED
+        elif tier == "untrusted":
+            assert s < TIER_UNTRUSTED
+    t.close()
