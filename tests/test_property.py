@@ -927,3 +927,23 @@ def test_trust_score_bounds_and_tier_consistency(updates):
         elif tier == "untrusted":
             assert s < TIER_UNTRUSTED
     t.close()
+
+
+@given(st.lists(st.tuples(st.sampled_from(["a.com", "b.org", "c.net"]),
+                          st.sampled_from(["en", "de", ""]),
+                          st.floats(0.0, 1.0)),
+                min_size=0, max_size=30))
+@settings(max_examples=20, deadline=None)
+def test_facets_counts_partition_results(rows):
+    """Facet counts always sum to the result count per dimension."""
+    from infomesh_amd.index.local_store import SearchHit
+    from infomesh_amd.search.facets import compute_facets
+
+    hits = [SearchHit(doc_id=i, url=f"http://{d}/p{i}", title="t",
+                      snippet="", bm25=1.0, language=lang, domain=d,
+                      crawled_at=1.7e9, score=s)
+            for i, (d, lang, s) in enumerate(rows)]
+    f = compute_facets(hits)
+    assert sum(f.get("domains", {}).values()) == len(hits)
+    langs = f.get("languages", {})
+    assert sum(langs.values()) <= len(hits)   # unlabeled rows excluded
