@@ -947,3 +947,31 @@ def test_facets_counts_partition_results(rows):
     assert sum(f.get("domains", {}).values()) == len(hits)
     langs = f.get("languages", {})
     assert sum(langs.values()) <= len(hits)   # unlabeled rows excluded
+
+
+@given(st.lists(st.text(min_size=1, max_size=15), min_size=1,
+                max_size=10),
+       st.integers(1, 4))
+@settings(max_examples=15, deadline=None)
+def test_query_cache_lru_ttl_invariants(keys, cap):
+    """QueryCache never exceeds capacity, returns exactly what was
+    put, and expires by TTL."""
+    import time as _t
+
+    from infomesh_amd.search.cache import QueryCache
+
+    c = QueryCache(max_entries=cap, ttl_s=0.05)
+    for i, k in enumerate(keys):
+        c.put(k, f"v{i}")
+    stats = c.stats() if hasattr(c, "stats") else {}
+    live = sum(1 for k in keys if c.get(k) is not None)
+    assert live <= cap
+    # returned values are the LAST put for that key
+    seen = {}
+    for i, k in enumerate(keys):
+        seen[k] = f"v{i}"
+    for k in set(keys):
+        v = c.get(k)
+        assert v is None or v == seen[k]
+    _t.sleep(0.06)
+    assert all(c.get(k) is None for k in keys)   # TTL expiry
