@@ -975,3 +975,35 @@ def test_query_cache_lru_ttl_invariants(keys, cap):
         assert v is None or v == seen[k]
     _t.sleep(0.06)
     assert all(c.get(k) is None for k in keys)   # TTL expiry
+
+
+@given(st.text(max_size=400))
+@settings(max_examples=40, deadline=None)
+def test_structured_extraction_total(raw):
+    """JSON-LD/OpenGraph extraction never raises on arbitrary HTML."""
+    from infomesh_amd.crawler.structured import extract_structured
+
+    sd = extract_structured(raw)
+    assert sd is not None
+
+
+@given(st.lists(st.tuples(st.text(max_size=40), st.text(max_size=200)),
+                min_size=1, max_size=5))
+@settings(max_examples=20, deadline=None)
+def test_warc_export_total_and_readable(pairs):
+    """WARC export writes one WARC/1.0 record per input and the file
+    stays parseable (headers + lengths consistent)."""
+    import tempfile
+    from pathlib import Path as P
+
+    from infomesh_amd.crawler.diff import warc_export
+
+    recs = [{"url": f"http://e.com/{u}"[:80], "content": b}
+            for u, b in pairs]
+    with tempfile.TemporaryDirectory() as d:
+        p = P(d) / "x.warc"
+        n = warc_export(p, recs)
+        data = p.read_text(errors="replace")
+    assert n == len(recs)
+    assert data.count("WARC/1.0") == len(recs)
+    assert data.count("WARC-Target-URI") == len(recs)
