@@ -1007,3 +1007,23 @@ def test_warc_export_total_and_readable(pairs):
     assert n == len(recs)
     assert data.count("WARC/1.0") == len(recs)
     assert data.count("WARC-Target-URI") == len(recs)
+
+
+@given(st.text(max_size=500))
+@settings(max_examples=40, deadline=None)
+def test_pdf_extract_total(raw):
+    """PDF text extraction never raises on arbitrary bytes-ish input
+    (invalid PDFs return empty/None, never crash)."""
+    from infomesh_amd.crawler.pdf import extract_pdf_text
+
+    out = extract_pdf_text(raw.encode("utf-8", "replace"))
+    assert out is None or isinstance(out, str)
+
+
+@given(st.text(max_size=300))
+@settings(max_examples=40, deadline=None)
+def test_spa_detection_total(html):
+    """SPA/JS-render detection is a total boolean function of HTML."""
+    from infomesh_amd.crawler.parser import looks_like_js_app
+
+    assert looks_like_js_app(html, html[:50]) in (True, False)
