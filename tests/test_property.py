@@ -1027,3 +1027,21 @@ def test_spa_detection_total(html):
     from infomesh_amd.crawler.parser import looks_like_js_app
 
     assert looks_like_js_app(html, html[:50]) in (True, False)
+
+
+@given(st.lists(st.integers(0, 2**63 - 1), min_size=1, max_size=64),
+       st.lists(st.integers(0, 2**63 - 1), min_size=1, max_size=64))
+@settings(max_examples=30, deadline=None)
+def test_simhash_index_query_consistency(fps, probes):
+    """SimHashIndex finds a fingerprint iff one within the Hamming
+    threshold exists (brute-force oracle)."""
+    from infomesh_amd.crawler.simhash import HAMMING_THRESHOLD, SimHashIndex
+
+    idx = SimHashIndex()
+    for i, f in enumerate(fps):
+        idx.add(f"u{i}", f)
+    for p in probes:
+        got = idx.find_near(p)
+        oracle = any(bin(p ^ f).count("1") <= HAMMING_THRESHOLD
+                     for f in fps)
+        assert (got is not None) == oracle
