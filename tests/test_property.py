@@ -1045,3 +1045,32 @@ def test_simhash_index_query_consistency(fps, probes):
         oracle = any(bin(p ^ f).count("1") <= HAMMING_THRESHOLD
                      for f in fps)
         assert (got is not None) == oracle
+
+
+@given(st.lists(st.tuples(st.sampled_from("abcde"),
+                          st.sampled_from("abcde")),
+                min_size=0, max_size=30))
+@settings(max_examples=20, deadline=None)
+def test_link_graph_authority_bounds(edges):
+    """Domain authority stays in [0,1] for any link structure and a
+    domain with strictly more distinct inlinks never scores lower than
+    a domain with none."""
+    from infomesh_amd.index.link_graph import LinkGraph
+
+    g = LinkGraph(":memory:")
+    for s, t in edges:
+        if s != t:
+            g.add_links(f"http://{s}.com/x", [f"http://{t}.com/y"])
+    doms = {f"{c}.com" for c in "abcde"}
+    scores = {d: g.domain_authority(d) for d in doms}
+    assert all(0.0 <= v <= 1.0 for v in scores.values())
+    indeg = {d: 0 for d in doms}
+    for s, t in edges:
+        if s != t:
+            indeg[f"{t}.com"] += 1
+    linked = [d for d in doms if indeg[d] > 0]
+    unlinked = [d for d in doms if indeg[d] == 0]
+    if linked and unlinked:
+        assert max(scores[d] for d in linked) >= \
+            max(scores[d] for d in unlinked) - 1e-9
+    g.close()
