@@ -1074,3 +1074,28 @@ def test_link_graph_authority_bounds(edges):
         assert max(scores[d] for d in linked) >= \
             max(scores[d] for d in unlinked) - 1e-9
     g.close()
+
+
+@given(st.lists(st.booleans(), min_size=1, max_size=25))
+@settings(max_examples=15, deadline=None)
+def test_recrawl_interval_adaptive_bounds(changes):
+    """Adaptive recrawl intervals stay in [1 h, 30 d] under any
+    change history, shrink on change and grow on stability."""
+    from infomesh_amd.index.local_store import Document, LocalStore
+
+    store = LocalStore(":memory:")
+    store.add_document(Document(url="http://r/x", title="t",
+                                text="recrawl target body"))
+    prev = 86400.0
+    for changed in changes:
+        store.update_recrawl("http://r/x", changed)
+        cur = float(store.conn.execute(
+            "SELECT recrawl_interval_s FROM documents WHERE url=?",
+            ("http://r/x",)).fetchone()[0])
+        assert 3600.0 <= cur <= 30 * 86400.0
+        if changed:
+            assert cur <= prev + 1e-6
+        else:
+            assert cur >= prev - 1e-6
+        prev = cur
+    store.close()
