@@ -1099,3 +1099,27 @@ def test_recrawl_interval_adaptive_bounds(changes):
             assert cur >= prev - 1e-6
         prev = cur
     store.close()
+
+
+@given(st.floats(0, 8), st.floats(0, 64), st.floats(0, 64),
+       st.floats(0, 1))
+@settings(max_examples=40, deadline=None)
+def test_governor_level_monotone_in_pressure(load, mem_gb, rss_gb,
+                                             hbm_free):
+    """Degrade level is monotone: strictly MORE pressure (higher load/
+    rss, lower available mem / free HBM) never yields a lower level."""
+    from infomesh_amd.utils.governor import (DegradeLevel,
+                                             ResourceGovernor,
+                                             ResourceSample)
+
+    g = ResourceGovernor()
+    s1 = ResourceSample(load_per_cpu=load, mem_available_gb=mem_gb,
+                        mem_total_gb=64.0, rss_gb=rss_gb,
+                        hbm_free_frac=hbm_free)
+    s2 = ResourceSample(load_per_cpu=load * 1.5 + 0.5,
+                        mem_available_gb=mem_gb * 0.5,
+                        mem_total_gb=64.0, rss_gb=rss_gb * 1.5 + 1.0,
+                        hbm_free_frac=hbm_free * 0.5)
+    l1, l2 = g._level_for(s1), g._level_for(s2)
+    assert isinstance(l1, DegradeLevel)
+    assert l2.value >= l1.value
