@@ -1123,3 +1123,19 @@ def test_governor_level_monotone_in_pressure(load, mem_gb, rss_gb,
     l1, l2 = g._level_for(s1), g._level_for(s2)
     assert isinstance(l1, DegradeLevel)
     assert l2.value >= l1.value
+
+
+@given(st.lists(st.integers(0, 99), min_size=1, max_size=4),
+       st.lists(st.integers(0, 99), min_size=1, max_size=4))
+@settings(max_examples=40, deadline=None)
+def test_version_compare_matches_tuple_order(a, b):
+    """is_newer agrees with tuple ordering for any dotted versions and
+    parse_version is total on junk."""
+    from infomesh_amd.utils.version_check import is_newer, parse_version
+
+    va = ".".join(map(str, a))
+    vb = ".".join(map(str, b))
+    assert is_newer(va, vb) == (tuple(a) > tuple(b)
+                                or (len(a) != len(b)
+                                    and parse_version(va) > parse_version(vb)))
+    assert isinstance(parse_version("garbage-1.x.?"), tuple)
