@@ -1139,3 +1139,21 @@ def test_version_compare_matches_tuple_order(a, b):
                                 or (len(a) != len(b)
                                     and parse_version(va) > parse_version(vb)))
     assert isinstance(parse_version("garbage-1.x.?"), tuple)
+
+
+@given(st.lists(st.text(min_size=1, max_size=20), min_size=1,
+                max_size=10),
+       st.text(max_size=60))
+@settings(max_examples=20, deadline=None)
+def test_extended_batch_and_translate_total(queries, q):
+    """batch_search maps each query through the search fn exactly once
+    and keyword_translate is total (identity without a mapping)."""
+    from infomesh_amd.search.extended import batch_search, keyword_translate
+
+    calls = []
+    out = batch_search(lambda s: (calls.append(s), [s.upper()])[1],
+                       queries)
+    assert len(out) == len(queries) and calls == list(queries)
+    assert isinstance(keyword_translate(q), str)
+    # whitespace-normalizing identity without a mapping
+    assert keyword_translate(q, {}) == " ".join(q.split())
