@@ -1157,3 +1157,24 @@ def test_extended_batch_and_translate_total(queries, q):
     assert isinstance(keyword_translate(q), str)
     # whitespace-normalizing identity without a mapping
     assert keyword_translate(q, {}) == " ".join(q.split())
+
+
+@given(st.lists(st.tuples(st.text(min_size=1, max_size=12),
+                          st.floats(-5, 5)),
+                min_size=1, max_size=15))
+@settings(max_examples=20, deadline=None)
+def test_related_search_tracker_bounds(events):
+    """RelatedSearchTracker suggestions are always drawn from the
+    recorded co-session queries and never exceed the cap."""
+    from infomesh_amd.search.nlp import RelatedSearchTracker
+
+    t = RelatedSearchTracker()
+    seen = set()
+    for q, _ in events:
+        t.record(q)
+        if q.strip():
+            seen.add(q.strip().lower())
+    for q, _ in events:
+        rel = t.related(q, limit=3)
+        assert len(rel) <= 3
+        assert all(r in seen for r in rel)
