@@ -1177,3 +1177,28 @@ def test_related_search_tracker_bounds(events):
         rel = t.related(q, limit=3)
         assert len(rel) <= 3
         assert all(r in seen for r in rel)
+
+
+@given(st.binary(min_size=8, max_size=32), st.text(min_size=1, max_size=20),
+       st.sampled_from(["admin", "operator", "reader"]),
+       st.binary(max_size=64))
+@settings(max_examples=25, deadline=None)
+def test_jwt_and_webhook_hmac_roundtrip_and_tamper(secret, subject,
+                                                   role, body):
+    """Tokens verify with the right secret, fail with a different one
+    or any tamper; webhook HMAC likewise."""
+    from infomesh_amd.utils.security_ext import (issue_token,
+                                                 sign_webhook,
+                                                 verify_token,
+                                                 verify_webhook)
+
+    tok = issue_token(secret, subject, role)
+    claims = verify_token(secret, tok)
+    assert claims and claims["sub"] == subject and claims["role"] == role
+    assert verify_token(secret + b"x", tok) is None
+    assert verify_token(secret, tok[:-2] + "zz") is None
+
+    h = sign_webhook(secret, body)
+    assert verify_webhook(secret, body, h)
+    assert not verify_webhook(secret, body + b"x", h)
+    assert not verify_webhook(secret + b"x", body, h)
