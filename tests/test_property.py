@@ -1202,3 +1202,24 @@ def test_jwt_and_webhook_hmac_roundtrip_and_tamper(secret, subject,
     assert verify_webhook(secret, body, h)
     assert not verify_webhook(secret, body + b"x", h)
     assert not verify_webhook(secret + b"x", body, h)
+
+
+@given(st.floats(-12, 14), st.integers(0, 30))
+@settings(max_examples=15, deadline=None)
+def test_timezone_claim_verification_total(offset_h, n_llm):
+    """verify_timezone_claim is total and reports a [0,1] off-peak
+    fraction with plausible=True under sparse evidence."""
+    from infomesh_amd.credits.ledger import Action, CreditLedger
+    from infomesh_amd.credits.timezone_verify import verify_timezone_claim
+
+    led = CreditLedger(":memory:", off_peak_fn=lambda ts: True)
+    base = 1_700_000_000.0
+    for i in range(n_llm):
+        led.record_action(Action.LLM_SUMMARIZE, 1.0,
+                          ts=base + i * 3600.0)
+    rep = verify_timezone_claim(led, offset_h)
+    assert set(rep) >= {"plausible", "off_peak_fraction", "n"}
+    assert 0.0 <= rep["off_peak_fraction"] <= 1.0
+    if rep["n"] < 20:
+        assert rep["plausible"]   # sparse evidence never convicts
+    led.close()
