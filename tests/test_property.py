@@ -1223,3 +1223,23 @@ def test_timezone_claim_verification_total(offset_h, n_llm):
     if rep["n"] < 20:
         assert rep["plausible"]   # sparse evidence never convicts
     led.close()
+
+
+@given(st.lists(st.tuples(st.booleans(), st.floats(1.0, 5000.0),
+                          st.booleans()),
+                min_size=1, max_size=40))
+@settings(max_examples=20, deadline=None)
+def test_crawl_speed_tuner_delay_bounds(events):
+    """Adaptive per-domain crawl delay stays positive/bounded, grows
+    after failures or 429s, and never drops below the base after
+    sustained errors."""
+    from infomesh_amd.crawler.intelligence import AdaptiveCrawlTuner
+
+    t = AdaptiveCrawlTuner(base_delay_s=1.0)
+    for ok, lat, throttled in events:
+        t.record("d.com", ok, latency_ms=lat,
+                 status=429 if throttled else 200)
+    d = t.delay_for("d.com")
+    assert 0.25 <= d <= 60.0
+    if all((not ok) or th for ok, _, th in events):
+        assert d >= 1.0   # all-bad history never speeds up
