@@ -1243,3 +1243,28 @@ def test_crawl_speed_tuner_delay_bounds(events):
     assert 0.25 <= d <= 60.0
     if all((not ok) or th for ok, _, th in events):
         assert d >= 1.0   # all-bad history never speeds up
+
+
+@given(st.text(max_size=120))
+@settings(max_examples=40, deadline=None)
+def test_intent_and_cjk_recommendation_total(q):
+    """Intent classification and tokenizer recommendation are total
+    with outputs from their documented enums."""
+    from infomesh_amd.search.cjk import recommend_tokenizer
+    from infomesh_amd.search.passage import classify_intent
+
+    intent = classify_intent(q)
+    assert isinstance(intent, str) and intent
+    tok = recommend_tokenizer(q)
+    assert tok in ("unicode61", "ascii", "porter", "trigram")
+
+
+def test_dx_generators_produce_docs(tmp_path):
+    """Tool-guide/changelog generators emit non-trivial markdown."""
+    from infomesh_amd.utils.dx import generate_tool_guide, write_docs
+
+    guide = generate_tool_guide()
+    assert "web_search" in guide and len(guide) > 200
+    files = write_docs(tmp_path)
+    assert files and all(p.exists() and p.stat().st_size > 0
+                         for p in files)
