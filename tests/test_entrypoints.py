@@ -691,3 +691,17 @@ def test_examples_serve_and_query_runs():
                          capture_output=True, text=True, timeout=120)
     assert out.returncode == 0, out.stderr[-800:]
     assert "health: True" in out.stdout
+
+
+def test_examples_mcp_stdio_runs():
+    import subprocess
+    import sys
+    from pathlib import Path
+
+    root = Path(__file__).resolve().parent.parent
+    out = subprocess.run([sys.executable, str(root / "examples" /
+                                              "mcp_stdio_client.py")],
+                         capture_output=True, text=True, timeout=120)
+    assert out.returncode == 0, out.stderr[-800:]
+    assert "web_search" in out.stdout
+    assert "mentions MFMA: True" in out.stdout
