@@ -705,3 +705,16 @@ def test_examples_mcp_stdio_runs():
     assert out.returncode == 0, out.stderr[-800:]
     assert "web_search" in out.stdout
     assert "mentions MFMA: True" in out.stdout
+
+
+def test_examples_rag_with_adapters_runs():
+    import subprocess
+    import sys
+    from pathlib import Path
+
+    root = Path(__file__).resolve().parent.parent
+    out = subprocess.run([sys.executable, str(root / "examples" /
+                                              "rag_with_adapters.py")],
+                         capture_output=True, text=True, timeout=120)
+    assert out.returncode == 0, out.stderr[-800:]
+    assert "retrieved:" in out.stdout and "chunks:" in out.stdout
