@@ -39,6 +39,10 @@ def main() -> int:
          [sys.executable, "examples/quickstart.py"]),
         ("serve+SDK example",
          [sys.executable, "examples/serve_and_query.py"]),
+        ("MCP stdio example",
+         [sys.executable, "examples/mcp_stdio_client.py"]),
+        ("adapter RAG example",
+         [sys.executable, "examples/rag_with_adapters.py"]),
         ("bench contract (1 rank, CPU sanity)",
          [sys.executable, "bench.py", "--steps", "2", "--warmup", "1",
           "--docs-per-gpu", "5000"]),
