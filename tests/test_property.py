@@ -1287,3 +1287,32 @@ def test_feed_monitor_tiers_and_dedup(feeds):
     assert len(listed) == len(urls)
     due = m.due_feeds() if hasattr(m, "due_feeds") else []
     assert all(isinstance(x, object) for x in due)
+
+
+@given(st.lists(st.text(min_size=1, max_size=15), min_size=1,
+                max_size=20),
+       st.integers(1, 5))
+@settings(max_examples=15, deadline=None)
+def test_suggest_prefix_consistency(titles, n):
+    """LocalStore.suggest only returns stored titles/terms matching
+    the prefix, capped at the limit."""
+    from infomesh_amd.index.local_store import Document, LocalStore
+
+    store = LocalStore(":memory:")
+    kept = []
+    for i, t in enumerate(titles):
+        rid = store.add_document(Document(url=f"http://s/{i}",
+                                          title=t,
+                                          text=f"body {i} unique{i}"))
+        if rid is not None:
+            kept.append(t)
+    if not kept:
+        store.close()
+        return
+    probe = kept[0][:2]
+    out = store.suggest(probe, limit=n)
+    assert len(out) <= n
+    for s in out:
+        assert probe.lower() in s.lower() or s.lower().startswith(
+            probe.lower())
+    store.close()
