@@ -108,6 +108,30 @@ class _Extractor(HTMLParser):
             self.chunks.append(text)
 
 
+_PAYWALL_SIGNALS = (
+    "subscribe to continue", "subscription required", "sign in to read",
+    "create a free account", "this content is for subscribers",
+    "to continue reading", "register to continue", "metered paywall",
+    "already a subscriber", "unlock this article", "premium content",
+    "paywall",
+)
+
+
+def is_paywall_content(text: str) -> bool:
+    """Heuristic paywall detection for fetched pages (reference:
+    services.py is_paywall_content): a short extraction carrying
+    subscription-wall phrasing. Total on any input."""
+    if not text:
+        return False
+    t = text.lower()
+    hits = sum(1 for s in _PAYWALL_SIGNALS if s in t)
+    if hits == 0:
+        return False
+    # a long article that merely MENTIONS subscriptions is fine; a
+    # short stub with wall phrasing is the signal
+    return len(text) < 2500 or hits >= 2
+
+
 def looks_like_js_app(html: str, text: str) -> bool:
     """SPA detection: tiny extracted text + JS framework signals
     (reference: crawler/js_detect.py:88-148, simplified)."""

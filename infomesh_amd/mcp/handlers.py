@@ -150,10 +150,12 @@ class Handlers:
                 doc = self.ctx.fetch_page(url)
         if doc is None:
             return {"url": url, "found": False}
+        from ..crawler.parser import is_paywall_content
         return {"url": doc.url, "found": True, "title": doc.title,
                 "language": doc.language, "crawled_at": doc.crawled_at,
                 "text": doc.text[:max_chars],
-                "truncated": len(doc.text) > max_chars}
+                "truncated": len(doc.text) > max_chars,
+                "paywalled": is_paywall_content(doc.text)}
 
     # ------------------------------------------------------------- crawl
     def crawl_url(self, url: str, force: bool = False, depth: int = 0,

@@ -1335,3 +1335,15 @@ def test_cpu_fp8_quant_bounded_error(vals, k):
         return   # quantizer is internal to the embed path
     err = (q.float() - x).abs()
     assert (err <= x.abs() * 0.08 + 0.02).all()
+
+
+@given(st.text(max_size=300))
+@settings(max_examples=40, deadline=None)
+def test_paywall_detection_total_and_sane(text):
+    """Paywall detection is total; clean long prose is never flagged,
+    a short subscription-wall stub always is."""
+    from infomesh_amd.crawler.parser import is_paywall_content
+
+    assert is_paywall_content(text) in (True, False)
+    assert not is_paywall_content("plain informative sentence. " * 200)
+    assert is_paywall_content("Subscribe to continue reading.")
