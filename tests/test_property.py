@@ -1347,3 +1347,16 @@ def test_paywall_detection_total_and_sane(text):
     assert is_paywall_content(text) in (True, False)
     assert not is_paywall_content("plain informative sentence. " * 200)
     assert is_paywall_content("Subscribe to continue reading.")
+
+
+@given(st.text(max_size=200), st.text(max_size=40))
+@settings(max_examples=30, deadline=None)
+def test_fast_snippet_total_and_marked(text, q):
+    """fast_snippet is total for hostile text/query combinations and
+    only ever emits <b> markers around query terms."""
+    from infomesh_amd.search.passage import fast_snippet
+
+    s = fast_snippet(q, text)
+    assert isinstance(s, str)
+    # no unbalanced markers
+    assert s.count("<b>") == s.count("</b>")
