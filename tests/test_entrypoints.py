@@ -718,3 +718,31 @@ def test_examples_rag_with_adapters_runs():
                          capture_output=True, text=True, timeout=120)
     assert out.returncode == 0, out.stderr[-800:]
     assert "retrieved:" in out.stdout and "chunks:" in out.stdout
+
+
+def test_admin_api_endpoint_surface():
+    """Every documented admin endpoint answers (reference parity:
+    local_api route table)."""
+    import httpx
+
+    from infomesh_amd.api.local_api import create_app
+
+    ctx = AppContext.create(config=Config(), with_worker=False,
+                            with_engine=False, in_memory=True)
+    try:
+        ctx.index_document(Document(url="http://a/1", title="t",
+                                    text="endpoint surface body"))
+        from fastapi.testclient import TestClient
+        app = create_app(ctx)
+        client = TestClient(app)
+        for path in ("/health", "/readiness", "/status", "/config",
+                     "/index/stats", "/credits/balance",
+                     "/network/peers", "/analytics", "/analytics/tools",
+                     "/index/compression", "/metrics", "/openapi-spec",
+                     "/dashboard"):
+            r = client.get(path)
+            assert r.status_code == 200, (path, r.status_code, r.text[:100])
+        r = client.get("/search", params={"q": "endpoint surface"})
+        assert r.status_code == 200 and r.json()
+    finally:
+        ctx.close()
